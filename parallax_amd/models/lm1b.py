@@ -1,0 +1,115 @@
+"""LM1B LSTM language model — the flagship hybrid (sparse+dense) workload.
+
+Reference config (examples/lm1b/language_model.py:16-100,
+language_model_graph.py:16-65): vocab 793,470; embedding 512 partitioned
+into >=32 shards via get_partitioner; 1-layer LSTM state 2048 projected to
+512; sampled softmax with 8192 candidates over softmax_w [V,512] /
+softmax_b [V]; Adagrad(0.2) with clip_by_global_norm(10) on the dense
+(LSTM) params; batch 128 x 20 unrolled steps.
+
+Sparse parameters (classified by the tagging pass, updated through the
+in-process PS): the embedding shards, softmax_w shards and softmax_b.
+Dense parameters (RCCL all-reduce): LSTM W/B/W_proj.
+"""
+
+import torch
+import torch.nn as nn
+
+from parallax_amd import optim
+from parallax_amd.ops.embedding import PartitionedEmbedding
+from parallax_amd.ops.lstm import ParallaxLSTM
+from parallax_amd.ops.sampled_softmax import sampled_softmax_loss
+from parallax_amd.ops.misc import clip_by_global_norm
+from parallax_amd.partitions import get_partitioner
+from parallax_amd.session import TrainGraph
+
+VOCAB_SIZE = 793470
+EMB_DIM = 512
+STATE_SIZE = 2048
+PROJ_SIZE = 512
+NUM_STEPS = 20
+BATCH_SIZE = 128
+NUM_SAMPLED = 8192
+
+
+class LM1BModel(nn.Module):
+    def __init__(self, vocab_size=VOCAB_SIZE, emb_dim=EMB_DIM,
+                 state_size=STATE_SIZE, proj_size=PROJ_SIZE,
+                 num_sampled=NUM_SAMPLED, min_partitions=32,
+                 dtype=torch.float32):
+        super().__init__()
+        self.vocab_size = vocab_size
+        self.num_sampled = num_sampled
+        partitioner = get_partitioner(min_partitions)
+        self.emb = PartitionedEmbedding(vocab_size, emb_dim, partitioner,
+                                        dtype=dtype, init_scale=0.1)
+        self.lstm = ParallaxLSTM(emb_dim, state_size, proj_size,
+                                 dtype=dtype)
+        self.softmax_w = PartitionedEmbedding(vocab_size, proj_size,
+                                              partitioner, dtype=dtype,
+                                              init_scale=proj_size ** -0.5)
+        self.softmax_b = nn.Parameter(torch.zeros(vocab_size, dtype=dtype))
+        from parallax_amd.tagging import mark_sparse
+        mark_sparse(self.softmax_b)
+
+    def forward(self, x, y, state=None):
+        """x, y: [B, T] int64.  Returns (mean loss per word, final state)."""
+        emb = self.emb(x)                      # [B, T, D]
+        hs, state = self.lstm(emb, state)      # [B, T, P]
+        flat_h = hs.reshape(-1, hs.shape[-1])
+        flat_y = y.reshape(-1)
+        losses = sampled_softmax_loss(self.softmax_w, self.softmax_b,
+                                      flat_y, flat_h, self.num_sampled,
+                                      self.vocab_size)
+        return losses.mean(), state
+
+    def dense_parameters(self):
+        return list(self.lstm.parameters())
+
+
+def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
+                     vocab_size=VOCAB_SIZE, lr=0.2, min_partitions=32,
+                     emb_dim=EMB_DIM, state_size=STATE_SIZE,
+                     proj_size=PROJ_SIZE, num_sampled=NUM_SAMPLED,
+                     max_grad_norm=10.0, dtype=torch.float32):
+    model = LM1BModel(vocab_size, emb_dim, state_size, proj_size,
+                      num_sampled, min_partitions, dtype=dtype)
+    optimizer = optim.Adagrad(model.parameters(), lr=lr,
+                              initial_accumulator_value=1.0)
+    dense_params = model.dense_parameters()
+
+    def step_fn(m, batch):
+        x, y, state = batch
+        loss, new_state = m(x, y, state)
+        return loss, {'state': (new_state[0].detach(),
+                                new_state[1].detach())}
+
+    graph = TrainGraph(model, optimizer, step_fn)
+
+    # clip_by_global_norm on the dense (LSTM) grads, applied between
+    # backward and the optimizer step (reference
+    # language_model_graph.py:53).  Installed as an engine pre-step hook by
+    # wrapping the optimizer step.
+    orig_step = optimizer.step
+
+    def step_with_clip(closure=None):
+        grads = [p.grad for p in dense_params if p.grad is not None
+                 and p.grad.layout == torch.strided]
+        if grads:
+            clip_by_global_norm(grads, max_grad_norm)
+        return orig_step(closure)
+
+    optimizer.step = step_with_clip
+    return graph
+
+
+def synthetic_batch(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
+                    vocab_size=VOCAB_SIZE, device='cpu', state=None,
+                    generator=None):
+    """Synthetic LM1B data (reference use_synthetic flag,
+    lm1b_distributed_driver.py:81-84: uniform random ids)."""
+    x = torch.randint(0, vocab_size, (batch_size, num_steps), device=device,
+                      generator=generator)
+    y = torch.randint(0, vocab_size, (batch_size, num_steps), device=device,
+                      generator=generator)
+    return (x, y, state)

@@ -1,0 +1,114 @@
+"""ResNet v1 (50/101/152) — the dense AR-mode benchmark workload.
+
+Reference: examples/tf_cnn_benchmarks/models/resnet_model.py:42-258
+(bottleneck v1 with projection shortcuts, BN decay .999, 1001-way softmax,
+224x224 synthetic ImageNet, per-GPU batch 64) driven by
+benchmark_cnn.py (SGD/Momentum, softmax-CE loss :290-302).
+
+MI355X-native choices: NHWC (channels_last) memory format — the fast
+layout for MIOpen convolutions on CDNA4 (the reference defaults to NCHW
+for CUDA); bf16 autocast for the conv/GEMM path; our fused
+softmax-CE kernel for the loss; all parameters dense -> the whole model
+exercises the bucketed RCCL all-reduce engine.
+"""
+
+import torch
+import torch.nn as nn
+
+from parallax_amd import optim
+from parallax_amd.ops.softmax_ce import softmax_cross_entropy
+from parallax_amd.session import TrainGraph
+
+_LAYERS = {50: (3, 4, 6, 3), 101: (3, 4, 23, 3), 152: (3, 8, 36, 3)}
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, cin, width, stride=1, bn_momentum=0.001):
+        super().__init__()
+        cout = width * self.expansion
+        self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(width, momentum=bn_momentum)
+        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
+                               bias=False)
+        self.bn2 = nn.BatchNorm2d(width, momentum=bn_momentum)
+        self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(cout, momentum=bn_momentum)
+        self.relu = nn.ReLU(inplace=True)
+        if stride != 1 or cin != cout:
+            self.downsample = nn.Sequential(
+                nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
+                nn.BatchNorm2d(cout, momentum=bn_momentum))
+        else:
+            self.downsample = None
+
+    def forward(self, x):
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        return self.relu(out + identity)
+
+
+class ResNet(nn.Module):
+    def __init__(self, depth=50, num_classes=1001, bn_momentum=0.001):
+        super().__init__()
+        layers = _LAYERS[depth]
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64, momentum=bn_momentum)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        cin, stages = 64, []
+        for i, (width, n) in enumerate(zip((64, 128, 256, 512), layers)):
+            blocks = []
+            for j in range(n):
+                stride = 2 if (i > 0 and j == 0) else 1
+                blocks.append(Bottleneck(cin, width, stride, bn_momentum))
+                cin = width * Bottleneck.expansion
+            stages.append(nn.Sequential(*blocks))
+        self.layer1, self.layer2, self.layer3, self.layer4 = stages
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        x = self.avgpool(x).flatten(1)
+        return self.fc(x)
+
+
+def build_resnet_graph(depth=50, num_classes=1001, batch_size=64, lr=0.1,
+                       momentum=0.9, weight_decay=1e-4, use_bf16=True,
+                       channels_last=True):
+    model = ResNet(depth, num_classes)
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
+    optimizer = optim.SGD(model.parameters(), lr=lr, momentum=momentum,
+                          weight_decay=weight_decay)
+    use_amp = use_bf16
+
+    def step_fn(m, batch):
+        x, y = batch
+        if channels_last and x.dim() == 4:
+            x = x.to(memory_format=torch.channels_last)
+        if use_amp and x.is_cuda:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                logits = m(x)
+            loss = softmax_cross_entropy(logits.float(), y)
+        else:
+            logits = m(x)
+            loss = softmax_cross_entropy(logits, y)
+        return loss, {}
+
+    return TrainGraph(model, optimizer, step_fn)
+
+
+def synthetic_batch(batch_size=64, image_size=224, num_classes=1001,
+                    device='cpu', generator=None):
+    """Synthetic ImageNet batch (reference benchmark_cnn.py:836-860)."""
+    x = torch.randn(batch_size, 3, image_size, image_size, device=device,
+                    generator=generator)
+    y = torch.randint(0, num_classes, (batch_size,), device=device,
+                      generator=generator)
+    return x, y

@@ -1,0 +1,74 @@
+"""parallax_amd.ops — hand-written CDNA4 HIP kernels + CPU reference impls.
+
+Dispatch policy (per the framework's no-silent-fallback rule):
+  * CUDA/HIP tensors  -> the native extension ``parallax_amd._C`` MUST be
+    importable; if it is missing the op RAISES instead of silently falling
+    back to eager PyTorch.  (Set PARALLAX_ALLOW_EAGER_GPU=1 to override for
+    debugging only.)
+  * CPU tensors       -> plain PyTorch reference implementations (also the
+    ground truth for the numerics tests in tests/test_ops_*.py).
+
+The kernels themselves live in parallax_amd/ops/hip/ and are built in-tree
+for gfx950 via ``python -m parallax_amd.ops.build`` (or setup.py / the
+repo-level __graft_entry__.build()).
+"""
+
+import os
+
+import torch
+
+_EXT = None
+_EXT_ERR = None
+
+
+def _try_load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from parallax_amd import _C  # built in-tree by ops/build.py
+        _EXT = _C
+    except ImportError as e:  # extension not built
+        _EXT_ERR = e
+        _EXT = None
+    return _EXT
+
+
+def native_available():
+    return _try_load_ext() is not None
+
+
+def ext():
+    """Return the native extension module, raising loudly on a GPU box where
+    it is missing (a GPU test passing on an eager fallback is a bug)."""
+    m = _try_load_ext()
+    if m is None:
+        if os.environ.get('PARALLAX_ALLOW_EAGER_GPU') == '1':
+            return None
+        raise RuntimeError(
+            'parallax_amd native extension (parallax_amd._C) is not built; '
+            'run `python -m parallax_amd.ops.build` (gfx950). '
+            'Original import error: %r' % (_EXT_ERR,))
+    return m
+
+
+def use_native(*tensors):
+    """True when the op should dispatch to the HIP kernel."""
+    on_gpu = any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+    if not on_gpu:
+        return False
+    ext()  # raises if missing (unless eager override)
+    return _EXT is not None
+
+
+from parallax_amd.ops.sparse import (unique_sum_rows, segment_sum_rows,
+                                     rows_gather)
+from parallax_amd.ops.embedding import (ParallaxEmbedding,
+                                        PartitionedEmbedding,
+                                        embedding_lookup)
+from parallax_amd.ops.layernorm import layer_norm, ParallaxLayerNorm
+from parallax_amd.ops.softmax_ce import softmax_cross_entropy
+from parallax_amd.ops.lstm import lstm_cell, ParallaxLSTM
+from parallax_amd.ops.sampled_softmax import sampled_softmax_loss
+from parallax_amd.ops.misc import (clip_by_global_norm, global_norm,
+                                   ema_update)

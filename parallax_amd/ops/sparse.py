@@ -1,0 +1,105 @@
+"""Indexed-slices primitives: dedup/segmented row-sum, row gather.
+
+MI355X-native equivalent of the reference's forked
+SparseConditionalAccumulator ``take_indexed_slices_grad(..., average_option)``
+(contract N2; reference graph_transform_lib.py:101-102,385-390,1115-1118):
+accumulate n sparse gradients by summing rows with equal indices, with an
+optional per-row count average (SPARSE_AVERAGE_BY_COUNTER) vs plain sum
+(SPARSE_NO_AVERAGE).
+
+GPU path: torch.sort (rocPRIM radix sort) for ordering + hand-written HIP
+segment-sum kernel over the sorted runs (ops/hip/sparse_reduce.hip).
+CPU path: the same algorithm in plain PyTorch (ground truth for tests).
+"""
+
+import torch
+
+from parallax_amd import ops as _ops
+
+SPARSE_AVERAGE_BY_COUNTER = 1
+SPARSE_NO_AVERAGE = 3
+
+
+def _sorted_segments(indices):
+    """Sort indices; return (sorted_vals, order, seg_starts, unique_vals).
+
+    seg_starts has one entry per unique index (start position in the sorted
+    array) plus a final n sentinel."""
+    sorted_vals, order = torch.sort(indices)
+    n = sorted_vals.numel()
+    if n == 0:
+        empty = torch.zeros(1, dtype=torch.long, device=indices.device)
+        return sorted_vals, order, empty, sorted_vals
+    mask = torch.ones(n, dtype=torch.bool, device=indices.device)
+    mask[1:] = sorted_vals[1:] != sorted_vals[:-1]
+    seg_starts = torch.nonzero(mask).flatten()
+    unique_vals = sorted_vals[seg_starts]
+    seg_starts = torch.cat([seg_starts,
+                            torch.tensor([n], dtype=torch.long,
+                                         device=indices.device)])
+    return sorted_vals, order, seg_starts, unique_vals
+
+
+def unique_sum_rows(indices, values, average=False):
+    """Deduplicate ``indices`` and segment-sum the corresponding rows of
+    ``values`` ([n, d] or [n]); returns (unique_indices[u], summed[u, d],
+    counts[u]).  ``average=True`` divides each output row by its count
+    (reference average_option=SPARSE_AVERAGE_BY_COUNTER)."""
+    assert indices.dim() == 1
+    squeeze = False
+    if values.dim() == 1:
+        values = values.unsqueeze(1)
+        squeeze = True
+    assert values.shape[0] == indices.shape[0]
+
+    if indices.numel() == 0:
+        out = values.new_zeros((0, values.shape[1]))
+        counts = torch.zeros(0, dtype=torch.long, device=indices.device)
+        return indices, out.squeeze(1) if squeeze else out, counts
+
+    sorted_vals, order, seg_starts, unique_vals = _sorted_segments(indices)
+    counts = seg_starts[1:] - seg_starts[:-1]
+    if _ops.use_native(values):
+        out = _ops.ext().segment_sum_rows(
+            values, order.to(torch.int32), seg_starts.to(torch.int32),
+            bool(average))
+    else:
+        out = segment_sum_rows_ref(values, order, seg_starts, average)
+    return unique_vals, (out.squeeze(1) if squeeze else out), counts
+
+
+def segment_sum_rows(values, order, seg_starts, average=False):
+    """Row segment-sum given precomputed sort ``order`` and segment
+    boundaries (see _sorted_segments).  out[s] = sum_k values[order[k]]
+    for k in [seg_starts[s], seg_starts[s+1])."""
+    if _ops.use_native(values):
+        return _ops.ext().segment_sum_rows(
+            values, order.to(torch.int32), seg_starts.to(torch.int32),
+            bool(average))
+    return segment_sum_rows_ref(values, order, seg_starts, average)
+
+
+def segment_sum_rows_ref(values, order, seg_starts, average=False):
+    """CPU/ground-truth implementation (fp32 accumulation)."""
+    u = seg_starts.numel() - 1
+    gathered = values.index_select(0, order).to(torch.float32)
+    out = torch.zeros((u, values.shape[1]), dtype=torch.float32,
+                      device=values.device)
+    seg_ids = torch.repeat_interleave(
+        torch.arange(u, device=values.device),
+        (seg_starts[1:] - seg_starts[:-1]))
+    out.index_add_(0, seg_ids, gathered)
+    if average:
+        counts = (seg_starts[1:] - seg_starts[:-1]).to(torch.float32)
+        out = out / counts.unsqueeze(1)
+    return out.to(values.dtype)
+
+
+def rows_gather(weight, indices):
+    """out[i] = weight[indices[i]] — the embedding-lookup gather
+    (hand-written HIP kernel on GPU; reference: TF Gather CUDA kernel,
+    examples/lm1b/language_model.py:26-40)."""
+    if _ops.use_native(weight):
+        return _ops.ext().rows_gather(weight, indices.to(torch.int64))
+    return weight.index_select(0, indices.reshape(-1)).view(
+        *indices.shape, weight.shape[1])

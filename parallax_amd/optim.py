@@ -1,0 +1,290 @@
+"""Fused optimizers — the reference's update-op vocabulary, MI355X-native.
+
+The reference's graph rewrites recognize a fixed set of variable-update ops
+(graph_transform_lib.py:56-75): dense {ApplyGradientDescent, ApplyMomentum,
+ApplyAdagrad, ApplyAdam, ApplyRMSProp, ...} and sparse {SparseApplyAdagrad,
+ScatterAdd/Sub, ...}.  Here each optimizer provides
+
+  * a fused DENSE step — one hand-written multi-tensor HIP launch over all
+    its parameters (ops/hip/multi_tensor.hip) on GPU, a torch reference on
+    CPU; and
+  * ``sparse_row_step(param, ids, grad_rows, counts)`` — the
+    SparseApplyAdagrad-equivalent: update ONLY the touched (deduplicated)
+    rows, with per-row optimizer state.  The sparse engines call this on
+    the owner rank of each shard.
+
+All sparse state (e.g. the Adagrad accumulator) is kept per-parameter at
+full shape so checkpoints can round-trip it.
+"""
+
+import math
+
+import torch
+from torch.optim.optimizer import Optimizer
+
+from parallax_amd import ops as _ops
+
+
+def _use_native(params):
+    return any(p.is_cuda for p in params) and _ops.native_available()
+
+
+class ParallaxOptimizer(Optimizer):
+    """Base: splits the fused dense path from the per-row sparse path."""
+
+    def supports_sparse(self):
+        return True
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            params, grads = [], []
+            for p in group['params']:
+                if p.grad is None:
+                    continue
+                if p.grad.layout != torch.strided:
+                    # sparse grads are handled by the engine via
+                    # sparse_row_step; a bare (non-engine) run coalesces here
+                    if p.grad._nnz() == 0:
+                        continue  # engine-swallowed (empty replacement grad)
+                    g = p.grad.coalesce()
+                    self.sparse_row_step(p, g.indices().flatten(),
+                                         g.values(), group)
+                    continue
+                params.append(p)
+                grads.append(p.grad)
+            if params:
+                self._dense_step(params, grads, group)
+        return loss
+
+    def _dense_step(self, params, grads, group):
+        raise NotImplementedError
+
+    def sparse_row_step(self, param, ids, grad_rows, group=None):
+        raise NotImplementedError
+
+    def group_of(self, param):
+        for group in self.param_groups:
+            for p in group['params']:
+                if p is param:
+                    return group
+        raise KeyError('param not in optimizer')
+
+
+class SGD(ParallaxOptimizer):
+    """ApplyGradientDescent / ApplyMomentum (+nesterov)
+    (reference graph_transform_lib.py:57,66)."""
+
+    def __init__(self, params, lr, momentum=0.0, nesterov=False,
+                 weight_decay=0.0):
+        defaults = dict(lr=lr, momentum=momentum, nesterov=nesterov,
+                        weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+    def _dense_step(self, params, grads, group):
+        lr, mom = group['lr'], group['momentum']
+        wd, nesterov = group['weight_decay'], group['nesterov']
+        if mom == 0.0:
+            if _use_native(params):
+                _ops.ext().multi_tensor_sgd(params, [g for g in grads],
+                                            lr, wd)
+                return
+            for p, g in zip(params, grads):
+                if wd:
+                    g = g.add(p, alpha=wd)
+                p.add_(g, alpha=-lr)
+            return
+        bufs = []
+        for p in params:
+            st = self.state[p]
+            if 'momentum_buffer' not in st:
+                st['momentum_buffer'] = torch.zeros_like(p)
+            bufs.append(st['momentum_buffer'])
+        if _use_native(params):
+            _ops.ext().multi_tensor_momentum(params, list(grads), bufs,
+                                             lr, mom, wd, nesterov)
+            return
+        for p, g, buf in zip(params, grads, bufs):
+            if wd:
+                g = g.add(p, alpha=wd)
+            buf.mul_(mom).add_(g)
+            if nesterov:
+                p.add_(g.add(buf, alpha=mom), alpha=-lr)
+            else:
+                p.add_(buf, alpha=-lr)
+
+    @torch.no_grad()
+    def sparse_row_step(self, param, ids, grad_rows, group=None):
+        group = group or self.group_of(param)
+        lr = group['lr']
+        if _use_native([param]):
+            _ops.ext().sparse_sgd(param, ids.to(torch.int64),
+                                  grad_rows.contiguous(), lr)
+            return
+        param.index_add_(0, ids, grad_rows.to(param.dtype), alpha=-lr)
+
+
+class Adagrad(ParallaxOptimizer):
+    """ApplyAdagrad (dense) + SparseApplyAdagrad (sparse rows)
+    (reference graph_transform_lib.py:59,74) — the LM1B optimizer
+    (language_model_graph.py:60-61)."""
+
+    def __init__(self, params, lr=0.2, initial_accumulator_value=1.0,
+                 eps=1e-10):
+        defaults = dict(lr=lr,
+                        initial_accumulator_value=initial_accumulator_value,
+                        eps=eps)
+        super().__init__(params, defaults)
+
+    def _accum(self, p, group):
+        st = self.state[p]
+        if 'sum' not in st:
+            st['sum'] = torch.full_like(
+                p, group['initial_accumulator_value'],
+                dtype=torch.float32)
+        return st['sum']
+
+    def _dense_step(self, params, grads, group):
+        lr, eps = group['lr'], group['eps']
+        accums = [self._accum(p, group) for p in params]
+        if _use_native(params):
+            _ops.ext().multi_tensor_adagrad(params, list(grads), accums,
+                                            lr, eps)
+            return
+        for p, g, acc in zip(params, grads, accums):
+            g32 = g.float()
+            acc.add_(g32 * g32)
+            p.add_((g32 / (acc.sqrt() + eps)).to(p.dtype), alpha=-lr)
+
+    @torch.no_grad()
+    def sparse_row_step(self, param, ids, grad_rows, group=None):
+        group = group or self.group_of(param)
+        lr, eps = group['lr'], group['eps']
+        acc = self._accum(param, group)
+        if _use_native([param]):
+            _ops.ext().sparse_adagrad(param, acc, ids.to(torch.int64),
+                                      grad_rows.contiguous(), lr, eps)
+            return
+        g32 = grad_rows.float()
+        acc.index_add_(0, ids, g32 * g32)
+        denom = acc.index_select(0, ids).sqrt() + eps
+        param.index_add_(0, ids, (g32 / denom).to(param.dtype), alpha=-lr)
+
+
+class Adam(ParallaxOptimizer):
+    """ApplyAdam (reference graph_transform_lib.py:67) — skip_thoughts
+    optimizer."""
+
+    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                 weight_decay=0.0):
+        defaults = dict(lr=lr, betas=betas, eps=eps,
+                        weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+    def _state(self, p):
+        st = self.state[p]
+        if 'step' not in st:
+            st['step'] = 0
+            st['exp_avg'] = torch.zeros_like(p, dtype=torch.float32)
+            st['exp_avg_sq'] = torch.zeros_like(p, dtype=torch.float32)
+        return st
+
+    def _dense_step(self, params, grads, group):
+        lr, (b1, b2), eps = group['lr'], group['betas'], group['eps']
+        wd = group['weight_decay']
+        ms, vs, steps = [], [], []
+        for p in params:
+            st = self._state(p)
+            st['step'] += 1
+            ms.append(st['exp_avg'])
+            vs.append(st['exp_avg_sq'])
+            steps.append(st['step'])
+        if _use_native(params):
+            _ops.ext().multi_tensor_adam(params, list(grads), ms, vs,
+                                         [float(s) for s in steps],
+                                         lr, b1, b2, eps, wd)
+            return
+        for p, g, m, v, t in zip(params, grads, ms, vs, steps):
+            g32 = g.float()
+            if wd:
+                g32 = g32.add(p.float(), alpha=wd)
+            m.mul_(b1).add_(g32, alpha=1 - b1)
+            v.mul_(b2).addcmul_(g32, g32, value=1 - b2)
+            bc1 = 1 - b1 ** t
+            bc2 = 1 - b2 ** t
+            step_size = lr / bc1
+            denom = (v / bc2).sqrt().add_(eps)
+            p.add_((m / denom).to(p.dtype), alpha=-step_size)
+
+    @torch.no_grad()
+    def sparse_row_step(self, param, ids, grad_rows, group=None):
+        """Lazy-Adam style: moments updated only on touched rows (the TF
+        sparse Adam apply)."""
+        group = group or self.group_of(param)
+        lr, (b1, b2), eps = group['lr'], group['betas'], group['eps']
+        st = self._state(param)
+        st['step'] += 1
+        t = st['step']
+        m, v = st['exp_avg'], st['exp_avg_sq']
+        g32 = grad_rows.float()
+        m_rows = m.index_select(0, ids).mul_(b1).add_(g32, alpha=1 - b1)
+        v_rows = v.index_select(0, ids).mul_(b2).addcmul_(g32, g32,
+                                                          value=1 - b2)
+        m.index_copy_(0, ids, m_rows)
+        v.index_copy_(0, ids, v_rows)
+        bc1, bc2 = 1 - b1 ** t, 1 - b2 ** t
+        denom = (v_rows / bc2).sqrt().add_(eps)
+        upd = (m_rows / bc1 / denom).to(param.dtype)
+        param.index_add_(0, ids, upd, alpha=-lr)
+
+
+class RMSProp(ParallaxOptimizer):
+    """ApplyRMSProp (reference graph_transform_lib.py:68) — tf_cnn_benchmarks
+    option (benchmark_cnn.py:805-818)."""
+
+    def __init__(self, params, lr, decay=0.9, momentum=0.9, eps=1.0):
+        defaults = dict(lr=lr, decay=decay, momentum=momentum, eps=eps)
+        super().__init__(params, defaults)
+
+    def _dense_step(self, params, grads, group):
+        lr, decay = group['lr'], group['decay']
+        mom, eps = group['momentum'], group['eps']
+        rms, bufs = [], []
+        for p in params:
+            st = self.state[p]
+            if 'square_avg' not in st:
+                st['square_avg'] = torch.zeros_like(p, dtype=torch.float32)
+                st['momentum_buffer'] = torch.zeros_like(
+                    p, dtype=torch.float32)
+            rms.append(st['square_avg'])
+            bufs.append(st['momentum_buffer'])
+        if _use_native(params):
+            _ops.ext().multi_tensor_rmsprop(params, list(grads), rms, bufs,
+                                            lr, decay, mom, eps)
+            return
+        for p, g, sq, buf in zip(params, grads, rms, bufs):
+            g32 = g.float()
+            sq.mul_(decay).addcmul_(g32, g32, value=1 - decay)
+            upd = g32 / (sq.sqrt() + eps)
+            buf.mul_(mom).add_(upd)
+            p.add_(buf.to(p.dtype), alpha=-lr)
+
+    @torch.no_grad()
+    def sparse_row_step(self, param, ids, grad_rows, group=None):
+        group = group or self.group_of(param)
+        lr, decay = group['lr'], group['decay']
+        mom, eps = group['momentum'], group['eps']
+        st = self.state[param]
+        if 'square_avg' not in st:
+            st['square_avg'] = torch.zeros_like(param, dtype=torch.float32)
+        sq = st['square_avg']
+        g32 = grad_rows.float()
+        sq_rows = sq.index_select(0, ids).mul_(decay).addcmul_(
+            g32, g32, value=1 - decay)
+        sq.index_copy_(0, ids, sq_rows)
+        upd = (g32 / (sq_rows.sqrt() + eps)).to(param.dtype)
+        param.index_add_(0, ids, upd, alpha=-lr)

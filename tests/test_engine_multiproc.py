@@ -1,0 +1,256 @@
+"""Multi-process CPU tests (gloo backend, world_size=2): the distributed
+paths the driver exercises on the 8-GPU node, validated here without a GPU
+(BASELINE config 1 class of tests).  Rendezvous over 127.0.0.1."""
+
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from parallax_amd.lib import get_free_port
+
+WORLD = 2
+
+
+def _init_env(rank, world, port):
+    os.environ['RANK'] = str(rank)
+    os.environ['LOCAL_RANK'] = str(rank)
+    os.environ['WORLD_SIZE'] = str(world)
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+
+
+def _spawn(fn, port, world=WORLD, timeout=180):
+    ctx = mp.get_context('spawn')
+    procs = []
+    err_q = ctx.SimpleQueue()
+    for r in range(world):
+        p = ctx.Process(target=_child_main, args=(fn, r, world, port, err_q))
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(timeout)
+    errs = []
+    while not err_q.empty():
+        errs.append(err_q.get())
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
+            errs.append('worker timed out')
+        elif p.exitcode != 0:
+            errs.append('worker exit %s' % p.exitcode)
+    assert not errs, '\n'.join(str(e) for e in errs)
+
+
+def _child_main(fn, rank, world, port, err_q):
+    try:
+        _init_env(rank, world, port)
+        fn(rank, world)
+    except Exception:
+        import traceback
+        err_q.put(traceback.format_exc())
+        raise
+    finally:
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+# ---------------------------------------------------------------------------
+# child bodies (module-level: picklable under spawn)
+# ---------------------------------------------------------------------------
+
+def _body_allreduce_engine(rank, world):
+    import torch.nn as nn
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.dense_engine import DenseAllReduceEngine
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    torch.manual_seed(123)  # identical init on both ranks
+    model = nn.Sequential(nn.Linear(4, 8), nn.Linear(8, 2))
+    infos = classify_params(model)
+    eng = DenseAllReduceEngine(comm, infos, bucket_bytes=64,  # force >1 bucket
+                               overlap=True)
+    eng.prepare_step()
+    torch.manual_seed(500 + rank)  # different data per rank
+    x = torch.randn(6, 4)
+    loss = model(x).pow(2).sum()
+    loss.backward()
+    eng.reduce_gradients()
+    # grads must now equal the average over both ranks: recompute both
+    grads_mine = [p.grad.clone() for p in model.parameters()]
+    ref_model = nn.Sequential(nn.Linear(4, 8), nn.Linear(8, 2))
+    ref_model.load_state_dict(model.state_dict())
+    expected = [torch.zeros_like(p) for p in ref_model.parameters()]
+    for r in range(world):
+        ref_model.zero_grad()
+        torch.manual_seed(500 + r)
+        xr = torch.randn(6, 4)
+        ref_model(xr).pow(2).sum().backward()
+        for e, p in zip(expected, ref_model.parameters()):
+            e += p.grad / world
+    for g, e in zip(grads_mine, expected):
+        assert torch.allclose(g, e, atol=1e-5), (g - e).abs().max()
+    eng.close()
+
+
+def _body_sparse_allgather(rank, world):
+    import torch.nn as nn
+    from parallax_amd import optim
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.sparse_engine import SparseAllGatherEngine
+    from parallax_amd.ops.embedding import ParallaxEmbedding
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    torch.manual_seed(7)
+    model = nn.Module()
+    model.emb = ParallaxEmbedding(30, 4)
+    opt = optim.Adagrad(model.parameters(), lr=0.5)
+    infos = classify_params(model)
+    eng = SparseAllGatherEngine(comm, infos, opt)
+    eng.prepare_step()
+    ids = torch.tensor([1 + rank, 5])  # rank0: {1,5}, rank1: {2,5}
+    out = model.emb(ids)
+    out.sum().backward()
+    eng.reduce_and_apply()
+    # both ranks must hold IDENTICAL tables afterwards
+    w = model.emb.weight.detach()
+    ws = [torch.zeros_like(w) for _ in range(world)]
+    torch.distributed.all_gather(ws, w)
+    assert torch.allclose(ws[0], ws[1], atol=1e-6)
+    # rows 1,2,5 updated; row 5 got grad from both ranks (summed)
+    torch.manual_seed(7)
+    ref = ParallaxEmbedding(30, 4)
+    assert not torch.allclose(w[5], ref.weight[5])
+    assert not torch.allclose(w[1], ref.weight[1])
+    assert not torch.allclose(w[2], ref.weight[2])
+    assert torch.allclose(w[9], ref.weight[9])
+    eng.close()
+
+
+def _body_sparse_ps(rank, world):
+    import torch.nn as nn
+    from parallax_amd import optim
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.sparse_engine import SparsePSEngine
+    from parallax_amd.ops.embedding import PartitionedEmbedding
+    from parallax_amd.partitions import FixedSizePartitioner
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    torch.manual_seed(11)
+    model = nn.Module()
+    model.emb = PartitionedEmbedding(40, 4, FixedSizePartitioner(4))
+    opt = optim.Adagrad(model.parameters(), lr=0.5)
+    infos = classify_params(model)
+    eng = SparsePSEngine(comm, infos, opt, replicate_variables=True)
+    # shards round-robin over 2 ranks
+    owners = [eng.owner[id(gi.param)] for gi in eng.infos]
+    assert owners == [0, 1, 0, 1]
+    eng.prepare_step()
+    ids = torch.tensor([0, 11 + rank, 25])  # shards 0,1,2; rank-varying row
+    out = model.emb(ids)
+    out.sum().backward()
+    eng.reduce_and_apply()
+    # mirrors must be consistent across ranks after the refresh
+    full = model.emb.full_weight()
+    fulls = [torch.zeros_like(full) for _ in range(world)]
+    torch.distributed.all_gather(fulls, full)
+    assert torch.allclose(fulls[0], fulls[1], atol=1e-6), \
+        (fulls[0] - fulls[1]).abs().max()
+    # updated rows: 0 (both ranks), 11, 12, 25
+    torch.manual_seed(11)
+    ref = PartitionedEmbedding(40, 4, FixedSizePartitioner(4))
+    ref_full = ref.full_weight()
+    for row in (0, 11, 12, 25):
+        assert not torch.allclose(full[row], ref_full[row]), row
+    for row in (3, 20, 39):
+        assert torch.allclose(full[row], ref_full[row]), row
+    eng.close()
+
+
+def _body_parallel_run_hybrid(rank, world):
+    """Full parallel_run worker path with a mixed dense+sparse model in
+    HYBRID mode; checks loss decreases and models stay in sync."""
+    import torch.nn as nn
+    import parallax_amd
+    from parallax_amd import optim
+    from parallax_amd.config import ParallaxConfig
+    from parallax_amd.session import TrainGraph
+    from parallax_amd.ops.embedding import PartitionedEmbedding
+    from parallax_amd.partitions import FixedSizePartitioner
+
+    def build():
+        torch.manual_seed(21 + rank)  # broadcast must fix divergent init
+        model = nn.Module()
+        model.emb = PartitionedEmbedding(40, 8, FixedSizePartitioner(4))
+        model.fc = nn.Linear(8, 2)
+        opt = optim.Adagrad(model.parameters(), lr=0.3)
+
+        def step_fn(m, batch):
+            ids, y = batch
+            out = m.fc(m.emb(ids))
+            loss = ((out - y) ** 2).mean()
+            return loss, {}
+
+        return TrainGraph(model, opt, step_fn)
+
+    sess, num_workers, worker_id, n_rep = parallax_amd.parallel_run(
+        build, None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    assert num_workers == world and worker_id == rank
+    torch.manual_seed(100 + rank)
+    ids = torch.randint(0, 40, (16,))
+    y = torch.randn(16, 2) * 0.1
+    first = None
+    for i in range(20):
+        out = sess.run((ids, y))
+        if first is None:
+            first = float(out['loss'])
+    assert float(out['loss']) < first
+    # dense params must be identical across ranks (same broadcast init +
+    # same averaged grads)
+    w = sess.graph.model.fc.weight.detach()
+    ws = [torch.zeros_like(w) for _ in range(world)]
+    torch.distributed.all_gather(ws, w)
+    assert torch.allclose(ws[0], ws[1], atol=1e-5)
+    # sparse mirrors consistent
+    full = sess.graph.model.emb.full_weight()
+    fs = [torch.zeros_like(full) for _ in range(world)]
+    torch.distributed.all_gather(fs, full)
+    assert torch.allclose(fs[0], fs[1], atol=1e-5)
+    sess.close()
+
+
+def _body_comm_collectives(rank, world):
+    from parallax_amd.parallel.comm import CommContext
+    comm = CommContext()
+    # all_gather_v with different lengths
+    t = torch.arange(3 + rank, dtype=torch.float32).reshape(-1, 1)
+    outs = comm.all_gather_v(t)
+    assert [o.shape[0] for o in outs] == [3, 4]
+    assert outs[1][3, 0] == 3.0
+    # all_to_all_v
+    sends = [torch.full((rank + 1, 2), float(10 * rank + d))
+             for d in range(world)]
+    recv = comm.all_to_all_v(sends)
+    # from rank r we receive (r+1) rows filled with 10*r + my_rank
+    for r in range(world):
+        assert recv[r].shape == (r + 1, 2)
+        assert torch.all(recv[r] == 10 * r + rank)
+
+
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize('body', [
+    _body_comm_collectives,
+    _body_allreduce_engine,
+    _body_sparse_allgather,
+    _body_sparse_ps,
+    _body_parallel_run_hybrid,
+], ids=['collectives', 'dense_allreduce', 'sparse_allgather', 'sparse_ps',
+        'parallel_run_hybrid'])
+def test_multiproc(body):
+    _spawn(body, get_free_port())

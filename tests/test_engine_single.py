@@ -1,0 +1,178 @@
+"""World-size-1 engine/session tests on CPU (gloo) — the full worker path
+of parallel_run without a spawner."""
+
+import os
+
+import pytest
+import torch
+import torch.nn as nn
+
+import parallax_amd
+from parallax_amd.config import ParallaxConfig, CheckPointConfig
+from parallax_amd.models.simple import build_simple_graph
+from parallax_amd.models.lm1b import build_lm1b_graph, synthetic_batch
+from parallax_amd.runner import _run_worker
+from parallax_amd.session import TrainGraph
+from parallax_amd import optim
+from parallax_amd.ops.embedding import ParallaxEmbedding
+
+
+def _worker_env(monkeypatch, free_port, world=1, rank=0):
+    monkeypatch.setenv('WORLD_SIZE', str(world))
+    monkeypatch.setenv('RANK', str(rank))
+    monkeypatch.setenv('MASTER_ADDR', '127.0.0.1')
+    monkeypatch.setenv('MASTER_PORT', str(free_port))
+
+
+@pytest.fixture(autouse=True)
+def _destroy_pg():
+    yield
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+def test_simple_loss_decreases(monkeypatch, free_port):
+    """The reference 'simple' example contract: loss decreases
+    (simple_driver.py)."""
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    sess, num_workers, worker_id, n_rep = parallax_amd.parallel_run(
+        lambda: build_simple_graph(lr=0.05), resource_info=None,
+        parallax_config=ParallaxConfig(run_option='HYBRID'))
+    assert (num_workers, worker_id, n_rep) == (1, 0, 1)
+    x = torch.randn(16, 2)
+    y = x @ torch.tensor([[2.0], [-1.0]]) + 0.5
+    losses = []
+    for i in range(60):
+        out = sess.run((x, y))
+        losses.append(float(out['loss']))
+    assert out['global_step'] == 60
+    assert losses[-1] < 0.05 * losses[0]
+    sess.close()
+
+
+def test_hybrid_sparse_model_single(monkeypatch, free_port):
+    """An embedding+linear model under HYBRID: sparse rows update through
+    the PS engine, dense through the optimizer."""
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+
+    def build():
+        model = nn.Module()
+        model.emb = ParallaxEmbedding(50, 8)
+        model.fc = nn.Linear(8, 2)
+        opt = optim.Adagrad(
+            list(model.parameters()), lr=0.3)
+
+        def step_fn(m, batch):
+            ids, y = batch
+            out = m.fc(m.emb(ids)).mean(dim=0, keepdim=True)
+            loss = ((out - y) ** 2).sum()
+            return loss, {}
+
+        return TrainGraph(model, opt, step_fn)
+
+    sess, *_ = parallax_amd.parallel_run(
+        build, None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    m = sess.graph.model
+    w_before = m.emb.weight.detach().clone()
+    ids = torch.tensor([1, 5, 5, 9])
+    y = torch.tensor([[1.0, -1.0]])
+    for _ in range(3):
+        sess.run((ids, y))
+    w_after = m.emb.weight.detach()
+    touched = torch.tensor([1, 5, 9])
+    untouched = torch.tensor([0, 2, 3, 49])
+    assert not torch.allclose(w_before[touched], w_after[touched])
+    assert torch.allclose(w_before[untouched], w_after[untouched])
+    assert sess.engine.mode == 'HYBRID'
+    assert sess.engine.sparse_engine is not None
+    sess.close()
+
+
+def test_degenerate_hybrid_falls_back_to_mpi(monkeypatch, free_port):
+    """HYBRID + all-dense model -> MPI mode (reference runner.py:93-94)."""
+    _worker_env(monkeypatch, free_port)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: build_simple_graph(), None,
+        parallax_config=ParallaxConfig(run_option='HYBRID'))
+    assert sess.engine.mode == 'MPI'
+    sess.close()
+
+
+def test_degenerate_hybrid_falls_back_to_ps(monkeypatch, free_port):
+    """HYBRID + all-sparse model -> PS mode (reference runner.py:103-104)."""
+    _worker_env(monkeypatch, free_port)
+
+    def build():
+        model = nn.Module()
+        model.emb = ParallaxEmbedding(20, 4)
+        opt = optim.SGD(model.parameters(), lr=0.1)
+
+        def step_fn(m, batch):
+            return m.emb(batch).sum(), {}
+
+        return TrainGraph(model, opt, step_fn)
+
+    sess, *_ = parallax_amd.parallel_run(
+        build, None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    assert sess.engine.mode == 'PS'
+    sess.close()
+
+
+def test_lm1b_tiny_loss_decreases(monkeypatch, free_port):
+    """Tiny-config LM1B: the full hybrid stack (partitioned embeddings,
+    sampled softmax, Adagrad, clipping) learns on a repeated batch."""
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: build_lm1b_graph(batch_size=4, num_steps=3, vocab_size=200,
+                                 emb_dim=16, state_size=32, proj_size=16,
+                                 num_sampled=32, min_partitions=4, lr=0.4),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    assert sess.engine.mode == 'HYBRID'
+    x = torch.randint(0, 200, (4, 3))
+    y = torch.randint(0, 200, (4, 3))
+    first = None
+    state = None
+    for i in range(30):
+        out = sess.run((x, y, state))
+        if first is None:
+            first = float(out['loss'])
+    assert float(out['loss']) < first
+    sess.close()
+
+
+def test_checkpoint_roundtrip(monkeypatch, free_port, tmp_path):
+    """Chief-only save on step cadence; restore resumes global_step and
+    weights (reference §5.4 contract)."""
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    ck = CheckPointConfig(ckpt_dir=str(tmp_path / 'ckpt'),
+                          save_ckpt_steps=5)
+    cfg = ParallaxConfig(run_option='HYBRID', ckpt_config=ck)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: build_simple_graph(lr=0.05), None, parallax_config=cfg)
+    x = torch.randn(8, 2)
+    y = torch.randn(8, 1)
+    for _ in range(7):
+        sess.run((x, y))
+    w_saved_run = sess.graph.model.w.detach().clone()
+    sess.close()
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+    # checkpoint exists and is keyed by user-visible names
+    files = os.listdir(str(tmp_path / 'ckpt'))
+    assert 'model.ckpt-5.pt' in files and 'checkpoint' in files
+    payload = torch.load(str(tmp_path / 'ckpt' / 'model.ckpt-5.pt'),
+                         weights_only=False)
+    assert set(payload['model'].keys()) == {'w', 'b'}
+
+    # new session restores at step 5
+    monkeypatch.setenv('MASTER_PORT', str(free_port + 1))
+    sess2, *_ = parallax_amd.parallel_run(
+        lambda: build_simple_graph(lr=0.05), None, parallax_config=cfg)
+    assert sess2.global_step == 5
+    sess2.close()

@@ -1,0 +1,295 @@
+"""Numerics tests for the ops layer on CPU (the same checks re-run on GPU in
+test_ops_gpu.py against the HIP kernels)."""
+
+import math
+
+import pytest
+import torch
+import torch.nn as nn
+
+from parallax_amd.ops.sparse import unique_sum_rows, rows_gather
+from parallax_amd.ops.embedding import (ParallaxEmbedding,
+                                        PartitionedEmbedding,
+                                        embedding_lookup)
+from parallax_amd.ops.lstm import lstm_cell, ParallaxLSTM
+from parallax_amd.ops.sampled_softmax import (sampled_softmax_loss,
+                                              log_uniform_candidates)
+from parallax_amd.ops.softmax_ce import softmax_cross_entropy
+from parallax_amd.ops.misc import clip_by_global_norm, global_norm, ema_update
+from parallax_amd.partitions import FixedSizePartitioner
+from parallax_amd import optim
+
+torch.manual_seed(0)
+
+
+# --- sparse reduce (contract N2) -------------------------------------------
+
+def test_unique_sum_rows_basic():
+    ids = torch.tensor([3, 1, 3, 7, 1, 1])
+    vals = torch.arange(12, dtype=torch.float32).reshape(6, 2)
+    u, s, c = unique_sum_rows(ids, vals)
+    assert u.tolist() == [1, 3, 7]
+    # id1: rows 1,4,5 -> [2+8+10, 3+9+11]; id3: rows 0,2; id7: row 3
+    assert s.tolist() == [[20.0, 23.0], [4.0, 6.0], [6.0, 7.0]]
+    assert c.tolist() == [3, 2, 1]
+
+
+def test_unique_sum_rows_average():
+    ids = torch.tensor([0, 0, 5])
+    vals = torch.tensor([[2.0], [4.0], [9.0]])
+    u, s, c = unique_sum_rows(ids, vals, average=True)
+    assert s.tolist() == [[3.0], [9.0]]
+
+
+def test_unique_sum_rows_empty():
+    ids = torch.zeros(0, dtype=torch.long)
+    vals = torch.zeros(0, 4)
+    u, s, c = unique_sum_rows(ids, vals)
+    assert u.numel() == 0 and s.shape == (0, 4)
+
+
+def test_rows_gather():
+    w = torch.randn(10, 4)
+    ids = torch.tensor([[1, 2], [9, 1]])
+    out = rows_gather(w, ids)
+    assert out.shape == (2, 2, 4)
+    assert torch.equal(out[1, 0], w[9])
+
+
+# --- embedding (indexed-slices grads) ---------------------------------------
+
+def test_embedding_sparse_grad():
+    emb = ParallaxEmbedding(20, 4)
+    ids = torch.tensor([1, 3, 1])
+    out = emb(ids)
+    out.sum().backward()
+    g = emb.weight.grad
+    assert g.layout == torch.sparse_coo
+    d = g.to_dense()
+    assert torch.allclose(d[1], torch.full((4,), 2.0))
+    assert torch.allclose(d[3], torch.ones(4))
+    assert d[0].abs().sum() == 0
+
+
+def test_partitioned_embedding_matches_full():
+    torch.manual_seed(1)
+    pe = PartitionedEmbedding(17, 6, FixedSizePartitioner(4))
+    full = pe.full_weight()
+    ids = torch.tensor([0, 4, 16, 5, 4])
+    out = pe(ids)
+    ref = full[ids]
+    assert torch.allclose(out, ref)
+    # grads: each shard gets an indexed-slices grad in LOCAL row ids
+    out.sum().backward()
+    shard_of = pe.shard_of(ids)
+    for k, p in enumerate(pe.parts):
+        if (shard_of == k).any():
+            assert p.grad is not None
+            assert p.grad.layout == torch.sparse_coo
+        # local indices within range
+        if p.grad is not None and p.grad._nnz() > 0:
+            assert int(p.grad._indices().max()) < p.shape[0]
+
+
+def test_partitioned_embedding_batch_shape():
+    pe = PartitionedEmbedding(30, 5, FixedSizePartitioner(3))
+    ids = torch.randint(0, 30, (2, 7))
+    assert pe(ids).shape == (2, 7, 5)
+
+
+# --- lstm -------------------------------------------------------------------
+
+def test_lstm_cell_matches_manual():
+    torch.manual_seed(2)
+    B, I, H = 4, 3, 5
+    x = torch.randn(B, I)
+    h = torch.randn(B, H)
+    c = torch.randn(B, H)
+    w = torch.randn(I + H, 4 * H, requires_grad=True)
+    b = torch.randn(4 * H, requires_grad=True)
+    h2, c2 = lstm_cell(x, h, c, w, b, forget_bias=1.0)
+    gates = torch.cat([x, h], -1) @ w + b
+    i_, g_, f_, o_ = torch.chunk(gates, 4, -1)
+    c_ref = torch.sigmoid(f_ + 1.0) * c + torch.sigmoid(i_) * torch.tanh(g_)
+    h_ref = torch.sigmoid(o_) * torch.tanh(c_ref)
+    assert torch.allclose(h2, h_ref, atol=1e-6)
+    assert torch.allclose(c2, c_ref, atol=1e-6)
+
+
+def test_lstm_module_shapes_and_backward():
+    m = ParallaxLSTM(8, 16, proj_size=4)
+    xs = torch.randn(2, 5, 8)
+    out, (h, c) = m(xs)
+    assert out.shape == (2, 5, 4)
+    assert h.shape == (2, 4) and c.shape == (2, 16)
+    out.sum().backward()
+    assert m.w.grad is not None and m.w_proj.grad is not None
+
+
+# --- softmax CE -------------------------------------------------------------
+
+def test_softmax_ce_matches_torch():
+    logits = torch.randn(6, 11)
+    labels = torch.randint(0, 11, (6,))
+    ours = softmax_cross_entropy(logits, labels, reduction='none')
+    ref = torch.nn.functional.cross_entropy(logits, labels, reduction='none')
+    assert torch.allclose(ours, ref, atol=1e-6)
+
+
+# --- sampled softmax ---------------------------------------------------------
+
+def test_log_uniform_sampler_distribution():
+    torch.manual_seed(3)
+    ids, expected = log_uniform_candidates(20000, 1000, 'cpu')
+    assert ids.min() >= 0 and ids.max() < 1000
+    # Zipf: id 0 should be sampled far more than id 500
+    c0 = (ids == 0).sum().item()
+    c500 = (ids == 500).sum().item()
+    assert c0 > 10 * max(c500, 1)
+
+
+def test_sampled_softmax_loss_backward_sparse():
+    torch.manual_seed(4)
+    V, D, N, S = 50, 8, 6, 16
+    w = nn.Parameter(torch.randn(V, D) * 0.1)
+    b = nn.Parameter(torch.zeros(V))
+    h = torch.randn(N, D, requires_grad=True)
+    labels = torch.randint(0, V, (N,))
+    losses = sampled_softmax_loss(w, b, labels, h, S, V)
+    assert losses.shape == (N,)
+    losses.mean().backward()
+    assert w.grad is not None and w.grad.layout == torch.sparse_coo
+    assert b.grad is not None and b.grad.layout == torch.sparse_coo
+    assert h.grad is not None
+    # touched rows only
+    touched = set(w.grad._indices().flatten().tolist())
+    assert len(touched) <= S + N
+
+
+# --- misc -------------------------------------------------------------------
+
+def test_clip_by_global_norm_matches_torch():
+    ts = [torch.randn(5, 5) for _ in range(3)]
+    refs = [t.clone() for t in ts]
+    norm = clip_by_global_norm(ts, 1.0)
+    ref_norm = torch.sqrt(sum(t.pow(2).sum() for t in refs))
+    assert torch.allclose(norm, ref_norm, atol=1e-5)
+    scale = 1.0 / max(ref_norm.item(), 1.0)
+    for t, r in zip(ts, refs):
+        assert torch.allclose(t, r * scale, atol=1e-5)
+    # below the clip: unchanged
+    small = [torch.full((2,), 0.01)]
+    clip_by_global_norm(small, 10.0)
+    assert torch.allclose(small[0], torch.full((2,), 0.01))
+
+
+def test_ema_update():
+    s = [torch.zeros(3)]
+    v = [torch.ones(3)]
+    ema_update(s, v, 0.9)
+    assert torch.allclose(s[0], torch.full((3,), 0.1))
+
+
+# --- optimizers (the update-op vocabulary) -----------------------------------
+
+def _quad_setup(opt_cls, **kw):
+    p = nn.Parameter(torch.tensor([1.0, -2.0, 3.0]))
+    opt = opt_cls([p], **kw)
+    return p, opt
+
+
+def test_sgd_dense():
+    p, opt = _quad_setup(optim.SGD, lr=0.1)
+    p.grad = torch.tensor([1.0, 1.0, 1.0])
+    opt.step()
+    assert torch.allclose(p.data, torch.tensor([0.9, -2.1, 2.9]))
+
+
+def test_momentum_nesterov_matches_torch():
+    torch.manual_seed(5)
+    p1 = nn.Parameter(torch.randn(10))
+    p2 = nn.Parameter(p1.detach().clone())
+    ours = optim.SGD([p1], lr=0.1, momentum=0.9, nesterov=True)
+    ref = torch.optim.SGD([p2], lr=0.1, momentum=0.9, nesterov=True)
+    for _ in range(5):
+        g = torch.randn(10)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        ours.step()
+        ref.step()
+    assert torch.allclose(p1.data, p2.data, atol=1e-6)
+
+
+def test_adagrad_dense_matches_torch():
+    p1 = nn.Parameter(torch.ones(4))
+    p2 = nn.Parameter(torch.ones(4))
+    ours = optim.Adagrad([p1], lr=0.5, initial_accumulator_value=0.1,
+                         eps=1e-10)
+    ref = torch.optim.Adagrad([p2], lr=0.5, initial_accumulator_value=0.1,
+                              eps=1e-10)
+    for i in range(4):
+        g = torch.full((4,), 0.5 + 0.1 * i)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        ours.step()
+        ref.step()
+    # torch adagrad uses lr/(sqrt+eps) with lr decay 0 — same formula
+    assert torch.allclose(p1.data, p2.data, atol=1e-5)
+
+
+def test_adagrad_sparse_rows_equals_dense():
+    """sparse_row_step on touched rows == dense step with a scattered grad
+    (the SparseApplyAdagrad contract)."""
+    w1 = nn.Parameter(torch.ones(6, 3))
+    w2 = nn.Parameter(torch.ones(6, 3))
+    o1 = optim.Adagrad([w1], lr=0.3)
+    o2 = optim.Adagrad([w2], lr=0.3)
+    ids = torch.tensor([1, 4])
+    rows = torch.tensor([[1.0, 2.0, 3.0], [0.5, 0.5, 0.5]])
+    o1.sparse_row_step(w1, ids, rows)
+    dense = torch.zeros(6, 3)
+    dense[1] = rows[0]
+    dense[4] = rows[1]
+    w2.grad = dense
+    o2.step()
+    # rows 1 and 4 must match; but dense adagrad also bumps accumulator of
+    # untouched rows by 0 -> identical
+    assert torch.allclose(w1.data, w2.data, atol=1e-6)
+
+
+def test_adam_dense_matches_torch():
+    p1 = nn.Parameter(torch.ones(4))
+    p2 = nn.Parameter(torch.ones(4))
+    ours = optim.Adam([p1], lr=0.01)
+    ref = torch.optim.Adam([p2], lr=0.01)
+    for i in range(5):
+        g = torch.full((4,), 0.3) * (i + 1)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        ours.step()
+        ref.step()
+    assert torch.allclose(p1.data, p2.data, atol=1e-5)
+
+
+def test_rmsprop_step_runs():
+    p, opt = _quad_setup(optim.RMSProp, lr=0.01, decay=0.9, momentum=0.9,
+                         eps=1.0)
+    for _ in range(3):
+        p.grad = torch.ones(3)
+        opt.step()
+    assert torch.isfinite(p.data).all()
+
+
+def test_optimizer_handles_bare_sparse_grad():
+    """Without an engine, a sparse COO grad must still be applied
+    (coalesced) by the optimizer itself."""
+    emb = ParallaxEmbedding(10, 4)
+    opt = optim.Adagrad(emb.parameters(), lr=0.5)
+    before = emb.weight.detach().clone()
+    out = emb(torch.tensor([2, 2, 7]))
+    out.sum().backward()
+    opt.step()
+    after = emb.weight.detach()
+    assert not torch.allclose(before[2], after[2])
+    assert not torch.allclose(before[7], after[7])
+    assert torch.allclose(before[0], after[0])
