@@ -25,10 +25,21 @@ def _try_load_ext():
     global _EXT, _EXT_ERR
     if _EXT is not None or _EXT_ERR is not None:
         return _EXT
+    import glob
+    import importlib.util
+    lib_dir = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), '_lib')
+    sos = glob.glob(os.path.join(lib_dir, 'parallax_amd_C*.so'))
+    if not sos:
+        _EXT_ERR = ImportError('no parallax_amd_C*.so under %s' % lib_dir)
+        return None
     try:
-        from parallax_amd import _C  # built in-tree by ops/build.py
-        _EXT = _C
-    except ImportError as e:  # extension not built
+        spec = importlib.util.spec_from_file_location('parallax_amd_C',
+                                                      sos[0])
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _EXT = mod
+    except Exception as e:  # pragma: no cover - load failure on GPU box
         _EXT_ERR = e
         _EXT = None
     return _EXT

@@ -1,0 +1,90 @@
+// parallax_amd native extension: pybind bindings for the hand-written
+// CDNA4 (gfx950) kernels.  Built in-tree by parallax_amd/ops/build.py with
+// PYTORCH_ROCM_ARCH=gfx950.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+// sparse_kernels.hip
+at::Tensor rows_gather(const at::Tensor& weight, const at::Tensor& ids);
+at::Tensor segment_sum_rows(const at::Tensor& values, const at::Tensor& order,
+                            const at::Tensor& seg_starts, bool average);
+
+// optim_kernels.hip
+void multi_tensor_sgd(std::vector<at::Tensor> params,
+                      std::vector<at::Tensor> grads, double lr, double wd);
+void multi_tensor_momentum(std::vector<at::Tensor> params,
+                           std::vector<at::Tensor> grads,
+                           std::vector<at::Tensor> bufs, double lr,
+                           double mom, double wd, bool nesterov);
+void multi_tensor_adagrad(std::vector<at::Tensor> params,
+                          std::vector<at::Tensor> grads,
+                          std::vector<at::Tensor> accums, double lr,
+                          double eps);
+void multi_tensor_adam(std::vector<at::Tensor> params,
+                       std::vector<at::Tensor> grads,
+                       std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                       std::vector<double> steps, double lr, double b1,
+                       double b2, double eps, double wd);
+void multi_tensor_rmsprop(std::vector<at::Tensor> params,
+                          std::vector<at::Tensor> grads,
+                          std::vector<at::Tensor> sqs,
+                          std::vector<at::Tensor> bufs, double lr,
+                          double decay, double mom, double eps);
+void multi_tensor_ema(std::vector<at::Tensor> shadows,
+                      std::vector<at::Tensor> values, double decay);
+at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors);
+void multi_tensor_scale_by_norm(std::vector<at::Tensor> tensors,
+                                at::Tensor norm, double clip);
+void sparse_adagrad(at::Tensor param, at::Tensor accum, at::Tensor ids,
+                    at::Tensor grad_rows, double lr, double eps);
+void sparse_sgd(at::Tensor param, at::Tensor ids, at::Tensor grad_rows,
+                double lr);
+
+// layernorm.hip
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
+                                      const at::Tensor& w,
+                                      const at::Tensor& b, double eps);
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
+                                      const at::Tensor& x,
+                                      const at::Tensor& w,
+                                      const at::Tensor& mean,
+                                      const at::Tensor& rstd);
+
+// softmax_ce.hip
+std::vector<at::Tensor> softmax_ce_fwd(const at::Tensor& logits,
+                                       const at::Tensor& labels);
+at::Tensor softmax_ce_bwd(const at::Tensor& logits, const at::Tensor& labels,
+                          const at::Tensor& lse, const at::Tensor& dloss);
+
+// lstm_cell.hip
+std::vector<at::Tensor> lstm_gates_fwd(const at::Tensor& gates,
+                                       const at::Tensor& c_prev,
+                                       double forget_bias);
+std::vector<at::Tensor> lstm_gates_bwd(
+    const at::Tensor& dh, const at::Tensor& dc_in, const at::Tensor& c_prev,
+    const at::Tensor& c_out, const at::Tensor& i_s, const at::Tensor& g_t,
+    const at::Tensor& f_s, const at::Tensor& o_s);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rows_gather", &rows_gather, "embedding row gather (HIP)");
+  m.def("segment_sum_rows", &segment_sum_rows,
+        "indexed-slices segmented row sum (HIP)");
+  m.def("multi_tensor_sgd", &multi_tensor_sgd);
+  m.def("multi_tensor_momentum", &multi_tensor_momentum);
+  m.def("multi_tensor_adagrad", &multi_tensor_adagrad);
+  m.def("multi_tensor_adam", &multi_tensor_adam);
+  m.def("multi_tensor_rmsprop", &multi_tensor_rmsprop);
+  m.def("multi_tensor_ema", &multi_tensor_ema);
+  m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
+  m.def("multi_tensor_scale_by_norm", &multi_tensor_scale_by_norm);
+  m.def("sparse_adagrad", &sparse_adagrad);
+  m.def("sparse_sgd", &sparse_sgd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("softmax_ce_fwd", &softmax_ce_fwd);
+  m.def("softmax_ce_bwd", &softmax_ce_bwd);
+  m.def("lstm_gates_fwd", &lstm_gates_fwd);
+  m.def("lstm_gates_bwd", &lstm_gates_bwd);
+}

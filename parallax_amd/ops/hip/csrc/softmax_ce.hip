@@ -1,0 +1,97 @@
+// Fused softmax cross-entropy with integer labels (the
+// ResNet/NMT loss, and the [B, 1+8192] CE inside sampled softmax).
+//
+// Forward: one block per row; ONE sweep computing row-max then sum-exp
+// (two block reductions), loss = lse - logit[label]; saves lse.
+// Backward: flat elementwise (softmax - onehot) * dloss.
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+template <typename T>
+__global__ void softmax_ce_fwd_kernel(const T* __restrict__ logits,
+                                      const int64_t* __restrict__ labels,
+                                      float* __restrict__ loss,
+                                      float* __restrict__ lse_out, long n,
+                                      long c) {
+  __shared__ float scratch[PX_BLOCK / WAVE];
+  for (long row = blockIdx.x; row < n; row += gridDim.x) {
+    const T* lr = logits + row * c;
+    float mx = -INFINITY;
+    for (long j = threadIdx.x; j < c; j += blockDim.x) {
+      mx = fmaxf(mx, to_f32<T>(lr[j]));
+    }
+    mx = block_reduce_max(mx, scratch);
+    float se = 0.f;
+    for (long j = threadIdx.x; j < c; j += blockDim.x) {
+      se += __expf(to_f32<T>(lr[j]) - mx);
+    }
+    se = block_reduce_sum(se, scratch);
+    if (threadIdx.x == 0) {
+      const float lse = mx + __logf(se);
+      lse_out[row] = lse;
+      loss[row] = lse - to_f32<T>(lr[labels[row]]);
+    }
+  }
+}
+
+std::vector<at::Tensor> softmax_ce_fwd(const at::Tensor& logits,
+                                       const at::Tensor& labels) {
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 &&
+              logits.is_contiguous());
+  TORCH_CHECK(labels.scalar_type() == at::kLong);
+  const long n = logits.size(0), c = logits.size(1);
+  auto loss = at::empty({n}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({n}, logits.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = (int)(n < PX_MAX_GRID ? (n > 0 ? n : 1) : PX_MAX_GRID);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, logits.scalar_type(), "softmax_ce_fwd", [&] {
+        softmax_ce_fwd_kernel<scalar_t><<<grid, PX_BLOCK, 0, stream>>>(
+            reinterpret_cast<const scalar_t*>(logits.data_ptr()),
+            labels.data_ptr<int64_t>(), loss.data_ptr<float>(),
+            lse.data_ptr<float>(), n, c);
+        HIP_CHECK_KERNEL();
+      });
+  return {loss.to(logits.dtype()), lse};
+}
+
+template <typename T>
+__global__ void softmax_ce_bwd_kernel(const T* __restrict__ logits,
+                                      const int64_t* __restrict__ labels,
+                                      const float* __restrict__ lse,
+                                      const T* __restrict__ dloss,
+                                      T* __restrict__ dlogits, long n,
+                                      long c) {
+  const long total = n * c;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long row = idx / c;
+    const long j = idx - row * c;
+    float p = __expf(to_f32<T>(logits[idx]) - lse[row]);
+    float g = p - (labels[row] == j ? 1.f : 0.f);
+    dlogits[idx] = from_f32<T>(g * to_f32<T>(dloss[row]));
+  }
+}
+
+at::Tensor softmax_ce_bwd(const at::Tensor& logits, const at::Tensor& labels,
+                          const at::Tensor& lse, const at::Tensor& dloss) {
+  const long n = logits.size(0), c = logits.size(1);
+  auto dlogits = at::empty_like(logits);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto dl = dloss.to(logits.dtype()).contiguous();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, logits.scalar_type(), "softmax_ce_bwd", [&] {
+        softmax_ce_bwd_kernel<scalar_t>
+            <<<px_grid(n * c), PX_BLOCK, 0, stream>>>(
+                reinterpret_cast<const scalar_t*>(logits.data_ptr()),
+                labels.data_ptr<int64_t>(), lse.data_ptr<float>(),
+                reinterpret_cast<const scalar_t*>(dl.data_ptr()),
+                reinterpret_cast<scalar_t*>(dlogits.data_ptr()), n, c);
+        HIP_CHECK_KERNEL();
+      });
+  return dlogits;
+}
