@@ -1,0 +1,151 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — the driver contract.
+
+``python bench.py --gpus N --steps K --warmup W [--model lm1b|resnet50]``
+
+Measures the reference's headline metrics (BASELINE.json / BASELINE.md) on
+MI355X with synthetic data and random-init weights:
+
+  * lm1b (default): LM1B LSTM LM, vocab 793,470, emb 512, state 2048 proj
+    512, sampled softmax 8192, batch 128 x 20 steps per GPU, Adagrad,
+    HYBRID mode (sparse tables through the in-process PS, dense LSTM
+    params through RCCL all-reduce).  Metric: words/sec (whole job).
+    Reference anchor: 277k wps on 48 TITAN Xp (Parallax-HYBRID).
+  * resnet50: ResNet-50 v1, 1001 classes, synthetic 224x224, batch 64 per
+    GPU, bf16 autocast, momentum SGD, AR (MPI) mode.  Metric: images/sec.
+    Reference anchor: 7,520 img/s on 48 TITAN Xp.
+
+Under the driver's multi-GPU launch (torch.distributed.run, one rank per
+GPU over RCCL) RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* come from the env.
+W untimed warmup steps, then EXACTLY K timed steps bracketed by
+barrier + torch.cuda.synchronize on both sides; elapsed = MAX over ranks;
+rank 0 prints ONE JSON line.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=30)
+    p.add_argument('--warmup', type=int, default=10)
+    p.add_argument('--model', choices=['lm1b', 'resnet50'], default='lm1b')
+    p.add_argument('--run-option', default='HYBRID')
+    p.add_argument('--batch-size', type=int, default=None)
+    p.add_argument('--partitions', type=int, default=None)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    n = int(os.environ.get('WORLD_SIZE', args.gpus))
+    os.environ.setdefault('WORLD_SIZE', str(n))
+    os.environ.setdefault('RANK', '0')
+    os.environ.setdefault('LOCAL_RANK', os.environ.get('RANK', '0'))
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    os.environ.setdefault('MASTER_PORT', '29517')
+    if args.partitions is not None:
+        os.environ['PARALLAX_PARTITIONS'] = str(args.partitions)
+
+    import parallax_amd
+    from parallax_amd.config import ParallaxConfig
+    cfg = ParallaxConfig(run_option=args.run_option,
+                         search_partitions=False)
+
+    on_gpu = torch.cuda.is_available()
+    if args.model == 'lm1b':
+        from parallax_amd.models import lm1b as M
+        batch = args.batch_size or M.BATCH_SIZE          # 128
+        seq = M.NUM_STEPS                                # 20
+        vocab = M.VOCAB_SIZE                             # 793470
+        build = lambda: M.build_lm1b_graph(
+            batch_size=batch, num_steps=seq, min_partitions=32)
+        metric, unit = 'LM1B words/sec', 'words/s'
+        per_step_items = batch * seq                     # words per rank
+        baseline = 277000.0
+        config_json = {'model': 'lm1b', 'vocab': vocab,
+                       'global_batch': batch * n, 'seq_len': seq,
+                       'num_sampled': M.NUM_SAMPLED,
+                       'parallelism': 'hybrid dp%d (dense AR + sparse PS)'
+                                      % n}
+        dtype = 'fp32'
+    else:
+        from parallax_amd.models import resnet as M
+        batch = args.batch_size or 64
+        build = lambda: M.build_resnet_graph(depth=50, batch_size=batch,
+                                             use_bf16=on_gpu,
+                                             channels_last=on_gpu)
+        metric, unit = 'ResNet-50 images/sec', 'images/s'
+        per_step_items = batch
+        baseline = 7520.0
+        config_json = {'model': 'resnet50', 'global_batch': batch * n,
+                       'image_size': 224, 'parallelism': 'dp%d (AR)' % n}
+        dtype = 'bf16' if on_gpu else 'fp32'
+
+    sess, num_workers, worker_id, _ = parallax_amd.parallel_run(
+        build, None, parallax_config=cfg)
+    device = sess.comm.device
+
+    def make_batch(state=None):
+        if args.model == 'lm1b':
+            return M.synthetic_batch(batch, seq, device=device, state=state)
+        return M.synthetic_batch(batch, device=device)
+
+    # warmup (untimed)
+    state = None
+    for _ in range(args.warmup):
+        out = sess.run(make_batch(state))
+        state = out.get('state')
+
+    sess.comm.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        out = sess.run(make_batch(state))
+        state = out.get('state')
+    if on_gpu:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    sess.comm.barrier()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64,
+                           device=device if sess.comm.backend == 'nccl'
+                           else 'cpu')
+    if num_workers > 1:
+        torch.distributed.all_reduce(elapsed,
+                                     op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(elapsed.item())
+
+    if worker_id == 0:
+        value = per_step_items * num_workers * args.steps / elapsed
+        result = {
+            'metric': metric,
+            'value': round(value, 2),
+            'unit': unit,
+            'n_gpus': num_workers,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': round(elapsed / args.steps * 1000.0, 3),
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': round(value / baseline, 4),
+            'dtype': dtype,
+            'data': 'synthetic',
+            'config': config_json,
+        }
+        print(json.dumps(result))
+    sess.close()
+
+
+if __name__ == '__main__':
+    main()
