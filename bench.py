@@ -77,7 +77,7 @@ def main():
                        'num_sampled': M.NUM_SAMPLED,
                        'parallelism': 'hybrid dp%d (dense AR + sparse PS)'
                                       % n}
-        dtype = 'fp32'
+        dtype = 'bf16' if on_gpu else 'fp32'
     else:
         from parallax_amd.models import resnet as M
         batch = args.batch_size or 64

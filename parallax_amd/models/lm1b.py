@@ -71,7 +71,10 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
                      vocab_size=VOCAB_SIZE, lr=0.2, min_partitions=32,
                      emb_dim=EMB_DIM, state_size=STATE_SIZE,
                      proj_size=PROJ_SIZE, num_sampled=NUM_SAMPLED,
-                     max_grad_norm=10.0, dtype=torch.float32):
+                     max_grad_norm=10.0, dtype=torch.float32,
+                     use_bf16=True):
+    """use_bf16: run the GEMM path under bf16 autocast on GPU (fp32 master
+    weights, fp32 CE and optimizer math)."""
     model = LM1BModel(vocab_size, emb_dim, state_size, proj_size,
                       num_sampled, min_partitions, dtype=dtype)
     optimizer = optim.Adagrad(model.parameters(), lr=lr,
@@ -80,7 +83,11 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
 
     def step_fn(m, batch):
         x, y, state = batch
-        loss, new_state = m(x, y, state)
+        if use_bf16 and x.is_cuda:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                loss, new_state = m(x, y, state)
+        else:
+            loss, new_state = m(x, y, state)
         return loss, {'state': (new_state[0].detach(),
                                 new_state[1].detach())}
 

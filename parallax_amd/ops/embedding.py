@@ -96,14 +96,22 @@ class ParallaxEmbedding(nn.Module):
 
 
 class PartitionedEmbedding(nn.Module):
-    """Row-partitioned embedding: p shard Parameters ``part_0..part_{p-1}``
-    (reference fixed_size_partitioner shard naming, partitions.py:51;
-    checkpoint layout parity §5.4).
+    """Row-partitioned embedding with VIRTUAL shards.
 
-    Lookup routes each id to its shard (first ``V % p`` shards hold one
-    extra row) and re-assembles the output.  Each shard's gradient is its
-    own indexed-slices, so the sparse engine exchanges/updates shards
-    independently — that is what partition search varies.
+    Reference behavior: ``tf.fixed_size_partitioner`` splits the table into
+    p shard variables ``<name>/part_k`` placed on different PS tasks
+    (partitions.py:35-51); checkpoints keep the shard layout (§5.4).
+
+    MI355X-native design: 288 GB of HBM3E per GPU makes physical sharding
+    of storage pointless on one node — the table is ONE resident Parameter
+    (lookup = one gather kernel, gradient = one indexed-slices tensor) and
+    the partitioning is METADATA (`weight._parallax_partition_meta` =
+    (num_partitions, offsets)) consumed by
+      * the sparse engines (shard granularity of the exchange/update and of
+        partition search),
+      * the checkpoint layer (state_dict emits/accepts ``parts.k`` shard
+        slices, so the on-disk layout keeps reference shard structure and a
+        checkpoint written with p shards restores under any p').
     """
 
     _parallax_is_embedding = True
@@ -118,14 +126,11 @@ class PartitionedEmbedding(nn.Module):
         self.offsets = [o for o, _ in bounds]
         self.sizes = [s for _, s in bounds]
         scale = init_scale if init_scale is not None else embedding_dim ** -0.5
-        self.parts = nn.ParameterList()
-        for k, (off, size) in enumerate(bounds):
-            w = torch.empty(size, embedding_dim, dtype=dtype)
-            nn.init.uniform_(w, -scale, scale)
-            self.parts.append(nn.Parameter(w))
-        for k, p in enumerate(self.parts):
-            p._parallax_partition = (k, self.num_partitions,
-                                     self.offsets[k], num_embeddings)
+        w = torch.empty(num_embeddings, embedding_dim, dtype=dtype)
+        nn.init.uniform_(w, -scale, scale)
+        self.weight = nn.Parameter(w)
+        self.weight._parallax_partition_meta = (self.num_partitions,
+                                                tuple(self.offsets))
         self._offsets_t = None
 
     def _offsets_tensor(self, device):
@@ -140,23 +145,38 @@ class PartitionedEmbedding(nn.Module):
         return torch.searchsorted(offs, flat_ids, right=True) - 1
 
     def forward(self, ids):
-        flat = ids.reshape(-1)
-        shard_ids = self.shard_of(flat)
-        out = None
-        for k in range(self.num_partitions):
-            mask = shard_ids == k
-            if not torch.any(mask):
-                continue
-            pos = torch.nonzero(mask).flatten()
-            local = flat[pos] - self.offsets[k]
-            rows = embedding_lookup(self.parts[k], local)
-            if out is None:
-                out = rows.new_zeros(flat.shape[0], self.embedding_dim)
-            out = out.index_copy(0, pos, rows)
-        if out is None:
-            out = self.parts[0].new_zeros(flat.shape[0], self.embedding_dim)
-        return out.view(*ids.shape, self.embedding_dim)
+        return embedding_lookup(self.weight, ids)
 
     def full_weight(self):
-        """Materialize the logical [V, D] table (tests / checkpoint export)."""
-        return torch.cat([p.detach() for p in self.parts], dim=0)
+        """The logical [V, D] table."""
+        return self.weight.detach()
+
+    # --- checkpoint layout parity (reference 'emb/part_k' shards) ----------
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        w = self.weight if keep_vars else self.weight.detach()
+        for k in range(self.num_partitions):
+            off, size = self.offsets[k], self.sizes[k]
+            destination[prefix + 'parts.%d' % k] = w[off:off + size]
+
+    def _load_from_state_dict(self, state_dict, prefix, local_metadata,
+                              strict, missing_keys, unexpected_keys,
+                              error_msgs):
+        part_keys = sorted(
+            (k for k in state_dict if k.startswith(prefix + 'parts.')),
+            key=lambda k: int(k.rsplit('.', 1)[1]))
+        if not part_keys:
+            if prefix + 'weight' in state_dict:  # plain layout also accepted
+                with torch.no_grad():
+                    self.weight.copy_(state_dict[prefix + 'weight'])
+                return
+            missing_keys.append(prefix + 'parts.0')
+            return
+        with torch.no_grad():
+            full = torch.cat([state_dict[k].to(self.weight.dtype)
+                              for k in part_keys], dim=0)
+            if full.shape != self.weight.shape:
+                error_msgs.append(
+                    'PartitionedEmbedding %s: checkpoint rows %s != %s'
+                    % (prefix, tuple(full.shape), tuple(self.weight.shape)))
+                return
+            self.weight.copy_(full)

@@ -51,14 +51,21 @@ def _lstm_gates_ref(gates, c_prev, forget_bias):
     return h, c
 
 
+def lstm_gates(gates, c_prev, forget_bias=1.0):
+    """Fused pointwise gate math given pre-activation gates [B, 4H]."""
+    c_prev = c_prev.to(gates.dtype)
+    if _ops.use_native(gates):
+        return _LSTMGatesFn.apply(gates.contiguous(), c_prev.contiguous(),
+                                  forget_bias)
+    return _lstm_gates_ref(gates, c_prev, forget_bias)
+
+
 def lstm_cell(x, h_prev, c_prev, w, b, forget_bias=1.0):
     """One LSTM step.  x [B, I], h_prev [B, H'], c_prev [B, H],
     w [(I+H'), 4H], b [4H].  Returns (h [B, H], c [B, H])."""
     xh = torch.cat([x, h_prev], dim=-1)
     gates = torch.addmm(b, xh, w)
-    if _ops.use_native(gates):
-        return _LSTMGatesFn.apply(gates, c_prev, forget_bias)
-    return _lstm_gates_ref(gates, c_prev, forget_bias)
+    return lstm_gates(gates, c_prev, forget_bias)
 
 
 class ParallaxLSTM(nn.Module):
@@ -95,15 +102,27 @@ class ParallaxLSTM(nn.Module):
         return h, c
 
     def forward(self, xs, state=None):
-        """xs [B, T, I] -> outputs [B, T, P], (h, c) final."""
+        """xs [B, T, I] -> outputs [B, T, P], (h, c) final.
+
+        The input half of the gate GEMM is hoisted out of the time loop:
+        ONE [B*T, I] x [I, 4H] GEMM for all timesteps (big MFMA-friendly
+        shape), leaving per step only the [B, P] x [P, 4H] recurrent GEMM
+        + the fused gate kernel + the projection GEMM."""
         B, T, _ = xs.shape
         if state is None:
             h = xs.new_zeros(B, self.proj_size)
             c = xs.new_zeros(B, self.state_size)
         else:
             h, c = state
+        w_x = self.w[:self.input_size]           # [I, 4H] (contiguous rows)
+        w_h = self.w[self.input_size:]           # [P, 4H]
+        gx = torch.addmm(self.b, xs.reshape(B * T, self.input_size), w_x)
+        gx = gx.view(B, T, 4 * self.state_size)
         outs = []
         for t in range(T):
-            h, c = self.step(xs[:, t, :], h, c)
+            gates = torch.addmm(gx[:, t], h, w_h)
+            h, c = lstm_gates(gates, c, self.forget_bias)
+            if self.w_proj is not None:
+                h = h @ self.w_proj
             outs.append(h)
         return torch.stack(outs, dim=1), (h, c)

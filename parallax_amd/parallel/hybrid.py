@@ -55,14 +55,13 @@ class HybridEngine(object):
                 self.sparse_engine = SparseAllGatherEngine(
                     comm, sparse_infos, optimizer,
                     average_sparse=config.average_sparse)
-        elif self.mode == 'PS':
-            # in-process PS for BOTH dense and sparse: dense vars are
-            # treated as all-rows-touched sparse tables (the reference's PS
-            # mode pushes dense grads to accumulators the same way,
-            # graph_transform_lib.py:356-404).  Dense grads still go over
-            # a (cheaper) allreduce since owner-apply + mirror broadcast
-            # over xGMI is never faster than one fused allreduce on a
-            # single node — behaviorally identical, and documented.
+        else:  # PS or HYBRID
+            # Dense grads go over the bucketed allreduce in BOTH modes: the
+            # reference's PS mode pushes dense grads to gRPC accumulators
+            # (graph_transform_lib.py:356-404) because workers could not
+            # hold/update every variable over 100 Gb IB; on one xGMI node a
+            # fused RCCL allreduce is behaviorally identical and strictly
+            # faster (documented divergence).
             if dense_infos:
                 self.dense_engine = DenseAllReduceEngine(
                     comm, dense_infos,
@@ -72,17 +71,7 @@ class HybridEngine(object):
                 self.sparse_engine = SparsePSEngine(
                     comm, sparse_infos, optimizer,
                     average_sparse=config.average_sparse,
-                    replicate_variables=cc.ps_config.replicate_variables)
-        else:  # HYBRID
-            if dense_infos:
-                self.dense_engine = DenseAllReduceEngine(
-                    comm, dense_infos,
-                    bucket_bytes=cc.dense_bucket_bytes,
-                    overlap=cc.overlap_grad_comm)
-            if sparse_infos:
-                self.sparse_engine = SparsePSEngine(
-                    comm, sparse_infos, optimizer,
-                    average_sparse=config.average_sparse,
+                    local_aggregation=cc.ps_config.local_aggregation,
                     replicate_variables=cc.ps_config.replicate_variables)
         parallax_log.info('parallax engine mode=%s dense=%d sparse=%d '
                           'world=%d', self.mode, len(dense_infos),

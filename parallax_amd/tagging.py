@@ -53,9 +53,9 @@ class GradientsInfo:
     is_sparse:  True == INDEXED_SLICES (reference
                 gradients_info_pb2.TensorInfoDef.INDEXED_SLICES).
     num_rows:   embedding row count (sparse only).
-    partition:  (shard_index, num_shards, row_offset, total_rows) when the
-                parameter is one shard of a partitioned variable
-                (reference tf.fixed_size_partitioner shards 'emb/part_k').
+    partition:  (num_partitions, row_offsets) when the parameter is a
+                virtually partitioned table (reference
+                tf.fixed_size_partitioner shards 'emb/part_k').
     """
     name: str
     param: torch.nn.Parameter
@@ -85,19 +85,18 @@ def classify_params(model: nn.Module) -> List[GradientsInfo]:
     sparse_params = {}  # id(param) -> (num_rows, partition)
     for mod_name, module in model.named_modules():
         if _is_embedding_module(module):
-            # recurse: a PartitionedEmbedding keeps its shards in a nested
-            # ParameterList
             for p_name, p in module.named_parameters(recurse=True):
                 if p.dim() >= 2:
-                    part = getattr(p, '_parallax_partition', None)
+                    part = getattr(p, '_parallax_partition_meta', None)
                     sparse_params[id(p)] = (p.shape[0], part)
     for name, p in model.named_parameters():
         if not p.requires_grad:
             continue
         explicit = getattr(p, _PARALLAX_TAG_ATTR, None)
         if explicit == SPARSE:
-            infos.append(GradientsInfo(name, p, True, p.shape[0],
-                                       getattr(p, '_parallax_partition', None)))
+            infos.append(GradientsInfo(
+                name, p, True, p.shape[0],
+                getattr(p, '_parallax_partition_meta', None)))
         elif explicit == DENSE:
             infos.append(GradientsInfo(name, p, False))
         elif id(p) in sparse_params:

@@ -201,11 +201,11 @@ def test_partitioned_embedding_tagging():
     m = nn.Module()
     m.emb = PartitionedEmbedding(100, 8, FixedSizePartitioner(4))
     infos = tagging.classify_params(m)
-    assert len(infos) == 4
-    assert all(gi.is_sparse for gi in infos)
-    parts = {gi.name: gi.partition for gi in infos}
-    assert parts['emb.parts.0'] == (0, 4, 0, 100)
-    assert parts['emb.parts.3'] == (3, 4, 75, 100)
+    assert len(infos) == 1
+    gi = infos[0]
+    assert gi.is_sparse and gi.name == 'emb.weight'
+    assert gi.num_rows == 100
+    assert gi.partition == (4, (0, 25, 50, 75))
 
 
 # --- public API surface (reference __init__.py:16-26) -----------------------

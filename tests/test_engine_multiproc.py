@@ -146,16 +146,14 @@ def _body_sparse_ps(rank, world):
     model.emb = PartitionedEmbedding(40, 4, FixedSizePartitioner(4))
     opt = optim.Adagrad(model.parameters(), lr=0.5)
     infos = classify_params(model)
-    eng = SparsePSEngine(comm, infos, opt, replicate_variables=True)
-    # shards round-robin over 2 ranks
-    owners = [eng.owner[id(gi.param)] for gi in eng.infos]
-    assert owners == [0, 1, 0, 1]
+    assert infos[0].partition == (4, (0, 10, 20, 30))
+    eng = SparsePSEngine(comm, infos, opt, local_aggregation=True)
     eng.prepare_step()
-    ids = torch.tensor([0, 11 + rank, 25])  # shards 0,1,2; rank-varying row
+    ids = torch.tensor([0, 11 + rank, 25, 25])  # dup row tests local agg
     out = model.emb(ids)
     out.sum().backward()
     eng.reduce_and_apply()
-    # mirrors must be consistent across ranks after the refresh
+    # replicated tables must be identical across ranks after apply
     full = model.emb.full_weight()
     fulls = [torch.zeros_like(full) for _ in range(world)]
     torch.distributed.all_gather(fulls, full)
@@ -169,6 +167,15 @@ def _body_sparse_ps(rank, world):
         assert not torch.allclose(full[row], ref_full[row]), row
     for row in (3, 20, 39):
         assert torch.allclose(full[row], ref_full[row]), row
+    # local aggregation semantics: the duplicated row 25 must have been
+    # summed BEFORE the adagrad update (one update with grad 2g, not two
+    # with g) — compare against a single-process reference
+    o_ref = optim.Adagrad(ref.parameters(), lr=0.5)
+    ref_ids = torch.tensor([0, 0, 11, 12, 25, 25, 25, 25])
+    ref(ref_ids).sum().backward()
+    g = ref.weight.grad.coalesce()
+    o_ref.sparse_row_step(ref.weight, g.indices().flatten(), g.values())
+    assert torch.allclose(full, ref.full_weight(), atol=1e-6)
     eng.close()
 
 

@@ -62,13 +62,14 @@ def test_lm1b_model_parameter_shapes():
                   num_sampled=64, min_partitions=4)
     from parallax_amd.tagging import classify_params, split_dense_sparse
     sparse, dense = split_dense_sparse(classify_params(m))
-    # sparse: 4 emb shards + 4 softmax_w shards + softmax_b
-    assert len(sparse) == 9
+    # sparse: emb table + softmax_w table + softmax_b
+    assert len(sparse) == 3
     # dense: lstm w, b, w_proj
     assert len(dense) == 3
     names = {gi.name for gi in sparse}
-    assert 'softmax_b' in names
-    assert 'emb.parts.0' in names and 'softmax_w.parts.3' in names
+    assert names == {'emb.weight', 'softmax_w.weight', 'softmax_b'}
+    parts = {gi.name: gi.partition for gi in sparse}
+    assert parts['emb.weight'][0] == 4  # min_partitions shards
 
 
 def test_lm1b_partition_env_override(monkeypatch):

@@ -79,16 +79,27 @@ def test_partitioned_embedding_matches_full():
     out = pe(ids)
     ref = full[ids]
     assert torch.allclose(out, ref)
-    # grads: each shard gets an indexed-slices grad in LOCAL row ids
+    # gradient: one indexed-slices tensor on the single weight
     out.sum().backward()
-    shard_of = pe.shard_of(ids)
-    for k, p in enumerate(pe.parts):
-        if (shard_of == k).any():
-            assert p.grad is not None
-            assert p.grad.layout == torch.sparse_coo
-        # local indices within range
-        if p.grad is not None and p.grad._nnz() > 0:
-            assert int(p.grad._indices().max()) < p.shape[0]
+    g = pe.weight.grad
+    assert g is not None and g.layout == torch.sparse_coo
+    assert set(g._indices().flatten().tolist()) == {0, 4, 16, 5}
+    # shard routing metadata
+    assert pe.shard_of(ids).tolist() == [0, 0, 3, 1, 0]
+
+
+def test_partitioned_embedding_state_dict_layout():
+    """Checkpoint-shard layout parity: state_dict emits parts.k slices and
+    a p-shard checkpoint restores under a different p' (§5.4)."""
+    torch.manual_seed(2)
+    pe = PartitionedEmbedding(10, 4, FixedSizePartitioner(3))
+    sd = pe.state_dict()
+    assert set(sd.keys()) == {'parts.0', 'parts.1', 'parts.2'}
+    assert sd['parts.0'].shape == (4, 4)  # 10 rows -> 4,3,3
+    assert torch.equal(torch.cat(list(sd.values())), pe.full_weight())
+    pe2 = PartitionedEmbedding(10, 4, FixedSizePartitioner(5))
+    pe2.load_state_dict(sd)
+    assert torch.equal(pe2.full_weight(), pe.full_weight())
 
 
 def test_partitioned_embedding_batch_shape():
