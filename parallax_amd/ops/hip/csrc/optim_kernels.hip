@@ -191,16 +191,33 @@ static void mt_launch(const std::vector<at::Tensor>& a,
   }
 }
 
+// The mt_* kernels sweep each tensor's STORAGE linearly, which is correct
+// for any non-overlapping dense layout (e.g. channels_last conv weights)
+// as long as every companion tensor (grad, state) shares the param's
+// strides — autograd and zeros_like both preserve them.
 static void check_contig(const std::vector<at::Tensor>& ts,
                          const char* name) {
   for (auto& t : ts) {
-    TORCH_CHECK(t.is_contiguous(), name, ": tensors must be contiguous");
+    TORCH_CHECK(t.is_non_overlapping_and_dense(), name,
+                ": tensors must be non-overlapping and dense");
+  }
+}
+
+static void check_same_layout(const std::vector<at::Tensor>& a,
+                              const std::vector<at::Tensor>& b,
+                              const char* name) {
+  TORCH_CHECK(a.size() == b.size(), name, ": list size mismatch");
+  for (size_t i = 0; i < a.size(); ++i) {
+    TORCH_CHECK(a[i].strides() == b[i].strides() &&
+                    a[i].sizes() == b[i].sizes(),
+                name, ": layout mismatch at tensor ", i);
   }
 }
 
 void multi_tensor_sgd(std::vector<at::Tensor> params,
                       std::vector<at::Tensor> grads, double lr, double wd) {
   check_contig(params, "mt_sgd");
+  check_same_layout(params, grads, "mt_sgd");
   mt_launch(params, grads, {}, {}, {}, SGDOp{(float)lr, (float)wd});
 }
 
@@ -209,6 +226,8 @@ void multi_tensor_momentum(std::vector<at::Tensor> params,
                            std::vector<at::Tensor> bufs, double lr,
                            double mom, double wd, bool nesterov) {
   check_contig(params, "mt_momentum");
+  check_same_layout(params, grads, "mt_momentum");
+  check_same_layout(params, bufs, "mt_momentum");
   mt_launch(params, grads, bufs, {}, {},
             MomentumOp{(float)lr, (float)mom, (float)wd, nesterov});
 }
@@ -218,6 +237,7 @@ void multi_tensor_adagrad(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> accums, double lr,
                           double eps) {
   check_contig(params, "mt_adagrad");
+  check_same_layout(params, grads, "mt_adagrad");
   mt_launch(params, grads, accums, {}, {},
             AdagradOp{(float)lr, (float)eps});
 }
@@ -229,6 +249,7 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        double lr, double b1, double b2, double eps,
                        double wd) {
   check_contig(params, "mt_adam");
+  check_same_layout(params, grads, "mt_adam");
   std::vector<float> e(steps.begin(), steps.end());
   mt_launch(params, grads, ms, vs, e,
             AdamOp{(float)lr, (float)b1, (float)b2, (float)eps, (float)wd});
@@ -240,6 +261,7 @@ void multi_tensor_rmsprop(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> bufs, double lr,
                           double decay, double mom, double eps) {
   check_contig(params, "mt_rmsprop");
+  check_same_layout(params, grads, "mt_rmsprop");
   mt_launch(params, grads, sqs, bufs, {},
             RMSPropOp{(float)lr, (float)decay, (float)mom, (float)eps});
 }

@@ -110,6 +110,27 @@ def test_multi_tensor_adam_vs_torch():
     assert torch.allclose(p1.detach().cpu(), p2.detach(), atol=1e-4)
 
 
+def test_multi_tensor_momentum_channels_last():
+    """Conv weights in channels_last (non-contiguous but dense) must go
+    through the multi-tensor kernel with matching-layout grads."""
+    from parallax_amd import optim
+    torch.manual_seed(33)
+    p1 = torch.nn.Parameter(
+        torch.randn(8, 4, 3, 3, device=DEV)
+        .to(memory_format=torch.channels_last))
+    p2 = torch.nn.Parameter(p1.detach().cpu().contiguous().clone())
+    o1 = optim.SGD([p1], lr=0.1, momentum=0.9)
+    o2 = torch.optim.SGD([p2], lr=0.1, momentum=0.9)
+    for _ in range(3):
+        g = torch.randn(8, 4, 3, 3)
+        p1.grad = g.to(DEV).to(memory_format=torch.channels_last)
+        p2.grad = g.clone()
+        o1.step()
+        o2.step()
+    assert torch.allclose(p1.detach().cpu().contiguous(), p2.detach(),
+                          atol=1e-5)
+
+
 def test_sparse_adagrad_rows_vs_cpu():
     from parallax_amd import optim
     torch.manual_seed(5)
