@@ -42,6 +42,10 @@ def parse_args():
     p.add_argument('--run-option', default='HYBRID')
     p.add_argument('--batch-size', type=int, default=None)
     p.add_argument('--partitions', type=int, default=None)
+    p.add_argument('--cuda-graph', dest='cuda_graph', default=True,
+                   action='store_true')
+    p.add_argument('--no-cuda-graph', dest='cuda_graph',
+                   action='store_false')
     return p.parse_args()
 
 
@@ -99,6 +103,15 @@ def main():
         if args.model == 'lm1b':
             return M.synthetic_batch(batch, seq, device=device, state=state)
         return M.synthetic_batch(batch, device=device)
+
+    if args.cuda_graph and on_gpu and args.model == 'lm1b':
+        try:
+            state0 = M.make_init_state(batch, device=device,
+                                       dtype=torch.bfloat16)
+            sess.enable_cuda_graph(make_batch(state0))
+        except Exception as e:
+            print('cuda-graph capture failed, falling back to eager: %r'
+                  % (e,), file=sys.stderr)
 
     # warmup (untimed)
     state = None

@@ -110,6 +110,16 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
     return graph
 
 
+def make_init_state(batch_size=BATCH_SIZE, state_size=STATE_SIZE,
+                    proj_size=PROJ_SIZE, device='cpu',
+                    dtype=torch.float32):
+    """Zero LSTM state (h [B, P], c [B, H]) — explicit tensors so the
+    hipGraph capture can feed the recurrent state back in-graph."""
+    h = torch.zeros(batch_size, proj_size, device=device, dtype=dtype)
+    c = torch.zeros(batch_size, state_size, device=device, dtype=dtype)
+    return (h, c)
+
+
 def synthetic_batch(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
                     vocab_size=VOCAB_SIZE, device='cpu', state=None,
                     generator=None):

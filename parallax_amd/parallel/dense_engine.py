@@ -156,6 +156,20 @@ class DenseAllReduceEngine(object):
     def prepare_step(self):
         self.zero_grad()
 
+    def set_overlap(self, overlap):
+        """Toggle hook-driven overlap (off under hipGraph capture: the
+        collective launches move after graph replay)."""
+        overlap = overlap and self.comm.world_size > 1
+        if overlap == self.overlap:
+            return
+        self.overlap = overlap
+        if overlap and not self._hooks:
+            self._register_hooks()
+        elif not overlap:
+            for h in self._hooks:
+                h.remove()
+            self._hooks = []
+
     def close(self):
         for h in self._hooks:
             h.remove()

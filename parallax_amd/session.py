@@ -92,8 +92,77 @@ class ParallaxSession(object):
             parallax_log.warning('profile export failed: %r', e)
         return out
 
+    # -- hipGraph capture of the fwd+bwd hot loop ---------------------------
+    # The LM1B step is launch-bound (hundreds of small kernels across 20
+    # unrolled LSTM steps); capturing forward+backward in ONE hipGraph and
+    # replaying it removes the per-kernel host gaps (guide: HIP streams and
+    # graphs instead of a tracing compiler).  Outside the graph remain the
+    # dynamic-shape parts: the sparse dedup/exchange, collectives, and the
+    # optimizer step.
+    def enable_cuda_graph(self, example_batch, warmup_iters=3):
+        """Capture fwd+bwd (+recurrent-state feedback) as a hipGraph over
+        static input buffers.  Requires static shapes and a CUDA device.
+        ``example_batch``: (tensors..., state_or_None) matching step_fn."""
+        assert self.comm.device.type == 'cuda', 'cuda graphs need a GPU'
+        g = self.graph
+        if self.engine.dense_engine:
+            self.engine.dense_engine.set_overlap(False)
+
+        # static input buffers (state tensors feed back inside the graph)
+        def _clone(t):
+            return t.clone() if isinstance(t, torch.Tensor) else t
+        self._static_batch = tuple(_clone(b) for b in example_batch)
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(warmup_iters):
+                self.engine.prepare_step()
+                g.optimizer.zero_grad(set_to_none=True)
+                loss, fetches = g.step_fn(g.model, self._static_batch)
+                loss.backward()
+        torch.cuda.current_stream().wait_stream(side)
+
+        # materialize dense grads so the captured zero_()s hit stable memory
+        dense_params = [p for p in g.model.parameters()
+                        if p.grad is not None
+                        and p.grad.layout == torch.strided]
+        self.engine.prepare_step()  # clear sparse capture slots pre-capture
+        self._graph_obj = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph_obj):
+            for p in dense_params:
+                p.grad.zero_()
+            loss, fetches = g.step_fn(g.model, self._static_batch)
+            loss.backward()
+            state = fetches.get('state') if fetches else None
+            if state is not None and isinstance(
+                    self._static_batch[-1], (tuple, list)):
+                for dst, src in zip(self._static_batch[-1], state):
+                    dst.copy_(src)
+        self._static_loss = loss
+        self._static_fetches = fetches
+        parallax_log.info('hipGraph capture enabled (%d dense grads, '
+                          '%d sparse slots)', len(dense_params),
+                          len(self.engine.sparse_engine.slots)
+                          if self.engine.sparse_engine else 0)
+
+    def _run_step_graphed(self, batch):
+        # copy dynamic inputs into the static buffers (state feeds back
+        # inside the graph; a caller-passed state is ignored after capture)
+        for dst, src in zip(self._static_batch, batch):
+            if isinstance(dst, torch.Tensor) and isinstance(
+                    src, torch.Tensor) and dst.data_ptr() != src.data_ptr():
+                dst.copy_(src, non_blocking=True)
+        self._graph_obj.replay()
+        self.engine.finish_step()
+        if self.graph.lr_scheduler is not None:
+            self.graph.lr_scheduler.step()
+        return self._static_loss, self._static_fetches
+
     # -- the step ------------------------------------------------------------
     def _run_step(self, batch):
+        if getattr(self, '_graph_obj', None) is not None:
+            return self._run_step_graphed(batch)
         g = self.graph
         self.engine.prepare_step()
         g.optimizer.zero_grad(set_to_none=True)
