@@ -64,6 +64,8 @@ def main():
     from parallax_amd.config import ParallaxConfig
     cfg = ParallaxConfig(run_option=args.run_option,
                          search_partitions=False)
+    # halve xGMI bytes: bf16 fusion buffers for the dense allreduce
+    cfg.communication_config.dense_comm_dtype = 'bf16'
 
     on_gpu = torch.cuda.is_available()
     if args.model == 'lm1b':

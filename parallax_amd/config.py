@@ -71,6 +71,9 @@ class CommunicationConfig(object):
         # and backward/comm overlap toggle.
         self.dense_bucket_bytes = dense_bucket_bytes
         self.overlap_grad_comm = overlap_grad_comm
+        # 'bf16' reduces dense grads in bf16 fusion buffers (halved xGMI
+        # bytes; fp32 write-back); None keeps the gradient dtype.
+        self.dense_comm_dtype = None
 
 
 class CheckPointConfig(object):

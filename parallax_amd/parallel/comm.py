@@ -100,6 +100,33 @@ class CommContext(object):
         dist.all_gather(outs, padded)
         return [outs[r][:counts[r]] for r in range(self.world_size)]
 
+    def all_gather_v_multi(self, tensors):
+        """Allgather SEVERAL tensors that share dim-0 length per rank
+        (ids/rows/counts of one sparse grad) with a SINGLE counts exchange.
+        Returns a list (per input tensor) of lists (per rank)."""
+        if self.world_size == 1:
+            return [[t] for t in tensors]
+        n_local = torch.tensor([tensors[0].shape[0]], dtype=torch.long,
+                               device=tensors[0].device)
+        counts_t = [torch.zeros_like(n_local)
+                    for _ in range(self.world_size)]
+        dist.all_gather(counts_t, n_local)
+        counts = [int(c.item()) for c in counts_t]
+        max_n = max(counts) if counts else 0
+        outs_all = []
+        for t in tensors:
+            assert t.shape[0] == int(n_local.item())
+            tail_shape = tuple(t.shape[1:])
+            padded = t.new_zeros((max_n,) + tail_shape)
+            if t.shape[0] > 0:
+                padded[:t.shape[0]] = t
+            outs = [torch.empty_like(padded)
+                    for _ in range(self.world_size)]
+            dist.all_gather(outs, padded)
+            outs_all.append([outs[r][:counts[r]]
+                             for r in range(self.world_size)])
+        return outs_all
+
     def all_to_all_v(self, tensors_per_rank):
         """Send tensors_per_rank[r] to rank r; receive one tensor from each
         rank (contract N7 PS transport).  Works on nccl (all_to_all) and

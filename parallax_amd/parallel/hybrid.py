@@ -50,7 +50,10 @@ class HybridEngine(object):
                 self.dense_engine = DenseAllReduceEngine(
                     comm, dense_infos,
                     bucket_bytes=cc.dense_bucket_bytes,
-                    overlap=cc.overlap_grad_comm)
+                    overlap=cc.overlap_grad_comm,
+                    comm_dtype=(torch.bfloat16
+                                if cc.dense_comm_dtype == 'bf16'
+                                and comm.device.type == 'cuda' else None))
             if sparse_infos:
                 self.sparse_engine = SparseAllGatherEngine(
                     comm, sparse_infos, optimizer,
@@ -66,7 +69,10 @@ class HybridEngine(object):
                 self.dense_engine = DenseAllReduceEngine(
                     comm, dense_infos,
                     bucket_bytes=cc.dense_bucket_bytes,
-                    overlap=cc.overlap_grad_comm)
+                    overlap=cc.overlap_grad_comm,
+                    comm_dtype=(torch.bfloat16
+                                if cc.dense_comm_dtype == 'bf16'
+                                and comm.device.type == 'cuda' else None))
             if sparse_infos:
                 self.sparse_engine = SparsePSEngine(
                     comm, sparse_infos, optimizer,

@@ -77,11 +77,17 @@ def _capture_hook(slot):
 
 
 class _SparseEngineBase(object):
-    def __init__(self, comm, sparse_infos, optimizer, average_sparse=False):
+    def __init__(self, comm, sparse_infos, optimizer, average_sparse=False,
+                 comm_dtype=torch.bfloat16):
         self.comm = comm
         self.infos = list(sparse_infos)
         self.optimizer = optimizer
         self.average_sparse = average_sparse
+        # bf16 payloads only on the RCCL/GPU path; CPU/gloo plumbing tests
+        # keep native dtype (and exact numerics)
+        if comm.device.type != 'cuda':
+            comm_dtype = None
+        self.comm_dtype = comm_dtype
         self.slots = [_SparseGradSlot(gi) for gi in self.infos]
         self._hooks = []
         for slot in self.slots:
@@ -105,12 +111,20 @@ class _SparseEngineBase(object):
 
     def _exchange_and_apply(self, slot, u_ids, u_rows, u_counts):
         """Allgather (ids, rows[, counts]) across ranks, dedup the union
-        (optionally count-averaged) and apply on every rank."""
+        (optionally count-averaged) and apply on every rank.  Row payloads
+        travel in ``comm_dtype`` (default bf16 — halves the xGMI bytes; the
+        reduction itself stays fp32 in the segment-sum kernel)."""
         if self.comm.world_size > 1:
-            all_ids = torch.cat(self.comm.all_gather_v(u_ids))
-            all_rows = torch.cat(self.comm.all_gather_v(u_rows))
+            send_rows = u_rows.to(self.comm_dtype) \
+                if self.comm_dtype is not None else u_rows
+            tensors = [u_ids, send_rows]
             if self.average_sparse:
-                all_counts = torch.cat(self.comm.all_gather_v(u_counts))
+                tensors.append(u_counts)
+            gathered = self.comm.all_gather_v_multi(tensors)
+            all_ids = torch.cat(gathered[0])
+            all_rows = torch.cat(gathered[1]).to(u_rows.dtype)
+            if self.average_sparse:
+                all_counts = torch.cat(gathered[2])
         else:
             all_ids, all_rows = u_ids, u_rows
             if self.average_sparse:
