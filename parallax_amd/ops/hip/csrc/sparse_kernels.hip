@@ -79,23 +79,30 @@ at::Tensor rows_gather(const at::Tensor& weight, const at::Tensor& ids) {
 // segment_sum_rows
 // ---------------------------------------------------------------------------
 
+// Flat (segment, column) indexing: one thread per output element with the
+// per-segment gather loop inside.  Segments from a deduplicated gradient
+// are short (few duplicates per id), so the inner loop is 1-3 iterations;
+// flat indexing keeps all 64 lanes busy for ANY d (the block-per-segment
+// form wastes 255/256 threads on d=1 bias tables).
 template <typename T>
 __global__ void segment_sum_rows_kernel(const T* __restrict__ values,
                                         const int* __restrict__ order,
                                         const int* __restrict__ seg_starts,
                                         T* __restrict__ out, long u, long d,
                                         bool average) {
-  for (long s = blockIdx.x; s < u; s += gridDim.x) {
+  const long total = u * d;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long s = idx / d;
+    const long c = idx - s * d;
     const int beg = seg_starts[s];
     const int end = seg_starts[s + 1];
-    const float inv = average ? 1.0f / (float)(end - beg) : 1.0f;
-    for (long c = threadIdx.x; c < d; c += blockDim.x) {
-      float acc = 0.0f;
-      for (int k = beg; k < end; ++k) {
-        acc += to_f32<T>(values[(long)order[k] * d + c]);
-      }
-      out[s * d + c] = from_f32<T>(acc * inv);
+    float acc = 0.0f;
+    for (int k = beg; k < end; ++k) {
+      acc += to_f32<T>(values[(long)order[k] * d + c]);
     }
+    if (average) acc *= 1.0f / (float)(end - beg);
+    out[idx] = from_f32<T>(acc);
   }
 }
 
@@ -116,7 +123,7 @@ at::Tensor segment_sum_rows(const at::Tensor& values, const at::Tensor& order,
       at::kBFloat16, at::kHalf, v.scalar_type(), "segment_sum_rows", [&] {
         using T = scalar_t;
         auto stream = at::cuda::getCurrentCUDAStream();
-        int grid = (int)(u < PX_MAX_GRID ? u : PX_MAX_GRID);
+        int grid = px_grid(u * d);
         segment_sum_rows_kernel<T>
             <<<grid, PX_BLOCK, 0, stream>>>(
                 reinterpret_cast<const T*>(v.data_ptr()),

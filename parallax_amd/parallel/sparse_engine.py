@@ -169,7 +169,15 @@ class SparsePSEngine(_SparseEngineBase):
             if s.ids is None:
                 continue
             if self.local_aggregation:
-                u_ids, u_rows, u_counts = unique_sum_rows(s.ids, s.rows)
+                u_ids, u_rows, u_counts = unique_sum_rows(
+                    s.ids, s.rows,
+                    average=(self.average_sparse
+                             and self.comm.world_size == 1))
+                if self.comm.world_size == 1:
+                    # already deduplicated (and averaged) — apply directly,
+                    # no second sort/segment pass
+                    self._apply_rows(s.info.param, u_ids, u_rows)
+                    continue
             else:
                 u_ids, u_rows = s.ids, s.rows
                 u_counts = torch.ones_like(u_ids)
