@@ -1,0 +1,196 @@
+"""CNN benchmark model family — the tf_cnn_benchmarks equivalents.
+
+Reference: examples/tf_cnn_benchmarks (models/*.py: ResNet-50/101/152
+v1/v2, VGG-16/19, AlexNet, GoogLeNet, Inception-v3, Overfeat, Trivial;
+registry in benchmark_cnn.py).  All dense models -> they exercise the
+bucketed RCCL all-reduce engine (MPI mode / degenerate HYBRID).
+
+``build_cnn_graph(model_name, ...)`` is the benchmark_cnn-style entry:
+synthetic ImageNet, softmax-CE (fused HIP kernel), SGD/Momentum/RMSProp
+choice (benchmark_cnn.py:805-818), NHWC (channels_last) + bf16 on GPU.
+"""
+
+import torch
+import torch.nn as nn
+
+from parallax_amd import optim
+from parallax_amd.models.resnet import ResNet
+from parallax_amd.ops.softmax_ce import softmax_cross_entropy
+from parallax_amd.session import TrainGraph
+
+
+def _conv_bn_relu(cin, cout, k, s=1, p=None):
+    if p is None:
+        p = k // 2
+    return nn.Sequential(
+        nn.Conv2d(cin, cout, k, stride=s, padding=p, bias=False),
+        nn.BatchNorm2d(cout, momentum=0.001), nn.ReLU(inplace=True))
+
+
+class VGG16(nn.Module):
+    def __init__(self, num_classes=1001):
+        super().__init__()
+        cfg = [64, 64, 'M', 128, 128, 'M', 256, 256, 256, 'M',
+               512, 512, 512, 'M', 512, 512, 512, 'M']
+        layers, cin = [], 3
+        for v in cfg:
+            if v == 'M':
+                layers.append(nn.MaxPool2d(2, 2))
+            else:
+                layers += [nn.Conv2d(cin, v, 3, padding=1),
+                           nn.ReLU(inplace=True)]
+                cin = v
+        self.features = nn.Sequential(*layers)
+        self.classifier = nn.Sequential(
+            nn.Flatten(), nn.Linear(512 * 7 * 7, 4096),
+            nn.ReLU(inplace=True), nn.Dropout(0.5), nn.Linear(4096, 4096),
+            nn.ReLU(inplace=True), nn.Dropout(0.5),
+            nn.Linear(4096, num_classes))
+
+    def forward(self, x):
+        return self.classifier(self.features(x))
+
+
+class AlexNet(nn.Module):
+    def __init__(self, num_classes=1001):
+        super().__init__()
+        self.features = nn.Sequential(
+            nn.Conv2d(3, 64, 11, 4, 2), nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, 2),
+            nn.Conv2d(64, 192, 5, padding=2), nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, 2),
+            nn.Conv2d(192, 384, 3, padding=1), nn.ReLU(inplace=True),
+            nn.Conv2d(384, 256, 3, padding=1), nn.ReLU(inplace=True),
+            nn.Conv2d(256, 256, 3, padding=1), nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, 2))
+        self.classifier = nn.Sequential(
+            nn.Flatten(), nn.Dropout(0.5), nn.Linear(256 * 6 * 6, 4096),
+            nn.ReLU(inplace=True), nn.Dropout(0.5), nn.Linear(4096, 4096),
+            nn.ReLU(inplace=True), nn.Linear(4096, num_classes))
+
+    def forward(self, x):
+        return self.classifier(self.features(x))
+
+
+class Overfeat(nn.Module):
+    def __init__(self, num_classes=1001):
+        super().__init__()
+        self.features = nn.Sequential(
+            nn.Conv2d(3, 96, 11, 4), nn.ReLU(inplace=True),
+            nn.MaxPool2d(2, 2),
+            nn.Conv2d(96, 256, 5), nn.ReLU(inplace=True),
+            nn.MaxPool2d(2, 2),
+            nn.Conv2d(256, 512, 3, padding=1), nn.ReLU(inplace=True),
+            nn.Conv2d(512, 1024, 3, padding=1), nn.ReLU(inplace=True),
+            nn.Conv2d(1024, 1024, 3, padding=1), nn.ReLU(inplace=True),
+            nn.MaxPool2d(2, 2))
+        self.classifier = nn.Sequential(
+            nn.Flatten(), nn.Linear(1024 * 6 * 6, 3072),
+            nn.ReLU(inplace=True), nn.Linear(3072, 4096),
+            nn.ReLU(inplace=True), nn.Linear(4096, num_classes))
+
+    def forward(self, x):
+        return self.classifier(self.features(x))
+
+
+class _Inception(nn.Module):
+    def __init__(self, cin, c1, c3r, c3, c5r, c5, pp):
+        super().__init__()
+        self.b1 = _conv_bn_relu(cin, c1, 1)
+        self.b3 = nn.Sequential(_conv_bn_relu(cin, c3r, 1),
+                                _conv_bn_relu(c3r, c3, 3))
+        self.b5 = nn.Sequential(_conv_bn_relu(cin, c5r, 1),
+                                _conv_bn_relu(c5r, c5, 5))
+        self.bp = nn.Sequential(nn.MaxPool2d(3, 1, 1),
+                                _conv_bn_relu(cin, pp, 1))
+
+    def forward(self, x):
+        return torch.cat([self.b1(x), self.b3(x), self.b5(x), self.bp(x)],
+                         dim=1)
+
+
+class GoogLeNet(nn.Module):
+    def __init__(self, num_classes=1001):
+        super().__init__()
+        self.stem = nn.Sequential(
+            _conv_bn_relu(3, 64, 7, 2), nn.MaxPool2d(3, 2, 1),
+            _conv_bn_relu(64, 64, 1), _conv_bn_relu(64, 192, 3),
+            nn.MaxPool2d(3, 2, 1))
+        self.blocks = nn.Sequential(
+            _Inception(192, 64, 96, 128, 16, 32, 32),
+            _Inception(256, 128, 128, 192, 32, 96, 64),
+            nn.MaxPool2d(3, 2, 1),
+            _Inception(480, 192, 96, 208, 16, 48, 64),
+            _Inception(512, 160, 112, 224, 24, 64, 64),
+            _Inception(512, 128, 128, 256, 24, 64, 64),
+            _Inception(512, 112, 144, 288, 32, 64, 64),
+            _Inception(528, 256, 160, 320, 32, 128, 128),
+            nn.MaxPool2d(3, 2, 1),
+            _Inception(832, 256, 160, 320, 32, 128, 128),
+            _Inception(832, 384, 192, 384, 48, 128, 128))
+        self.head = nn.Sequential(nn.AdaptiveAvgPool2d(1), nn.Flatten(),
+                                  nn.Dropout(0.4),
+                                  nn.Linear(1024, num_classes))
+
+    def forward(self, x):
+        return self.head(self.blocks(self.stem(x)))
+
+
+class TrivialModel(nn.Module):
+    """tf_cnn_benchmarks 'trivial' (pipeline-overhead probe)."""
+
+    def __init__(self, num_classes=1001, image_size=224):
+        super().__init__()
+        self.net = nn.Sequential(nn.Flatten(),
+                                 nn.Linear(3 * image_size * image_size, 1),
+                                 nn.Linear(1, num_classes))
+
+    def forward(self, x):
+        return self.net(x)
+
+
+MODEL_REGISTRY = {
+    'resnet50': lambda nc: ResNet(50, nc),
+    'resnet101': lambda nc: ResNet(101, nc),
+    'resnet152': lambda nc: ResNet(152, nc),
+    'vgg16': lambda nc: VGG16(nc),
+    'alexnet': lambda nc: AlexNet(nc),
+    'overfeat': lambda nc: Overfeat(nc),
+    'googlenet': lambda nc: GoogLeNet(nc),
+    'trivial': lambda nc: TrivialModel(nc),
+}
+
+
+def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
+                    optimizer_name='momentum', lr=0.1, momentum=0.9,
+                    weight_decay=1e-4, use_bf16=True, channels_last=True):
+    model = MODEL_REGISTRY[model_name](num_classes)
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
+    if optimizer_name == 'momentum':
+        opt = optim.SGD(model.parameters(), lr=lr, momentum=momentum,
+                        weight_decay=weight_decay)
+    elif optimizer_name == 'sgd':
+        opt = optim.SGD(model.parameters(), lr=lr,
+                        weight_decay=weight_decay)
+    elif optimizer_name == 'rmsprop':
+        opt = optim.RMSProp(model.parameters(), lr=lr, decay=0.9,
+                            momentum=momentum, eps=1.0)
+    else:
+        raise ValueError('optimizer must be momentum|sgd|rmsprop '
+                         '(benchmark_cnn.py:805-818)')
+
+    def step_fn(m, batch):
+        x, y = batch
+        if channels_last and x.dim() == 4:
+            x = x.to(memory_format=torch.channels_last)
+        if use_bf16 and x.is_cuda:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                logits = m(x)
+            loss = softmax_cross_entropy(logits.float(), y)
+        else:
+            logits = m(x)
+            loss = softmax_cross_entropy(logits, y)
+        return loss, {}
+
+    return TrainGraph(model, opt, step_fn)

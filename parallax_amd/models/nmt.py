@@ -1,0 +1,207 @@
+"""NMT models — hybrid PS+AR workload with partitioned src/tgt embeddings.
+
+Reference: examples/nmt (~8k LoC GNMT/attention seq2seq: LSTM enc-dec with
+Luong attention, BPE vocabs, ``parallax.shard.shard`` input pipeline and
+``get_partitioner`` embeddings — model_helper.py:310-311,
+iterator_utils.py:103).  BASELINE.json config 4 names the Transformer
+variant ("NMT Transformer hybrid PS+AR, sparse src/tgt embeddings
+partitioned"); both are provided:
+
+  * NMTTransformer — encoder-decoder Transformer; src/tgt embeddings are
+    PartitionedEmbedding tables (sparse/PS plane), projection + attention
+    + FFN params dense (RCCL all-reduce plane); LayerNorm is the
+    hand-written HIP kernel; loss is the fused softmax-CE over the shared
+    output embedding (tied weights, indexed-slices grads).
+  * GNMTAttention — the reference's architecture: LSTM encoder + LSTM
+    decoder with Luong-style multiplicative attention, fused gate kernels.
+"""
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from parallax_amd import optim
+from parallax_amd.ops.embedding import PartitionedEmbedding, embedding_lookup
+from parallax_amd.ops.layernorm import ParallaxLayerNorm
+from parallax_amd.ops.lstm import ParallaxLSTM
+from parallax_amd.ops.sampled_softmax import sampled_softmax_loss
+from parallax_amd.ops.softmax_ce import softmax_cross_entropy
+from parallax_amd.partitions import get_partitioner
+from parallax_amd.session import TrainGraph
+
+SRC_VOCAB = 32000
+TGT_VOCAB = 32000
+
+
+class _MHA(nn.Module):
+    def __init__(self, d, heads):
+        super().__init__()
+        self.d = d
+        self.h = heads
+        self.qkv = nn.Linear(d, 3 * d, bias=False)
+        self.kv = nn.Linear(d, 2 * d, bias=False)
+        self.q = nn.Linear(d, d, bias=False)
+        self.out = nn.Linear(d, d, bias=False)
+
+    def forward(self, x, mem=None, causal=False):
+        B, T, D = x.shape
+        if mem is None:
+            q, k, v = self.qkv(x).chunk(3, dim=-1)
+        else:
+            q = self.q(x)
+            k, v = self.kv(mem).chunk(2, dim=-1)
+        S = k.shape[1]
+
+        def split(t, n):
+            return t.view(B, n, self.h, D // self.h).transpose(1, 2)
+
+        q, k, v = split(q, T), split(k, S), split(v, S)
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+        return self.out(o.transpose(1, 2).reshape(B, T, D))
+
+
+class _Block(nn.Module):
+    def __init__(self, d, heads, ffn, cross=False):
+        super().__init__()
+        self.ln1 = ParallaxLayerNorm(d)
+        self.attn = _MHA(d, heads)
+        self.cross = cross
+        if cross:
+            self.ln_x = ParallaxLayerNorm(d)
+            self.xattn = _MHA(d, heads)
+        self.ln2 = ParallaxLayerNorm(d)
+        self.fc1 = nn.Linear(d, ffn)
+        self.fc2 = nn.Linear(ffn, d)
+
+    def forward(self, x, mem=None, causal=False):
+        x = x + self.attn(self.ln1(x), causal=causal)
+        if self.cross:
+            x = x + self.xattn(self.ln_x(x), mem=mem)
+        x = x + self.fc2(F.gelu(self.fc1(self.ln2(x))))
+        return x
+
+
+class NMTTransformer(nn.Module):
+    def __init__(self, src_vocab=SRC_VOCAB, tgt_vocab=TGT_VOCAB, d_model=512,
+                 heads=8, ffn=2048, enc_layers=6, dec_layers=6,
+                 min_partitions=8, max_len=512):
+        super().__init__()
+        part = get_partitioner(min_partitions)
+        self.d_model = d_model
+        self.tgt_vocab = tgt_vocab
+        self.src_emb = PartitionedEmbedding(src_vocab, d_model, part,
+                                            init_scale=0.05)
+        self.tgt_emb = PartitionedEmbedding(tgt_vocab, d_model, part,
+                                            init_scale=0.05)
+        self.pos = nn.Parameter(torch.zeros(max_len, d_model))
+        from parallax_amd.tagging import mark_dense
+        mark_dense(self.pos)
+        self.enc = nn.ModuleList(
+            [_Block(d_model, heads, ffn) for _ in range(enc_layers)])
+        self.dec = nn.ModuleList(
+            [_Block(d_model, heads, ffn, cross=True)
+             for _ in range(dec_layers)])
+        self.ln_f = ParallaxLayerNorm(d_model)
+
+    def forward(self, src, tgt_in, tgt_out):
+        scale = math.sqrt(self.d_model)
+        e = self.src_emb(src) * scale + self.pos[:src.shape[1]]
+        for blk in self.enc:
+            e = blk(e)
+        d = self.tgt_emb(tgt_in) * scale + self.pos[:tgt_in.shape[1]]
+        for blk in self.dec:
+            d = blk(d, mem=e, causal=True)
+        d = self.ln_f(d)
+        # Tied output softmax over the partitioned tgt table, via sampled
+        # softmax so the table's gradient stays indexed-slices (a full
+        # dense logits GEMM would produce a dense [V, D] grad and push the
+        # whole table onto the all-reduce plane).
+        flat_d = d.reshape(-1, self.d_model)
+        flat_y = tgt_out.reshape(-1)
+        losses = sampled_softmax_loss(self.tgt_emb, None, flat_y, flat_d,
+                                      num_sampled=min(4096,
+                                                      self.tgt_vocab // 2),
+                                      num_classes=self.tgt_vocab)
+        return losses.mean()
+
+
+class GNMTAttention(nn.Module):
+    """Reference-architecture seq2seq: LSTM encoder, LSTM decoder with
+    Luong multiplicative attention (nmt/model.py attention_architecture)."""
+
+    def __init__(self, src_vocab=SRC_VOCAB, tgt_vocab=TGT_VOCAB,
+                 emb_dim=512, units=512, min_partitions=8):
+        super().__init__()
+        part = get_partitioner(min_partitions)
+        self.src_emb = PartitionedEmbedding(src_vocab, emb_dim, part,
+                                            init_scale=0.05)
+        self.tgt_emb = PartitionedEmbedding(tgt_vocab, emb_dim, part,
+                                            init_scale=0.05)
+        self.encoder = ParallaxLSTM(emb_dim, units)
+        self.decoder = ParallaxLSTM(emb_dim + units, units)
+        self.att_w = nn.Parameter(torch.empty(units, units).uniform_(-0.05,
+                                                                     0.05))
+        self.proj = nn.Linear(2 * units, units)
+        self.units = units
+        self.tgt_vocab = tgt_vocab
+
+    def forward(self, src, tgt_in, tgt_out):
+        enc_out, (h, c) = self.encoder(self.src_emb(src))
+        B, S, U = enc_out.shape
+        T = tgt_in.shape[1]
+        emb = self.tgt_emb(tgt_in)
+        ctx = enc_out.new_zeros(B, U)
+        dec_h, dec_c = h, c
+        outs = []
+        for t in range(T):
+            x = torch.cat([emb[:, t], ctx], dim=-1)
+            dec_h, dec_c = self.decoder.step(x, dec_h, dec_c)
+            scores = torch.einsum('bu,uv,bsv->bs', dec_h, self.att_w,
+                                  enc_out)
+            alpha = torch.softmax(scores, dim=-1)
+            ctx = torch.einsum('bs,bsu->bu', alpha, enc_out)
+            outs.append(torch.tanh(self.proj(
+                torch.cat([dec_h, ctx], dim=-1))))
+        dec = torch.stack(outs, dim=1).reshape(B * T, U)
+        losses = sampled_softmax_loss(self.tgt_emb, None,
+                                      tgt_out.reshape(-1), dec,
+                                      num_sampled=min(1024,
+                                                      self.tgt_vocab // 2),
+                                      num_classes=self.tgt_vocab)
+        return losses.mean()
+
+
+def build_nmt_graph(arch='transformer', src_vocab=SRC_VOCAB,
+                    tgt_vocab=TGT_VOCAB, d_model=512, heads=8, ffn=2048,
+                    enc_layers=6, dec_layers=6, min_partitions=8, lr=1e-3,
+                    use_bf16=True, **kw):
+    if arch == 'transformer':
+        model = NMTTransformer(src_vocab, tgt_vocab, d_model, heads, ffn,
+                               enc_layers, dec_layers, min_partitions)
+    else:
+        model = GNMTAttention(src_vocab, tgt_vocab, d_model,
+                              kw.get('units', d_model), min_partitions)
+    optimizer = optim.Adam(model.parameters(), lr=lr)
+
+    def step_fn(m, batch):
+        src, tgt_in, tgt_out = batch
+        if use_bf16 and src.is_cuda:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                loss = m(src, tgt_in, tgt_out)
+        else:
+            loss = m(src, tgt_in, tgt_out)
+        return loss, {}
+
+    return TrainGraph(model, optimizer, step_fn)
+
+
+def synthetic_batch(batch_size=64, src_len=48, tgt_len=48,
+                    src_vocab=SRC_VOCAB, tgt_vocab=TGT_VOCAB, device='cpu'):
+    src = torch.randint(0, src_vocab, (batch_size, src_len), device=device)
+    tgt_in = torch.randint(0, tgt_vocab, (batch_size, tgt_len),
+                           device=device)
+    tgt_out = torch.randint(0, tgt_vocab, (batch_size, tgt_len),
+                            device=device)
+    return src, tgt_in, tgt_out

@@ -1,0 +1,110 @@
+"""Tests for the NMT / skip_thoughts / CNN-family models (CPU, tiny
+configs)."""
+
+import pytest
+import torch
+
+import parallax_amd
+from parallax_amd.config import ParallaxConfig
+from parallax_amd.models import nmt, skip_thoughts, cnn
+from parallax_amd.tagging import classify_params, split_dense_sparse
+
+
+@pytest.fixture(autouse=True)
+def _destroy_pg():
+    yield
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+def _worker_env(monkeypatch, port):
+    monkeypatch.setenv('WORLD_SIZE', '1')
+    monkeypatch.setenv('RANK', '0')
+    monkeypatch.setenv('MASTER_ADDR', '127.0.0.1')
+    monkeypatch.setenv('MASTER_PORT', str(port))
+
+
+def test_nmt_transformer_tagging_and_step(monkeypatch, free_port):
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: nmt.build_nmt_graph(
+            arch='transformer', src_vocab=500, tgt_vocab=500, d_model=32,
+            heads=4, ffn=64, enc_layers=2, dec_layers=2, min_partitions=2,
+            use_bf16=False),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    assert sess.engine.mode == 'HYBRID'
+    sparse, dense = split_dense_sparse(
+        classify_params(sess.graph.model))
+    names = {gi.name for gi in sparse}
+    assert names == {'src_emb.weight', 'tgt_emb.weight'}
+    assert len(dense) > 10
+    batch = nmt.synthetic_batch(4, 6, 7, 500, 500)
+    first = None
+    for _ in range(8):
+        out = sess.run(batch)
+        if first is None:
+            first = float(out['loss'])
+    assert torch.isfinite(out['loss'])
+    assert float(out['loss']) < first
+    sess.close()
+
+
+def test_gnmt_attention_step(monkeypatch, free_port):
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: nmt.build_nmt_graph(
+            arch='gnmt', src_vocab=300, tgt_vocab=300, d_model=24,
+            units=24, min_partitions=2, use_bf16=False),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    batch = nmt.synthetic_batch(3, 5, 5, 300, 300)
+    out = sess.run(batch)
+    assert torch.isfinite(out['loss'])
+    sess.close()
+
+
+def test_skip_thoughts_step(monkeypatch, free_port):
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: skip_thoughts.build_skip_thoughts_graph(
+            vocab=400, emb_dim=16, units=24, num_sampled=32,
+            min_partitions=2, use_bf16=False),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    batch = skip_thoughts.synthetic_batch(4, 6, 400)
+    first = None
+    for _ in range(6):
+        out = sess.run(batch)
+        if first is None:
+            first = float(out['loss'])
+    assert float(out['loss']) < first
+    sess.close()
+
+
+@pytest.mark.parametrize('name', ['vgg16', 'alexnet', 'overfeat',
+                                  'googlenet', 'trivial'])
+def test_cnn_family_forward(name):
+    torch.manual_seed(0)
+    m = cnn.MODEL_REGISTRY[name](11)
+    size = 231 if name == 'overfeat' else 224
+    x = torch.randn(2, 3, size, size)
+    out = m(x)
+    assert out.shape == (2, 11)
+
+
+def test_cnn_graph_optimizers():
+    g = cnn.build_cnn_graph('trivial', num_classes=11,
+                            optimizer_name='rmsprop', use_bf16=False,
+                            channels_last=False)
+    from parallax_amd import optim as pxo
+    assert isinstance(g.optimizer, pxo.RMSProp)
+    x = torch.randn(2, 3, 224, 224)
+    y = torch.randint(0, 11, (2,))
+    loss, _ = g.step_fn(g.model, (x, y))
+    loss.backward()
+    g.optimizer.step()
+    assert torch.isfinite(loss)
+    with pytest.raises(ValueError):
+        cnn.build_cnn_graph('trivial', optimizer_name='adagrad')
