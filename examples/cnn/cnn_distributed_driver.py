@@ -1,0 +1,68 @@
+"""CNN benchmark driver — tf_cnn_benchmarks counterpart.
+
+Reference: examples/tf_cnn_benchmarks/CNNBenchmark_distributed_driver.py
+(+ benchmark_cnn.py): model registry, synthetic ImageNet, images/sec
+logging, momentum/sgd/rmsprop.
+"""
+
+import argparse
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                '..', '..'))
+
+import parallax_amd
+from parallax_amd.models import cnn, resnet
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument('--model', default='resnet50',
+                    choices=sorted(cnn.MODEL_REGISTRY))
+    ap.add_argument('--batch_size', type=int, default=64)
+    ap.add_argument('--optimizer', default='momentum',
+                    choices=['momentum', 'sgd', 'rmsprop'])
+    ap.add_argument('--learning_rate', type=float, default=0.1)
+    ap.add_argument('--max_steps', type=int, default=1000000)
+    ap.add_argument('--log_frequency', type=int, default=10)
+    ap.add_argument('--run_option', default='MPI')
+    ap.add_argument('--params_stat', action='store_true')
+    ap.add_argument('--resource_info', default=os.path.join(
+        os.path.dirname(os.path.abspath(__file__)), 'resource_info'))
+    args = ap.parse_args()
+
+    sess, num_workers, worker_id, _ = parallax_amd.parallel_run(
+        lambda: cnn.build_cnn_graph(args.model,
+                                    batch_size=args.batch_size,
+                                    optimizer_name=args.optimizer,
+                                    lr=args.learning_rate),
+        args.resource_info,
+        parallax_config=parallax_amd.Config(run_option=args.run_option))
+    device = sess.comm.device
+    if args.params_stat and worker_id == 0:
+        stats = sess.params_stat()
+        parallax_amd.log.info('params: %d tensors, %.1f M elements, '
+                              '%.1f MB grads/step', stats['num_tensors'],
+                              stats['num_elements'] / 1e6,
+                              stats['grad_bytes'] / 1e6)
+    prev = time.time()
+    for step in range(args.max_steps):
+        x, y = resnet.synthetic_batch(args.batch_size, device=device)
+        out = sess.run((x, y))
+        if step % args.log_frequency == 0 and worker_id == 0:
+            now = time.time()
+            ips = args.log_frequency * args.batch_size * num_workers \
+                / max(now - prev, 1e-9)
+            parallax_amd.log.info(
+                'step %d, images/sec = %.1f, loss = %.4f',
+                out['global_step'], ips, float(out['loss']))
+            prev = now
+    sess.close()
+
+
+if __name__ == '__main__':
+    main()

@@ -72,9 +72,12 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
                      emb_dim=EMB_DIM, state_size=STATE_SIZE,
                      proj_size=PROJ_SIZE, num_sampled=NUM_SAMPLED,
                      max_grad_norm=10.0, dtype=torch.float32,
-                     use_bf16=True):
+                     use_bf16=True, ema_decay=0.999):
     """use_bf16: run the GEMM path under bf16 autocast on GPU (fp32 master
-    weights, fp32 CE and optimizer math)."""
+    weights, fp32 CE and optimizer math).
+    ema_decay: exponential moving average of the LSTM params (reference
+    language_model_graph.py:63-65, decay .999); shadows live as model
+    buffers (-> broadcast + checkpointed); None disables."""
     model = LM1BModel(vocab_size, emb_dim, state_size, proj_size,
                       num_sampled, min_partitions, dtype=dtype)
     optimizer = optim.Adagrad(model.parameters(), lr=lr,
@@ -97,6 +100,12 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
     # backward and the optimizer step (reference
     # language_model_graph.py:53).  Installed as an engine pre-step hook by
     # wrapping the optimizer step.
+    n_ema = 0
+    if ema_decay is not None:
+        for i, p in enumerate(dense_params):
+            model.register_buffer('ema_%d' % i, p.detach().clone())
+        n_ema = len(dense_params)
+
     orig_step = optimizer.step
 
     def step_with_clip(closure=None):
@@ -104,7 +113,13 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
                  and p.grad.layout == torch.strided]
         if grads:
             clip_by_global_norm(grads, max_grad_norm)
-        return orig_step(closure)
+        out = orig_step(closure)
+        if n_ema:
+            from parallax_amd.ops.misc import ema_update
+            shadows = [getattr(model, 'ema_%d' % i) for i in range(n_ema)]
+            ema_update(shadows, [p.detach() for p in dense_params],
+                       ema_decay)
+        return out
 
     optimizer.step = step_with_clip
     return graph

@@ -226,6 +226,8 @@ def _run_worker(build_fn, config):
         list(graph.model.named_parameters()) +
         list(graph.model.named_buffers()))
     engine = HybridEngine(comm, infos, graph.optimizer, config)
+    if config.export_graph_path and worker_id == 0:
+        _export_graph(config.export_graph_path, graph, infos, engine, config)
     stat_queue = _connect_worker_queue()
     sess = ParallaxSession(graph, engine, comm, config, num_workers,
                            worker_id, stat_queue=stat_queue)
@@ -233,6 +235,37 @@ def _run_worker(build_fn, config):
     # (sess, num_workers, worker_id, num_replicas_per_worker); one process
     # per GPU -> exactly one replica per worker.
     return sess, num_workers, worker_id, 1
+
+
+def _export_graph(path, graph, infos, engine, config):
+    """Dump the transformed-training-plan description (the reference's
+    export_graph_path wrote the transformed MetaGraphDef,
+    mpi/runner.py:164-168): model structure, the dense/sparse tagging
+    table, engine mode and the allreduce bucket plan."""
+    import json
+    os.makedirs(path, exist_ok=True)
+    with open(os.path.join(path, 'model.txt'), 'w') as f:
+        f.write(repr(graph.model))
+    plan = {
+        'mode': engine.mode,
+        'run_option': config.run_option,
+        'average_sparse': config.average_sparse,
+        'params': [{
+            'name': gi.name,
+            'shape': list(gi.param.shape),
+            'dtype': str(gi.param.dtype),
+            'grad': 'INDEXED_SLICES' if gi.is_sparse else 'DENSE',
+            'partition': list(gi.partition) if gi.partition else None,
+        } for gi in infos],
+    }
+    if engine.dense_engine is not None:
+        plan['allreduce_buckets'] = [
+            {'numel': b.numel,
+             'params': [p.shape[0] if p.dim() else 1 for p in b.params]}
+            for b in engine.dense_engine._buckets]
+    with open(os.path.join(path, 'comm_plan.json'), 'w') as f:
+        json.dump(plan, f, indent=2, default=str)
+    parallax_log.info('exported training plan to %s', path)
 
 
 def _move_optimizer_state(optimizer, device):

@@ -210,6 +210,25 @@ class ParallaxSession(object):
                 parallax_log.warning('stat report failed: %r', e)
             self._stat_sent = True
 
+    def params_stat(self):
+        """Gradient/parameter byte counters (reference --params_stat,
+        benchmark_cnn.py:785-796,825-827)."""
+        from parallax_amd.tagging import classify_params, split_dense_sparse
+        infos = classify_params(self.graph.model)
+        sparse, dense = split_dense_sparse(infos)
+        num_el = sum(gi.param.numel() for gi in infos)
+        grad_bytes = sum(gi.param.numel() * gi.param.element_size()
+                         for gi in dense)
+        return {
+            'num_tensors': len(infos),
+            'num_dense': len(dense),
+            'num_sparse': len(sparse),
+            'num_elements': num_el,
+            'param_bytes': sum(gi.param.numel() * gi.param.element_size()
+                               for gi in infos),
+            'grad_bytes': grad_bytes,
+        }
+
     def close(self):
         self.engine.close()
 
