@@ -33,7 +33,8 @@ class PSConfig(object):
         self.replicate_variables = replicate_variables
         self.local_aggregation = local_aggregation
         self.boundary_among_servers = boundary_among_servers
-        self.boundary_between_workers_and_servers = boundary_between_workers_and_servers
+        self.boundary_between_workers_and_servers = \
+            boundary_between_workers_and_servers
 
 
 class MPIConfig(object):

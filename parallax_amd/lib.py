@@ -158,7 +158,8 @@ def create_profile_directory(profile_dir, resource_info):
 
 
 def redirect_stdout_stderr(redirect_path, worker_id):
-    """Per-process log capture (reference redirect_path, ps/runner.py:34-46)."""
+    """Per-process log capture (reference redirect_path,
+    ps/runner.py:34-46)."""
     os.makedirs(redirect_path, exist_ok=True)
     out = open(os.path.join(redirect_path, 'worker_%d.out' % worker_id), 'a')
     err = open(os.path.join(redirect_path, 'worker_%d.err' % worker_id), 'a')

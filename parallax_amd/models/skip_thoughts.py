@@ -3,7 +3,8 @@
 Reference: examples/skip_thoughts (~1.5k LoC): GRU encoder over word
 embeddings, two GRU decoders reconstructing the previous/next sentence,
 Adam optimizer, in-graph sharding via
-``shard.create_num_shards_and_shard_id()`` (skip_thoughts/ops/input_ops.py:92-101).
+``shard.create_num_shards_and_shard_id()``
+(skip_thoughts/ops/input_ops.py:92-101).
 Here: same architecture with a PartitionedEmbedding vocabulary (sparse
 plane) and dense GRU/projection params (all-reduce plane).
 """

@@ -44,7 +44,8 @@ class CommContext(object):
         if device is None:
             if torch.cuda.is_available():
                 local = int(os.environ.get('LOCAL_RANK', self.rank))
-                device = torch.device('cuda', local % torch.cuda.device_count())
+                device = torch.device(
+                    'cuda', local % torch.cuda.device_count())
                 torch.cuda.set_device(device)
             else:
                 device = torch.device('cpu')

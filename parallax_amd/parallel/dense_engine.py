@@ -63,7 +63,8 @@ class DenseAllReduceEngine(object):
             p = gi.param
             esize = p.element_size() if self.comm_dtype is None \
                 else torch.tensor([], dtype=self.comm_dtype).element_size()
-            if cur.numel > 0 and (cur.numel + p.numel()) * esize > bucket_bytes:
+            if cur.numel > 0 \
+                    and (cur.numel + p.numel()) * esize > bucket_bytes:
                 self._buckets.append(cur)
                 cur = _Bucket()
             self._param_bucket[id(p)] = (len(self._buckets), len(cur.params))

@@ -50,7 +50,8 @@ class _ShardedIterable(object):
 
 
 class ShardedSampler(object):
-    """torch Sampler yielding dataset indices i with i % num_shards == shard_id.
+    """torch Sampler yielding dataset indices i with
+    i % num_shards == shard_id.
     Usable as DataLoader(sampler=ShardedSampler(len(ds)))."""
 
     def __init__(self, num_samples, num_shards=None, shard_id=None):
