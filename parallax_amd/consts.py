@@ -41,5 +41,9 @@ REMOTE_PARALLAX_ROOT = os.path.join(
 #     session_context.py:28-29) ---------------------------------------------
 NUM_ITERATIONS_FOR_TEST = 200
 NUM_ITERATIONS_FOR_WARMUP = 200
-COLLECT_STAT_START_STEP = 50
-COLLECT_STAT_END_STEP = 100
+# step window timed for partition search (reference
+# session_context.py:28-29); env-overridable for fast harness tests
+COLLECT_STAT_START_STEP = int(os.environ.get('PARALLAX_COLLECT_STAT_START',
+                                             '50'))
+COLLECT_STAT_END_STEP = int(os.environ.get('PARALLAX_COLLECT_STAT_END',
+                                           '100'))

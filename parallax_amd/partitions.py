@@ -98,13 +98,14 @@ class PartitionStatCollector(object):
     the process-management wrapper lives in runner.py.
     """
 
-    def __init__(self, p_to_test, min_partitions=None):
+    def __init__(self, p_to_test, min_partitions=None, max_trials=12):
         self.p_to_test = p_to_test
         self.prev_p = None
         self.prev_exec_time = None
         self.exec_time_list = []
         self.p_list = []
         self.start = None
+        self.max_trials = max_trials
         if min_partitions is None:
             min_partitions = int(os.environ.get(PARALLAX_MIN_PARTITIONS, '1'))
         self.min_partitions = min_partitions
@@ -135,6 +136,8 @@ class PartitionStatCollector(object):
                     stop = True
             else:
                 self.p_to_test *= 2
+            if len(self.p_list) >= self.max_trials:
+                stop = True  # noise guard beyond the reference walk
             self.prev_p = curr_p
             self.prev_exec_time = exec_time
         else:
