@@ -212,6 +212,10 @@ def _run_worker(build_fn, config):
         redirect_stdout_stderr(config.redirect_path, worker_id)
 
     comm = CommContext(rank=worker_id, world_size=num_workers)
+    # Per-rank RNG streams: data/candidate sampling must differ across
+    # ranks (the rank-0 broadcast below unifies the model init).
+    seed = int(os.environ.get('PARALLAX_SEED', '1234'))
+    torch.manual_seed(seed + worker_id)
     graph = build_fn()
     if not isinstance(graph, TrainGraph):
         model, optimizer, step_fn = graph
