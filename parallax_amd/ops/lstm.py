@@ -68,6 +68,91 @@ def lstm_cell(x, h_prev, c_prev, w, b, forget_bias=1.0):
     return lstm_gates(gates, c_prev, forget_bias)
 
 
+class _LSTMSeqFn(torch.autograd.Function):
+    """Sequence-level projected-LSTM autograd: the whole T-step unroll as
+    ONE Function so backward can (a) run 3 launches per step (fused gate
+    kernel + 2 GEMMs) instead of autograd's ~8, and (b) batch BOTH weight
+    gradients into single [K, B*T] x [B*T, N] GEMMs at the end — the
+    per-step dW accumulation GEMMs of eager autograd disappear.
+
+    Inputs: gx [B,T,4H] (pre-computed input projections + bias), h0 [B,P],
+    c0 [B,H], w_h [P,4H], w_proj [H,P].  Outputs: hp [B,T,P], c_T.
+    h0/c0 receive no grads (recurrent state is detached between steps —
+    truncated BPTT, reference language_model_graph.py:76-87).
+    """
+
+    @staticmethod
+    def forward(ctx, gx, h0, c0, w_h, w_proj, forget_bias):
+        ext = _ops.ext()
+        B, T, G = gx.shape
+        H = G // 4
+        P = w_proj.shape[1]
+        # one compute-dtype cast per sequence (autocast would cast per use)
+        ctx.w_dtype = w_h.dtype
+        w_h = w_h.to(gx.dtype)
+        w_proj = w_proj.to(gx.dtype)
+        h_in = torch.empty(T, B, P, dtype=gx.dtype, device=gx.device)
+        s_all = torch.empty(T, B, H, dtype=gx.dtype, device=gx.device)
+        c_all = torch.empty(T + 1, B, H, dtype=gx.dtype, device=gx.device)
+        acts = torch.empty(4, T, B, H, dtype=torch.float32,
+                           device=gx.device)
+        c_all[0] = c0.to(gx.dtype)
+        hp = h0.to(gx.dtype)
+        outs = torch.empty(B, T, P, dtype=gx.dtype, device=gx.device)
+        for t in range(T):
+            h_in[t] = hp
+            gates = torch.addmm(gx[:, t], hp, w_h)
+            h_t, c_t, i_s, g_t, f_s, o_s = ext.lstm_gates_fwd(
+                gates, c_all[t], float(forget_bias))
+            s_all[t] = h_t
+            c_all[t + 1] = c_t
+            acts[0, t], acts[1, t] = i_s, g_t
+            acts[2, t], acts[3, t] = f_s, o_s
+            hp = h_t @ w_proj
+            outs[:, t] = hp
+        ctx.save_for_backward(h_in, s_all, c_all, acts, w_h, w_proj)
+        ctx.fbias = forget_bias
+        return outs, c_all[T].clone()
+
+    @staticmethod
+    def backward(ctx, d_outs, d_cT):
+        ext = _ops.ext()
+        h_in, s_all, c_all, acts, w_h, w_proj = ctx.saved_tensors
+        T, B, P = h_in.shape
+        H = s_all.shape[2]
+        d_outs = d_outs.contiguous()
+        dG = torch.empty(T, B, 4 * H, dtype=d_outs.dtype,
+                         device=d_outs.device)
+        dHP = torch.empty(T, B, P, dtype=d_outs.dtype,
+                          device=d_outs.device)
+        w_h_t = w_h.t().contiguous()
+        w_proj_t = w_proj.t().contiguous()
+        carry_dhp = None
+        if d_cT is None:
+            dc = torch.zeros(B, H, dtype=d_outs.dtype,
+                             device=d_outs.device)
+        else:
+            dc = d_cT.to(d_outs.dtype).contiguous()
+        for t in range(T - 1, -1, -1):
+            dhp = d_outs[:, t]
+            if carry_dhp is not None:
+                dhp = dhp + carry_dhp
+            dHP[t] = dhp
+            ds = dhp @ w_proj_t
+            dgates, dc = ext.lstm_gates_bwd(
+                ds.contiguous(), dc, c_all[t], c_all[t + 1],
+                acts[0, t], acts[1, t], acts[2, t], acts[3, t])
+            dG[t] = dgates
+            carry_dhp = dgates @ w_h_t
+        # batched weight grads: one GEMM each over all timesteps
+        dG_flat = dG.reshape(T * B, 4 * H)
+        dw_h = h_in.reshape(T * B, P).t() @ dG_flat
+        dw_proj = s_all.reshape(T * B, H).t() @ dHP.reshape(T * B, P)
+        dgx = dG.permute(1, 0, 2)
+        return dgx, None, None, dw_h.to(ctx.w_dtype), \
+            dw_proj.to(ctx.w_dtype), None
+
+
 class ParallaxLSTM(nn.Module):
     """Single-layer projected LSTM (LM1B architecture: state_size H with a
     projection H -> P applied to h, so the recurrent input is P-dim).
@@ -118,6 +203,11 @@ class ParallaxLSTM(nn.Module):
         w_h = self.w[self.input_size:]           # [P, 4H]
         gx = torch.addmm(self.b, xs.reshape(B * T, self.input_size), w_x)
         gx = gx.view(B, T, 4 * self.state_size)
+        if (self.w_proj is not None and _ops.native_available()
+                and gx.is_cuda):
+            outs, c = _LSTMSeqFn.apply(gx, h, c, w_h, self.w_proj,
+                                       self.forget_bias)
+            return outs, (outs[:, -1], c)
         outs = []
         for t in range(T):
             gates = torch.addmm(gx[:, t], h, w_h)

@@ -212,6 +212,9 @@ def _run_worker(build_fn, config):
         redirect_stdout_stderr(config.redirect_path, worker_id)
 
     comm = CommContext(rank=worker_id, world_size=num_workers)
+    if comm.device.type == 'cuda':
+        # let MIOpen's find phase pick tuned conv solvers per shape
+        torch.backends.cudnn.benchmark = True
     # Per-rank RNG streams: data/candidate sampling must differ across
     # ranks (the rank-0 broadcast below unifies the model init).
     seed = int(os.environ.get('PARALLAX_SEED', '1234'))

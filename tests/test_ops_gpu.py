@@ -248,6 +248,33 @@ def test_lstm_gates_fwd_bwd_vs_ref():
     assert torch.allclose(c0.grad.cpu(), c0c.grad, atol=1e-5)
 
 
+def test_lstm_sequence_fn_matches_loop():
+    """The sequence-level LSTM Function (batched weight grads) must match
+    the eager per-step loop reference (CPU fp32) in outputs and grads."""
+    from parallax_amd.ops.lstm import ParallaxLSTM
+    torch.manual_seed(12)
+    B, T, I, H, P = 8, 6, 12, 20, 12
+    m_gpu = ParallaxLSTM(I, H, proj_size=P).to(DEV)
+    m_cpu = ParallaxLSTM(I, H, proj_size=P)
+    m_cpu.load_state_dict({k: v.cpu() for k, v in
+                           m_gpu.state_dict().items()})
+    xs = torch.randn(B, T, I)
+    h0 = torch.randn(B, P)
+    c0 = torch.randn(B, H)
+    out_g, (h_g, c_g) = m_gpu(xs.to(DEV), (h0.to(DEV), c0.to(DEV)))
+    out_c, (h_c, c_c) = m_cpu(xs, (h0, c0))
+    assert torch.allclose(out_g.cpu(), out_c, atol=1e-5)
+    assert torch.allclose(c_g.cpu(), c_c, atol=1e-5)
+    dloss = torch.randn(B, T, P)
+    (out_g * dloss.to(DEV)).sum().backward()
+    (out_c * dloss).sum().backward()
+    assert torch.allclose(m_gpu.w.grad.cpu(), m_cpu.w.grad, atol=1e-3,
+                          rtol=1e-4)
+    assert torch.allclose(m_gpu.b.grad.cpu(), m_cpu.b.grad, atol=1e-3)
+    assert torch.allclose(m_gpu.w_proj.grad.cpu(), m_cpu.w_proj.grad,
+                          atol=1e-3, rtol=1e-4)
+
+
 # --- end-to-end -------------------------------------------------------------
 
 def test_embedding_end_to_end_gpu():
