@@ -41,6 +41,9 @@ void sparse_adagrad(at::Tensor param, at::Tensor accum, at::Tensor ids,
                     at::Tensor grad_rows, double lr, double eps);
 void sparse_sgd(at::Tensor param, at::Tensor ids, at::Tensor grad_rows,
                 double lr);
+void multi_tensor_pack(std::vector<at::Tensor> tensors, at::Tensor flat);
+void multi_tensor_unpack(std::vector<at::Tensor> tensors, at::Tensor flat,
+                         double scale);
 
 // layernorm.hip
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
@@ -81,6 +84,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_scale_by_norm", &multi_tensor_scale_by_norm);
   m.def("sparse_adagrad", &sparse_adagrad);
   m.def("sparse_sgd", &sparse_sgd);
+  m.def("multi_tensor_pack", &multi_tensor_pack);
+  m.def("multi_tensor_unpack", &multi_tensor_unpack);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);

@@ -406,3 +406,90 @@ void sparse_sgd(at::Tensor param, at::Tensor ids, at::Tensor grad_rows,
         HIP_CHECK_KERNEL();
       });
 }
+
+// --- fusion-bucket pack/unpack ---------------------------------------------
+// One launch packs a bucket's per-param gradients into the flat bf16/fp32
+// fusion buffer (and one unpacks the reduced buffer back, scaled by
+// 1/world) — replaces per-parameter copy/div kernel pairs in the dense
+// all-reduce engine (2 x n_params launches -> 2 per <=32-tensor chunk).
+
+template <typename TSrc, typename TDst>
+__global__ void mt_pack_kernel(MTMeta meta, TDst* __restrict__ flat) {
+  const long total = meta.cum[meta.n];
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int ti = mt_find(meta, idx);
+    const long j = idx - meta.cum[ti];
+    flat[idx] = from_f32<TDst>(
+        to_f32<TSrc>(reinterpret_cast<const TSrc*>(meta.a[ti])[j]));
+  }
+}
+
+template <typename TSrc, typename TDst>
+__global__ void mt_unpack_kernel(MTMeta meta, const TSrc* __restrict__ flat,
+                                 float scale) {
+  const long total = meta.cum[meta.n];
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int ti = mt_find(meta, idx);
+    const long j = idx - meta.cum[ti];
+    reinterpret_cast<TDst*>(meta.a[ti])[j] =
+        from_f32<TDst>(to_f32<TSrc>(flat[idx]) * scale);
+  }
+}
+
+template <bool PACK>
+static void mt_pack_unpack(std::vector<at::Tensor>& tensors,
+                           at::Tensor& flat, double scale) {
+  check_contig(tensors, "mt_pack");
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  size_t i = 0;
+  long base = 0;
+  while (i < tensors.size()) {
+    MTMeta meta;
+    meta.n = 0;
+    meta.cum[0] = 0;
+    while (i < tensors.size() && meta.n < MT_MAX) {
+      int k = meta.n;
+      meta.a[k] = tensors[i].data_ptr();
+      meta.cum[k + 1] = meta.cum[k] + tensors[i].numel();
+      meta.n++;
+      i++;
+    }
+    const long total = meta.cum[meta.n];
+    AT_DISPATCH_FLOATING_TYPES_AND2(
+        at::kBFloat16, at::kHalf, tensors[0].scalar_type(), "mt_pack_t",
+        [&] {
+          using TP = scalar_t;
+          AT_DISPATCH_FLOATING_TYPES_AND2(
+              at::kBFloat16, at::kHalf, flat.scalar_type(), "mt_pack_f",
+              [&] {
+                using TF = scalar_t;
+                if (PACK) {
+                 hipLaunchKernelGGL(( mt_pack_kernel<TP, TF>)
+                      , dim3(px_grid(total)), dim3(PX_BLOCK), 0, stream, 
+                          meta,
+                          reinterpret_cast<TF*>(flat.data_ptr()) + base);
+                } else {
+                 hipLaunchKernelGGL(( mt_unpack_kernel<TF, TP>)
+                      , dim3(px_grid(total)), dim3(PX_BLOCK), 0, stream, 
+                          meta,
+                          reinterpret_cast<const TF*>(flat.data_ptr())
+                              + base,
+                          (float)scale);
+                }
+                HIP_CHECK_KERNEL();
+              });
+        });
+    base += total;
+  }
+}
+
+void multi_tensor_pack(std::vector<at::Tensor> tensors, at::Tensor flat) {
+  mt_pack_unpack<true>(tensors, flat, 1.0);
+}
+
+void multi_tensor_unpack(std::vector<at::Tensor> tensors, at::Tensor flat,
+                         double scale) {
+  mt_pack_unpack<false>(tensors, flat, scale);
+}

@@ -96,13 +96,30 @@ class DenseAllReduceEngine(object):
     def _on_grad_ready(self, p):
         bi, slot = self._param_bucket[id(p)]
         b = self._buckets[bi]
-        flat = self._ensure_flat(b, p)
-        off = b.offsets[slot]
-        flat[off:off + p.numel()].copy_(
-            p.grad.detach().reshape(-1).to(flat.dtype))
         b.pending -= 1
         if b.pending == 0:
+            self._pack(b)
             self._launch(b)
+
+    def _pack(self, b):
+        """Copy the bucket's gradients into its flat fusion buffer — ONE
+        multi_tensor_pack launch per <=32-tensor chunk on GPU (with the
+        bf16 cast fused), per-tensor copies otherwise."""
+        flat = self._ensure_flat(b, b.params[0])
+        grads = [p.grad for p in b.params]
+        from parallax_amd import ops as _ops
+        if (flat.is_cuda and _ops.native_available()
+                and all(g is not None for g in grads)
+                and len({g.dtype for g in grads}) == 1):
+            _ops.ext().multi_tensor_pack(
+                [g.detach() for g in grads], flat)
+            return
+        for slot, p in enumerate(b.params):
+            if p.grad is None:
+                continue
+            off = b.offsets[slot]
+            flat[off:off + p.numel()].copy_(
+                p.grad.detach().reshape(-1).to(flat.dtype))
 
     def _launch(self, b):
         if self.comm.world_size == 1:
@@ -125,15 +142,8 @@ class DenseAllReduceEngine(object):
             return
         if not self.overlap:
             for b in self._buckets:
-                flat = None
-                for slot, p in enumerate(b.params):
-                    if p.grad is None:
-                        continue
-                    flat = self._ensure_flat(b, p)
-                    off = b.offsets[slot]
-                    flat[off:off + p.numel()].copy_(
-                        p.grad.detach().reshape(-1).to(flat.dtype))
-                if flat is not None:
+                if any(p.grad is not None for p in b.params):
+                    self._pack(b)
                     self._launch(b)
         denom = float(self.comm.world_size) if self.average else 1.0
         for b in self._buckets:
@@ -144,6 +154,14 @@ class DenseAllReduceEngine(object):
                 b.work = None
             if self.comm.comm_stream is not None:
                 torch.cuda.current_stream().wait_stream(self.comm.comm_stream)
+            grads = [p.grad for p in b.params]
+            from parallax_amd import ops as _ops
+            if (b.flat.is_cuda and _ops.native_available()
+                    and all(g is not None for g in grads)
+                    and len({g.dtype for g in grads}) == 1):
+                _ops.ext().multi_tensor_unpack(
+                    [g.detach() for g in grads], b.flat, 1.0 / denom)
+                continue
             for slot, p in enumerate(b.params):
                 if p.grad is None:
                     continue
