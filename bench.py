@@ -38,7 +38,8 @@ def parse_args():
     p.add_argument('--gpus', type=int, default=1)
     p.add_argument('--steps', type=int, default=30)
     p.add_argument('--warmup', type=int, default=10)
-    p.add_argument('--model', choices=['lm1b', 'resnet50'], default='lm1b')
+    p.add_argument('--model', choices=['lm1b', 'resnet50', 'nmt'],
+                   default='lm1b')
     p.add_argument('--run-option', default='HYBRID')
     p.add_argument('--batch-size', type=int, default=None)
     p.add_argument('--partitions', type=int, default=None)
@@ -84,6 +85,18 @@ def main():
                        'parallelism': 'hybrid dp%d (dense AR + sparse PS)'
                                       % n}
         dtype = 'bf16' if on_gpu else 'fp32'
+    elif args.model == 'nmt':
+        from parallax_amd.models import nmt as M
+        batch = args.batch_size or 64
+        tgt_len = 48
+        build = lambda: M.build_nmt_graph(arch='transformer')
+        metric, unit = 'NMT tgt words/sec', 'words/s'
+        per_step_items = batch * tgt_len
+        baseline = None  # no published reference number
+        config_json = {'model': 'nmt-transformer', 'global_batch': batch * n,
+                       'src_len': 48, 'tgt_len': tgt_len,
+                       'parallelism': 'hybrid dp%d' % n}
+        dtype = 'bf16' if on_gpu else 'fp32'
     else:
         from parallax_amd.models import resnet as M
         batch = args.batch_size or 64
@@ -104,6 +117,8 @@ def main():
     def make_batch(state=None):
         if args.model == 'lm1b':
             return M.synthetic_batch(batch, seq, device=device, state=state)
+        if args.model == 'nmt':
+            return M.synthetic_batch(batch, device=device)
         return M.synthetic_batch(batch, device=device)
 
     if args.cuda_graph and on_gpu and args.model == 'lm1b':
@@ -153,7 +168,8 @@ def main():
             'ms_per_step': round(elapsed / args.steps * 1000.0, 3),
             'higher_is_better': True,
             'scaling': 'weak',
-            'vs_baseline': round(value / baseline, 4),
+            'vs_baseline': (round(value / baseline, 4)
+                            if baseline else None),
             'dtype': dtype,
             'data': 'synthetic',
             'config': config_json,

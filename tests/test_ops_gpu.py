@@ -248,6 +248,28 @@ def test_lstm_gates_fwd_bwd_vs_ref():
     assert torch.allclose(c0.grad.cpu(), c0c.grad, atol=1e-5)
 
 
+def test_multi_tensor_pack_unpack_roundtrip():
+    """Bucket pack (fp32 grads -> bf16 flat) + unpack (scaled write-back)
+    must round-trip, including >32-tensor buckets (chunking) and
+    channels_last layouts (storage-order sweep)."""
+    e = _ext()
+    torch.manual_seed(40)
+    tensors = [torch.randn(37, 5, device=DEV) for _ in range(35)]
+    tensors.append(torch.randn(4, 3, 2, 2, device=DEV)
+                   .to(memory_format=torch.channels_last))
+    total = sum(t.numel() for t in tensors)
+    flat = torch.zeros(total, device=DEV, dtype=torch.bfloat16)
+    e.multi_tensor_pack([t.detach() for t in tensors], flat)
+    outs = [torch.zeros_like(t) for t in tensors]
+    # preserve layouts
+    outs[-1] = outs[-1].to(memory_format=torch.channels_last)
+    e.multi_tensor_unpack(outs, flat, 0.5)
+    for t, o in zip(tensors, outs):
+        assert torch.allclose(o, (t.to(torch.bfloat16).float() * 0.5)
+                              .to(t.dtype), atol=1e-2), \
+            (o - t * 0.5).abs().max()
+
+
 def test_lstm_sequence_fn_matches_loop():
     """The sequence-level LSTM Function (batched weight grads) must match
     the eager per-step loop reference (CPU fp32) in outputs and grads."""
