@@ -128,7 +128,13 @@ def _spawn_workers(resource_info, config, p_to_test, stat_addr):
                 os.killpg(os.getpgid(p.pid), signal.SIGTERM)
             except (ProcessLookupError, OSError):
                 pass
+        if signum is not None:  # invoked as a signal handler: fail-stop
+            sys.exit(128 + signum)
 
+    # fail-stop process-group cleanup on interrupt (reference cleanup_mpi /
+    # cleanup_ps, mpi/runner.py:122-130, ps/runner.py:186-192)
+    signal.signal(signal.SIGINT, cleanup)
+    signal.signal(signal.SIGTERM, cleanup)
     return procs, cleanup
 
 
