@@ -60,6 +60,16 @@ std::vector<at::Tensor> softmax_ce_fwd(const at::Tensor& logits,
                                        const at::Tensor& labels);
 at::Tensor softmax_ce_bwd(const at::Tensor& logits, const at::Tensor& labels,
                           const at::Tensor& lse, const at::Tensor& dloss);
+std::vector<at::Tensor> sampled_ce_fwd(const at::Tensor& true_logits,
+                                       const at::Tensor& sampled,
+                                       const at::Tensor& row_corr,
+                                       const at::Tensor& col_corr);
+std::vector<at::Tensor> sampled_ce_bwd(const at::Tensor& true_logits,
+                                       const at::Tensor& sampled,
+                                       const at::Tensor& row_corr,
+                                       const at::Tensor& col_corr,
+                                       const at::Tensor& lse,
+                                       const at::Tensor& dloss);
 
 // lstm_cell.hip
 std::vector<at::Tensor> lstm_gates_fwd(const at::Tensor& gates,
@@ -90,6 +100,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);
   m.def("softmax_ce_bwd", &softmax_ce_bwd);
+  m.def("sampled_ce_fwd", &sampled_ce_fwd);
+  m.def("sampled_ce_bwd", &sampled_ce_bwd);
   m.def("lstm_gates_fwd", &lstm_gates_fwd);
   m.def("lstm_gates_bwd", &lstm_gates_bwd);
 }

@@ -17,8 +17,37 @@ import math
 
 import torch
 
+from parallax_amd import ops as _ops
 from parallax_amd.ops.embedding import embedding_lookup
 from parallax_amd.ops.softmax_ce import softmax_cross_entropy
+
+
+def _ops_native():
+    return _ops.native_available()
+
+
+class _SampledCEFn(torch.autograd.Function):
+    """Fused sampled-softmax CE (ops/hip/csrc/softmax_ce.hip):
+    loss[i] = LSE_j([t_i+rc_i | s_ij+cc_j]) - (t_i + rc_i)."""
+
+    @staticmethod
+    def forward(ctx, true_logits, sampled_logits, row_corr, col_corr):
+        loss, lse = _ops.ext().sampled_ce_fwd(true_logits, sampled_logits,
+                                              row_corr, col_corr)
+        ctx.save_for_backward(true_logits, sampled_logits, row_corr,
+                              col_corr, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        true_logits, sampled_logits, row_corr, col_corr, lse = \
+            ctx.saved_tensors
+        d_true, d_sampled = _ops.ext().sampled_ce_bwd(
+            true_logits, sampled_logits, row_corr, col_corr, lse,
+            dloss.contiguous())
+        d_row = d_true.to(row_corr.dtype)
+        d_col = d_sampled.sum(dim=0).to(col_corr.dtype)
+        return d_true, d_sampled, d_row, d_col
 
 
 def log_uniform_candidates(num_sampled, num_classes, device,
@@ -59,23 +88,29 @@ def sampled_softmax_loss(weight, bias, labels, inputs, num_sampled,
     sampled_w = rows(weight, sampled_ids)             # [S, D]
     true_logits = (inputs * true_w).sum(-1)           # [N]
     sampled_logits = inputs @ sampled_w.t()           # [N, S]
-    if bias is not None:
-        true_logits = true_logits + embedding_lookup(bias, labels)
-        sampled_logits = sampled_logits + embedding_lookup(bias, sampled_ids)
 
-    # subtract log expected counts (TF sampled-softmax correction)
-    true_logits = true_logits - torch.log(
+    # per-row / per-column logit corrections: + bias - log(expected_count)
+    # (TF sampled-softmax correction); tiny [N] / [S] vectors
+    row_corr = -torch.log(
         _expected_count(labels, num_classes, num_sampled).clamp_min(1e-20))
-    sampled_logits = sampled_logits - torch.log(
-        sampled_expected.clamp_min(1e-20))
+    col_corr = -torch.log(sampled_expected.clamp_min(1e-20))
+    if bias is not None:
+        row_corr = row_corr + embedding_lookup(bias, labels)
+        col_corr = col_corr + embedding_lookup(bias, sampled_ids)
 
+    if (not remove_accidental_hits and inputs.is_cuda
+            and _ops_native()):
+        # fused kernel: CE over [true+row_corr | sampled+col_corr] without
+        # materializing corrected copies of the [N, S] matrix
+        return _SampledCEFn.apply(true_logits, sampled_logits.contiguous(),
+                                  row_corr, col_corr)
+
+    true_logits = true_logits + row_corr.to(true_logits.dtype)
+    sampled_logits = sampled_logits + col_corr.to(sampled_logits.dtype)
     if remove_accidental_hits:
         hits = labels.unsqueeze(1) == sampled_ids.unsqueeze(0)  # [N, S]
         sampled_logits = sampled_logits.masked_fill(hits, -1e9)
 
-    # The fused CE kernel accumulates max/sum-exp in fp32 internally, so
-    # logits can stay in the GEMM dtype (bf16 under autocast) — avoids
-    # materializing an [N, 1+S] fp32 copy (84 MB at the LM1B shape).
     logits = torch.cat([true_logits.unsqueeze(1).to(sampled_logits.dtype),
                         sampled_logits], dim=1)
     target = torch.zeros(labels.shape[0], dtype=torch.long, device=device)

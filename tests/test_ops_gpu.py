@@ -248,6 +248,35 @@ def test_lstm_gates_fwd_bwd_vs_ref():
     assert torch.allclose(c0.grad.cpu(), c0c.grad, atol=1e-5)
 
 
+def test_sampled_ce_fused_vs_composed():
+    """Fused sampled-softmax CE kernel vs the composed fp32 reference
+    (add corrections, cat, cross_entropy) — values and all four grads."""
+    from parallax_amd.ops.sampled_softmax import _SampledCEFn
+    torch.manual_seed(44)
+    N, S = 64, 257
+    t = torch.randn(N, device=DEV, requires_grad=True)
+    sm = torch.randn(N, S, device=DEV, requires_grad=True)
+    rc = torch.randn(N, device=DEV, requires_grad=True)
+    cc = torch.randn(S, device=DEV, requires_grad=True)
+    loss = _SampledCEFn.apply(t, sm, rc, cc)
+
+    t2 = t.detach().cpu().clone().requires_grad_(True)
+    sm2 = sm.detach().cpu().clone().requires_grad_(True)
+    rc2 = rc.detach().cpu().clone().requires_grad_(True)
+    cc2 = cc.detach().cpu().clone().requires_grad_(True)
+    logits = torch.cat([(t2 + rc2).unsqueeze(1), sm2 + cc2], dim=1)
+    ref = torch.nn.functional.cross_entropy(
+        logits, torch.zeros(N, dtype=torch.long), reduction='none')
+    assert torch.allclose(loss.cpu(), ref.detach(), atol=1e-4)
+    dl = torch.randn(N)
+    loss.backward(dl.to(DEV))
+    ref.backward(dl)
+    assert torch.allclose(t.grad.cpu(), t2.grad, atol=1e-4)
+    assert torch.allclose(sm.grad.cpu(), sm2.grad, atol=1e-4)
+    assert torch.allclose(rc.grad.cpu(), rc2.grad, atol=1e-4)
+    assert torch.allclose(cc.grad.cpu(), cc2.grad, atol=1e-3)
+
+
 def test_multi_tensor_pack_unpack_roundtrip():
     """Bucket pack (fp32 grads -> bf16 flat) + unpack (scaled write-back)
     must round-trip, including >32-tensor buckets (chunking) and
