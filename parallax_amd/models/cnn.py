@@ -28,10 +28,12 @@ def _conv_bn_relu(cin, cout, k, s=1, p=None):
 
 
 class VGG16(nn.Module):
+    CFG = [64, 64, 'M', 128, 128, 'M', 256, 256, 256, 'M',
+           512, 512, 512, 'M', 512, 512, 512, 'M']
+
     def __init__(self, num_classes=1001):
         super().__init__()
-        cfg = [64, 64, 'M', 128, 128, 'M', 256, 256, 256, 'M',
-               512, 512, 512, 'M', 512, 512, 512, 'M']
+        cfg = self.CFG
         layers, cin = [], 3
         for v in cfg:
             if v == 'M':
@@ -49,6 +51,11 @@ class VGG16(nn.Module):
 
     def forward(self, x):
         return self.classifier(self.features(x))
+
+
+class VGG19(VGG16):
+    CFG = [64, 64, 'M', 128, 128, 'M', 256, 256, 256, 256, 'M',
+           512, 512, 512, 512, 'M', 512, 512, 512, 512, 'M']
 
 
 class AlexNet(nn.Module):
@@ -153,7 +160,11 @@ MODEL_REGISTRY = {
     'resnet50': lambda nc: ResNet(50, nc),
     'resnet101': lambda nc: ResNet(101, nc),
     'resnet152': lambda nc: ResNet(152, nc),
+    'resnet50_v2': lambda nc: ResNet(50, nc, version=2),
+    'resnet101_v2': lambda nc: ResNet(101, nc, version=2),
+    'resnet152_v2': lambda nc: ResNet(152, nc, version=2),
     'vgg16': lambda nc: VGG16(nc),
+    'vgg19': lambda nc: VGG19(nc),
     'alexnet': lambda nc: AlexNet(nc),
     'overfeat': lambda nc: Overfeat(nc),
     'googlenet': lambda nc: GoogLeNet(nc),

@@ -51,29 +51,68 @@ class Bottleneck(nn.Module):
         return self.relu(out + identity)
 
 
-class ResNet(nn.Module):
-    def __init__(self, depth=50, num_classes=1001, bn_momentum=0.001):
+class BottleneckV2(nn.Module):
+    """Pre-activation bottleneck (ResNet v2, reference
+    resnet_model.py:129-180)."""
+
+    expansion = 4
+
+    def __init__(self, cin, width, stride=1, bn_momentum=0.001):
         super().__init__()
+        cout = width * self.expansion
+        self.bn1 = nn.BatchNorm2d(cin, momentum=bn_momentum)
+        self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
+        self.bn2 = nn.BatchNorm2d(width, momentum=bn_momentum)
+        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
+                               bias=False)
+        self.bn3 = nn.BatchNorm2d(width, momentum=bn_momentum)
+        self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
+        self.relu = nn.ReLU(inplace=True)
+        if stride != 1 or cin != cout:
+            self.downsample = nn.Conv2d(cin, cout, 1, stride=stride,
+                                        bias=False)
+        else:
+            self.downsample = None
+
+    def forward(self, x):
+        pre = self.relu(self.bn1(x))
+        identity = x if self.downsample is None else self.downsample(pre)
+        out = self.conv1(pre)
+        out = self.conv2(self.relu(self.bn2(out)))
+        out = self.conv3(self.relu(self.bn3(out)))
+        return out + identity
+
+
+class ResNet(nn.Module):
+    def __init__(self, depth=50, num_classes=1001, bn_momentum=0.001,
+                 version=1):
+        super().__init__()
+        self.version = version
         layers = _LAYERS[depth]
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = nn.BatchNorm2d(64, momentum=bn_momentum)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        block = Bottleneck if version == 1 else BottleneckV2
         cin, stages = 64, []
         for i, (width, n) in enumerate(zip((64, 128, 256, 512), layers)):
             blocks = []
             for j in range(n):
                 stride = 2 if (i > 0 and j == 0) else 1
-                blocks.append(Bottleneck(cin, width, stride, bn_momentum))
-                cin = width * Bottleneck.expansion
+                blocks.append(block(cin, width, stride, bn_momentum))
+                cin = width * block.expansion
             stages.append(nn.Sequential(*blocks))
         self.layer1, self.layer2, self.layer3, self.layer4 = stages
+        if version == 2:
+            self.bn_final = nn.BatchNorm2d(cin, momentum=bn_momentum)
         self.avgpool = nn.AdaptiveAvgPool2d(1)
-        self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
+        self.fc = nn.Linear(512 * block.expansion, num_classes)
 
     def forward(self, x):
         x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        if self.version == 2:
+            x = self.relu(self.bn_final(x))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)
 
