@@ -51,7 +51,9 @@ def main():
                               stats['grad_bytes'] / 1e6)
     prev = time.time()
     for step in range(args.max_steps):
-        x, y = resnet.synthetic_batch(args.batch_size, device=device)
+        x, y = resnet.synthetic_batch(
+            args.batch_size, image_size=cnn.image_size_for(args.model),
+            device=device)
         out = sess.run((x, y))
         if step % args.log_frequency == 0 and worker_id == 0:
             now = time.time()

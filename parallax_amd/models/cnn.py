@@ -205,3 +205,143 @@ def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
         return loss, {}
 
     return TrainGraph(model, opt, step_fn)
+
+
+# --- Inception v3 (reference models/inception_model.py) ---------------------
+
+class _IncA(nn.Module):
+    def __init__(self, cin, pool_ch):
+        super().__init__()
+        self.b1 = _conv_bn_relu(cin, 64, 1)
+        self.b5 = nn.Sequential(_conv_bn_relu(cin, 48, 1),
+                                _conv_bn_relu(48, 64, 5))
+        self.b3 = nn.Sequential(_conv_bn_relu(cin, 64, 1),
+                                _conv_bn_relu(64, 96, 3),
+                                _conv_bn_relu(96, 96, 3))
+        self.bp = nn.Sequential(nn.AvgPool2d(3, 1, 1),
+                                _conv_bn_relu(cin, pool_ch, 1))
+
+    def forward(self, x):
+        return torch.cat([self.b1(x), self.b5(x), self.b3(x), self.bp(x)],
+                         1)
+
+
+class _IncB(nn.Module):  # grid reduction 35->17
+    def __init__(self, cin):
+        super().__init__()
+        self.b3 = _conv_bn_relu(cin, 384, 3, s=2, p=0)
+        self.bd = nn.Sequential(_conv_bn_relu(cin, 64, 1),
+                                _conv_bn_relu(64, 96, 3),
+                                _conv_bn_relu(96, 96, 3, s=2, p=0))
+        self.mp = nn.MaxPool2d(3, 2)
+
+    def forward(self, x):
+        return torch.cat([self.b3(x), self.bd(x), self.mp(x)], 1)
+
+
+class _IncC(nn.Module):  # factorized 7x7
+    def __init__(self, cin, ch7):
+        super().__init__()
+        def c7(cin_, cout, horiz):
+            k, p = ((1, 7), (0, 3)) if horiz else ((7, 1), (3, 0))
+            return nn.Sequential(nn.Conv2d(cin_, cout, k, padding=p,
+                                           bias=False),
+                                 nn.BatchNorm2d(cout, momentum=0.001),
+                                 nn.ReLU(inplace=True))
+        self.b1 = _conv_bn_relu(cin, 192, 1)
+        self.b7 = nn.Sequential(_conv_bn_relu(cin, ch7, 1),
+                                c7(ch7, ch7, True), c7(ch7, 192, False))
+        self.b77 = nn.Sequential(_conv_bn_relu(cin, ch7, 1),
+                                 c7(ch7, ch7, False), c7(ch7, ch7, True),
+                                 c7(ch7, ch7, False), c7(ch7, 192, True))
+        self.bp = nn.Sequential(nn.AvgPool2d(3, 1, 1),
+                                _conv_bn_relu(cin, 192, 1))
+
+    def forward(self, x):
+        return torch.cat([self.b1(x), self.b7(x), self.b77(x),
+                          self.bp(x)], 1)
+
+
+class _IncD(nn.Module):  # grid reduction 17->8
+    def __init__(self, cin):
+        super().__init__()
+        self.b3 = nn.Sequential(_conv_bn_relu(cin, 192, 1),
+                                _conv_bn_relu(192, 320, 3, s=2, p=0))
+        self.b7 = nn.Sequential(
+            _conv_bn_relu(cin, 192, 1),
+            nn.Conv2d(192, 192, (1, 7), padding=(0, 3), bias=False),
+            nn.BatchNorm2d(192, momentum=0.001), nn.ReLU(inplace=True),
+            nn.Conv2d(192, 192, (7, 1), padding=(3, 0), bias=False),
+            nn.BatchNorm2d(192, momentum=0.001), nn.ReLU(inplace=True),
+            _conv_bn_relu(192, 192, 3, s=2, p=0))
+        self.mp = nn.MaxPool2d(3, 2)
+
+    def forward(self, x):
+        return torch.cat([self.b3(x), self.b7(x), self.mp(x)], 1)
+
+
+class _IncE(nn.Module):  # expanded 8x8 blocks
+    def __init__(self, cin):
+        super().__init__()
+        self.b1 = _conv_bn_relu(cin, 320, 1)
+        self.b3_stem = _conv_bn_relu(cin, 384, 1)
+        self.b3_a = nn.Sequential(
+            nn.Conv2d(384, 384, (1, 3), padding=(0, 1), bias=False),
+            nn.BatchNorm2d(384, momentum=0.001), nn.ReLU(inplace=True))
+        self.b3_b = nn.Sequential(
+            nn.Conv2d(384, 384, (3, 1), padding=(1, 0), bias=False),
+            nn.BatchNorm2d(384, momentum=0.001), nn.ReLU(inplace=True))
+        self.bd_stem = nn.Sequential(_conv_bn_relu(cin, 448, 1),
+                                     _conv_bn_relu(448, 384, 3))
+        self.bd_a = nn.Sequential(
+            nn.Conv2d(384, 384, (1, 3), padding=(0, 1), bias=False),
+            nn.BatchNorm2d(384, momentum=0.001), nn.ReLU(inplace=True))
+        self.bd_b = nn.Sequential(
+            nn.Conv2d(384, 384, (3, 1), padding=(1, 0), bias=False),
+            nn.BatchNorm2d(384, momentum=0.001), nn.ReLU(inplace=True))
+        self.bp = nn.Sequential(nn.AvgPool2d(3, 1, 1),
+                                _conv_bn_relu(cin, 192, 1))
+
+    def forward(self, x):
+        s3 = self.b3_stem(x)
+        sd = self.bd_stem(x)
+        return torch.cat([self.b1(x), self.b3_a(s3), self.b3_b(s3),
+                          self.bd_a(sd), self.bd_b(sd), self.bp(x)], 1)
+
+
+class InceptionV3(nn.Module):
+    """Inception-v3 (299x299 input)."""
+
+    def __init__(self, num_classes=1001):
+        super().__init__()
+        self.stem = nn.Sequential(
+            _conv_bn_relu(3, 32, 3, s=2, p=0),
+            _conv_bn_relu(32, 32, 3, p=0),
+            _conv_bn_relu(32, 64, 3),
+            nn.MaxPool2d(3, 2),
+            _conv_bn_relu(64, 80, 1),
+            _conv_bn_relu(80, 192, 3, p=0),
+            nn.MaxPool2d(3, 2))
+        self.blocks = nn.Sequential(
+            _IncA(192, 32), _IncA(256, 64), _IncA(288, 64),
+            _IncB(288),
+            _IncC(768, 128), _IncC(768, 160), _IncC(768, 160),
+            _IncC(768, 192),
+            _IncD(768),
+            _IncE(1280), _IncE(2048))
+        self.head = nn.Sequential(nn.AdaptiveAvgPool2d(1), nn.Flatten(),
+                                  nn.Dropout(0.2),
+                                  nn.Linear(2048, num_classes))
+
+    def forward(self, x):
+        return self.head(self.blocks(self.stem(x)))
+
+
+MODEL_REGISTRY['inception3'] = lambda nc: InceptionV3(nc)
+
+# per-model input resolution (reference models/*.py image_size)
+MODEL_IMAGE_SIZE = {'overfeat': 231, 'inception3': 299}
+
+
+def image_size_for(model_name):
+    return MODEL_IMAGE_SIZE.get(model_name, 224)

@@ -83,12 +83,12 @@ def test_skip_thoughts_step(monkeypatch, free_port):
     sess.close()
 
 
-@pytest.mark.parametrize('name', ['vgg16', 'alexnet', 'overfeat',
-                                  'googlenet', 'trivial'])
+@pytest.mark.parametrize('name', ['vgg16', 'vgg19', 'alexnet', 'overfeat',
+                                  'googlenet', 'inception3', 'trivial'])
 def test_cnn_family_forward(name):
     torch.manual_seed(0)
     m = cnn.MODEL_REGISTRY[name](11)
-    size = 231 if name == 'overfeat' else 224
+    size = cnn.image_size_for(name)
     x = torch.randn(2, 3, size, size)
     out = m(x)
     assert out.shape == (2, 11)
