@@ -137,6 +137,8 @@ std::vector<at::Tensor> sampled_ce_fwd(const at::Tensor& true_logits,
                                        const at::Tensor& col_corr) {
   TORCH_CHECK(sampled.is_cuda() && sampled.dim() == 2 &&
               sampled.is_contiguous());
+  TORCH_CHECK(true_logits.scalar_type() == sampled.scalar_type(),
+              "sampled_ce: true/sampled dtype mismatch");
   const long n = sampled.size(0), s = sampled.size(1);
   auto loss = at::empty({n}, sampled.options().dtype(at::kFloat));
   auto lse = at::empty({n}, sampled.options().dtype(at::kFloat));
@@ -186,6 +188,8 @@ std::vector<at::Tensor> sampled_ce_bwd(const at::Tensor& true_logits,
                                        const at::Tensor& col_corr,
                                        const at::Tensor& lse,
                                        const at::Tensor& dloss) {
+  TORCH_CHECK(true_logits.scalar_type() == sampled.scalar_type(),
+              "sampled_ce_bwd: true/sampled dtype mismatch");
   const long n = sampled.size(0), s = sampled.size(1);
   auto d_true = at::empty_like(true_logits);
   auto d_sampled = at::empty_like(sampled);

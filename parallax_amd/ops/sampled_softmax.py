@@ -101,8 +101,11 @@ def sampled_softmax_loss(weight, bias, labels, inputs, num_sampled,
     if (not remove_accidental_hits and inputs.is_cuda
             and _ops_native()):
         # fused kernel: CE over [true+row_corr | sampled+col_corr] without
-        # materializing corrected copies of the [N, S] matrix
-        return _SampledCEFn.apply(true_logits, sampled_logits.contiguous(),
+        # materializing corrected copies of the [N, S] matrix.  Note:
+        # autocast computes the row-wise sum in fp32, so true_logits must
+        # be cast to the GEMM dtype the kernel dispatches on.
+        return _SampledCEFn.apply(true_logits.to(sampled_logits.dtype),
+                                  sampled_logits.contiguous(),
                                   row_corr, col_corr)
 
     true_logits = true_logits + row_corr.to(true_logits.dtype)
