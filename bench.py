@@ -148,6 +148,12 @@ def main():
     t1 = time.perf_counter()
     sess.comm.barrier()
 
+    final_loss = float(out['loss'])
+    if not (final_loss == final_loss and abs(final_loss) < 1e30):
+        print('BENCH INVALID: non-finite loss %r after timed steps'
+              % final_loss, file=sys.stderr)
+        sys.exit(1)
+
     elapsed = torch.tensor([t1 - t0], dtype=torch.float64,
                            device=device if sess.comm.backend == 'nccl'
                            else 'cpu')
