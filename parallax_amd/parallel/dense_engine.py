@@ -140,11 +140,23 @@ class DenseAllReduceEngine(object):
         without (or world_size==1) reduce everything now."""
         if self.comm.world_size == 1:
             return
-        if not self.overlap:
-            for b in self._buckets:
-                if any(p.grad is not None for p in b.params):
-                    self._pack(b)
-                    self._launch(b)
+        self.launch_async()
+        self.wait_and_unpack()
+
+    def launch_async(self):
+        """Pack + launch every bucket's all-reduce (async, comm stream)
+        without waiting — lets finish_step overlap the sparse exchange
+        with the dense collectives."""
+        if self.comm.world_size == 1 or self.overlap:
+            return  # overlap mode already launched from the hooks
+        for b in self._buckets:
+            if any(p.grad is not None for p in b.params):
+                self._pack(b)
+                self._launch(b)
+
+    def wait_and_unpack(self):
+        if self.comm.world_size == 1:
+            return
         denom = float(self.comm.world_size) if self.average else 1.0
         for b in self._buckets:
             if b.flat is None:

@@ -91,12 +91,15 @@ class HybridEngine(object):
             self.sparse_engine.prepare_step()
 
     def finish_step(self):
-        """After backward: finish dense comm, run sparse exchange + row
-        updates, then the fused dense optimizer step."""
+        """After backward: launch the dense all-reduces (async on the comm
+        stream), run the sparse exchange + row updates UNDER them, then
+        wait/unpack and take the fused dense optimizer step."""
         if self.dense_engine:
-            self.dense_engine.reduce_gradients()
+            self.dense_engine.launch_async()
         if self.sparse_engine:
             self.sparse_engine.reduce_and_apply()
+        if self.dense_engine:
+            self.dense_engine.wait_and_unpack()
         self.optimizer.step()
 
     def close(self):

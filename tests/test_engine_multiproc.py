@@ -249,6 +249,36 @@ def _body_comm_collectives(rank, world):
         assert torch.all(recv[r] == 10 * r + rank)
 
 
+def _body_lm1b_tiny_full_stack(rank, world):
+    """Tiny LM1B through the COMPLETE per-step protocol on several ranks:
+    partitioned tables, sampled softmax, clip, EMA, dense AR, sparse PS —
+    the collective-order rehearsal for the 8-GPU node."""
+    import parallax_amd
+    from parallax_amd.config import ParallaxConfig
+    from parallax_amd.models import lm1b as M
+
+    sess, num_workers, worker_id, _ = parallax_amd.parallel_run(
+        lambda: M.build_lm1b_graph(batch_size=4, num_steps=3,
+                                   vocab_size=300, emb_dim=16,
+                                   state_size=32, proj_size=16,
+                                   num_sampled=32, min_partitions=4,
+                                   lr=0.3),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    state = None
+    for i in range(8):
+        out = sess.run(M.synthetic_batch(4, 3, 300, state=state))
+        state = out.get('state')
+        assert float(out['loss']) == float(out['loss'])
+    # all replicated state identical across ranks at the end
+    m = sess.graph.model
+    for t in (m.emb.weight.detach(), m.softmax_b.detach(),
+              m.lstm.w.detach(), m.ema_0.detach()):
+        ts = [torch.zeros_like(t) for _ in range(world)]
+        torch.distributed.all_gather(ts, t)
+        assert torch.allclose(ts[0], ts[-1], atol=1e-5)
+    sess.close()
+
+
 # ---------------------------------------------------------------------------
 
 @pytest.mark.parametrize('body', [
@@ -261,3 +291,8 @@ def _body_comm_collectives(rank, world):
         'parallel_run_hybrid'])
 def test_multiproc(body):
     _spawn(body, get_free_port())
+
+
+def test_lm1b_full_stack_world4():
+    _spawn(_body_lm1b_tiny_full_stack, get_free_port(), world=4,
+           timeout=300)
