@@ -79,6 +79,15 @@ std::vector<at::Tensor> lstm_gates_bwd(
     const at::Tensor& dh, const at::Tensor& dc_in, const at::Tensor& c_prev,
     const at::Tensor& c_out, const at::Tensor& i_s, const at::Tensor& g_t,
     const at::Tensor& f_s, const at::Tensor& o_s);
+void lstm_gates_fwd_out(const at::Tensor& gates, const at::Tensor& c_prev,
+                        double forget_bias, at::Tensor h_out,
+                        at::Tensor c_out, at::Tensor i_s, at::Tensor g_t,
+                        at::Tensor f_s, at::Tensor o_s);
+void lstm_gates_bwd_out(const at::Tensor& dh, const at::Tensor& dc_in,
+                        const at::Tensor& c_prev, const at::Tensor& c_out,
+                        const at::Tensor& i_s, const at::Tensor& g_t,
+                        const at::Tensor& f_s, const at::Tensor& o_s,
+                        at::Tensor dgates_out, at::Tensor dc_prev_out);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rows_gather", &rows_gather, "embedding row gather (HIP)");
@@ -104,4 +113,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sampled_ce_bwd", &sampled_ce_bwd);
   m.def("lstm_gates_fwd", &lstm_gates_fwd);
   m.def("lstm_gates_bwd", &lstm_gates_bwd);
+  m.def("lstm_gates_fwd_out", &lstm_gates_fwd_out);
+  m.def("lstm_gates_bwd_out", &lstm_gates_bwd_out);
 }

@@ -138,3 +138,59 @@ std::vector<at::Tensor> lstm_gates_bwd(const at::Tensor& dh,
       });
   return {dgates, dc_prev};
 }
+
+// out-variants: write into caller-provided buffers (the sequence-level
+// LSTM Function pre-allocates [T, B, H] stacks and passes per-step slices,
+// so the time loop issues ZERO extra copy kernels).
+
+void lstm_gates_fwd_out(const at::Tensor& gates, const at::Tensor& c_prev,
+                        double forget_bias, at::Tensor h_out,
+                        at::Tensor c_out, at::Tensor i_s, at::Tensor g_t,
+                        at::Tensor f_s, at::Tensor o_s) {
+  const long B = gates.size(0);
+  const long H = gates.size(1) / 4;
+  TORCH_CHECK(gates.is_contiguous() && c_prev.is_contiguous() &&
+              h_out.is_contiguous() && c_out.is_contiguous());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, gates.scalar_type(), "lstm_gates_fwd_out",
+      [&] {
+        lstm_gates_fwd_kernel<scalar_t>
+            <<<px_grid(B * H), PX_BLOCK, 0, stream>>>(
+                reinterpret_cast<const scalar_t*>(gates.data_ptr()),
+                reinterpret_cast<const scalar_t*>(c_prev.data_ptr()),
+                reinterpret_cast<scalar_t*>(h_out.data_ptr()),
+                reinterpret_cast<scalar_t*>(c_out.data_ptr()),
+                i_s.data_ptr<float>(), g_t.data_ptr<float>(),
+                f_s.data_ptr<float>(), o_s.data_ptr<float>(), B * H, H,
+                (float)forget_bias);
+        HIP_CHECK_KERNEL();
+      });
+}
+
+void lstm_gates_bwd_out(const at::Tensor& dh, const at::Tensor& dc_in,
+                        const at::Tensor& c_prev, const at::Tensor& c_out,
+                        const at::Tensor& i_s, const at::Tensor& g_t,
+                        const at::Tensor& f_s, const at::Tensor& o_s,
+                        at::Tensor dgates_out, at::Tensor dc_prev_out) {
+  const long B = dh.size(0), H = dh.size(1);
+  TORCH_CHECK(dh.is_contiguous() && dc_in.is_contiguous() &&
+              dgates_out.is_contiguous() && dc_prev_out.is_contiguous());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, dh.scalar_type(), "lstm_gates_bwd_out",
+      [&] {
+        lstm_gates_bwd_kernel<scalar_t>
+            <<<px_grid(B * H), PX_BLOCK, 0, stream>>>(
+                reinterpret_cast<const scalar_t*>(dh.data_ptr()),
+                reinterpret_cast<const scalar_t*>(dc_in.data_ptr()),
+                reinterpret_cast<const scalar_t*>(c_prev.data_ptr()),
+                reinterpret_cast<const scalar_t*>(c_out.data_ptr()),
+                i_s.data_ptr<float>(), g_t.data_ptr<float>(),
+                f_s.data_ptr<float>(), o_s.data_ptr<float>(),
+                reinterpret_cast<scalar_t*>(dgates_out.data_ptr()),
+                reinterpret_cast<scalar_t*>(dc_prev_out.data_ptr()), B * H,
+                H);
+        HIP_CHECK_KERNEL();
+      });
+}

@@ -124,7 +124,10 @@ at::Tensor segment_sum_rows(const at::Tensor& values, const at::Tensor& order,
       at::kBFloat16, at::kHalf, v.scalar_type(), "segment_sum_rows", [&] {
         using T = scalar_t;
         auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-        int grid = px_grid(u * d);
+        // one element per thread: the per-element chain is two DEPENDENT
+        // loads (order[k] -> values row), so latency must be hidden by
+        // thread-level parallelism — do not cap the grid at PX_MAX_GRID
+        int grid = px_grid(u * d, PX_BLOCK, 131072);
        hipLaunchKernelGGL(( segment_sum_rows_kernel<T>)
             , dim3(grid), dim3(PX_BLOCK), 0, stream, 
                 reinterpret_cast<const T*>(v.data_ptr()),

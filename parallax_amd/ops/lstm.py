@@ -91,62 +91,61 @@ class _LSTMSeqFn(torch.autograd.Function):
         ctx.w_dtype = w_h.dtype
         w_h = w_h.to(gx.dtype)
         w_proj = w_proj.to(gx.dtype)
-        h_in = torch.empty(T, B, P, dtype=gx.dtype, device=gx.device)
+        # pre-allocated stacks; every per-step kernel/GEMM writes straight
+        # into its slice (out-variant gate kernels + mm(out=...)) — the
+        # time loop issues NO copy kernels
+        hp_all = torch.empty(T + 1, B, P, dtype=gx.dtype, device=gx.device)
         s_all = torch.empty(T, B, H, dtype=gx.dtype, device=gx.device)
         c_all = torch.empty(T + 1, B, H, dtype=gx.dtype, device=gx.device)
         acts = torch.empty(4, T, B, H, dtype=torch.float32,
                            device=gx.device)
         c_all[0] = c0.to(gx.dtype)
-        hp = h0.to(gx.dtype)
-        outs = torch.empty(B, T, P, dtype=gx.dtype, device=gx.device)
+        hp_all[0] = h0.to(gx.dtype)
         for t in range(T):
-            h_in[t] = hp
-            gates = torch.addmm(gx[:, t], hp, w_h)
-            h_t, c_t, i_s, g_t, f_s, o_s = ext.lstm_gates_fwd(
-                gates, c_all[t], float(forget_bias))
-            s_all[t] = h_t
-            c_all[t + 1] = c_t
-            acts[0, t], acts[1, t] = i_s, g_t
-            acts[2, t], acts[3, t] = f_s, o_s
-            hp = h_t @ w_proj
-            outs[:, t] = hp
-        ctx.save_for_backward(h_in, s_all, c_all, acts, w_h, w_proj)
+            gates = torch.addmm(gx[:, t], hp_all[t], w_h)
+            ext.lstm_gates_fwd_out(gates, c_all[t], float(forget_bias),
+                                   s_all[t], c_all[t + 1], acts[0, t],
+                                   acts[1, t], acts[2, t], acts[3, t])
+            torch.mm(s_all[t], w_proj, out=hp_all[t + 1])
+        ctx.save_for_backward(hp_all, s_all, c_all, acts, w_h, w_proj)
         ctx.fbias = forget_bias
-        return outs, c_all[T].clone()
+        # outs view [B, T, P]: non-contiguous transpose of hp_all[1:]; the
+        # consumer (reshape for the softmax GEMM) makes it contiguous once
+        return hp_all[1:].transpose(0, 1), c_all[T].clone()
 
     @staticmethod
     def backward(ctx, d_outs, d_cT):
         ext = _ops.ext()
-        h_in, s_all, c_all, acts, w_h, w_proj = ctx.saved_tensors
-        T, B, P = h_in.shape
-        H = s_all.shape[2]
-        d_outs = d_outs.contiguous()
-        dG = torch.empty(T, B, 4 * H, dtype=d_outs.dtype,
-                         device=d_outs.device)
-        dHP = torch.empty(T, B, P, dtype=d_outs.dtype,
-                          device=d_outs.device)
+        hp_all, s_all, c_all, acts, w_h, w_proj = ctx.saved_tensors
+        T, B, H = s_all.shape
+        P = hp_all.shape[2]
+        dG = torch.empty(T, B, 4 * H, dtype=w_h.dtype, device=w_h.device)
+        dHP = torch.empty(T, B, P, dtype=w_h.dtype, device=w_h.device)
+        ds = torch.empty(B, H, dtype=w_h.dtype, device=w_h.device)
+        dc_a = torch.empty(B, H, dtype=w_h.dtype, device=w_h.device)
+        dc_b = torch.empty(B, H, dtype=w_h.dtype, device=w_h.device)
         w_h_t = w_h.t().contiguous()
         w_proj_t = w_proj.t().contiguous()
-        carry_dhp = None
         if d_cT is None:
-            dc = torch.zeros(B, H, dtype=d_outs.dtype,
-                             device=d_outs.device)
+            dc_a.zero_()
         else:
-            dc = d_cT.to(d_outs.dtype).contiguous()
+            dc_a.copy_(d_cT)
+        dc, dc_next = dc_a, dc_b
         for t in range(T - 1, -1, -1):
-            dhp = d_outs[:, t]
-            if carry_dhp is not None:
-                dhp = dhp + carry_dhp
-            dHP[t] = dhp
-            ds = dhp @ w_proj_t
-            dgates, dc = ext.lstm_gates_bwd(
-                ds.contiguous(), dc, c_all[t], c_all[t + 1],
-                acts[0, t], acts[1, t], acts[2, t], acts[3, t])
-            dG[t] = dgates
-            carry_dhp = dgates @ w_h_t
+            if t == T - 1:
+                dHP[t].copy_(d_outs[:, t])
+            else:
+                # dhp = upstream slice + recurrent carry (dG[t+1] @ w_h^T)
+                torch.add(d_outs[:, t].to(w_h.dtype),
+                          torch.mm(dG[t + 1], w_h_t), out=dHP[t])
+            torch.mm(dHP[t], w_proj_t, out=ds)
+            ext.lstm_gates_bwd_out(ds, dc, c_all[t], c_all[t + 1],
+                                   acts[0, t], acts[1, t], acts[2, t],
+                                   acts[3, t], dG[t], dc_next)
+            dc, dc_next = dc_next, dc
         # batched weight grads: one GEMM each over all timesteps
         dG_flat = dG.reshape(T * B, 4 * H)
-        dw_h = h_in.reshape(T * B, P).t() @ dG_flat
+        dw_h = hp_all[:T].reshape(T * B, P).t() @ dG_flat
         dw_proj = s_all.reshape(T * B, H).t() @ dHP.reshape(T * B, P)
         dgx = dG.permute(1, 0, 2)
         return dgx, None, None, dw_h.to(ctx.w_dtype), \
