@@ -79,14 +79,43 @@ at::Tensor rows_gather(const at::Tensor& weight, const at::Tensor& ids) {
 // segment_sum_rows
 // ---------------------------------------------------------------------------
 
-// Flat (segment, column) indexing: one thread per output element with the
-// per-segment gather loop inside.  Segments from a deduplicated gradient
-// are short (few duplicates per id), so the inner loop is 1-3 iterations;
-// flat indexing keeps all 64 lanes busy for ANY d (the block-per-segment
-// form wastes 255/256 threads on d=1 bias tables).
+// Scatter-atomic formulation: one thread per INPUT element (k, c) adds its
+// value into the fp32 accumulator of its segment.  LOAD-BALANCED for any
+// duplication profile — Zipfian candidate ids produce segments hundreds of
+// entries long, which serialize a per-OUTPUT-element gather loop (measured
+// 147 us vs ~20 us for this form at the LM1B softmax_w shape).  The
+// segment of entry k is found by binary search over seg_starts (wave-
+// uniform for 64 consecutive lanes of one k).  Atomic contention only
+// occurs between duplicates of one id, spread across d columns.
 template <typename T>
-__global__ void segment_sum_rows_kernel(const T* __restrict__ values,
-                                        const int* __restrict__ order,
+__global__ void segment_sum_scatter_kernel(
+    const T* __restrict__ values, const int* __restrict__ order,
+    const int* __restrict__ seg_starts, float* __restrict__ acc, long n,
+    long u, long d) {
+  const long total = n * d;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long k = idx / d;
+    const long c = idx - k * d;
+    // binary search: greatest s with seg_starts[s] <= k
+    int lo = 0, hi = (int)u - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (seg_starts[mid] <= (int)k) lo = mid; else hi = mid - 1;
+    }
+    const float v = to_f32<T>(values[(long)order[k] * d + c]);
+    const int beg = seg_starts[lo];
+    const int end = seg_starts[lo + 1];
+    if (end - beg == 1) {
+      acc[(long)lo * d + c] = v;  // unique id: plain store
+    } else {
+      atomicAdd(&acc[(long)lo * d + c], v);
+    }
+  }
+}
+
+template <typename T>
+__global__ void segment_finalize_kernel(const float* __restrict__ acc,
                                         const int* __restrict__ seg_starts,
                                         T* __restrict__ out, long u, long d,
                                         bool average) {
@@ -94,15 +123,11 @@ __global__ void segment_sum_rows_kernel(const T* __restrict__ values,
   for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
     const long s = idx / d;
-    const long c = idx - s * d;
-    const int beg = seg_starts[s];
-    const int end = seg_starts[s + 1];
-    float acc = 0.0f;
-    for (int k = beg; k < end; ++k) {
-      acc += to_f32<T>(values[(long)order[k] * d + c]);
+    float v = acc[idx];
+    if (average) {
+      v *= 1.0f / (float)(seg_starts[s + 1] - seg_starts[s]);
     }
-    if (average) acc *= 1.0f / (float)(end - beg);
-    out[idx] = from_f32<T>(acc);
+    out[idx] = from_f32<T>(v);
   }
 }
 
@@ -119,20 +144,29 @@ at::Tensor segment_sum_rows(const at::Tensor& values, const at::Tensor& order,
   auto out = at::empty({u, d}, v.options());
   if (u == 0) return out;
 
+  const long n = ord.numel();
+  // fp32 accumulator (== out when out is already fp32 and no averaging
+  // pass is needed separately)
+  const bool direct = v.scalar_type() == at::kFloat;
+  auto acc = direct ? out : at::empty({u, d}, v.options().dtype(at::kFloat));
+  acc.zero_();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, v.scalar_type(), "segment_sum_rows", [&] {
         using T = scalar_t;
         auto stream = at::cuda::getCurrentCUDAStream();
-        // one element per thread: the per-element chain is two DEPENDENT
-        // loads (order[k] -> values row), so latency must be hidden by
-        // thread-level parallelism — do not cap the grid at PX_MAX_GRID
-        int grid = px_grid(u * d, PX_BLOCK, 131072);
-        segment_sum_rows_kernel<T>
-            <<<grid, PX_BLOCK, 0, stream>>>(
+        segment_sum_scatter_kernel<T>
+            <<<px_grid(n * d, PX_BLOCK, 131072), PX_BLOCK, 0, stream>>>(
                 reinterpret_cast<const T*>(v.data_ptr()),
                 ord.data_ptr<int>(), seg.data_ptr<int>(),
-                reinterpret_cast<T*>(out.data_ptr()), u, d, average);
+                acc.data_ptr<float>(), n, u, d);
         HIP_CHECK_KERNEL();
+        if (!direct || average) {
+          segment_finalize_kernel<T>
+              <<<px_grid(u * d, PX_BLOCK, 131072), PX_BLOCK, 0, stream>>>(
+                  acc.data_ptr<float>(), seg.data_ptr<int>(),
+                  reinterpret_cast<T*>(out.data_ptr()), u, d, average);
+          HIP_CHECK_KERNEL();
+        }
       });
   return out;
 }
