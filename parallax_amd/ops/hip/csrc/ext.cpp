@@ -45,6 +45,9 @@ void multi_tensor_pack(std::vector<at::Tensor> tensors, at::Tensor flat);
 void multi_tensor_unpack(std::vector<at::Tensor> tensors, at::Tensor flat,
                          double scale);
 
+// gemm.hip
+at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
+
 // layernorm.hip
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
                                       const at::Tensor& w,
@@ -105,6 +108,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sparse_sgd", &sparse_sgd);
   m.def("multi_tensor_pack", &multi_tensor_pack);
   m.def("multi_tensor_unpack", &multi_tensor_unpack);
+  m.def("pxgemm_bt", &pxgemm_bt,
+        "hand-written MFMA bf16 GEMM C = A @ B^T");
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);

@@ -1,0 +1,192 @@
+// Hand-written CDNA4 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T, bf16 in /
+// bf16 out, fp32 accumulate.  The building block for the NHWC conv path
+// (a 1x1 convolution IS this GEMM with A = activations [N*H*W, Cin] and
+// B = weights [Cout, Cin]) and for framework-owned projection GEMMs.
+//
+// Structure (guide cdna_hip_programming.md §5, the LDS-staged ladder):
+//   * 128x128 output tile per 256-thread block (4 waves, each a 64x64
+//     sub-tile of 4x4 mfma_f32_16x16x32_bf16 fragments, 64 fp32 acc
+//     VGPRs/lane);
+//   * BK=64 K-step, both operand tiles staged to LDS by
+//     global_load_lds_dwordx4 (16 B/lane direct-to-LDS DMA), double
+//     buffered;
+//   * XOR swizzle byte^=((row&7)<<4) against ds_read_b128 bank conflicts
+//     (guide G4/T2) applied on the glds SOURCE address + the LDS read
+//     (rule 21: the destination stays lane-linear);
+//   * 2-phase schedule: issue next tile's DMA, compute current from LDS
+//     behind s_setprio(1) (T3 minimum form + T5);
+//   * XCD-aware bijective block swizzle (T1) for L2 locality.
+// Both operands are K-major (B passed TRANSPOSED, i.e. [N,K] row-major,
+// the natural layout of torch Linear/conv weights), so every MFMA
+// fragment is 8 contiguous bf16 per lane.
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define PXG_THREADS 256
+
+// LDS tile: [128 rows][64 k] bf16 = 16 KiB; byte row stride 128.
+// swizzle: byte ^= ((row & 7) << 4)   (16-byte granularity, involution)
+__device__ __forceinline__ int swz(int row, int kbyte) {
+  return row * (BK * 2) + (kbyte ^ ((row & 7) << 4));
+}
+
+__global__ __launch_bounds__(PXG_THREADS, 2)
+void pxgemm_bt_kernel(const ushort* __restrict__ A,
+                      const ushort* __restrict__ B,
+                      ushort* __restrict__ C, long M, long N, long K,
+                      int nbm, int nbn) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [2 buffers][A|B][16 KiB]
+#define LDSA(b) (smem + (b) * 32768)
+#define LDSB(b) (smem + (b) * 32768 + 16384)
+
+  // XCD-aware bijective remap of the flat block id (guide T1)
+  int bid = blockIdx.x;
+  {
+    const int nwg = nbm * nbn;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, pos = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int bm = (bid / nbn) * BM;
+  const int bn = (bid % nbn) * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  // wave grid 2x2: each wave owns a 64x64 sub-tile
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int fr = lane & 15;        // fragment row/col within 16
+  const int ko = (lane >> 4) * 8;  // fragment k offset (8 bf16)
+
+  f32x4 acc[4][4] = {};
+
+  // ---- glds staging: each thread DMAs 16 B per instruction; a 16 KiB
+  // tile needs 4 instructions.  LDS dest is lane-linear
+  // (base + flat_byte), source address carries the inverse swizzle.
+  const long a_row0 = bm, b_row0 = bn;
+  // flat element position for this thread's i-th piece:
+  //   byte p = (tid + i*256) * 16 ; row = p / 128 ; kbyte = p % 128
+  int p_row[4], p_kb[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int p = (tid + i * PXG_THREADS) * 16;
+    p_row[i] = p / (BK * 2);
+    p_kb[i] = (p % (BK * 2)) ^ ((p_row[i] & 7) << 4);  // inverse swizzle
+  }
+
+  const int ktiles = (int)(K / BK);
+
+#define STAGE(buf, kt)                                                    \
+  do {                                                                     \
+    const long kbase = (long)(kt) * BK;                                    \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+      const ushort* srcA = A + (a_row0 + p_row[i]) * K + kbase +           \
+                           (p_kb[i] >> 1);                                 \
+      const ushort* srcB = B + (b_row0 + p_row[i]) * K + kbase +           \
+                           (p_kb[i] >> 1);                                 \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) void*)srcA,             \
+          (__attribute__((address_space(3))) void*)(LDSA(buf) +            \
+                                                    (tid + i * 256) * 16), \
+          16, 0, 0);                                                       \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) void*)srcB,             \
+          (__attribute__((address_space(3))) void*)(LDSB(buf) +            \
+                                                    (tid + i * 256) * 16), \
+          16, 0, 0);                                                       \
+    }                                                                      \
+  } while (0)
+
+  STAGE(0, 0);
+  __builtin_amdgcn_s_waitcnt(/*vmcnt(0) lgkmcnt(0)*/ 0);
+  __builtin_amdgcn_s_barrier();
+
+  int cur = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    if (kt + 1 < ktiles) {
+      STAGE(cur ^ 1, kt + 1);
+    }
+    // compute current tile from LDS
+#pragma unroll
+    for (int kk = 0; kk < BK / 32; ++kk) {
+      bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int ar = wm + i * 16 + fr;
+        a_frag[i] = *(const bf16x8*)(LDSA(cur) +
+                                     swz(ar, (kk * 32 + ko) * 2));
+        const int br = wn + i * 16 + fr;
+        b_frag[i] = *(const bf16x8*)(LDSB(cur) +
+                                     swz(br, (kk * 32 + ko) * 2));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // drain this iteration's DMA and cross the barrier
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: C/D fragment map for 16x16x32: col = lane&15,
+  // row = (lane>>4)*4 + reg (guide §3); bf16 convert + store
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = bm + wm + i * 16 + crow0 + r;
+        const long col = bn + wn + j * 16 + fr;
+        __hip_bfloat16 v = __float2bfloat16(acc[i][j][r]);
+        C[row * N + col] = *reinterpret_cast<ushort*>(&v);
+      }
+    }
+  }
+}
+
+at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt) {
+  TORCH_CHECK(a.is_cuda() && bt.is_cuda(), "pxgemm_bt: GPU only");
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 &&
+              bt.scalar_type() == at::kBFloat16, "pxgemm_bt: bf16 only");
+  TORCH_CHECK(a.dim() == 2 && bt.dim() == 2 && a.is_contiguous() &&
+              bt.is_contiguous());
+  const long M = a.size(0), K = a.size(1), N = bt.size(0);
+  TORCH_CHECK(bt.size(1) == K, "pxgemm_bt: K mismatch");
+  TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0,
+              "pxgemm_bt v1 needs M%128==0, N%128==0, K%64==0 (pad)");
+  auto c = at::empty({M, N}, a.options());
+  const int nbm = (int)(M / BM), nbn = (int)(N / BN);
+  const int smem = 64 * 1024;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&pxgemm_bt_kernel),
+      hipFuncAttributeMaxDynamicSharedMemorySize, smem);
+  pxgemm_bt_kernel<<<nbm * nbn, PXG_THREADS, smem, stream>>>(
+      reinterpret_cast<const ushort*>(a.data_ptr()),
+      reinterpret_cast<const ushort*>(bt.data_ptr()),
+      reinterpret_cast<ushort*>(c.data_ptr()), M, N, K, nbm, nbn);
+  HIP_CHECK_KERNEL();
+  return c;
+}
