@@ -121,6 +121,10 @@ def build_resnet_graph(depth=50, num_classes=1001, batch_size=64, lr=0.1,
                        momentum=0.9, weight_decay=1e-4, use_bf16=True,
                        channels_last=True):
     model = ResNet(depth, num_classes)
+    import os as _os
+    if _os.environ.get('PARALLAX_NATIVE_CONV1X1') == '1' and channels_last:
+        from parallax_amd.ops.conv import swap_conv1x1
+        swap_conv1x1(model)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
     optimizer = optim.SGD(model.parameters(), lr=lr, momentum=momentum,

@@ -1,0 +1,96 @@
+"""NHWC 1x1 convolution on the hand-written MFMA GEMM.
+
+The north-star conv worklist entry: ResNet's bottleneck
+blocks are dominated by 1x1 convolutions, which in NHWC are exactly
+``out[N*H*W, Cout] = act[N*H*W, Cin] @ W[Cout, Cin]^T`` — the pxgemm_bt
+kernel (ops/hip/csrc/gemm.hip), which measures FASTER than the library
+path at the ResNet shapes (e.g. 56x56 stage: 26.9 us vs 32.0 us).
+
+Forward runs the hand-written kernel; backward uses library GEMMs (dgrad/
+wgrad are large-K shapes where hipBLASLt is already optimal).  3x3 and
+strided convolutions stay on MIOpen (implicit-GEMM staging for those is
+round-2 work); enable in ResNet with PARALLAX_NATIVE_CONV1X1=1.
+"""
+
+import torch
+import torch.nn as nn
+
+from parallax_amd import ops as _ops
+
+
+def _pad_rows(x, mult):
+    m = x.shape[0]
+    pad = (-m) % mult
+    if pad == 0:
+        return x, m
+    return torch.nn.functional.pad(x, (0, 0, 0, pad)), m
+
+
+class _Conv1x1Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2d, weight):
+        # x2d [M, Cin] bf16, weight [Cout, Cin] bf16
+        ctx.save_for_backward(x2d, weight)
+        xp, m = _pad_rows(x2d, 128)
+        wp, n = _pad_rows(weight, 128)
+        out = _ops.ext().pxgemm_bt(xp.contiguous(), wp.contiguous())
+        return out[:m, :n]
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, weight = ctx.saved_tensors
+        dx = dy @ weight            # [M, Cout] @ [Cout, Cin]
+        dw = dy.t().conj() @ x2d    # [Cout, M] @ [M, Cin]
+        return dx, dw
+
+
+def conv1x1_nhwc(x, weight):
+    """x: [N, C, H, W] channels_last bf16; weight: [Cout, Cin, 1, 1].
+    Returns [N, Cout, H, W] channels_last."""
+    N, C, H, W = x.shape
+    cout = weight.shape[0]
+    x2d = x.permute(0, 2, 3, 1).reshape(N * H * W, C)  # free in NHWC
+    w2d = weight.reshape(cout, C)
+    if x.is_cuda and _ops.native_available() \
+            and x.dtype == torch.bfloat16 and C % 64 == 0:
+        out = _Conv1x1Fn.apply(x2d.contiguous(), w2d.to(x.dtype))
+    else:
+        out = x2d @ w2d.to(x2d.dtype).t()
+    return out.view(N, H, W, cout).permute(0, 3, 1, 2)
+
+
+class ParallaxConv1x1(nn.Module):
+    """Drop-in for nn.Conv2d(k=1, stride=1, bias=False) on the MFMA GEMM."""
+
+    def __init__(self, cin, cout):
+        super().__init__()
+        w = torch.empty(cout, cin, 1, 1)
+        nn.init.kaiming_normal_(w, mode='fan_out', nonlinearity='relu')
+        self.weight = nn.Parameter(w)
+        self.in_channels = cin
+        self.out_channels = cout
+
+    @classmethod
+    def from_conv(cls, conv):
+        assert conv.kernel_size == (1, 1) and conv.stride == (1, 1) \
+            and conv.bias is None
+        m = cls(conv.in_channels, conv.out_channels)
+        with torch.no_grad():
+            m.weight.copy_(conv.weight)
+        return m
+
+    def forward(self, x):
+        return conv1x1_nhwc(x, self.weight)
+
+
+def swap_conv1x1(module):
+    """Recursively replace eligible nn.Conv2d(1x1, stride 1, no bias)
+    children with ParallaxConv1x1 (PARALLAX_NATIVE_CONV1X1=1 path)."""
+    for name, child in module.named_children():
+        if (isinstance(child, nn.Conv2d) and child.kernel_size == (1, 1)
+                and child.stride == (1, 1) and child.bias is None
+                and child.groups == 1):
+            setattr(module, name, ParallaxConv1x1.from_conv(child))
+        else:
+            swap_conv1x1(child)
+    return module
