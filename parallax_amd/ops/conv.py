@@ -34,7 +34,9 @@ class _Conv1x1Fn(torch.autograd.Function):
         xp, m = _pad_rows(x2d, 128)
         wp, n = _pad_rows(weight, 128)
         out = _ops.ext().pxgemm_bt(xp.contiguous(), wp.contiguous())
-        return out[:m, :n]
+        # slicing padded N leaves a stride-gap tensor; downstream NHWC
+        # views need dense rows
+        return out[:m, :n].contiguous()
 
     @staticmethod
     def backward(ctx, dy):

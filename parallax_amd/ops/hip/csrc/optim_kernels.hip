@@ -203,13 +203,23 @@ static void check_contig(const std::vector<at::Tensor>& ts,
   }
 }
 
+// strides of size-1 dims are meaningless (a [cout,cin,1,1] conv weight in
+// channels_last and its contiguous grad share the same storage order) —
+// compare layouts over dims with extent > 1 only.
+static bool same_dense_layout(const at::Tensor& x, const at::Tensor& y) {
+  if (x.sizes() != y.sizes()) return false;
+  for (int64_t i = 0; i < x.dim(); ++i) {
+    if (x.size(i) > 1 && x.stride(i) != y.stride(i)) return false;
+  }
+  return true;
+}
+
 static void check_same_layout(const std::vector<at::Tensor>& a,
                               const std::vector<at::Tensor>& b,
                               const char* name) {
   TORCH_CHECK(a.size() == b.size(), name, ": list size mismatch");
   for (size_t i = 0; i < a.size(); ++i) {
-    TORCH_CHECK(a[i].strides() == b[i].strides() &&
-                    a[i].sizes() == b[i].sizes(),
+    TORCH_CHECK(same_dense_layout(a[i], b[i]),
                 name, ": layout mismatch at tensor ", i);
   }
 }
