@@ -381,12 +381,15 @@ def test_resnet_native_conv1x1_swap():
     from parallax_amd.ops.conv import ParallaxConv1x1
     os.environ['PARALLAX_NATIVE_CONV1X1'] = '1'
     try:
-        g = build_resnet_graph(depth=50, batch_size=2)
+        g = build_resnet_graph(depth=50, batch_size=16)
         n_swapped = sum(1 for m in g.model.modules()
                         if isinstance(m, ParallaxConv1x1))
         assert n_swapped > 30, n_swapped
         g.model.to(DEV)
-        x, y = synthetic_batch(2, device=DEV)
+        # batch >= 16: MIOpen (ROCm 7.2) BatchNorm SEGFAULTS on tiny
+        # bf16 channels_last inputs like [2, 512, 7, 7] — library bug
+        # reproduced with pure nn.BatchNorm2d, see doc/design.md
+        x, y = synthetic_batch(16, device=DEV)
         loss, _ = g.step_fn(g.model, (x, y))
         loss.backward()
         assert torch.isfinite(loss)
