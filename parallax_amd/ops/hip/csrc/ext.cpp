@@ -47,6 +47,8 @@ void multi_tensor_unpack(std::vector<at::Tensor> tensors, at::Tensor flat,
 
 // gemm.hip
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
+at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
+                         const c10::optional<at::Tensor>& addend);
 
 // layernorm.hip
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
@@ -110,6 +112,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_unpack", &multi_tensor_unpack);
   m.def("pxgemm_bt", &pxgemm_bt,
         "hand-written MFMA bf16 GEMM C = A @ B^T");
+  m.def("pxgemm_bt_add", &pxgemm_bt_add,
+        "MFMA GEMM with fused bf16 addend epilogue");
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);

@@ -41,11 +41,14 @@ __device__ __forceinline__ int swz(int row, int kbyte) {
   return row * (BK * 2) + (kbyte ^ ((row & 7) << 4));
 }
 
+// Cadd: optional bf16 addend with arbitrary row stride (elements),
+// fused into the epilogue (the addmm beta=1 case: gates = gx[:,t] + h@W).
 __global__ __launch_bounds__(PXG_THREADS, 2)
 void pxgemm_bt_kernel(const ushort* __restrict__ A,
                       const ushort* __restrict__ B,
                       ushort* __restrict__ C, long M, long N, long K,
-                      int nbm, int nbn) {
+                      int nbm, int nbn, const ushort* __restrict__ Cadd,
+                      long add_rstride) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // [2 buffers][A|B][16 KiB]
 #define LDSA(b) (smem + (b) * 32768)
@@ -160,14 +163,22 @@ void pxgemm_bt_kernel(const ushort* __restrict__ A,
       for (int r = 0; r < 4; ++r) {
         const long row = bm + wm + i * 16 + crow0 + r;
         const long col = bn + wn + j * 16 + fr;
-        __hip_bfloat16 v = __float2bfloat16(acc[i][j][r]);
+        float fv = acc[i][j][r];
+        if (Cadd != nullptr) {
+          __hip_bfloat16 av =
+              *reinterpret_cast<const __hip_bfloat16*>(
+                  &Cadd[row * add_rstride + col]);
+          fv += __bfloat162float(av);
+        }
+        __hip_bfloat16 v = __float2bfloat16(fv);
         C[row * N + col] = *reinterpret_cast<ushort*>(&v);
       }
     }
   }
 }
 
-at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt) {
+at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
+                         const c10::optional<at::Tensor>& addend) {
   TORCH_CHECK(a.is_cuda() && bt.is_cuda(), "pxgemm_bt: GPU only");
   TORCH_CHECK(a.scalar_type() == at::kBFloat16 &&
               bt.scalar_type() == at::kBFloat16, "pxgemm_bt: bf16 only");
@@ -184,10 +195,25 @@ at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt) {
   hipFuncSetAttribute(
       reinterpret_cast<const void*>(&pxgemm_bt_kernel),
       hipFuncAttributeMaxDynamicSharedMemorySize, smem);
+  const ushort* add_ptr = nullptr;
+  long add_rs = 0;
+  if (addend.has_value()) {
+    const auto& ad = addend.value();
+    TORCH_CHECK(ad.scalar_type() == at::kBFloat16 && ad.dim() == 2 &&
+                ad.size(0) == M && ad.size(1) == N && ad.stride(1) == 1,
+                "pxgemm_bt: bad addend");
+    add_ptr = reinterpret_cast<const ushort*>(ad.data_ptr());
+    add_rs = ad.stride(0);
+  }
  hipLaunchKernelGGL(( pxgemm_bt_kernel), dim3(nbm * nbn), dim3(PXG_THREADS), smem, stream, 
       reinterpret_cast<const ushort*>(a.data_ptr()),
       reinterpret_cast<const ushort*>(bt.data_ptr()),
-      reinterpret_cast<ushort*>(c.data_ptr()), M, N, K, nbm, nbn);
+      reinterpret_cast<ushort*>(c.data_ptr()), M, N, K, nbm, nbn,
+      add_ptr, add_rs);
   HIP_CHECK_KERNEL();
   return c;
+}
+
+at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt) {
+  return pxgemm_bt_add(a, bt, c10::nullopt);
 }
