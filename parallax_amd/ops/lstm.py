@@ -105,8 +105,10 @@ class _LSTMSeqFn(torch.autograd.Function):
         # gates shape [128,512]x[512,8192] (7.7 vs 18.8 us) with the
         # gx[:, t] addend fused into its epilogue; it wants B^T (one
         # 16 MB transpose per sequence, amortized over T steps)
+        import os as _os
         use_px = (gx.dtype == torch.bfloat16 and B % 128 == 0
-                  and (4 * H) % 128 == 0 and P % 64 == 0)
+                  and (4 * H) % 128 == 0 and P % 64 == 0
+                  and _os.environ.get('PARALLAX_PX_LSTM_FWD', '1') == '1')
         w_hT = w_h.t().contiguous() if use_px else None
         for t in range(T):
             if use_px:
@@ -143,8 +145,10 @@ class _LSTMSeqFn(torch.autograd.Function):
         dc, dc_next = dc_a, dc_b
         # ds = dhp @ w_proj^T maps DIRECTLY onto pxgemm_bt (w_proj is
         # [H, P] row-major = the transposed operand): 7.6 vs 19.3 us
+        import os as _os
         use_px = (w_h.dtype == torch.bfloat16 and B % 128 == 0
-                  and H % 128 == 0 and P % 64 == 0)
+                  and H % 128 == 0 and P % 64 == 0
+                  and _os.environ.get('PARALLAX_PX_LSTM_BWD', '1') == '1')
         for t in range(T - 1, -1, -1):
             if t == T - 1:
                 dHP[t].copy_(d_outs[:, t])
