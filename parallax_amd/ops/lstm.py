@@ -101,14 +101,16 @@ class _LSTMSeqFn(torch.autograd.Function):
                            device=gx.device)
         c_all[0] = c0.to(gx.dtype)
         hp_all[0] = h0.to(gx.dtype)
-        # our skinny-M MFMA GEMM beats hipBLASLt 2.4x on the recurrent
-        # gates shape [128,512]x[512,8192] (7.7 vs 18.8 us) with the
-        # gx[:, t] addend fused into its epilogue; it wants B^T (one
-        # 16 MB transpose per sequence, amortized over T steps)
+        # pxgemm beats hipBLASLt 2.4x on the recurrent gates shape in
+        # ISOLATION (7.7 vs 18.8 us) but measured ~4%% SLOWER end-to-end
+        # inside the captured step (A/B fwd=0/bwd=0 756.7k vs fwd=1/bwd=1
+        # 727.4k words/s) — dependent-chain pipelining favors the library
+        # kernels here, so the substitution is opt-in
+        # (PARALLAX_PX_LSTM_FWD/BWD=1)
         import os as _os
         use_px = (gx.dtype == torch.bfloat16 and B % 128 == 0
                   and (4 * H) % 128 == 0 and P % 64 == 0
-                  and _os.environ.get('PARALLAX_PX_LSTM_FWD', '1') == '1')
+                  and _os.environ.get('PARALLAX_PX_LSTM_FWD', '0') == '1')
         w_hT = w_h.t().contiguous() if use_px else None
         for t in range(T):
             if use_px:
@@ -148,7 +150,7 @@ class _LSTMSeqFn(torch.autograd.Function):
         import os as _os
         use_px = (w_h.dtype == torch.bfloat16 and B % 128 == 0
                   and H % 128 == 0 and P % 64 == 0
-                  and _os.environ.get('PARALLAX_PX_LSTM_BWD', '1') == '1')
+                  and _os.environ.get('PARALLAX_PX_LSTM_BWD', '0') == '1')
         for t in range(T - 1, -1, -1):
             if t == T - 1:
                 dHP[t].copy_(d_outs[:, t])
