@@ -141,10 +141,53 @@ class ParallaxSession(object):
                     dst.copy_(src)
         self._static_loss = loss
         self._static_fetches = fetches
+        self._opt_graph = None
+        self._opt_graph_state = 'pending'
+        self._opt_graph_steps = 0
         parallax_log.info('hipGraph capture enabled (%d dense grads, '
                           '%d sparse slots)', len(dense_params),
                           len(self.engine.sparse_engine.slots)
                           if self.engine.sparse_engine else 0)
+
+    def _maybe_capture_opt_graph(self):
+        """Capture the (shape-static) optimizer step — fused dense update,
+        clip, EMA — as a SECOND hipGraph once its state tensors exist
+        (after a few eager steps).  The gradient buffers are stable under
+        the main graph, and at world>1 the unpack kernel writes the
+        reduced grads into those same buffers before this replays."""
+        if self._opt_graph_state == 'disabled':
+            return False
+        if self._opt_graph_state == 'ready':
+            return True
+        from parallax_amd import optim as _optim
+        if isinstance(self.graph.optimizer, _optim.Adam):
+            # Adam's bias correction is a per-step HOST scalar — a capture
+            # would freeze it; keep Adam eager
+            self._opt_graph_state = 'disabled'
+            return False
+        self._opt_graph_steps += 1
+        if self._opt_graph_steps < 3:
+            return False
+        try:
+            # the warm call below EXECUTES this step's real update; the
+            # capture that follows only records — so return False this
+            # step (the update already happened) and replay from the next
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self.graph.optimizer.step()
+            torch.cuda.current_stream().wait_stream(side)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self.graph.optimizer.step()
+            self._opt_graph = g
+            self._opt_graph_state = 'just_captured'
+            parallax_log.info('optimizer-step hipGraph captured')
+        except Exception as e:
+            parallax_log.warning('optimizer graph capture failed (%r); '
+                                 'staying eager', e)
+            self._opt_graph_state = 'disabled'
+        return False
 
     def _run_step_graphed(self, batch):
         # copy dynamic inputs into the static buffers (state feeds back
@@ -154,7 +197,22 @@ class ParallaxSession(object):
                     src, torch.Tensor) and dst.data_ptr() != src.data_ptr():
                 dst.copy_(src, non_blocking=True)
         self._graph_obj.replay()
-        self.engine.finish_step()
+        eng = self.engine
+        if eng.dense_engine:
+            eng.dense_engine.launch_async()
+        if eng.sparse_engine:
+            eng.sparse_engine.reduce_and_apply()
+        if eng.dense_engine:
+            eng.dense_engine.wait_and_unpack()
+        if self._maybe_capture_opt_graph():
+            self._opt_graph.replay()
+        else:
+            # eager optimizer (the capture step's warm call IS this
+            # step's update — _maybe_capture_opt_graph returns False then)
+            if self._opt_graph_state != 'just_captured':
+                self.graph.optimizer.step()
+            else:
+                self._opt_graph_state = 'ready'
         if self.graph.lr_scheduler is not None:
             self.graph.lr_scheduler.step()
         return self._static_loss, self._static_fetches
