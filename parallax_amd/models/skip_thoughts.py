@@ -33,13 +33,10 @@ class GRUCell(nn.Module):
         self.b = nn.Parameter(torch.zeros(3 * hidden))
 
     def forward(self, x, h):
+        from parallax_amd.ops.gru import gru_gates
         xg = x @ self.w_x + self.b
         hg = h @ self.w_h
-        H = self.hidden
-        r = torch.sigmoid(xg[:, :H] + hg[:, :H])
-        z = torch.sigmoid(xg[:, H:2 * H] + hg[:, H:2 * H])
-        n = torch.tanh(xg[:, 2 * H:] + r * hg[:, 2 * H:])
-        return (1 - z) * n + z * h
+        return gru_gates(xg, hg, h)
 
 
 class GRU(nn.Module):

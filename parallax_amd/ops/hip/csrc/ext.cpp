@@ -45,6 +45,14 @@ void multi_tensor_pack(std::vector<at::Tensor> tensors, at::Tensor flat);
 void multi_tensor_unpack(std::vector<at::Tensor> tensors, at::Tensor flat,
                          double scale);
 
+// gru_cell.hip
+std::vector<at::Tensor> gru_gates_fwd(const at::Tensor& xg,
+                                      const at::Tensor& hg,
+                                      const at::Tensor& h_prev);
+std::vector<at::Tensor> gru_gates_bwd(
+    const at::Tensor& dh, const at::Tensor& h_prev, const at::Tensor& r_s,
+    const at::Tensor& z_s, const at::Tensor& n_t, const at::Tensor& hgn_s);
+
 // gemm.hip
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
@@ -110,6 +118,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sparse_sgd", &sparse_sgd);
   m.def("multi_tensor_pack", &multi_tensor_pack);
   m.def("multi_tensor_unpack", &multi_tensor_unpack);
+  m.def("gru_gates_fwd", &gru_gates_fwd);
+  m.def("gru_gates_bwd", &gru_gates_bwd);
   m.def("pxgemm_bt", &pxgemm_bt,
         "hand-written MFMA bf16 GEMM C = A @ B^T");
   m.def("pxgemm_bt_add", &pxgemm_bt_add,

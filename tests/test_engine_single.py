@@ -121,6 +121,48 @@ def test_degenerate_hybrid_falls_back_to_ps(monkeypatch, free_port):
     sess.close()
 
 
+def test_ps_mode_mixed_model(monkeypatch, free_port):
+    """Explicit run_option='PS': dense grads still go over the (fused)
+    allreduce plane, sparse over the PS engine (doc/design.md divergence
+    2); both engines active for a mixed model."""
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+
+    def build():
+        model = nn.Module()
+        model.emb = ParallaxEmbedding(30, 4)
+        model.fc = nn.Linear(4, 2)
+        opt = optim.Adagrad(model.parameters(), lr=0.2)
+
+        def step_fn(m, batch):
+            ids, y = batch
+            return ((m.fc(m.emb(ids)) - y) ** 2).mean(), {}
+
+        return TrainGraph(model, opt, step_fn)
+
+    sess, *_ = parallax_amd.parallel_run(
+        build, None, parallax_config=ParallaxConfig(run_option='PS'))
+    assert sess.engine.mode == 'PS'
+    assert sess.engine.dense_engine is not None
+    assert sess.engine.sparse_engine is not None
+    out = sess.run((torch.tensor([1, 2, 2]), torch.randn(3, 2)))
+    assert torch.isfinite(out['loss'])
+    sess.close()
+
+
+def test_profile_directory_scaffolding(tmp_path):
+    """lib.create_profile_directory (reference lib.py:333-358): per-host/
+    worker run_meta dirs + task_info manifest."""
+    from parallax_amd.lib import create_profile_directory
+    ri = {'worker': [{'hostname': 'h0', 'gpus': [0, 1]},
+                     {'hostname': 'h1', 'gpus': [0]}]}
+    create_profile_directory(str(tmp_path), ri)
+    assert (tmp_path / 'task_info').exists()
+    assert (tmp_path / 'h0' / 'worker:0' / 'run_meta').is_dir()
+    assert (tmp_path / 'h1' / 'worker:1' / 'run_meta').is_dir()
+    assert 'worker:0:h0:0,1' in (tmp_path / 'task_info').read_text()
+
+
 def test_lm1b_tiny_loss_decreases(monkeypatch, free_port):
     """Tiny-config LM1B: the full hybrid stack (partitioned embeddings,
     sampled softmax, Adagrad, clipping) learns on a repeated batch."""
