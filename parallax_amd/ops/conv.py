@@ -29,14 +29,11 @@ def _pad_rows(x, mult):
 class _Conv1x1Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2d, weight):
-        # x2d [M, Cin] bf16, weight [Cout, Cin] bf16
+        # x2d [M, Cin] bf16, weight [Cout, Cin] bf16; the kernel handles
+        # arbitrary M/N with clamped edge-tile loads + guarded stores, so
+        # no padding copies
         ctx.save_for_backward(x2d, weight)
-        xp, m = _pad_rows(x2d, 128)
-        wp, n = _pad_rows(weight, 128)
-        out = _ops.ext().pxgemm_bt(xp.contiguous(), wp.contiguous())
-        # slicing padded N leaves a stride-gap tensor; downstream NHWC
-        # views need dense rows
-        return out[:m, :n].contiguous()
+        return _ops.ext().pxgemm_bt(x2d.contiguous(), weight.contiguous())
 
     @staticmethod
     def backward(ctx, dy):

@@ -83,11 +83,19 @@ void pxgemm_bt_kernel(const ushort* __restrict__ A,
   // flat element position for this thread's i-th piece:
   //   byte p = (tid + i*256) * 16 ; row = p / 128 ; kbyte = p % 128
   int p_row[4], p_kb[4];
+  long a_src_row[4], b_src_row[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const int p = (tid + i * PXG_THREADS) * 16;
     p_row[i] = p / (BK * 2);
     p_kb[i] = (p % (BK * 2)) ^ ((p_row[i] & 7) << 4);  // inverse swizzle
+    // edge tiles: clamp out-of-range source rows to the last valid row —
+    // the stale fragments they produce are discarded by the guarded
+    // epilogue stores, so NO host-side padding copies are needed
+    long ar = a_row0 + p_row[i];
+    long br = b_row0 + p_row[i];
+    a_src_row[i] = ar < M ? ar : M - 1;
+    b_src_row[i] = br < N ? br : N - 1;
   }
 
   const int ktiles = (int)(K / BK);
@@ -96,9 +104,9 @@ void pxgemm_bt_kernel(const ushort* __restrict__ A,
   do {                                                                     \
     const long kbase = (long)(kt) * BK;                                    \
     _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
-      const ushort* srcA = A + (a_row0 + p_row[i]) * K + kbase +           \
+      const ushort* srcA = A + a_src_row[i] * K + kbase +                  \
                            (p_kb[i] >> 1);                                 \
-      const ushort* srcB = B + (b_row0 + p_row[i]) * K + kbase +           \
+      const ushort* srcB = B + b_src_row[i] * K + kbase +                  \
                            (p_kb[i] >> 1);                                 \
       __builtin_amdgcn_global_load_lds(                                    \
           (const __attribute__((address_space(1))) void*)srcA,             \
@@ -163,6 +171,7 @@ void pxgemm_bt_kernel(const ushort* __restrict__ A,
       for (int r = 0; r < 4; ++r) {
         const long row = bm + wm + i * 16 + crow0 + r;
         const long col = bn + wn + j * 16 + fr;
+        if (row >= M || col >= N) continue;
         float fv = acc[i][j][r];
         if (Cadd != nullptr) {
           __hip_bfloat16 av =
@@ -186,10 +195,11 @@ at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
               bt.is_contiguous());
   const long M = a.size(0), K = a.size(1), N = bt.size(0);
   TORCH_CHECK(bt.size(1) == K, "pxgemm_bt: K mismatch");
-  TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0,
-              "pxgemm_bt v1 needs M%128==0, N%128==0, K%64==0 (pad)");
+  TORCH_CHECK(K % BK == 0, "pxgemm_bt needs K % 64 == 0");
+  TORCH_CHECK(M >= 1 && N >= 16, "pxgemm_bt: degenerate shape");
   auto c = at::empty({M, N}, a.options());
-  const int nbm = (int)(M / BM), nbn = (int)(N / BN);
+  const int nbm = (int)((M + BM - 1) / BM);
+  const int nbn = (int)((N + BN - 1) / BN);
   const int smem = 64 * 1024;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   hipFuncSetAttribute(

@@ -60,6 +60,7 @@ def bench(M, N, K, iters=30):
 if __name__ == '__main__':
     ok = True
     for shape in ((128, 128, 64), (256, 256, 128), (512, 384, 192),
+                  (100, 70, 64), (3136, 512, 2048),  # edge tiles
                   (2560, 8192, 512), (4096, 4096, 4096)):
         ok &= check(*shape)
     print('CORRECTNESS:', 'PASS' if ok else 'FAIL', flush=True)
