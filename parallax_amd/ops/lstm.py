@@ -155,9 +155,9 @@ class _LSTMSeqFn(torch.autograd.Function):
             if t == T - 1:
                 dHP[t].copy_(d_outs[:, t])
             else:
-                # dhp = upstream slice + recurrent carry (dG[t+1] @ w_h^T)
-                torch.add(d_outs[:, t].to(w_h.dtype),
-                          torch.mm(dG[t + 1], w_h_t), out=dHP[t])
+                # dhp = upstream slice + recurrent carry, fused as ONE
+                # beta=1 GEMM: dHP[t] = d_outs[:, t] + dG[t+1] @ w_h^T
+                torch.addmm(d_outs[:, t], dG[t + 1], w_h_t, out=dHP[t])
             if use_px:
                 ds = ext.pxgemm_bt(dHP[t], w_proj)
             else:
