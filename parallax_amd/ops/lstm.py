@@ -155,9 +155,11 @@ class _LSTMSeqFn(torch.autograd.Function):
             if t == T - 1:
                 dHP[t].copy_(d_outs[:, t])
             else:
-                # dhp = upstream slice + recurrent carry, fused as ONE
-                # beta=1 GEMM: dHP[t] = d_outs[:, t] + dG[t+1] @ w_h^T
-                torch.addmm(d_outs[:, t], dG[t + 1], w_h_t, out=dHP[t])
+                # dhp = upstream slice + recurrent carry; NOTE an addmm
+                # (beta=1) fusion of the add measured ~2% SLOWER overall
+                # (hipBLASLt picks a worse kernel on the strided-C path)
+                torch.add(d_outs[:, t].to(w_h.dtype),
+                          torch.mm(dG[t + 1], w_h_t), out=dHP[t])
             if use_px:
                 ds = ext.pxgemm_bt(dHP[t], w_proj)
             else:
