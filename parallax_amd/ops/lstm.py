@@ -4,12 +4,14 @@ Reference analog: the LM1B model's per-step ``xw_plus_b`` [1024 -> 8192] +
 sigmoid/tanh gate arithmetic + 2048->512 projection, 20 steps unrolled
 (examples/lm1b/language_model.py:51-62,76-87).
 
-MI355X-native split: the two GEMMs (input/hidden projection and the output
-projection) go through torch.matmul (hipBLASLt MFMA path — a plain library
-GEMM); ALL the pointwise gate math (sigmoid x3, tanh x2, the cell-state
-FMA and their backward) is ONE hand-written HIP kernel pair
+MI355X-native split: the GEMMs (input projection hoisted out of the time
+loop, the recurrent/projection GEMMs, the batched weight-grad GEMMs) run
+on hipBLASLt by default (the hand-written pxgemm substitution is opt-in,
+see _LSTMSeqFn); ALL the pointwise gate math (sigmoid x3, tanh x2, the
+cell-state FMA and their backward) is ONE hand-written HIP kernel pair
 (ops/hip/lstm_cell.hip) so the [B, 8K] gate tensor is read/written once
-instead of ~10 elementwise launches.
+instead of ~10 elementwise launches.  ParallaxLSTM.forward dispatches the
+whole unroll to the sequence-level Function (_LSTMSeqFn) on GPU.
 
 Gate layout in the fused gates tensor: [i | g (cell cand) | f | o] chunks of
 hidden_size each, matching ``torch.chunk(gates, 4, -1)`` below.

@@ -9,8 +9,10 @@ gradients arrive as indexed slices — on the full-vocab softmax table that
 is exactly what makes the hybrid (sparse-PS) path pay: per step only
 (batch*T + num_sampled) rows of the 793k-row table carry gradient.
 
-The [B, D] x [S, D]^T candidate GEMM runs on hipBLASLt MFMA; the CE over
-[B, 1+S] uses the fused softmax-CE kernel.
+The [B, D] x [S, D]^T candidate GEMM runs on hipBLASLt MFMA; the CE +
+per-row/column logit corrections run in ONE fused kernel (sampled_ce_*,
+ops/hip/csrc/softmax_ce.hip) without materializing corrected copies of
+the [B, S] matrix.
 """
 
 import math
