@@ -249,6 +249,41 @@ def _body_comm_collectives(rank, world):
         assert torch.all(recv[r] == 10 * r + rank)
 
 
+def _body_average_sparse(rank, world):
+    """average_sparse=True: the update uses the occurrence-count average
+    of duplicate rows ACROSS ranks (reference average_option
+    SPARSE_AVERAGE_BY_COUNTER, graph_transform_lib.py:385-390) — counts
+    must survive the local pre-aggregation."""
+    import torch.nn as nn
+    from parallax_amd import optim
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.sparse_engine import SparsePSEngine
+    from parallax_amd.ops.embedding import ParallaxEmbedding
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    torch.manual_seed(77)
+    model = nn.Module()
+    model.emb = ParallaxEmbedding(20, 4)
+    opt = optim.SGD(model.parameters(), lr=1.0)
+    eng = SparsePSEngine(comm, classify_params(model), opt,
+                         average_sparse=True, local_aggregation=True)
+    eng.prepare_step()
+    # rank 0 uses row 5 three times, rank 1 once -> avg divisor 4
+    n = 3 if rank == 0 else 1
+    ids = torch.full((n,), 5, dtype=torch.long)
+    before = model.emb.weight[5].detach().clone()
+    out = model.emb(ids)
+    out.sum().backward()
+    eng.reduce_and_apply()
+    after = model.emb.weight[5].detach()
+    # each use contributes grad of ones; sum = 4*ones, count = 4 ->
+    # averaged grad = ones; SGD lr=1 -> delta = -1
+    assert torch.allclose(after - before, -torch.ones(4), atol=1e-5), \
+        (after - before)
+    eng.close()
+
+
 def _body_lm1b_tiny_full_stack(rank, world):
     """Tiny LM1B through the COMPLETE per-step protocol on several ranks:
     partitioned tables, sampled softmax, clip, EMA, dense AR, sparse PS —
@@ -296,3 +331,7 @@ def test_multiproc(body):
 def test_lm1b_full_stack_world4():
     _spawn(_body_lm1b_tiny_full_stack, get_free_port(), world=4,
            timeout=300)
+
+
+def test_average_sparse_world2():
+    _spawn(_body_average_sparse, get_free_port())
