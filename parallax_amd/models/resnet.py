@@ -125,6 +125,9 @@ def build_resnet_graph(depth=50, num_classes=1001, batch_size=64, lr=0.1,
     if _os.environ.get('PARALLAX_NATIVE_CONV1X1') == '1' and channels_last:
         from parallax_amd.ops.conv import swap_conv1x1
         swap_conv1x1(model)
+    if _os.environ.get('PARALLAX_NATIVE_CONV3X3') == '1' and channels_last:
+        from parallax_amd.ops.conv import swap_conv3x3
+        swap_conv3x3(model)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
     optimizer = optim.SGD(model.parameters(), lr=lr, momentum=momentum,

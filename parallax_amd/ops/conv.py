@@ -93,3 +93,86 @@ def swap_conv1x1(module):
         else:
             swap_conv1x1(child)
     return module
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    """Forward: implicit-GEMM MFMA kernel over a host-padded NHWC input.
+    Backward: library dgrad/wgrad (aten.convolution_backward -> MIOpen)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride):
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        N, C, H, W = x.shape
+        cout = weight.shape[0]
+        oh = (H + 2 - 3) // stride + 1
+        ow = (W + 2 - 3) // stride + 1
+        # halo pad once per call (glds cannot predicate edge reads)
+        x_nhwc = x.permute(0, 2, 3, 1)            # view (channels_last)
+        xp = torch.nn.functional.pad(x_nhwc, (0, 0, 1, 1, 1, 1))
+        w2d = (weight.permute(0, 2, 3, 1).reshape(cout, 9 * C)
+               .to(x.dtype).contiguous())
+        out = _ops.ext().pxconv3x3_fwd(xp.contiguous(), w2d, N, oh, ow, C,
+                                       stride, H + 2, W + 2)
+        return out.view(N, oh, ow, cout).permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        s = ctx.stride
+        dx, dw, _ = torch.ops.aten.convolution_backward(
+            dy.to(x.dtype), x, weight.to(x.dtype), None, [s, s], [1, 1],
+            [1, 1], False, [0, 0], 1, [True, True, False])
+        return dx, dw.to(weight.dtype), None
+
+
+def conv3x3_nhwc(x, weight, stride=1):
+    """3x3 pad-1 NHWC convolution; forward on the hand-written MFMA
+    implicit-GEMM kernel when eligible (bf16, Cin power-of-two >= 64)."""
+    C = x.shape[1]
+    if (x.is_cuda and _ops.native_available()
+            and x.dtype == torch.bfloat16 and C >= 64
+            and (C & (C - 1)) == 0):
+        return _Conv3x3Fn.apply(x, weight, stride)
+    return torch.nn.functional.conv2d(x, weight.to(x.dtype),
+                                      stride=stride, padding=1)
+
+
+class ParallaxConv3x3(nn.Module):
+    """Drop-in for nn.Conv2d(k=3, padding=1, bias=False)."""
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__()
+        w = torch.empty(cout, cin, 3, 3)
+        nn.init.kaiming_normal_(w, mode='fan_out', nonlinearity='relu')
+        self.weight = nn.Parameter(w)
+        self.stride = stride
+        self.in_channels = cin
+        self.out_channels = cout
+
+    @classmethod
+    def from_conv(cls, conv):
+        assert conv.kernel_size == (3, 3) and conv.padding == (1, 1) \
+            and conv.bias is None
+        m = cls(conv.in_channels, conv.out_channels, conv.stride[0])
+        with torch.no_grad():
+            m.weight.copy_(conv.weight)
+        return m
+
+    def forward(self, x):
+        return conv3x3_nhwc(x, self.weight, self.stride)
+
+
+def swap_conv3x3(module):
+    """Replace eligible nn.Conv2d(3x3, pad 1, no bias, power-of-two
+    Cin >= 64) children with ParallaxConv3x3
+    (PARALLAX_NATIVE_CONV3X3=1 path)."""
+    for name, child in module.named_children():
+        if (isinstance(child, nn.Conv2d) and child.kernel_size == (3, 3)
+                and child.padding == (1, 1) and child.bias is None
+                and child.groups == 1 and child.in_channels >= 64
+                and (child.in_channels & (child.in_channels - 1)) == 0):
+            setattr(module, name, ParallaxConv3x3.from_conv(child))
+        else:
+            swap_conv3x3(child)
+    return module

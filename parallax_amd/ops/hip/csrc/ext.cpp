@@ -57,6 +57,9 @@ std::vector<at::Tensor> gru_gates_bwd(
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
                          const c10::optional<at::Tensor>& addend);
+at::Tensor pxconv3x3_fwd(const at::Tensor& xp, const at::Tensor& w2d,
+                         long n_img, long oh, long ow, long cin,
+                         long stride, long HP, long WP);
 
 // layernorm.hip
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
@@ -124,6 +127,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hand-written MFMA bf16 GEMM C = A @ B^T");
   m.def("pxgemm_bt_add", &pxgemm_bt_add,
         "MFMA GEMM with fused bf16 addend epilogue");
+  m.def("pxconv3x3_fwd", &pxconv3x3_fwd,
+        "implicit-GEMM 3x3 NHWC conv forward (MFMA)");
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);

@@ -486,3 +486,28 @@ def test_gru_gates_fwd_bwd_vs_ref():
     assert torch.allclose(xg.grad.cpu(), xg2.grad, atol=1e-4)
     assert torch.allclose(hg.grad.cpu(), hg2.grad, atol=1e-4)
     assert torch.allclose(hp.grad.cpu(), hp2.grad, atol=1e-4)
+
+
+def test_conv3x3_nhwc_vs_torch():
+    """Implicit-GEMM 3x3 NHWC conv fwd (strides 1 and 2) vs conv2d."""
+    from parallax_amd.ops.conv import conv3x3_nhwc
+    torch.manual_seed(61)
+    for (N, C, H, W, COUT, S) in ((2, 64, 16, 16, 128, 1),
+                                  (2, 128, 14, 14, 128, 2),
+                                  (3, 64, 13, 15, 192, 1),
+                                  (2, 256, 8, 8, 64, 2)):
+        x = (torch.randn(N, C, H, W, device=DEV, dtype=torch.bfloat16)
+             .to(memory_format=torch.channels_last).requires_grad_(True))
+        w = torch.randn(COUT, C, 3, 3, device=DEV, dtype=torch.bfloat16,
+                        requires_grad=True) * 0.1
+        w.retain_grad()
+        y = conv3x3_nhwc(x, w, stride=S)
+        ref = torch.nn.functional.conv2d(x.float(), w.float(), stride=S,
+                                         padding=1)
+        rel = ((y.float() - ref).abs().max()
+               / ref.abs().max().clamp_min(1e-6))
+        assert float(rel) < 0.05, (N, C, H, W, COUT, S, float(rel))
+        dy = torch.randn_like(ref)
+        y.backward(dy.to(torch.bfloat16))
+        assert torch.isfinite(x.grad.float()).all()
+        assert torch.isfinite(w.grad.float()).all()
