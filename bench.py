@@ -121,11 +121,14 @@ def main():
             return M.synthetic_batch(batch, device=device)
         return M.synthetic_batch(batch, device=device)
 
-    if args.cuda_graph and on_gpu and args.model == 'lm1b':
+    if args.cuda_graph and on_gpu:
         try:
-            state0 = M.make_init_state(batch, device=device,
-                                       dtype=torch.bfloat16)
-            sess.enable_cuda_graph(make_batch(state0))
+            if args.model == 'lm1b':
+                state0 = M.make_init_state(batch, device=device,
+                                           dtype=torch.bfloat16)
+                sess.enable_cuda_graph(make_batch(state0))
+            else:  # resnet50 / nmt: static shapes, no recurrent state
+                sess.enable_cuda_graph(make_batch())
         except Exception as e:
             print('cuda-graph capture failed, falling back to eager: %r'
                   % (e,), file=sys.stderr)
