@@ -121,6 +121,14 @@ def main():
             return M.synthetic_batch(batch, device=device)
         return M.synthetic_batch(batch, device=device)
 
+    # eager warmup FIRST (lets MIOpen's benchmark/find phase pick tuned
+    # conv algorithms before any capture freezes the choices), then
+    # capture, then a short graphed re-warmup.
+    state = None
+    for _ in range(args.warmup):
+        out = sess.run(make_batch(state))
+        state = out.get('state')
+
     if args.cuda_graph and on_gpu:
         try:
             if args.model == 'lm1b':
@@ -132,12 +140,9 @@ def main():
         except Exception as e:
             print('cuda-graph capture failed, falling back to eager: %r'
                   % (e,), file=sys.stderr)
-
-    # warmup (untimed)
-    state = None
-    for _ in range(args.warmup):
-        out = sess.run(make_batch(state))
-        state = out.get('state')
+        for _ in range(max(3, args.warmup // 3)):
+            out = sess.run(make_batch(state))
+            state = out.get('state')
 
     sess.comm.barrier()
     if on_gpu:
