@@ -12,6 +12,8 @@ softmax-CE kernel for the loss; all parameters dense -> the whole model
 exercises the bucketed RCCL all-reduce engine.
 """
 
+import os as _os
+
 import torch
 import torch.nn as nn
 
@@ -22,6 +24,20 @@ from parallax_amd.session import TrainGraph
 _LAYERS = {50: (3, 4, 6, 3), 101: (3, 4, 23, 3), 152: (3, 8, 36, 3)}
 
 
+def _bn(c, bn_momentum, relu_after=False):
+    """nn.BatchNorm2d, or the hand-written NHWC BN with the following
+    ReLU fused when PARALLAX_NATIVE_BN=1 (ops/batchnorm.py)."""
+    if _os.environ.get('PARALLAX_NATIVE_BN') == '1':
+        from parallax_amd.ops.batchnorm import ParallaxBatchNorm2d
+        return ParallaxBatchNorm2d(c, momentum=bn_momentum,
+                                   relu=relu_after)
+    return nn.BatchNorm2d(c, momentum=bn_momentum)
+
+
+def _bn_fused():
+    return _os.environ.get('PARALLAX_NATIVE_BN') == '1'
+
+
 class Bottleneck(nn.Module):
     expansion = 4
 
@@ -29,24 +45,27 @@ class Bottleneck(nn.Module):
         super().__init__()
         cout = width * self.expansion
         self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width, momentum=bn_momentum)
+        self.bn1 = _bn(width, bn_momentum, relu_after=True)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(width, momentum=bn_momentum)
+        self.bn2 = _bn(width, bn_momentum, relu_after=True)
         self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(cout, momentum=bn_momentum)
+        self.bn3 = _bn(cout, bn_momentum)
         self.relu = nn.ReLU(inplace=True)
+        # with fused BN+ReLU the separate activation after bn1/bn2 is a
+        # no-op; keep it as Identity to preserve the module structure
+        self.act = nn.Identity() if _bn_fused() else self.relu
         if stride != 1 or cin != cout:
             self.downsample = nn.Sequential(
                 nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(cout, momentum=bn_momentum))
+                _bn(cout, bn_momentum))
         else:
             self.downsample = None
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.act(self.bn1(self.conv1(x)))
+        out = self.act(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
         return self.relu(out + identity)
 

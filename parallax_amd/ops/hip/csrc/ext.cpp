@@ -53,6 +53,21 @@ std::vector<at::Tensor> gru_gates_bwd(
     const at::Tensor& dh, const at::Tensor& h_prev, const at::Tensor& r_s,
     const at::Tensor& z_s, const at::Tensor& n_t, const at::Tensor& hgn_s);
 
+// batchnorm.hip
+std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& beta,
+                                      c10::optional<at::Tensor> run_mean,
+                                      c10::optional<at::Tensor> run_var,
+                                      double momentum, double eps,
+                                      bool relu);
+std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
+                                      const at::Tensor& x2d,
+                                      const at::Tensor& mean,
+                                      const at::Tensor& invstd,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& beta, bool relu);
+
 // gemm.hip
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
@@ -121,6 +136,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sparse_sgd", &sparse_sgd);
   m.def("multi_tensor_pack", &multi_tensor_pack);
   m.def("multi_tensor_unpack", &multi_tensor_unpack);
+  m.def("batchnorm_fwd", &batchnorm_fwd);
+  m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("gru_gates_fwd", &gru_gates_fwd);
   m.def("gru_gates_bwd", &gru_gates_bwd);
   m.def("pxgemm_bt", &pxgemm_bt,
