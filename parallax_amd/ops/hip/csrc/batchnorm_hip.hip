@@ -21,7 +21,8 @@
 #define BN_MAXCPT 8  // max channels per thread (C <= 2048 at 256 threads)
 
 template <typename T, bool XHAT>
-__global__ void bn_colsum_kernel(const T* __restrict__ x,
+__global__ __launch_bounds__(PX_BLOCK)
+void bn_colsum_kernel(const T* __restrict__ x,
                                  const T* __restrict__ dy,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
@@ -70,11 +71,28 @@ __global__ void bn_colsum_kernel(const T* __restrict__ x,
       acc2[k] += v * v;
     }
   }
+  // block-level LDS accumulation first: ONE global atomic per
+  // (block, channel), not per thread — 512 blocks x C addresses instead
+  // of 131k threads hammering C cells
+  __shared__ float lsum1[256 * BN_MAXCPT];
+  __shared__ float lsum2[256 * BN_MAXCPT];
+  for (int c = tid; c < C; c += (int)blockDim.x) {
+    lsum1[c] = 0.f;
+    lsum2[c] = 0.f;
+  }
+  __syncthreads();
 #pragma unroll
   for (int k = 0; k < BN_MAXCPT; ++k) {
-    if (k < cpt && (acc1[k] != 0.f || acc2[k] != 0.f)) {
-      atomicAdd(&out1[own_c[k]], acc1[k]);
-      atomicAdd(&out2[own_c[k]], acc2[k]);
+    if (k < cpt) {
+      atomicAdd(&lsum1[own_c[k]], acc1[k]);
+      atomicAdd(&lsum2[own_c[k]], acc2[k]);
+    }
+  }
+  __syncthreads();
+  for (int c = tid; c < C; c += (int)blockDim.x) {
+    if (lsum1[c] != 0.f || lsum2[c] != 0.f) {
+      atomicAdd(&out1[c], lsum1[c]);
+      atomicAdd(&out2[c], lsum2[c]);
     }
   }
 }
@@ -170,7 +188,7 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x2d.scalar_type(), "bn_fwd", [&] {
        hipLaunchKernelGGL(( bn_colsum_kernel<scalar_t, false>)
-            , dim3(px_grid(M * C)), dim3(PX_BLOCK), 0, stream, 
+            , dim3(px_grid(M * C, PX_BLOCK, 512)), dim3(PX_BLOCK), 0, stream, 
                 reinterpret_cast<const scalar_t*>(x2d.data_ptr()), nullptr,
                 nullptr, nullptr, nullptr, nullptr, s1.data_ptr<float>(),
                 s2.data_ptr<float>(), M, C, false);
@@ -211,7 +229,7 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x2d.scalar_type(), "bn_bwd", [&] {
        hipLaunchKernelGGL(( bn_colsum_kernel<scalar_t, true>)
-            , dim3(px_grid(M * C)), dim3(PX_BLOCK), 0, stream, 
+            , dim3(px_grid(M * C, PX_BLOCK, 512)), dim3(PX_BLOCK), 0, stream, 
                 reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
                 reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
