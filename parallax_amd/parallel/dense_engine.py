@@ -159,6 +159,20 @@ class DenseAllReduceEngine(object):
             return
         denom = float(self.comm.world_size) if self.average else 1.0
         for b in self._buckets:
+            if self.overlap and b.pending > 0 \
+                    and any(p.grad is not None for p in b.params):
+                # a dense param produced no grad this step: the bucket
+                # never became ready and collective order would diverge
+                # across ranks — fail loudly (reference TF graphs always
+                # compute every registered gradient; so must step_fn)
+                missing = [i for i, p in enumerate(b.params)
+                           if p.grad is None]
+                raise RuntimeError(
+                    'dense all-reduce bucket incomplete: %d parameter(s) '
+                    'received no gradient this step (every dense '
+                    'parameter must be used in every step, or be marked '
+                    'requires_grad=False / excluded from the optimizer); '
+                    'missing slots %s' % (len(missing), missing[:8]))
             if b.flat is None:
                 continue
             if b.work is not None:
