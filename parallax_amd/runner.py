@@ -163,10 +163,27 @@ def _run_master(config):
             procs, cleanup = _spawn_workers(resource_info, config,
                                             p_to_test, stat_addr)
             if not search_p:
-                rc = procs[0].wait()
-                for p in procs[1:]:
-                    p.wait()
-                break
+                # fail-stop: any worker dying non-zero kills the job
+                # (reference master wait()s the first process,
+                # runner.py:123-124; polling all is strictly safer)
+                rc = 0
+                running = list(procs)
+                while running:
+                    time.sleep(0.5)
+                    for p in list(running):
+                        r = p.poll()
+                        if r is None:
+                            continue
+                        running.remove(p)
+                        if r != 0:
+                            parallax_log.error(
+                                'worker pid %d exited rc=%d; stopping '
+                                'the job', p.pid, r)
+                            cleanup(None, None)
+                            rc = r
+                            running = []
+                            break
+                sys.exit(rc if rc else 0)
             # collect one exec time per worker, watching for deaths
             times, alive = [], True
             while len(times) < num_workers and alive:
