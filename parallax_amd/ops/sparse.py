@@ -20,18 +20,31 @@ SPARSE_AVERAGE_BY_COUNTER = 1
 SPARSE_NO_AVERAGE = 3
 
 
-def _sorted_segments(indices):
+def _sorted_segments_presort(indices):
+    """Device-async half of _sorted_segments: sort + boundary mask, NO
+    host sync.  Several tables' presorts can be queued before any of
+    their (syncing) boundary extractions run."""
+    sorted_vals, order = torch.sort(indices)
+    n = sorted_vals.numel()
+    if n == 0:
+        return sorted_vals, order, None
+    mask = torch.ones(n, dtype=torch.bool, device=indices.device)
+    mask[1:] = sorted_vals[1:] != sorted_vals[:-1]
+    return sorted_vals, order, mask
+
+
+def _sorted_segments(indices, presorted=None):
     """Sort indices; return (sorted_vals, order, seg_starts, unique_vals).
 
     seg_starts has one entry per unique index (start position in the sorted
-    array) plus a final n sentinel."""
-    sorted_vals, order = torch.sort(indices)
+    array) plus a final n sentinel.  The ``torch.nonzero`` here is the one
+    host sync of the pipeline (dynamic unique count)."""
+    sorted_vals, order, mask = (presorted if presorted is not None
+                                else _sorted_segments_presort(indices))
     n = sorted_vals.numel()
     if n == 0:
         empty = torch.zeros(1, dtype=torch.long, device=indices.device)
         return sorted_vals, order, empty, sorted_vals
-    mask = torch.ones(n, dtype=torch.bool, device=indices.device)
-    mask[1:] = sorted_vals[1:] != sorted_vals[:-1]
     seg_starts = torch.nonzero(mask).flatten()
     unique_vals = sorted_vals[seg_starts]
     seg_starts = torch.cat([seg_starts,
@@ -40,11 +53,12 @@ def _sorted_segments(indices):
     return sorted_vals, order, seg_starts, unique_vals
 
 
-def unique_sum_rows(indices, values, average=False):
+def unique_sum_rows(indices, values, average=False, presorted=None):
     """Deduplicate ``indices`` and segment-sum the corresponding rows of
     ``values`` ([n, d] or [n]); returns (unique_indices[u], summed[u, d],
     counts[u]).  ``average=True`` divides each output row by its count
-    (reference average_option=SPARSE_AVERAGE_BY_COUNTER)."""
+    (reference average_option=SPARSE_AVERAGE_BY_COUNTER); ``presorted``
+    takes a _sorted_segments_presort result to pipeline several tables."""
     assert indices.dim() == 1
     squeeze = False
     if values.dim() == 1:
@@ -57,7 +71,8 @@ def unique_sum_rows(indices, values, average=False):
         counts = torch.zeros(0, dtype=torch.long, device=indices.device)
         return indices, out.squeeze(1) if squeeze else out, counts
 
-    sorted_vals, order, seg_starts, unique_vals = _sorted_segments(indices)
+    sorted_vals, order, seg_starts, unique_vals = \
+        _sorted_segments(indices, presorted)
     counts = seg_starts[1:] - seg_starts[:-1]
     if _ops.use_native(values):
         out = _ops.ext().segment_sum_rows(

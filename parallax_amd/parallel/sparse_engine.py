@@ -165,14 +165,20 @@ class SparsePSEngine(_SparseEngineBase):
                 'tables (see sparse_engine.py module docstring)')
 
     def reduce_and_apply(self):
-        for s in self.slots:
-            if s.ids is None:
-                continue
+        live = [s for s in self.slots if s.ids is not None]
+        if self.local_aggregation:
+            # phase-split across tables: issue every table's sort/compare
+            # first, THEN the host-syncing boundary extraction — each
+            # sync then overlaps the other tables' queued GPU work
+            from parallax_amd.ops.sparse import _sorted_segments_presort
+            pre = [_sorted_segments_presort(s.ids) for s in live]
+        for i, s in enumerate(live):
             if self.local_aggregation:
                 u_ids, u_rows, u_counts = unique_sum_rows(
                     s.ids, s.rows,
                     average=(self.average_sparse
-                             and self.comm.world_size == 1))
+                             and self.comm.world_size == 1),
+                    presorted=pre[i])
                 if self.comm.world_size == 1:
                     # already deduplicated (and averaged) — apply directly,
                     # no second sort/segment pass
