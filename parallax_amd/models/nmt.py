@@ -23,7 +23,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from parallax_amd import optim
-from parallax_amd.ops.embedding import PartitionedEmbedding, embedding_lookup
+from parallax_amd.ops.embedding import PartitionedEmbedding
 from parallax_amd.ops.layernorm import ParallaxLayerNorm
 from parallax_amd.ops.lstm import ParallaxLSTM
 from parallax_amd.ops.sampled_softmax import sampled_softmax_loss

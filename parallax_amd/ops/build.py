@@ -8,8 +8,6 @@ travels with the source tree to the GPU box.
 
 import glob
 import os
-import shutil
-import sys
 
 PKG_DIR = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 LIB_DIR = os.path.join(PKG_DIR, '_lib')

@@ -13,14 +13,13 @@ Reference behavior being reproduced:
 
 MI355X-native design: one nn.Module whose weight gradient is a torch sparse
 COO tensor (duplicate indices allowed; dedup happens in the engine's
-segment-sum kernel, contract N2).  The partitioned variant holds p shard
-Parameters so the sparse engines can place/exchange shards independently.
+segment-sum kernel, contract N2).  The partitioned variant keeps ONE
+resident table with VIRTUAL shard metadata — see PartitionedEmbedding.
 """
 
 import torch
 import torch.nn as nn
 
-from parallax_amd import ops as _ops
 from parallax_amd.ops.sparse import rows_gather
 
 

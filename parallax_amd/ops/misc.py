@@ -9,8 +9,6 @@ GPU path: multi-tensor fused L2-norm + scale kernels (ops/hip/multi_tensor.hip)
 — one launch over all grads instead of per-tensor reduce chains.
 """
 
-import math
-
 import torch
 
 from parallax_amd import ops as _ops

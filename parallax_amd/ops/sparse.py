@@ -8,7 +8,7 @@ optional per-row count average (SPARSE_AVERAGE_BY_COUNTER) vs plain sum
 (SPARSE_NO_AVERAGE).
 
 GPU path: torch.sort (rocPRIM radix sort) for ordering + hand-written HIP
-segment-sum kernel over the sorted runs (ops/hip/sparse_reduce.hip).
+scatter-atomic segment-sum kernel (ops/hip/csrc/sparse_kernels.hip).
 CPU path: the same algorithm in plain PyTorch (ground truth for tests).
 """
 

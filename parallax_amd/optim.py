@@ -17,8 +17,6 @@ All sparse state (e.g. the Adagrad accumulator) is kept per-parameter at
 full shape so checkpoints can round-trip it.
 """
 
-import math
-
 import torch
 from torch.optim.optimizer import Optimizer
 

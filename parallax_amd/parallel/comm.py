@@ -20,7 +20,6 @@ import torch
 import torch.distributed as dist
 
 from parallax_amd import consts
-from parallax_amd.lib import parallax_log
 
 
 class CommContext(object):

@@ -23,8 +23,6 @@ Design for 8xMI355X over xGMI:
 import torch
 import torch.distributed as dist
 
-from parallax_amd.lib import parallax_log
-
 
 class _Bucket(object):
     __slots__ = ('params', 'offsets', 'numel', 'flat', 'pending', 'work',

@@ -21,7 +21,7 @@ import time
 import numpy as np
 
 from parallax_amd.consts import (PARALLAX_MIN_PARTITIONS,
-                                 PARALLAX_PARTITIONS, PARALLAX_SEARCH)
+                                 PARALLAX_PARTITIONS)
 from parallax_amd.lib import parallax_log
 
 

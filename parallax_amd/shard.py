@@ -8,7 +8,6 @@ filter parameterized by env-derived (num_shards, shard_id), fixed at worker
 start instead of graph-patch time.
 """
 
-import itertools
 import os
 
 from parallax_amd import consts

@@ -21,8 +21,8 @@ available as ``classify_by_probe`` for models whose structure hides the
 lookup.
 """
 
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from dataclasses import dataclass
+from typing import List, Optional
 
 import torch
 import torch.nn as nn
