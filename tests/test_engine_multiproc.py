@@ -247,6 +247,17 @@ def _body_comm_collectives(rank, world):
     for r in range(world):
         assert recv[r].shape == (r + 1, 2)
         assert torch.all(recv[r] == 10 * r + rank)
+    # exchange_rows (the N7 (ids, rows) convenience)
+    ids = [torch.arange(rank + 1) + 100 * rank for _ in range(world)]
+    rows = [torch.full((rank + 1, 3), float(rank)) for _ in range(world)]
+    rids, rrows = comm.exchange_rows(ids, rows)
+    for r in range(world):
+        assert rids[r].shape[0] == r + 1
+        assert torch.all(rrows[r] == float(r))
+    # all_gather_v_multi single counts exchange
+    multi = comm.all_gather_v_multi([t, t + 1.0])
+    assert torch.allclose(torch.cat(multi[1][0:1])[0],
+                          torch.cat(multi[0][0:1])[0] + 1.0)
 
 
 def _body_average_sparse(rank, world):
