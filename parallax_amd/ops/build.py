@@ -21,8 +21,10 @@ def build(verbose=True):
     from torch.utils import cpp_extension
 
     os.makedirs(LIB_DIR, exist_ok=True)
-    sources = sorted(glob.glob(os.path.join(CSRC, '*.cpp')) +
-                     glob.glob(os.path.join(CSRC, '*.hip')))
+    sources = sorted(
+        glob.glob(os.path.join(CSRC, '*.cpp')) +
+        [s for s in glob.glob(os.path.join(CSRC, '*.hip'))
+         if not s.endswith('_hip.hip')])  # torch-hipify build artifacts
     assert sources, 'no kernel sources found under %s' % CSRC
     mod = cpp_extension.load(
         name=EXT_NAME,
