@@ -27,6 +27,12 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                        std::vector<double> steps, double lr, double b1,
                        double b2, double eps, double wd);
+void multi_tensor_adam_dev(std::vector<at::Tensor> params,
+                           std::vector<at::Tensor> grads,
+                           std::vector<at::Tensor> ms,
+                           std::vector<at::Tensor> vs, at::Tensor step_dev,
+                           double lr, double b1, double b2, double eps,
+                           double wd);
 void multi_tensor_rmsprop(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> grads,
                           std::vector<at::Tensor> sqs,
@@ -128,6 +134,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_momentum", &multi_tensor_momentum);
   m.def("multi_tensor_adagrad", &multi_tensor_adagrad);
   m.def("multi_tensor_adam", &multi_tensor_adam);
+  m.def("multi_tensor_adam_dev", &multi_tensor_adam_dev);
   m.def("multi_tensor_rmsprop", &multi_tensor_rmsprop);
   m.def("multi_tensor_ema", &multi_tensor_ema);
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm);

@@ -95,13 +95,15 @@ struct AdagradOp {
 
 struct AdamOp {
   float lr, b1, b2, eps, wd;
+  const float* t_dev;  // device step counter (hipGraph-capturable); when
+                       // null the host scalar e[ti] is used instead
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* mv = reinterpret_cast<float*>(m.c[ti]);
     float* vv = reinterpret_cast<float*>(m.d[ti]);
-    const float t = m.e[ti];
+    const float t = t_dev ? *t_dev : m.e[ti];
     float gv = to_f32<T>(g[j]) + wd * to_f32<T>(p[j]);
     float m1 = mv[j] * b1 + (1.f - b1) * gv;
     float v1 = vv[j] * b2 + (1.f - b2) * gv * gv;
@@ -262,7 +264,29 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
   check_same_layout(params, grads, "mt_adam");
   std::vector<float> e(steps.begin(), steps.end());
   mt_launch(params, grads, ms, vs, e,
-            AdamOp{(float)lr, (float)b1, (float)b2, (float)eps, (float)wd});
+            AdamOp{(float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                   nullptr});
+}
+
+// hipGraph-capturable Adam: the step count lives in a 0-dim fp32 DEVICE
+// tensor the kernel dereferences at run time, so a captured optimizer
+// graph replays with fresh bias correction every step (the host-scalar
+// variant above bakes t into the kernel args at capture time).
+void multi_tensor_adam_dev(std::vector<at::Tensor> params,
+                           std::vector<at::Tensor> grads,
+                           std::vector<at::Tensor> ms,
+                           std::vector<at::Tensor> vs, at::Tensor step_dev,
+                           double lr, double b1, double b2, double eps,
+                           double wd) {
+  check_contig(params, "mt_adam_dev");
+  check_same_layout(params, grads, "mt_adam_dev");
+  TORCH_CHECK(step_dev.is_cuda() &&
+                  step_dev.scalar_type() == at::kFloat &&
+                  step_dev.numel() == 1,
+              "mt_adam_dev: step_dev must be a 0-dim fp32 CUDA tensor");
+  mt_launch(params, grads, ms, vs, {},
+            AdamOp{(float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                   step_dev.data_ptr<float>()});
 }
 
 void multi_tensor_rmsprop(std::vector<at::Tensor> params,

@@ -182,6 +182,13 @@ class Adam(ParallaxOptimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        # hipGraph-capturable dense path: one 0-dim fp32 device step
+        # counter per param group, shared by its dense params (the dense
+        # engine guarantees they all step together); the kernel
+        # dereferences it at run time, so a captured optimizer graph
+        # replays with fresh bias correction.
+        self._dev_steps = {}
+        self._dev_step_params = set()
 
     def _state(self, p):
         st = self.state[p]
@@ -190,6 +197,25 @@ class Adam(ParallaxOptimizer):
             st['exp_avg'] = torch.zeros_like(p, dtype=torch.float32)
             st['exp_avg_sq'] = torch.zeros_like(p, dtype=torch.float32)
         return st
+
+    # -- device-step bookkeeping (checkpoint keeps integer 'step') ----------
+    def _sync_steps_from_device(self):
+        """Under graph replay only the device counter advances; fold it
+        back into the per-param integer 'step' before serializing."""
+        for gi, buf in self._dev_steps.items():
+            t = int(buf.item())
+            for p in self.param_groups[gi]['params']:
+                if p in self._dev_step_params and p in self.state:
+                    self.state[p]['step'] = t
+
+    def state_dict(self):
+        self._sync_steps_from_device()
+        return super().state_dict()
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        self._dev_steps = {}  # re-seeded from restored steps on next step
+        self._dev_step_params = set()
 
     def _dense_step(self, params, grads, group):
         lr, (b1, b2), eps = group['lr'], group['betas'], group['eps']
@@ -202,9 +228,17 @@ class Adam(ParallaxOptimizer):
             vs.append(st['exp_avg_sq'])
             steps.append(st['step'])
         if _use_native(params):
-            _ops.ext().multi_tensor_adam(params, list(grads), ms, vs,
-                                         [float(s) for s in steps],
-                                         lr, b1, b2, eps, wd)
+            gi = next(i for i, g in enumerate(self.param_groups)
+                      if g is group)
+            buf = self._dev_steps.get(gi)
+            if buf is None:
+                buf = torch.tensor(float(steps[0] - 1), dtype=torch.float32,
+                                   device=params[0].device)
+                self._dev_steps[gi] = buf
+            self._dev_step_params.update(params)
+            buf.add_(1.0)
+            _ops.ext().multi_tensor_adam_dev(params, list(grads), ms, vs,
+                                             buf, lr, b1, b2, eps, wd)
             return
         for p, g, m, v, t in zip(params, grads, ms, vs, steps):
             g32 = g.float()

@@ -159,12 +159,8 @@ class ParallaxSession(object):
             return False
         if self._opt_graph_state == 'ready':
             return True
-        from parallax_amd import optim as _optim
-        if isinstance(self.graph.optimizer, _optim.Adam):
-            # Adam's bias correction is a per-step HOST scalar — a capture
-            # would freeze it; keep Adam eager
-            self._opt_graph_state = 'disabled'
-            return False
+        # (Adam is capturable too: its bias-correction step count lives in
+        # a 0-dim device tensor the kernel dereferences — optim.Adam)
         self._opt_graph_steps += 1
         if self._opt_graph_steps < 3:
             return False

@@ -110,6 +110,46 @@ def test_multi_tensor_adam_vs_torch():
     assert torch.allclose(p1.detach().cpu(), p2.detach(), atol=1e-4)
 
 
+def test_adam_graphed_bias_correction_advances():
+    """Adam's step count is a device tensor (multi_tensor_adam_dev) so a
+    captured optimizer hipGraph must replay with FRESH bias correction —
+    replays must match an eager torch.optim.Adam step-for-step (a frozen
+    t would diverge from step 2 on)."""
+    from parallax_amd import optim
+    torch.manual_seed(7)
+    p1 = torch.nn.Parameter(torch.randn(777, device=DEV))
+    p2 = torch.nn.Parameter(p1.detach().cpu().clone())
+    o1 = optim.Adam([p1], lr=0.05)
+    o2 = torch.optim.Adam([p2], lr=0.05)
+    grads = [torch.randn(777) for _ in range(8)]
+    gbuf = torch.empty(777, device=DEV)
+    p1.grad = gbuf
+    # one eager step so optimizer state + dev counter exist
+    gbuf.copy_(grads[0])
+    o1.step()
+    # warm-up on a side stream, then capture step()
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        gbuf.copy_(grads[1])
+        o1.step()
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        o1.step()  # records; does not execute
+    for i in range(2, 8):
+        gbuf.copy_(grads[i])
+        g.replay()
+    for i in range(8):
+        p2.grad = grads[i].clone()
+        o2.step()
+    torch.cuda.synchronize()
+    assert torch.allclose(p1.detach().cpu(), p2.detach(), atol=1e-4)
+    # checkpoint sync folds the device counter back into integer steps
+    sd = o1.state_dict()
+    assert sd['state'][0]['step'] == 8
+
+
 def test_multi_tensor_momentum_channels_last():
     """Conv weights in channels_last (non-contiguous but dense) must go
     through the multi-tensor kernel with matching-layout grads."""
