@@ -159,6 +159,10 @@ class ParallaxSession(object):
             return False
         if self._opt_graph_state == 'ready':
             return True
+        import os
+        if os.environ.get('PARALLAX_OPT_GRAPH', '1') == '0':  # debug/A-B
+            self._opt_graph_state = 'disabled'
+            return False
         # (Adam is capturable too: its bias-correction step count lives in
         # a 0-dim device tensor the kernel dereferences — optim.Adam)
         self._opt_graph_steps += 1
