@@ -4,9 +4,12 @@
 // single sum/sumsq sweep, wave64 shuffle + cross-wave LDS reduction, then
 // the normalize+affine write.  Saves (mean, rstd) fp32 per row.
 // Backward: per-row dx with the two standard row statistics
-// (mean(dy*w), mean(dy*w*xhat)); dw/db accumulated as per-block partials
-// in a [grid, D] fp32 workspace, reduced by a second kernel — no global
-// atomics on the hot path.
+// (mean(dy*w), mean(dy*w*xhat)).  dw/db: for d <= 4096 each block
+// accumulates fp32 partials in LDS (2*d floats) over its rows and flushes
+// ONCE per column with a global fp32 atomicAdd (the BN-colsum recipe —
+// grid*d atomics total, no per-row global RMW), then one tiny kernel
+// casts to the weight dtype.  Larger d falls back to a [grid, D]
+// workspace + column-sum second kernel.
 
 #include <torch/extension.h>
 #include <ATen/ATen.h>
@@ -109,6 +112,67 @@ __global__ void layernorm_bwd_kernel(
   }
 }
 
+// LDS-resident dw/db accumulation (d <= LN_LDS_MAX_D columns): one atomic
+// flush per (block, column) instead of a per-row global read-modify-write.
+#define LN_LDS_MAX_D 4096
+
+template <typename T>
+__global__ void layernorm_bwd_lds_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ rstd, T* __restrict__ dx,
+    float* __restrict__ dw32, float* __restrict__ db32, long n, long d) {
+  __shared__ float scratch[PX_BLOCK / WAVE];
+  extern __shared__ float smem[];  // [2 * d]: dw then db partials
+  float* sdw = smem;
+  float* sdb = smem + d;
+  for (long c = threadIdx.x; c < d; c += blockDim.x) {
+    sdw[c] = 0.f;
+    sdb[c] = 0.f;
+  }
+  __syncthreads();
+  for (long row = blockIdx.x; row < n; row += gridDim.x) {
+    const T* dyr = dy + row * d;
+    const T* xr = x + row * d;
+    const float m = mean[row];
+    const float r = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+    for (long c = threadIdx.x; c < d; c += blockDim.x) {
+      float dyv = to_f32<T>(dyr[c]);
+      float xh = (to_f32<T>(xr[c]) - m) * r;
+      float dyw = dyv * to_f32<T>(w[c]);
+      s1 += dyw;
+      s2 += dyw * xh;
+      sdw[c] += dyv * xh;  // LDS: same thread owns column c all rows
+      sdb[c] += dyv;
+    }
+    s1 = block_reduce_sum(s1, scratch) / d;
+    s2 = block_reduce_sum(s2, scratch) / d;
+    T* dxr = dx + row * d;
+    for (long c = threadIdx.x; c < d; c += blockDim.x) {
+      float xh = (to_f32<T>(xr[c]) - m) * r;
+      float dyw = to_f32<T>(dyr[c]) * to_f32<T>(w[c]);
+      dxr[c] = from_f32<T>(r * (dyw - s1 - xh * s2));
+    }
+  }
+  __syncthreads();
+  for (long c = threadIdx.x; c < d; c += blockDim.x) {
+    atomicAdd(&dw32[c], sdw[c]);
+    atomicAdd(&db32[c], sdb[c]);
+  }
+}
+
+template <typename T>
+__global__ void cast2_kernel(const float* __restrict__ a32,
+                             const float* __restrict__ b32,
+                             T* __restrict__ a, T* __restrict__ b, long d) {
+  for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < d;
+       c += (long)gridDim.x * blockDim.x) {
+    a[c] = from_f32<T>(a32[c]);
+    b[c] = from_f32<T>(b32[c]);
+  }
+}
+
 template <typename T>
 __global__ void colsum_kernel(const float* __restrict__ part, T* __restrict__
                               out, long rows, long d) {
@@ -130,10 +194,35 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
   auto dx = at::empty_like(x);
   auto dw = at::empty_like(w);
   auto db = at::empty_like(w);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (d <= LN_LDS_MAX_D) {
+    const int grid = (int)(n < 512 ? (n > 0 ? n : 1) : 512);
+    auto dw32 = at::zeros({d}, x.options().dtype(at::kFloat));
+    auto db32 = at::zeros({d}, x.options().dtype(at::kFloat));
+    const size_t lds = 2 * (size_t)d * sizeof(float);
+    AT_DISPATCH_FLOATING_TYPES_AND2(
+        at::kBFloat16, at::kHalf, x.scalar_type(), "layernorm_bwd", [&] {
+          layernorm_bwd_lds_kernel<scalar_t>
+              <<<grid, PX_BLOCK, lds, stream>>>(
+                  reinterpret_cast<const scalar_t*>(dy.data_ptr()),
+                  reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                  reinterpret_cast<const scalar_t*>(
+                      w.contiguous().data_ptr()),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                  dw32.data_ptr<float>(), db32.data_ptr<float>(), n, d);
+          HIP_CHECK_KERNEL();
+          cast2_kernel<scalar_t><<<px_grid(d), PX_BLOCK, 0, stream>>>(
+              dw32.data_ptr<float>(), db32.data_ptr<float>(),
+              reinterpret_cast<scalar_t*>(dw.data_ptr()),
+              reinterpret_cast<scalar_t*>(db.data_ptr()), d);
+          HIP_CHECK_KERNEL();
+        });
+    return {dx, dw, db};
+  }
   const int grid = (int)(n < 256 ? (n > 0 ? n : 1) : 256);
   auto dw_part = at::zeros({grid, d}, x.options().dtype(at::kFloat));
   auto db_part = at::zeros({grid, d}, x.options().dtype(at::kFloat));
-  auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x.scalar_type(), "layernorm_bwd", [&] {
         layernorm_bwd_kernel<scalar_t><<<grid, PX_BLOCK, 0, stream>>>(

@@ -220,6 +220,35 @@ def test_layernorm_fwd_bwd_vs_torch():
     assert torch.allclose(b.grad.cpu(), b2.grad, atol=1e-3)
 
 
+def test_layernorm_bwd_large_d_and_bf16():
+    """d=512 runs the LDS-accumulated dw/db path; d=6000 (> LN_LDS_MAX_D)
+    exercises the [grid, D]-workspace fallback.  Both vs fp32 torch."""
+    from parallax_amd.ops.layernorm import layer_norm
+    torch.manual_seed(17)
+    for d, dtype, atol in ((6000, torch.float32, 1e-3),
+                           (512, torch.bfloat16, 5e-2)):
+        x = torch.randn(300, d, device=DEV, dtype=dtype,
+                        requires_grad=True)
+        w = torch.randn(d, device=DEV, dtype=dtype, requires_grad=True)
+        b = torch.randn(d, device=DEV, dtype=dtype, requires_grad=True)
+        y = layer_norm(x, (d,), w, b)
+        x2 = x.detach().cpu().float().requires_grad_(True)
+        w2 = w.detach().cpu().float().requires_grad_(True)
+        b2 = b.detach().cpu().float().requires_grad_(True)
+        y2 = torch.nn.functional.layer_norm(x2, (d,), w2, b2)
+        dy = torch.randn(300, d)
+        y.backward(dy.to(DEV).to(dtype))
+        y2.backward(dy)
+        assert torch.allclose(y.detach().cpu().float(), y2.detach(),
+                              atol=atol)
+        assert torch.allclose(x.grad.cpu().float(), x2.grad, atol=atol)
+        # dw/db sum over 300 rows — scale tolerance with the reduction
+        assert torch.allclose(w.grad.cpu().float(), w2.grad,
+                              atol=atol * 30, rtol=1e-2)
+        assert torch.allclose(b.grad.cpu().float(), b2.grad,
+                              atol=atol * 30, rtol=1e-2)
+
+
 # --- softmax CE -------------------------------------------------------------
 
 def test_softmax_ce_fwd_bwd_vs_torch():
