@@ -114,7 +114,12 @@ __global__ void layernorm_bwd_kernel(
 
 // LDS-resident dw/db accumulation (d <= LN_LDS_MAX_D columns): one atomic
 // flush per (block, column) instead of a per-row global read-modify-write.
+// The flush stripes over LN_NBUF partial buffers (blockIdx % LN_NBUF) —
+// with one buffer, 512 blocks hammering the same d cells serialize ~512
+// deep per cell; striping cuts the contention chain to grid/LN_NBUF.
+// The cast kernel folds the stripes.
 #define LN_LDS_MAX_D 4096
+#define LN_NBUF 16
 
 template <typename T>
 __global__ void layernorm_bwd_lds_kernel(
@@ -156,20 +161,29 @@ __global__ void layernorm_bwd_lds_kernel(
     }
   }
   __syncthreads();
+  float* dwb = dw32 + (blockIdx.x % LN_NBUF) * d;
+  float* dbb = db32 + (blockIdx.x % LN_NBUF) * d;
   for (long c = threadIdx.x; c < d; c += blockDim.x) {
-    atomicAdd(&dw32[c], sdw[c]);
-    atomicAdd(&db32[c], sdb[c]);
+    atomicAdd(&dwb[c], sdw[c]);
+    atomicAdd(&dbb[c], sdb[c]);
   }
 }
 
+// Fold the LN_NBUF stripes and cast to the weight dtype in one launch.
 template <typename T>
-__global__ void cast2_kernel(const float* __restrict__ a32,
-                             const float* __restrict__ b32,
-                             T* __restrict__ a, T* __restrict__ b, long d) {
+__global__ void fold_cast2_kernel(const float* __restrict__ a32,
+                                  const float* __restrict__ b32,
+                                  T* __restrict__ a, T* __restrict__ b,
+                                  long d) {
   for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < d;
        c += (long)gridDim.x * blockDim.x) {
-    a[c] = from_f32<T>(a32[c]);
-    b[c] = from_f32<T>(b32[c]);
+    float sa = 0.f, sb = 0.f;
+    for (int k = 0; k < LN_NBUF; ++k) {
+      sa += a32[k * d + c];
+      sb += b32[k * d + c];
+    }
+    a[c] = from_f32<T>(sa);
+    b[c] = from_f32<T>(sb);
   }
 }
 
@@ -197,8 +211,8 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
   auto stream = at::cuda::getCurrentCUDAStream();
   if (d <= LN_LDS_MAX_D) {
     const int grid = (int)(n < 512 ? (n > 0 ? n : 1) : 512);
-    auto dw32 = at::zeros({d}, x.options().dtype(at::kFloat));
-    auto db32 = at::zeros({d}, x.options().dtype(at::kFloat));
+    auto dw32 = at::zeros({LN_NBUF * d}, x.options().dtype(at::kFloat));
+    auto db32 = at::zeros({LN_NBUF * d}, x.options().dtype(at::kFloat));
     const size_t lds = 2 * (size_t)d * sizeof(float);
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::kBFloat16, at::kHalf, x.scalar_type(), "layernorm_bwd", [&] {
@@ -212,7 +226,7 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
                   reinterpret_cast<scalar_t*>(dx.data_ptr()),
                   dw32.data_ptr<float>(), db32.data_ptr<float>(), n, d);
           HIP_CHECK_KERNEL();
-          cast2_kernel<scalar_t><<<px_grid(d), PX_BLOCK, 0, stream>>>(
+          fold_cast2_kernel<scalar_t><<<px_grid(d), PX_BLOCK, 0, stream>>>(
               dw32.data_ptr<float>(), db32.data_ptr<float>(),
               reinterpret_cast<scalar_t*>(dw.data_ptr()),
               reinterpret_cast<scalar_t*>(db.data_ptr()), d);
