@@ -4,7 +4,11 @@ on an NMT-like dense param set (~55M fp32 params in ~150 tensors).
 Run on a GPU box:  python scripts/bench_optstep.py
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
 
 import torch
 
