@@ -25,16 +25,38 @@ struct MTMeta {
   void* d[MT_MAX];    // state 2 (v / momentum buf)
   float e[MT_MAX];    // per-tensor scalar (adam step count)
   long cum[MT_MAX + 1];
+  long vcum[MT_MAX + 1];  // cumulative count of 4-element vector chunks
   int n;
 };
 
-__device__ __forceinline__ int mt_find(const MTMeta& m, long idx) {
-  int lo = 0, hi = m.n - 1;
+__device__ __forceinline__ int mt_find_in(const long* cum, int n, long idx) {
+  int lo = 0, hi = n - 1;
   while (lo < hi) {
     int mid = (lo + hi) >> 1;
-    if (idx < m.cum[mid + 1]) hi = mid; else lo = mid + 1;
+    if (idx < cum[mid + 1]) hi = mid; else lo = mid + 1;
   }
   return lo;
+}
+
+__device__ __forceinline__ int mt_find(const MTMeta& m, long idx) {
+  return mt_find_in(m.cum, m.n, idx);
+}
+
+// 4-wide vector of T with natural vector alignment (float4 -> dwordx4
+// loads, bf16x4 -> dwordx2) for the vectorized multi-tensor path.
+template <typename T>
+struct alignas(4 * sizeof(T)) V4 {
+  T v[4];
+};
+
+template <typename T>
+__device__ __forceinline__ V4<T> ld4(const T* p, long j) {
+  return *reinterpret_cast<const V4<T>*>(p + j);
+}
+
+template <typename T>
+__device__ __forceinline__ void st4(T* p, long j, const V4<T>& x) {
+  *reinterpret_cast<V4<T>*>(p + j) = x;
 }
 
 // Functor-driven elementwise multi-tensor kernel.  OP provides
@@ -50,53 +72,141 @@ __global__ void mt_kernel(MTMeta meta, OP op) {
   }
 }
 
+// Vectorized variant: grid-stride over 4-element chunks (vcum index
+// space, one binary search per chunk, dwordx4/x2 accesses); the <=3
+// tail elements of each tensor are finished by block 0's first lanes.
+// Requires every tensor pointer 16-byte aligned (checked on host).
+template <typename T, typename OP>
+__global__ void mt_kernel_v4(MTMeta meta, OP op) {
+  const long vtotal = meta.vcum[meta.n];
+  for (long vidx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+       vidx < vtotal; vidx += (long)gridDim.x * blockDim.x) {
+    const int ti = mt_find_in(meta.vcum, meta.n, vidx);
+    const long j = 4 * (vidx - meta.vcum[ti]);
+    op.template apply4<T>(meta, ti, j);
+  }
+  if (blockIdx.x == 0 && threadIdx.x < 4 * (unsigned)meta.n) {
+    const int ti = threadIdx.x >> 2;
+    const long numel = meta.cum[ti + 1] - meta.cum[ti];
+    const long j = (numel & ~3L) + (threadIdx.x & 3);
+    if (j < numel) op.template apply<T>(meta, ti, j);
+  }
+}
+
 // --- op functors -----------------------------------------------------------
+//
+// Each op provides the scalar per-element math (elem), the scalar global
+// apply (tails + misaligned fallback), and apply4 — explicit V4 loads /
+// stores (dwordx4 for fp32, dwordx2 for bf16) with the math unrolled over
+// the four lanes.  `vec` gates the vectorized kernel.
 
 struct SGDOp {
+  static constexpr bool vec = true;
   float lr, wd;
+  __device__ __forceinline__ float elem(float pv, float gv) const {
+    return pv - lr * (gv + wd * pv);
+  }
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
-    float gv = to_f32<T>(g[j]) + wd * to_f32<T>(p[j]);
-    p[j] = from_f32<T>(to_f32<T>(p[j]) - lr * gv);
+    p[j] = from_f32<T>(elem(to_f32<T>(p[j]), to_f32<T>(g[j])));
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    T* p = reinterpret_cast<T*>(m.a[ti]);
+    const T* g = reinterpret_cast<const T*>(m.b[ti]);
+    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      pv.v[k] = from_f32<T>(elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k])));
+    st4(p, j, pv);
   }
 };
 
 struct MomentumOp {
+  static constexpr bool vec = true;
   float lr, mom, wd;
   bool nesterov;
+  __device__ __forceinline__ float elem(float pv, float gv,
+                                        float& buf) const {
+    float gw = gv + wd * pv;
+    float b = buf * mom + gw;
+    buf = b;
+    float upd = nesterov ? (gw + mom * b) : b;
+    return pv - lr * upd;
+  }
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* buf = reinterpret_cast<float*>(m.c[ti]);
-    float gv = to_f32<T>(g[j]) + wd * to_f32<T>(p[j]);
-    float b = buf[j] * mom + gv;
-    buf[j] = b;
-    float upd = nesterov ? (gv + mom * b) : b;
-    p[j] = from_f32<T>(to_f32<T>(p[j]) - lr * upd);
+    p[j] = from_f32<T>(elem(to_f32<T>(p[j]), to_f32<T>(g[j]), buf[j]));
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    T* p = reinterpret_cast<T*>(m.a[ti]);
+    const T* g = reinterpret_cast<const T*>(m.b[ti]);
+    float* buf = reinterpret_cast<float*>(m.c[ti]);
+    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    V4<float> bv = ld4(buf, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      pv.v[k] = from_f32<T>(
+          elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]), bv.v[k]));
+    st4(p, j, pv);
+    st4(buf, j, bv);
   }
 };
 
 struct AdagradOp {
+  static constexpr bool vec = true;
   float lr, eps;
+  __device__ __forceinline__ float elem(float pv, float gv,
+                                        float& acc) const {
+    float a = acc + gv * gv;
+    acc = a;
+    return pv - lr * gv / (sqrtf(a) + eps);
+  }
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* acc = reinterpret_cast<float*>(m.c[ti]);
-    float gv = to_f32<T>(g[j]);
-    float a = acc[j] + gv * gv;
-    acc[j] = a;
-    p[j] = from_f32<T>(to_f32<T>(p[j]) - lr * gv / (sqrtf(a) + eps));
+    p[j] = from_f32<T>(elem(to_f32<T>(p[j]), to_f32<T>(g[j]), acc[j]));
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    T* p = reinterpret_cast<T*>(m.a[ti]);
+    const T* g = reinterpret_cast<const T*>(m.b[ti]);
+    float* acc = reinterpret_cast<float*>(m.c[ti]);
+    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    V4<float> av = ld4(acc, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      pv.v[k] = from_f32<T>(
+          elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]), av.v[k]));
+    st4(p, j, pv);
+    st4(acc, j, av);
   }
 };
 
 struct AdamOp {
+  static constexpr bool vec = true;
   float lr, b1, b2, eps, wd;
   const float* t_dev;  // device step counter (hipGraph-capturable); when
                        // null the host scalar e[ti] is used instead
+  __device__ __forceinline__ float elem(float pv, float gv, float& mv,
+                                        float& vv, float bc1,
+                                        float bc2) const {
+    float gw = gv + wd * pv;
+    float m1 = mv * b1 + (1.f - b1) * gw;
+    float v1 = vv * b2 + (1.f - b2) * gw * gw;
+    mv = m1;
+    vv = v1;
+    float denom = sqrtf(v1 / bc2) + eps;
+    return pv - (lr / bc1) * m1 / denom;
+  }
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
@@ -104,49 +214,98 @@ struct AdamOp {
     float* mv = reinterpret_cast<float*>(m.c[ti]);
     float* vv = reinterpret_cast<float*>(m.d[ti]);
     const float t = t_dev ? *t_dev : m.e[ti];
-    float gv = to_f32<T>(g[j]) + wd * to_f32<T>(p[j]);
-    float m1 = mv[j] * b1 + (1.f - b1) * gv;
-    float v1 = vv[j] * b2 + (1.f - b2) * gv * gv;
-    mv[j] = m1;
-    vv[j] = v1;
     const float bc1 = 1.f - powf(b1, t);
     const float bc2 = 1.f - powf(b2, t);
-    float denom = sqrtf(v1 / bc2) + eps;
-    p[j] = from_f32<T>(to_f32<T>(p[j]) - (lr / bc1) * m1 / denom);
+    p[j] = from_f32<T>(
+        elem(to_f32<T>(p[j]), to_f32<T>(g[j]), mv[j], vv[j], bc1, bc2));
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    T* p = reinterpret_cast<T*>(m.a[ti]);
+    const T* g = reinterpret_cast<const T*>(m.b[ti]);
+    float* mp = reinterpret_cast<float*>(m.c[ti]);
+    float* vp = reinterpret_cast<float*>(m.d[ti]);
+    const float t = t_dev ? *t_dev : m.e[ti];
+    const float bc1 = 1.f - powf(b1, t);
+    const float bc2 = 1.f - powf(b2, t);
+    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    V4<float> mv = ld4(mp, j), vv = ld4(vp, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      pv.v[k] = from_f32<T>(elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]),
+                                 mv.v[k], vv.v[k], bc1, bc2));
+    st4(p, j, pv);
+    st4(mp, j, mv);
+    st4(vp, j, vv);
   }
 };
 
 struct RMSPropOp {
+  static constexpr bool vec = true;
   float lr, decay, mom, eps;
+  __device__ __forceinline__ float elem(float pv, float gv, float& sq,
+                                        float& buf) const {
+    float s = sq * decay + (1.f - decay) * gv * gv;
+    sq = s;
+    float upd = gv / (sqrtf(s) + eps);
+    float b = buf * mom + upd;
+    buf = b;
+    return pv - lr * b;
+  }
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* sq = reinterpret_cast<float*>(m.c[ti]);
     float* buf = reinterpret_cast<float*>(m.d[ti]);
-    float gv = to_f32<T>(g[j]);
-    float s = sq[j] * decay + (1.f - decay) * gv * gv;
-    sq[j] = s;
-    float upd = gv / (sqrtf(s) + eps);
-    float b = buf[j] * mom + upd;
-    buf[j] = b;
-    p[j] = from_f32<T>(to_f32<T>(p[j]) - lr * b);
+    p[j] = from_f32<T>(
+        elem(to_f32<T>(p[j]), to_f32<T>(g[j]), sq[j], buf[j]));
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    T* p = reinterpret_cast<T*>(m.a[ti]);
+    const T* g = reinterpret_cast<const T*>(m.b[ti]);
+    float* sp = reinterpret_cast<float*>(m.c[ti]);
+    float* bp = reinterpret_cast<float*>(m.d[ti]);
+    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    V4<float> sv = ld4(sp, j), bv = ld4(bp, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      pv.v[k] = from_f32<T>(elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]),
+                                 sv.v[k], bv.v[k]));
+    st4(p, j, pv);
+    st4(sp, j, sv);
+    st4(bp, j, bv);
   }
 };
 
 struct EMAOp {
+  static constexpr bool vec = true;
   float decay;
+  __device__ __forceinline__ float elem(float sv, float vv) const {
+    return decay * sv + (1.f - decay) * vv;
+  }
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* shadow = reinterpret_cast<T*>(m.a[ti]);
     const T* val = reinterpret_cast<const T*>(m.b[ti]);
-    shadow[j] = from_f32<T>(decay * to_f32<T>(shadow[j]) +
-                            (1.f - decay) * to_f32<T>(val[j]));
+    shadow[j] = from_f32<T>(elem(to_f32<T>(shadow[j]), to_f32<T>(val[j])));
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    T* shadow = reinterpret_cast<T*>(m.a[ti]);
+    const T* val = reinterpret_cast<const T*>(m.b[ti]);
+    V4<T> sv = ld4(shadow, j), vv = ld4(val, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      sv.v[k] = from_f32<T>(elem(to_f32<T>(sv.v[k]), to_f32<T>(vv.v[k])));
+    st4(shadow, j, sv);
   }
 };
 
 struct ScaleOp {
-  const float* norm;  // device scalar
+  static constexpr bool vec = false;  // single pass over one list; scalar
+  const float* norm;                  // device scalar
   float clip;
   template <typename T>
   __device__ void apply(const MTMeta& m, int ti, long j) const {
@@ -172,6 +331,11 @@ static void mt_launch(const std::vector<at::Tensor>& a,
     MTMeta meta;
     meta.n = 0;
     meta.cum[0] = 0;
+    meta.vcum[0] = 0;
+    bool aligned = true;
+    auto chk = [&aligned](const void* p) {
+      if (p && (reinterpret_cast<uintptr_t>(p) & 15)) aligned = false;
+    };
     while (i < a.size() && meta.n < MT_MAX) {
       int k = meta.n;
       meta.a[k] = a[i].data_ptr();
@@ -180,12 +344,27 @@ static void mt_launch(const std::vector<at::Tensor>& a,
       meta.d[k] = d.empty() ? nullptr : d[i].data_ptr();
       meta.e[k] = e.empty() ? 0.f : e[i];
       meta.cum[k + 1] = meta.cum[k] + a[i].numel();
+      meta.vcum[k + 1] = meta.vcum[k] + a[i].numel() / 4;
+      chk(meta.a[k]);
+      chk(meta.b[k]);
+      chk(meta.c[k]);
+      chk(meta.d[k]);
       meta.n++;
       i++;
     }
     const long total = meta.cum[meta.n];
+    const long vtotal = meta.vcum[meta.n];
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::kBFloat16, at::kHalf, a[0].scalar_type(), "mt_launch", [&] {
+          if constexpr (OP::vec) {
+            if (aligned) {
+              mt_kernel_v4<scalar_t, OP>
+                  <<<px_grid(vtotal > 0 ? vtotal : 1), PX_BLOCK, 0,
+                     stream>>>(meta, op);
+              HIP_CHECK_KERNEL();
+              return;
+            }
+          }
           mt_kernel<scalar_t, OP>
               <<<px_grid(total), PX_BLOCK, 0, stream>>>(meta, op);
           HIP_CHECK_KERNEL();
