@@ -47,6 +47,11 @@ def parse_args():
                    action='store_true')
     p.add_argument('--no-cuda-graph', dest='cuda_graph',
                    action='store_false')
+    p.add_argument('--pure-bf16', dest='pure_bf16', default=True,
+                   action='store_true',
+                   help='nmt: bf16 body + fused fp32 master weights')
+    p.add_argument('--no-pure-bf16', dest='pure_bf16',
+                   action='store_false')
     return p.parse_args()
 
 
@@ -89,7 +94,8 @@ def main():
         from parallax_amd.models import nmt as M
         batch = args.batch_size or 64
         tgt_len = 48
-        build = lambda: M.build_nmt_graph(arch='transformer')
+        build = lambda: M.build_nmt_graph(
+            arch='transformer', pure_bf16=args.pure_bf16 and on_gpu)
         metric, unit = 'NMT tgt words/sec', 'words/s'
         per_step_items = batch * tgt_len
         baseline = None  # no published reference number

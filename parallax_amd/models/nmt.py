@@ -36,13 +36,20 @@ TGT_VOCAB = 32000
 
 
 class _MHA(nn.Module):
-    def __init__(self, d, heads):
+    """Self- (fused qkv) or cross- (separate q / kv) attention.  Only the
+    projections the instance actually runs are created — an unused dense
+    parameter would produce no gradient and trip the dense engine's
+    incomplete-bucket guard at world_size > 1."""
+
+    def __init__(self, d, heads, cross=False):
         super().__init__()
         self.d = d
         self.h = heads
-        self.qkv = nn.Linear(d, 3 * d, bias=False)
-        self.kv = nn.Linear(d, 2 * d, bias=False)
-        self.q = nn.Linear(d, d, bias=False)
+        if cross:
+            self.kv = nn.Linear(d, 2 * d, bias=False)
+            self.q = nn.Linear(d, d, bias=False)
+        else:
+            self.qkv = nn.Linear(d, 3 * d, bias=False)
         self.out = nn.Linear(d, d, bias=False)
 
     def forward(self, x, mem=None, causal=False):
@@ -70,7 +77,7 @@ class _Block(nn.Module):
         self.cross = cross
         if cross:
             self.ln_x = ParallaxLayerNorm(d)
-            self.xattn = _MHA(d, heads)
+            self.xattn = _MHA(d, heads, cross=True)
         self.ln2 = ParallaxLayerNorm(d)
         self.fc1 = nn.Linear(d, ffn)
         self.fc2 = nn.Linear(ffn, d)
@@ -107,10 +114,14 @@ class NMTTransformer(nn.Module):
 
     def forward(self, src, tgt_in, tgt_out):
         scale = math.sqrt(self.d_model)
-        e = self.src_emb(src) * scale + self.pos[:src.shape[1]]
+        # lookups come from the (always-fp32) sparse tables; cast to the
+        # body dtype (bf16 in pure-bf16 mode, no-op otherwise)
+        body = self.pos.dtype
+        e = self.src_emb(src).to(body) * scale + self.pos[:src.shape[1]]
         for blk in self.enc:
             e = blk(e)
-        d = self.tgt_emb(tgt_in) * scale + self.pos[:tgt_in.shape[1]]
+        d = self.tgt_emb(tgt_in).to(body) * scale \
+            + self.pos[:tgt_in.shape[1]]
         for blk in self.dec:
             d = blk(d, mem=e, causal=True)
         d = self.ln_f(d)
@@ -176,18 +187,27 @@ class GNMTAttention(nn.Module):
 def build_nmt_graph(arch='transformer', src_vocab=SRC_VOCAB,
                     tgt_vocab=TGT_VOCAB, d_model=512, heads=8, ffn=2048,
                     enc_layers=6, dec_layers=6, min_partitions=8, lr=1e-3,
-                    use_bf16=True, **kw):
+                    use_bf16=True, pure_bf16=False, **kw):
+    """pure_bf16 (transformer only): the dense body lives in bf16 — no
+    per-step autocast weight-cast kernels — with fp32 update precision via
+    the optimizers' fused master-weight path; the sparse embedding tables
+    stay fp32 (row updates keep full precision)."""
     if arch == 'transformer':
         model = NMTTransformer(src_vocab, tgt_vocab, d_model, heads, ffn,
                                enc_layers, dec_layers, min_partitions)
+        if pure_bf16:
+            model.to(torch.bfloat16)
+            model.src_emb.float()
+            model.tgt_emb.float()
     else:
+        pure_bf16 = False
         model = GNMTAttention(src_vocab, tgt_vocab, d_model,
                               kw.get('units', d_model), min_partitions)
     optimizer = optim.Adam(model.parameters(), lr=lr)
 
     def step_fn(m, batch):
         src, tgt_in, tgt_out = batch
-        if use_bf16 and src.is_cuda:
+        if use_bf16 and not pure_bf16 and src.is_cuda:
             with torch.autocast('cuda', dtype=torch.bfloat16):
                 loss = m(src, tgt_in, tgt_out)
         else:

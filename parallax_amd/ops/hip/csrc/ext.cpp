@@ -13,31 +13,35 @@ at::Tensor segment_sum_rows(const at::Tensor& values, const at::Tensor& order,
 
 // optim_kernels.hip
 void multi_tensor_sgd(std::vector<at::Tensor> params,
-                      std::vector<at::Tensor> grads, double lr, double wd);
+                      std::vector<at::Tensor> grads, double lr, double wd,
+                      std::vector<at::Tensor> masters);
 void multi_tensor_momentum(std::vector<at::Tensor> params,
                            std::vector<at::Tensor> grads,
                            std::vector<at::Tensor> bufs, double lr,
-                           double mom, double wd, bool nesterov);
+                           double mom, double wd, bool nesterov,
+                           std::vector<at::Tensor> masters);
 void multi_tensor_adagrad(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> grads,
                           std::vector<at::Tensor> accums, double lr,
-                          double eps);
+                          double eps, std::vector<at::Tensor> masters);
 void multi_tensor_adam(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> grads,
                        std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                        std::vector<double> steps, double lr, double b1,
-                       double b2, double eps, double wd);
+                       double b2, double eps, double wd,
+                       std::vector<at::Tensor> masters);
 void multi_tensor_adam_dev(std::vector<at::Tensor> params,
                            std::vector<at::Tensor> grads,
                            std::vector<at::Tensor> ms,
                            std::vector<at::Tensor> vs, at::Tensor step_dev,
                            double lr, double b1, double b2, double eps,
-                           double wd);
+                           double wd, std::vector<at::Tensor> masters);
 void multi_tensor_rmsprop(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> grads,
                           std::vector<at::Tensor> sqs,
                           std::vector<at::Tensor> bufs, double lr,
-                          double decay, double mom, double eps);
+                          double decay, double mom, double eps,
+                          std::vector<at::Tensor> masters);
 void multi_tensor_ema(std::vector<at::Tensor> shadows,
                       std::vector<at::Tensor> values, double decay);
 at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors);

@@ -23,6 +23,10 @@ struct MTMeta {
   void* b[MT_MAX];    // grad
   void* c[MT_MAX];    // state 1 (momentum / accum / m / square_avg)
   void* d[MT_MAX];    // state 2 (v / momentum buf)
+  void* f[MT_MAX];    // fp32 master weight (nullptr when param IS fp32):
+                      // the update reads/writes the master and stores the
+                      // rounded copy into the (bf16) param — pure-bf16
+                      // training without per-step cast kernels
   float e[MT_MAX];    // per-tensor scalar (adam step count)
   long cum[MT_MAX + 1];
   long vcum[MT_MAX + 1];  // cumulative count of 4-element vector chunks
@@ -57,6 +61,53 @@ __device__ __forceinline__ V4<T> ld4(const T* p, long j) {
 template <typename T>
 __device__ __forceinline__ void st4(T* p, long j, const V4<T>& x) {
   *reinterpret_cast<V4<T>*>(p + j) = x;
+}
+
+// Param read/write through the optional fp32 master (meta.f[ti]).
+template <typename T>
+__device__ __forceinline__ float ldp(const MTMeta& m, int ti, long j,
+                                     const T* p) {
+  const float* mw = reinterpret_cast<const float*>(m.f[ti]);
+  return mw ? mw[j] : to_f32<T>(p[j]);
+}
+
+template <typename T>
+__device__ __forceinline__ void stp(const MTMeta& m, int ti, long j, T* p,
+                                    float r) {
+  float* mw = reinterpret_cast<float*>(m.f[ti]);
+  if (mw) mw[j] = r;
+  p[j] = from_f32<T>(r);
+}
+
+template <typename T>
+__device__ __forceinline__ void ldp4(const MTMeta& m, int ti, long j,
+                                     const T* p, float out[4]) {
+  const float* mw = reinterpret_cast<const float*>(m.f[ti]);
+  if (mw) {
+    V4<float> v = ld4(mw, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) out[k] = v.v[k];
+  } else {
+    V4<T> v = ld4(p, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) out[k] = to_f32<T>(v.v[k]);
+  }
+}
+
+template <typename T>
+__device__ __forceinline__ void stp4(const MTMeta& m, int ti, long j, T* p,
+                                     const float r[4]) {
+  float* mw = reinterpret_cast<float*>(m.f[ti]);
+  if (mw) {
+    V4<float> v;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) v.v[k] = r[k];
+    st4(mw, j, v);
+  }
+  V4<T> v;
+#pragma unroll
+  for (int k = 0; k < 4; ++k) v.v[k] = from_f32<T>(r[k]);
+  st4(p, j, v);
 }
 
 // Functor-driven elementwise multi-tensor kernel.  OP provides
@@ -110,17 +161,18 @@ struct SGDOp {
   __device__ void apply(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
-    p[j] = from_f32<T>(elem(to_f32<T>(p[j]), to_f32<T>(g[j])));
+    stp(m, ti, j, p, elem(ldp(m, ti, j, p), to_f32<T>(g[j])));
   }
   template <typename T>
   __device__ void apply4(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
-    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    float pw[4];
+    ldp4(m, ti, j, p, pw);
+    V4<T> gv = ld4(g, j);
 #pragma unroll
-    for (int k = 0; k < 4; ++k)
-      pv.v[k] = from_f32<T>(elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k])));
-    st4(p, j, pv);
+    for (int k = 0; k < 4; ++k) pw[k] = elem(pw[k], to_f32<T>(gv.v[k]));
+    stp4(m, ti, j, p, pw);
   }
 };
 
@@ -141,20 +193,21 @@ struct MomentumOp {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* buf = reinterpret_cast<float*>(m.c[ti]);
-    p[j] = from_f32<T>(elem(to_f32<T>(p[j]), to_f32<T>(g[j]), buf[j]));
+    stp(m, ti, j, p, elem(ldp(m, ti, j, p), to_f32<T>(g[j]), buf[j]));
   }
   template <typename T>
   __device__ void apply4(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* buf = reinterpret_cast<float*>(m.c[ti]);
-    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    float pw[4];
+    ldp4(m, ti, j, p, pw);
+    V4<T> gv = ld4(g, j);
     V4<float> bv = ld4(buf, j);
 #pragma unroll
     for (int k = 0; k < 4; ++k)
-      pv.v[k] = from_f32<T>(
-          elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]), bv.v[k]));
-    st4(p, j, pv);
+      pw[k] = elem(pw[k], to_f32<T>(gv.v[k]), bv.v[k]);
+    stp4(m, ti, j, p, pw);
     st4(buf, j, bv);
   }
 };
@@ -173,20 +226,21 @@ struct AdagradOp {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* acc = reinterpret_cast<float*>(m.c[ti]);
-    p[j] = from_f32<T>(elem(to_f32<T>(p[j]), to_f32<T>(g[j]), acc[j]));
+    stp(m, ti, j, p, elem(ldp(m, ti, j, p), to_f32<T>(g[j]), acc[j]));
   }
   template <typename T>
   __device__ void apply4(const MTMeta& m, int ti, long j) const {
     T* p = reinterpret_cast<T*>(m.a[ti]);
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* acc = reinterpret_cast<float*>(m.c[ti]);
-    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    float pw[4];
+    ldp4(m, ti, j, p, pw);
+    V4<T> gv = ld4(g, j);
     V4<float> av = ld4(acc, j);
 #pragma unroll
     for (int k = 0; k < 4; ++k)
-      pv.v[k] = from_f32<T>(
-          elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]), av.v[k]));
-    st4(p, j, pv);
+      pw[k] = elem(pw[k], to_f32<T>(gv.v[k]), av.v[k]);
+    stp4(m, ti, j, p, pw);
     st4(acc, j, av);
   }
 };
@@ -216,8 +270,8 @@ struct AdamOp {
     const float t = t_dev ? *t_dev : m.e[ti];
     const float bc1 = 1.f - powf(b1, t);
     const float bc2 = 1.f - powf(b2, t);
-    p[j] = from_f32<T>(
-        elem(to_f32<T>(p[j]), to_f32<T>(g[j]), mv[j], vv[j], bc1, bc2));
+    stp(m, ti, j, p,
+        elem(ldp(m, ti, j, p), to_f32<T>(g[j]), mv[j], vv[j], bc1, bc2));
   }
   template <typename T>
   __device__ void apply4(const MTMeta& m, int ti, long j) const {
@@ -228,13 +282,14 @@ struct AdamOp {
     const float t = t_dev ? *t_dev : m.e[ti];
     const float bc1 = 1.f - powf(b1, t);
     const float bc2 = 1.f - powf(b2, t);
-    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    float pw[4];
+    ldp4(m, ti, j, p, pw);
+    V4<T> gv = ld4(g, j);
     V4<float> mv = ld4(mp, j), vv = ld4(vp, j);
 #pragma unroll
     for (int k = 0; k < 4; ++k)
-      pv.v[k] = from_f32<T>(elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]),
-                                 mv.v[k], vv.v[k], bc1, bc2));
-    st4(p, j, pv);
+      pw[k] = elem(pw[k], to_f32<T>(gv.v[k]), mv.v[k], vv.v[k], bc1, bc2);
+    stp4(m, ti, j, p, pw);
     st4(mp, j, mv);
     st4(vp, j, vv);
   }
@@ -258,8 +313,8 @@ struct RMSPropOp {
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* sq = reinterpret_cast<float*>(m.c[ti]);
     float* buf = reinterpret_cast<float*>(m.d[ti]);
-    p[j] = from_f32<T>(
-        elem(to_f32<T>(p[j]), to_f32<T>(g[j]), sq[j], buf[j]));
+    stp(m, ti, j, p,
+        elem(ldp(m, ti, j, p), to_f32<T>(g[j]), sq[j], buf[j]));
   }
   template <typename T>
   __device__ void apply4(const MTMeta& m, int ti, long j) const {
@@ -267,13 +322,14 @@ struct RMSPropOp {
     const T* g = reinterpret_cast<const T*>(m.b[ti]);
     float* sp = reinterpret_cast<float*>(m.c[ti]);
     float* bp = reinterpret_cast<float*>(m.d[ti]);
-    V4<T> pv = ld4(p, j), gv = ld4(g, j);
+    float pw[4];
+    ldp4(m, ti, j, p, pw);
+    V4<T> gv = ld4(g, j);
     V4<float> sv = ld4(sp, j), bv = ld4(bp, j);
 #pragma unroll
     for (int k = 0; k < 4; ++k)
-      pv.v[k] = from_f32<T>(elem(to_f32<T>(pv.v[k]), to_f32<T>(gv.v[k]),
-                                 sv.v[k], bv.v[k]));
-    st4(p, j, pv);
+      pw[k] = elem(pw[k], to_f32<T>(gv.v[k]), sv.v[k], bv.v[k]);
+    stp4(m, ti, j, p, pw);
     st4(sp, j, sv);
     st4(bp, j, bv);
   }
@@ -323,8 +379,11 @@ static void mt_launch(const std::vector<at::Tensor>& a,
                       const std::vector<at::Tensor>& b,
                       const std::vector<at::Tensor>& c,
                       const std::vector<at::Tensor>& d,
-                      const std::vector<float>& e, OP op) {
+                      const std::vector<float>& e, OP op,
+                      const std::vector<at::Tensor>& f = {}) {
   TORCH_CHECK(!a.empty());
+  TORCH_CHECK(f.empty() || f.size() == a.size(),
+              "mt_launch: masters list size mismatch");
   auto stream = at::cuda::getCurrentCUDAStream();
   size_t i = 0;
   while (i < a.size()) {
@@ -342,6 +401,7 @@ static void mt_launch(const std::vector<at::Tensor>& a,
       meta.b[k] = b.empty() ? nullptr : b[i].data_ptr();
       meta.c[k] = c.empty() ? nullptr : c[i].data_ptr();
       meta.d[k] = d.empty() ? nullptr : d[i].data_ptr();
+      meta.f[k] = f.empty() ? nullptr : f[i].data_ptr();
       meta.e[k] = e.empty() ? 0.f : e[i];
       meta.cum[k + 1] = meta.cum[k] + a[i].numel();
       meta.vcum[k + 1] = meta.vcum[k] + a[i].numel() / 4;
@@ -349,6 +409,7 @@ static void mt_launch(const std::vector<at::Tensor>& a,
       chk(meta.b[k]);
       chk(meta.c[k]);
       chk(meta.d[k]);
+      chk(meta.f[k]);
       meta.n++;
       i++;
     }
@@ -405,32 +466,57 @@ static void check_same_layout(const std::vector<at::Tensor>& a,
   }
 }
 
+// Optional fp32 master weights: pass one fp32 tensor per (bf16) param, or
+// an empty list for the plain path.
+static void check_masters(const std::vector<at::Tensor>& params,
+                          const std::vector<at::Tensor>& masters,
+                          const char* name) {
+  if (masters.empty()) return;
+  TORCH_CHECK(masters.size() == params.size(), name,
+              ": masters list size mismatch");
+  for (size_t i = 0; i < masters.size(); ++i) {
+    TORCH_CHECK(masters[i].scalar_type() == at::kFloat &&
+                    masters[i].numel() == params[i].numel(),
+                name, ": master ", i, " must be fp32 with the param's "
+                "numel");
+    TORCH_CHECK(masters[i].is_non_overlapping_and_dense(), name,
+                ": master ", i, " must be dense");
+  }
+}
+
 void multi_tensor_sgd(std::vector<at::Tensor> params,
-                      std::vector<at::Tensor> grads, double lr, double wd) {
+                      std::vector<at::Tensor> grads, double lr, double wd,
+                      std::vector<at::Tensor> masters) {
   check_contig(params, "mt_sgd");
   check_same_layout(params, grads, "mt_sgd");
-  mt_launch(params, grads, {}, {}, {}, SGDOp{(float)lr, (float)wd});
+  check_masters(params, masters, "mt_sgd");
+  mt_launch(params, grads, {}, {}, {}, SGDOp{(float)lr, (float)wd},
+            masters);
 }
 
 void multi_tensor_momentum(std::vector<at::Tensor> params,
                            std::vector<at::Tensor> grads,
                            std::vector<at::Tensor> bufs, double lr,
-                           double mom, double wd, bool nesterov) {
+                           double mom, double wd, bool nesterov,
+                           std::vector<at::Tensor> masters) {
   check_contig(params, "mt_momentum");
   check_same_layout(params, grads, "mt_momentum");
   check_same_layout(params, bufs, "mt_momentum");
+  check_masters(params, masters, "mt_momentum");
   mt_launch(params, grads, bufs, {}, {},
-            MomentumOp{(float)lr, (float)mom, (float)wd, nesterov});
+            MomentumOp{(float)lr, (float)mom, (float)wd, nesterov},
+            masters);
 }
 
 void multi_tensor_adagrad(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> grads,
                           std::vector<at::Tensor> accums, double lr,
-                          double eps) {
+                          double eps, std::vector<at::Tensor> masters) {
   check_contig(params, "mt_adagrad");
   check_same_layout(params, grads, "mt_adagrad");
+  check_masters(params, masters, "mt_adagrad");
   mt_launch(params, grads, accums, {}, {},
-            AdagradOp{(float)lr, (float)eps});
+            AdagradOp{(float)lr, (float)eps}, masters);
 }
 
 void multi_tensor_adam(std::vector<at::Tensor> params,
@@ -438,13 +524,15 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> ms,
                        std::vector<at::Tensor> vs, std::vector<double> steps,
                        double lr, double b1, double b2, double eps,
-                       double wd) {
+                       double wd, std::vector<at::Tensor> masters) {
   check_contig(params, "mt_adam");
   check_same_layout(params, grads, "mt_adam");
+  check_masters(params, masters, "mt_adam");
   std::vector<float> e(steps.begin(), steps.end());
   mt_launch(params, grads, ms, vs, e,
             AdamOp{(float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
-                   nullptr});
+                   nullptr},
+            masters);
 }
 
 // hipGraph-capturable Adam: the step count lives in a 0-dim fp32 DEVICE
@@ -456,27 +544,32 @@ void multi_tensor_adam_dev(std::vector<at::Tensor> params,
                            std::vector<at::Tensor> ms,
                            std::vector<at::Tensor> vs, at::Tensor step_dev,
                            double lr, double b1, double b2, double eps,
-                           double wd) {
+                           double wd, std::vector<at::Tensor> masters) {
   check_contig(params, "mt_adam_dev");
   check_same_layout(params, grads, "mt_adam_dev");
+  check_masters(params, masters, "mt_adam_dev");
   TORCH_CHECK(step_dev.is_cuda() &&
                   step_dev.scalar_type() == at::kFloat &&
                   step_dev.numel() == 1,
               "mt_adam_dev: step_dev must be a 0-dim fp32 CUDA tensor");
   mt_launch(params, grads, ms, vs, {},
             AdamOp{(float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
-                   step_dev.data_ptr<float>()});
+                   step_dev.data_ptr<float>()},
+            masters);
 }
 
 void multi_tensor_rmsprop(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> grads,
                           std::vector<at::Tensor> sqs,
                           std::vector<at::Tensor> bufs, double lr,
-                          double decay, double mom, double eps) {
+                          double decay, double mom, double eps,
+                          std::vector<at::Tensor> masters) {
   check_contig(params, "mt_rmsprop");
   check_same_layout(params, grads, "mt_rmsprop");
+  check_masters(params, masters, "mt_rmsprop");
   mt_launch(params, grads, sqs, bufs, {},
-            RMSPropOp{(float)lr, (float)decay, (float)mom, (float)eps});
+            RMSPropOp{(float)lr, (float)decay, (float)mom, (float)eps},
+            masters);
 }
 
 void multi_tensor_ema(std::vector<at::Tensor> shadows,

@@ -86,8 +86,11 @@ def sampled_softmax_loss(weight, bias, labels, inputs, num_sampled,
             return w(ids)
         return embedding_lookup(w, ids)
 
-    true_w = rows(weight, labels)                     # [N, D]
-    sampled_w = rows(weight, sampled_ids)             # [S, D]
+    # table rows are fp32; cast to the input dtype (no-op when equal) so
+    # pure-bf16 bodies matmul without autocast — the .to() backward casts
+    # the row grads back to fp32 before the indexed-slices grad forms
+    true_w = rows(weight, labels).to(inputs.dtype)    # [N, D]
+    sampled_w = rows(weight, sampled_ids).to(inputs.dtype)  # [S, D]
     true_logits = (inputs * true_w).sum(-1)           # [N]
     sampled_logits = inputs @ sampled_w.t()           # [N, S]
 

@@ -65,6 +65,51 @@ class ParallaxOptimizer(Optimizer):
     def sparse_row_step(self, param, ids, grad_rows, group=None):
         raise NotImplementedError
 
+    # -- fp32 master weights (pure-bf16 training) ----------------------------
+    def _ensure_masters(self, params):
+        """fp32 master copies for non-fp32 params: the fused kernels read/
+        update the master and write the rounded copy into the bf16 param,
+        so a pure-bf16 model trains with fp32 update precision and no
+        per-step cast kernels.  Returns (masters, has_any): an empty list
+        when every param is fp32 (zero-overhead common case), else one
+        entry per param (the param itself stands in for fp32 entries)."""
+        if not any(p.dtype != torch.float32 for p in params):
+            return [], False
+        out = []
+        for p in params:
+            if p.dtype == torch.float32:
+                out.append(p)
+                continue
+            st = self.state[p]
+            if 'master' not in st:
+                m = torch.empty_like(p.detach(), dtype=torch.float32)
+                m.copy_(p.detach())
+                st['master'] = m
+            out.append(st['master'])
+        return out, True
+
+    def load_state_dict(self, state_dict):
+        """torch's default load casts floating state to the param's dtype —
+        that would round fp32 masters/accumulators to bf16.  Restore state
+        tensors verbatim (device move only)."""
+        super().load_state_dict(state_dict)
+        # undo any dtype casts torch applied: our state is authored at the
+        # correct dtype, so restore from the raw payload
+        id_map = {}
+        for sg, g in zip(state_dict['param_groups'], self.param_groups):
+            for old_id, p in zip(sg['params'], g['params']):
+                id_map[old_id] = p
+        for old_id, saved in state_dict['state'].items():
+            p = id_map.get(old_id)
+            if p is None:
+                continue
+            st = self.state[p]
+            for k, v in saved.items():
+                if torch.is_tensor(v) and k in st \
+                        and torch.is_tensor(st[k]) \
+                        and st[k].dtype != v.dtype:
+                    st[k] = v.to(device=p.device)
+
     def group_of(self, param):
         for group in self.param_groups:
             for p in group['params']:
@@ -86,34 +131,43 @@ class SGD(ParallaxOptimizer):
     def _dense_step(self, params, grads, group):
         lr, mom = group['lr'], group['momentum']
         wd, nesterov = group['weight_decay'], group['nesterov']
+        masters, has_m = self._ensure_masters(params)
         if mom == 0.0:
             if _use_native(params):
                 _ops.ext().multi_tensor_sgd(params, [g for g in grads],
-                                            lr, wd)
+                                            lr, wd, masters)
                 return
-            for p, g in zip(params, grads):
+            for i, (p, g) in enumerate(zip(params, grads)):
+                w = masters[i] if has_m else p
+                g = g.to(w.dtype)
                 if wd:
-                    g = g.add(p, alpha=wd)
-                p.add_(g, alpha=-lr)
+                    g = g.add(w, alpha=wd)
+                w.add_(g, alpha=-lr)
+                if w is not p:
+                    p.copy_(w)
             return
         bufs = []
         for p in params:
             st = self.state[p]
             if 'momentum_buffer' not in st:
-                st['momentum_buffer'] = torch.zeros_like(p)
+                st['momentum_buffer'] = torch.zeros_like(
+                    p, dtype=torch.float32)
             bufs.append(st['momentum_buffer'])
         if _use_native(params):
             _ops.ext().multi_tensor_momentum(params, list(grads), bufs,
-                                             lr, mom, wd, nesterov)
+                                             lr, mom, wd, nesterov,
+                                             masters)
             return
-        for p, g, buf in zip(params, grads, bufs):
+        for i, (p, g, buf) in enumerate(zip(params, grads, bufs)):
+            w = masters[i] if has_m else p
+            g = g.to(buf.dtype)
             if wd:
-                g = g.add(p, alpha=wd)
+                g = g.add(w.to(buf.dtype), alpha=wd)
             buf.mul_(mom).add_(g)
-            if nesterov:
-                p.add_(g.add(buf, alpha=mom), alpha=-lr)
-            else:
-                p.add_(buf, alpha=-lr)
+            upd = g.add(buf, alpha=mom) if nesterov else buf
+            w.add_(upd.to(w.dtype), alpha=-lr)
+            if w is not p:
+                p.copy_(w)
 
     @torch.no_grad()
     def sparse_row_step(self, param, ids, grad_rows, group=None):
@@ -149,14 +203,18 @@ class Adagrad(ParallaxOptimizer):
     def _dense_step(self, params, grads, group):
         lr, eps = group['lr'], group['eps']
         accums = [self._accum(p, group) for p in params]
+        masters, has_m = self._ensure_masters(params)
         if _use_native(params):
             _ops.ext().multi_tensor_adagrad(params, list(grads), accums,
-                                            lr, eps)
+                                            lr, eps, masters)
             return
-        for p, g, acc in zip(params, grads, accums):
+        for i, (p, g, acc) in enumerate(zip(params, grads, accums)):
+            w = masters[i] if has_m else p
             g32 = g.float()
             acc.add_(g32 * g32)
-            p.add_((g32 / (acc.sqrt() + eps)).to(p.dtype), alpha=-lr)
+            w.add_((g32 / (acc.sqrt() + eps)).to(w.dtype), alpha=-lr)
+            if w is not p:
+                p.copy_(w)
 
     @torch.no_grad()
     def sparse_row_step(self, param, ids, grad_rows, group=None):
@@ -227,6 +285,7 @@ class Adam(ParallaxOptimizer):
             ms.append(st['exp_avg'])
             vs.append(st['exp_avg_sq'])
             steps.append(st['step'])
+        masters, has_m = self._ensure_masters(params)
         if _use_native(params):
             gi = next(i for i, g in enumerate(self.param_groups)
                       if g is group)
@@ -238,19 +297,24 @@ class Adam(ParallaxOptimizer):
             self._dev_step_params.update(params)
             buf.add_(1.0)
             _ops.ext().multi_tensor_adam_dev(params, list(grads), ms, vs,
-                                             buf, lr, b1, b2, eps, wd)
+                                             buf, lr, b1, b2, eps, wd,
+                                             masters)
             return
-        for p, g, m, v, t in zip(params, grads, ms, vs, steps):
+        for i, (p, g, m, v, t) in enumerate(zip(params, grads, ms, vs,
+                                                steps)):
+            w = masters[i] if has_m else p
             g32 = g.float()
             if wd:
-                g32 = g32.add(p.float(), alpha=wd)
+                g32 = g32.add(w.float(), alpha=wd)
             m.mul_(b1).add_(g32, alpha=1 - b1)
             v.mul_(b2).addcmul_(g32, g32, value=1 - b2)
             bc1 = 1 - b1 ** t
             bc2 = 1 - b2 ** t
             step_size = lr / bc1
             denom = (v / bc2).sqrt().add_(eps)
-            p.add_((m / denom).to(p.dtype), alpha=-step_size)
+            w.add_((m / denom).to(w.dtype), alpha=-step_size)
+            if w is not p:
+                p.copy_(w)
 
     @torch.no_grad()
     def sparse_row_step(self, param, ids, grad_rows, group=None):
@@ -294,16 +358,20 @@ class RMSProp(ParallaxOptimizer):
                     p, dtype=torch.float32)
             rms.append(st['square_avg'])
             bufs.append(st['momentum_buffer'])
+        masters, has_m = self._ensure_masters(params)
         if _use_native(params):
             _ops.ext().multi_tensor_rmsprop(params, list(grads), rms, bufs,
-                                            lr, decay, mom, eps)
+                                            lr, decay, mom, eps, masters)
             return
-        for p, g, sq, buf in zip(params, grads, rms, bufs):
+        for i, (p, g, sq, buf) in enumerate(zip(params, grads, rms, bufs)):
+            w = masters[i] if has_m else p
             g32 = g.float()
             sq.mul_(decay).addcmul_(g32, g32, value=1 - decay)
             upd = g32 / (sq.sqrt() + eps)
             buf.mul_(mom).add_(upd)
-            p.add_(buf.to(p.dtype), alpha=-lr)
+            w.add_(buf.to(w.dtype), alpha=-lr)
+            if w is not p:
+                p.copy_(w)
 
     @torch.no_grad()
     def sparse_row_step(self, param, ids, grad_rows, group=None):

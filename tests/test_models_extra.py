@@ -51,6 +51,41 @@ def test_nmt_transformer_tagging_and_step(monkeypatch, free_port):
     sess.close()
 
 
+def test_nmt_pure_bf16_masters(monkeypatch, free_port):
+    """pure_bf16: bf16 body + fp32 master weights in the optimizer, fp32
+    sparse tables.  The eager (CPU) master path must learn and keep the
+    bf16 param an exact rounded copy of its fp32 master."""
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: nmt.build_nmt_graph(
+            arch='transformer', src_vocab=300, tgt_vocab=300, d_model=32,
+            heads=4, ffn=64, enc_layers=1, dec_layers=1, min_partitions=2,
+            use_bf16=False, pure_bf16=True, lr=3e-3),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    m = sess.graph.model
+    assert m.pos.dtype == torch.bfloat16
+    assert m.src_emb.weight.dtype == torch.float32  # sparse plane stays
+    batch = nmt.synthetic_batch(4, 5, 6, 300, 300)
+    first = None
+    for _ in range(10):
+        out = sess.run(batch)
+        if first is None:
+            first = float(out['loss'])
+    assert torch.isfinite(out['loss'])
+    assert float(out['loss']) < first
+    opt = sess.graph.optimizer
+    mw = opt.state[m.pos].get('master')
+    assert mw is not None and mw.dtype == torch.float32
+    assert torch.equal(m.pos.detach(), mw.to(torch.bfloat16))
+    # masters round-trip through the checkpoint payload uncast
+    sd = opt.state_dict()
+    opt.load_state_dict(sd)
+    mw2 = opt.state[m.pos]['master']
+    assert mw2.dtype == torch.float32 and torch.equal(mw2, mw)
+    sess.close()
+
+
 def test_gnmt_attention_step(monkeypatch, free_port):
     _worker_env(monkeypatch, free_port)
     torch.manual_seed(0)
