@@ -80,7 +80,8 @@ def main():
         seq = M.NUM_STEPS                                # 20
         vocab = M.VOCAB_SIZE                             # 793470
         build = lambda: M.build_lm1b_graph(
-            batch_size=batch, num_steps=seq, min_partitions=32)
+            batch_size=batch, num_steps=seq, min_partitions=32,
+            pure_bf16=args.pure_bf16 and on_gpu)
         metric, unit = 'LM1B words/sec', 'words/s'
         per_step_items = batch * seq                     # words per rank
         baseline = 277000.0

@@ -55,6 +55,9 @@ class LM1BModel(nn.Module):
     def forward(self, x, y, state=None):
         """x, y: [B, T] int64.  Returns (mean loss per word, final state)."""
         emb = self.emb(x)                      # [B, T, D]
+        body = self.lstm.w.dtype               # bf16 in pure-bf16 mode
+        if emb.dtype != body and not torch.is_autocast_enabled():
+            emb = emb.to(body)
         hs, state = self.lstm(emb, state)      # [B, T, P]
         flat_h = hs.reshape(-1, hs.shape[-1])
         flat_y = y.reshape(-1)
@@ -72,21 +75,27 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
                      emb_dim=EMB_DIM, state_size=STATE_SIZE,
                      proj_size=PROJ_SIZE, num_sampled=NUM_SAMPLED,
                      max_grad_norm=10.0, dtype=torch.float32,
-                     use_bf16=True, ema_decay=0.999):
+                     use_bf16=True, ema_decay=0.999, pure_bf16=False):
     """use_bf16: run the GEMM path under bf16 autocast on GPU (fp32 master
     weights, fp32 CE and optimizer math).
+    pure_bf16: the dense LSTM body lives in bf16 (no per-step autocast
+    weight casts); fp32 update precision comes from the optimizers'
+    fused master-weight path; the sparse tables and EMA shadows stay
+    fp32.
     ema_decay: exponential moving average of the LSTM params (reference
     language_model_graph.py:63-65, decay .999); shadows live as model
     buffers (-> broadcast + checkpointed); None disables."""
     model = LM1BModel(vocab_size, emb_dim, state_size, proj_size,
                       num_sampled, min_partitions, dtype=dtype)
+    if pure_bf16:
+        model.lstm.to(torch.bfloat16)
     optimizer = optim.Adagrad(model.parameters(), lr=lr,
                               initial_accumulator_value=1.0)
     dense_params = model.dense_parameters()
 
     def step_fn(m, batch):
         x, y, state = batch
-        if use_bf16 and x.is_cuda:
+        if use_bf16 and not pure_bf16 and x.is_cuda:
             with torch.autocast('cuda', dtype=torch.bfloat16):
                 loss, new_state = m(x, y, state)
         else:
@@ -103,7 +112,10 @@ def build_lm1b_graph(batch_size=BATCH_SIZE, num_steps=NUM_STEPS,
     n_ema = 0
     if ema_decay is not None:
         for i, p in enumerate(dense_params):
-            model.register_buffer('ema_%d' % i, p.detach().clone())
+            # fp32 shadows even over a bf16 body: .999-decay drift would
+            # be swallowed by bf16 rounding
+            model.register_buffer('ema_%d' % i,
+                                  p.detach().clone().float())
         n_ema = len(dense_params)
 
     orig_step = optimizer.step

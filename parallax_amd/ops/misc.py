@@ -43,10 +43,14 @@ def clip_by_global_norm(tensors, clip_norm):
 
 
 def ema_update(shadow_tensors, tensors, decay):
-    """shadow = decay * shadow + (1-decay) * value (TF EMA apply)."""
+    """shadow = decay * shadow + (1-decay) * value (TF EMA apply).
+    Shadows may be fp32 over bf16 values (pure-bf16 bodies keep fp32
+    shadows so the .999-decay drift is not swallowed by bf16 rounding)."""
+    shadow_tensors = list(shadow_tensors)
+    tensors = [t.to(s.dtype) if t.dtype != s.dtype else t
+               for s, t in zip(shadow_tensors, tensors)]
     if _ops.use_native(*tensors):
-        _ops.ext().multi_tensor_ema(list(shadow_tensors), list(tensors),
-                                    float(decay))
+        _ops.ext().multi_tensor_ema(shadow_tensors, tensors, float(decay))
         return
     with torch.no_grad():
         for s, t in zip(shadow_tensors, tensors):

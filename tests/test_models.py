@@ -84,3 +84,36 @@ def test_lm1b_synthetic_batch_shapes():
                                         vocab_size=100)
     assert x.shape == (4, 5) and y.shape == (4, 5) and st is None
     assert int(x.max()) < 100
+
+
+def test_lm1b_pure_bf16_learns(monkeypatch, free_port):
+    """pure_bf16: bf16 LSTM body, fp32 tables/shadows, master-weight
+    Adagrad — the eager CPU path must learn on a repeated batch."""
+    monkeypatch.setenv('WORLD_SIZE', '1')
+    monkeypatch.setenv('RANK', '0')
+    monkeypatch.setenv('MASTER_ADDR', '127.0.0.1')
+    monkeypatch.setenv('MASTER_PORT', str(free_port))
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: build_lm1b_graph(batch_size=4, num_steps=3, vocab_size=200,
+                                 emb_dim=16, state_size=32, proj_size=16,
+                                 num_sampled=32, min_partitions=4, lr=0.4,
+                                 pure_bf16=True),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    m = sess.graph.model
+    assert m.lstm.w.dtype == torch.bfloat16
+    assert m.emb.weight.dtype == torch.float32
+    assert m.ema_0.dtype == torch.float32
+    x = torch.randint(0, 200, (4, 3))
+    y = torch.randint(0, 200, (4, 3))
+    first = None
+    for _ in range(30):
+        out = sess.run((x, y, None))
+        if first is None:
+            first = float(out['loss'])
+    assert float(out['loss']) < first
+    assert sess.graph.optimizer.state[m.lstm.w]['master'].dtype \
+        == torch.float32
+    sess.close()
+    import torch.distributed as dist
+    dist.destroy_process_group()
