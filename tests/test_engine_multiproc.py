@@ -295,10 +295,12 @@ def _body_average_sparse(rank, world):
     eng.close()
 
 
-def _body_lm1b_tiny_full_stack(rank, world):
+def _body_lm1b_tiny_full_stack(rank, world, pure_bf16=False):
     """Tiny LM1B through the COMPLETE per-step protocol on several ranks:
     partitioned tables, sampled softmax, clip, EMA, dense AR, sparse PS —
-    the collective-order rehearsal for the 8-GPU node."""
+    the collective-order rehearsal for the 8-GPU node.  pure_bf16
+    additionally rehearses bf16 dense grads over the wire (gloo here,
+    RCCL on the node) with fp32 masters."""
     import parallax_amd
     from parallax_amd.config import ParallaxConfig
     from parallax_amd.models import lm1b as M
@@ -308,7 +310,7 @@ def _body_lm1b_tiny_full_stack(rank, world):
                                    vocab_size=300, emb_dim=16,
                                    state_size=32, proj_size=16,
                                    num_sampled=32, min_partitions=4,
-                                   lr=0.3),
+                                   lr=0.3, pure_bf16=pure_bf16),
         None, parallax_config=ParallaxConfig(run_option='HYBRID'))
     state = None
     for i in range(8):
@@ -321,8 +323,12 @@ def _body_lm1b_tiny_full_stack(rank, world):
               m.lstm.w.detach(), m.ema_0.detach()):
         ts = [torch.zeros_like(t) for _ in range(world)]
         torch.distributed.all_gather(ts, t)
-        assert torch.allclose(ts[0], ts[-1], atol=1e-5)
+        assert torch.allclose(ts[0].float(), ts[-1].float(), atol=1e-5)
     sess.close()
+
+
+def _body_lm1b_pure_bf16(rank, world):
+    _body_lm1b_tiny_full_stack(rank, world, pure_bf16=True)
 
 
 # ---------------------------------------------------------------------------
@@ -342,6 +348,10 @@ def test_multiproc(body):
 def test_lm1b_full_stack_world4():
     _spawn(_body_lm1b_tiny_full_stack, get_free_port(), world=4,
            timeout=300)
+
+
+def test_lm1b_pure_bf16_world2():
+    _spawn(_body_lm1b_pure_bf16, get_free_port(), world=2, timeout=300)
 
 
 def test_average_sparse_world2():
