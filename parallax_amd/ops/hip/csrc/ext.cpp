@@ -44,6 +44,7 @@ void multi_tensor_rmsprop(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> masters);
 void multi_tensor_ema(std::vector<at::Tensor> shadows,
                       std::vector<at::Tensor> values, double decay);
+void multi_tensor_zero(std::vector<at::Tensor> tensors);
 at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors);
 void multi_tensor_scale_by_norm(std::vector<at::Tensor> tensors,
                                 at::Tensor norm, double clip);
@@ -141,6 +142,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_adam_dev", &multi_tensor_adam_dev);
   m.def("multi_tensor_rmsprop", &multi_tensor_rmsprop);
   m.def("multi_tensor_ema", &multi_tensor_ema);
+  m.def("multi_tensor_zero", &multi_tensor_zero);
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
   m.def("multi_tensor_scale_by_norm", &multi_tensor_scale_by_norm);
   m.def("sparse_adagrad", &sparse_adagrad);

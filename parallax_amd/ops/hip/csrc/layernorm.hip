@@ -211,8 +211,10 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
   auto stream = at::cuda::getCurrentCUDAStream();
   if (d <= LN_LDS_MAX_D) {
     const int grid = (int)(n < 512 ? (n > 0 ? n : 1) : 512);
-    auto dw32 = at::zeros({LN_NBUF * d}, x.options().dtype(at::kFloat));
-    auto db32 = at::zeros({LN_NBUF * d}, x.options().dtype(at::kFloat));
+    // one allocation + one fill for both stripe buffers
+    auto ws = at::zeros({2, LN_NBUF * d}, x.options().dtype(at::kFloat));
+    auto dw32 = ws[0];
+    auto db32 = ws[1];
     const size_t lds = 2 * (size_t)d * sizeof(float);
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::kBFloat16, at::kHalf, x.scalar_type(), "layernorm_bwd", [&] {

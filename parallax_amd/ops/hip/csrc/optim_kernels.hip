@@ -359,6 +359,21 @@ struct EMAOp {
   }
 };
 
+struct ZeroOp {
+  static constexpr bool vec = true;
+  template <typename T>
+  __device__ void apply(const MTMeta& m, int ti, long j) const {
+    reinterpret_cast<T*>(m.a[ti])[j] = from_f32<T>(0.f);
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    V4<T> z;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) z.v[k] = from_f32<T>(0.f);
+    st4(reinterpret_cast<T*>(m.a[ti]), j, z);
+  }
+};
+
 struct ScaleOp {
   static constexpr bool vec = false;  // single pass over one list; scalar
   const float* norm;                  // device scalar
@@ -576,6 +591,13 @@ void multi_tensor_ema(std::vector<at::Tensor> shadows,
                       std::vector<at::Tensor> values, double decay) {
   check_contig(shadows, "mt_ema");
   mt_launch(shadows, values, {}, {}, {}, EMAOp{(float)decay});
+}
+
+// Fused zero of a tensor list (one launch per 32 tensors) — replaces the
+// per-tensor zero_() storm at the top of a captured step.
+void multi_tensor_zero(std::vector<at::Tensor> tensors) {
+  check_contig(tensors, "mt_zero");
+  mt_launch(tensors, {}, {}, {}, {}, ZeroOp{});
 }
 
 // --- global norm -----------------------------------------------------------

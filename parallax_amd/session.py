@@ -130,8 +130,7 @@ class ParallaxSession(object):
         self.engine.prepare_step()  # clear sparse capture slots pre-capture
         self._graph_obj = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self._graph_obj):
-            for p in dense_params:
-                p.grad.zero_()
+            self._zero_dense_grads(dense_params)
             loss, fetches = g.step_fn(g.model, self._static_batch)
             loss.backward()
             state = fetches.get('state') if fetches else None
@@ -148,6 +147,22 @@ class ParallaxSession(object):
                           '%d sparse slots)', len(dense_params),
                           len(self.engine.sparse_engine.slots)
                           if self.engine.sparse_engine else 0)
+
+    @staticmethod
+    def _zero_dense_grads(dense_params):
+        """Zero the static grad buffers at the top of the captured step:
+        ONE fused launch per <=32 same-dtype tensors instead of a
+        per-tensor zero_() storm."""
+        from parallax_amd import ops as _ops
+        if not _ops.native_available():
+            for p in dense_params:
+                p.grad.zero_()
+            return
+        by_dtype = {}
+        for p in dense_params:
+            by_dtype.setdefault(p.grad.dtype, []).append(p.grad.detach())
+        for grads in by_dtype.values():
+            _ops.ext().multi_tensor_zero(grads)
 
     def _maybe_capture_opt_graph(self):
         """Capture the (shape-static) optimizer step — fused dense update,
