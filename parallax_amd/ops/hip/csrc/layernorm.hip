@@ -17,7 +17,12 @@
 
 #include "common.h"
 
-template <typename T>
+static inline bool ln_aligned16(const at::Tensor& t) {
+  return (reinterpret_cast<uintptr_t>(t.data_ptr()) & 15) == 0;
+}
+
+
+template <typename T, bool VEC>
 __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
                                      const T* __restrict__ w,
                                      const T* __restrict__ b,
@@ -29,10 +34,22 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
   for (long row = blockIdx.x; row < n; row += gridDim.x) {
     const T* xr = x + row * d;
     float s = 0.f, ss = 0.f;
-    for (long c = threadIdx.x; c < d; c += blockDim.x) {
-      float v = to_f32<T>(xr[c]);
-      s += v;
-      ss += v * v;
+    if (VEC) {  // d % 4 == 0, 16B-aligned rows (checked on host)
+      for (long c = threadIdx.x * 4; c < d; c += (long)blockDim.x * 4) {
+        PxV4<T> v = px_ld4(xr, c);
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          float f = to_f32<T>(v.v[k]);
+          s += f;
+          ss += f * f;
+        }
+      }
+    } else {
+      for (long c = threadIdx.x; c < d; c += blockDim.x) {
+        float v = to_f32<T>(xr[c]);
+        s += v;
+        ss += v * v;
+      }
     }
     s = block_reduce_sum(s, scratch);
     ss = block_reduce_sum(ss, scratch);
@@ -45,9 +62,23 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
       rstd[row] = r;
     }
     T* yr = y + row * d;
-    for (long c = threadIdx.x; c < d; c += blockDim.x) {
-      float v = (to_f32<T>(xr[c]) - m) * r;
-      yr[c] = from_f32<T>(v * to_f32<T>(w[c]) + to_f32<T>(b[c]));
+    if (VEC) {
+      for (long c = threadIdx.x * 4; c < d; c += (long)blockDim.x * 4) {
+        PxV4<T> xv = px_ld4(xr, c), wv = px_ld4(w, c), bv = px_ld4(b, c);
+        PxV4<T> yv;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          float v = (to_f32<T>(xv.v[k]) - m) * r;
+          yv.v[k] = from_f32<T>(v * to_f32<T>(wv.v[k]) +
+                                to_f32<T>(bv.v[k]));
+        }
+        px_st4(yr, c, yv);
+      }
+    } else {
+      for (long c = threadIdx.x; c < d; c += blockDim.x) {
+        float v = (to_f32<T>(xr[c]) - m) * r;
+        yr[c] = from_f32<T>(v * to_f32<T>(w[c]) + to_f32<T>(b[c]));
+      }
     }
   }
 }
@@ -62,15 +93,24 @@ std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
   auto rstd = at::empty({n}, x.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   int grid = (int)(n < PX_MAX_GRID ? n : PX_MAX_GRID);
+  auto wc = w.contiguous();
+  auto bc = b.contiguous();
+  const bool vec = (d % 4 == 0) && ln_aligned16(x) && ln_aligned16(wc) &&
+                   ln_aligned16(bc);
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x.scalar_type(), "layernorm_fwd", [&] {
-        layernorm_fwd_kernel<scalar_t><<<grid, PX_BLOCK, 0, stream>>>(
-            reinterpret_cast<const scalar_t*>(x.data_ptr()),
-            reinterpret_cast<const scalar_t*>(w.contiguous().data_ptr()),
-            reinterpret_cast<const scalar_t*>(b.contiguous().data_ptr()),
-            reinterpret_cast<scalar_t*>(y.data_ptr()),
-            mean.data_ptr<float>(), rstd.data_ptr<float>(), n, d,
-            (float)eps);
+        auto launch = [&](auto vtag) {
+          layernorm_fwd_kernel<scalar_t, decltype(vtag)::value>
+              <<<grid, PX_BLOCK, 0, stream>>>(
+                  reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                  reinterpret_cast<const scalar_t*>(wc.data_ptr()),
+                  reinterpret_cast<const scalar_t*>(bc.data_ptr()),
+                  reinterpret_cast<scalar_t*>(y.data_ptr()),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(), n, d,
+                  (float)eps);
+        };
+        if (vec) launch(std::true_type{});
+        else launch(std::false_type{});
         HIP_CHECK_KERNEL();
       });
   return {y, mean, rstd};
@@ -121,7 +161,7 @@ __global__ void layernorm_bwd_kernel(
 #define LN_LDS_MAX_D 4096
 #define LN_NBUF 16
 
-template <typename T>
+template <typename T, bool VEC>
 __global__ void layernorm_bwd_lds_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const T* __restrict__ w, const float* __restrict__ mean,
@@ -142,22 +182,54 @@ __global__ void layernorm_bwd_lds_kernel(
     const float m = mean[row];
     const float r = rstd[row];
     float s1 = 0.f, s2 = 0.f;
-    for (long c = threadIdx.x; c < d; c += blockDim.x) {
-      float dyv = to_f32<T>(dyr[c]);
-      float xh = (to_f32<T>(xr[c]) - m) * r;
-      float dyw = dyv * to_f32<T>(w[c]);
-      s1 += dyw;
-      s2 += dyw * xh;
-      sdw[c] += dyv * xh;  // LDS: same thread owns column c all rows
-      sdb[c] += dyv;
+    if (VEC) {
+      for (long c = threadIdx.x * 4; c < d; c += (long)blockDim.x * 4) {
+        PxV4<T> dyv4 = px_ld4(dyr, c), xv4 = px_ld4(xr, c),
+                wv4 = px_ld4(w, c);
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          float dyv = to_f32<T>(dyv4.v[k]);
+          float xh = (to_f32<T>(xv4.v[k]) - m) * r;
+          float dyw = dyv * to_f32<T>(wv4.v[k]);
+          s1 += dyw;
+          s2 += dyw * xh;
+          sdw[c + k] += dyv * xh;
+          sdb[c + k] += dyv;
+        }
+      }
+    } else {
+      for (long c = threadIdx.x; c < d; c += blockDim.x) {
+        float dyv = to_f32<T>(dyr[c]);
+        float xh = (to_f32<T>(xr[c]) - m) * r;
+        float dyw = dyv * to_f32<T>(w[c]);
+        s1 += dyw;
+        s2 += dyw * xh;
+        sdw[c] += dyv * xh;  // LDS: same thread owns column c all rows
+        sdb[c] += dyv;
+      }
     }
     s1 = block_reduce_sum(s1, scratch) / d;
     s2 = block_reduce_sum(s2, scratch) / d;
     T* dxr = dx + row * d;
-    for (long c = threadIdx.x; c < d; c += blockDim.x) {
-      float xh = (to_f32<T>(xr[c]) - m) * r;
-      float dyw = to_f32<T>(dyr[c]) * to_f32<T>(w[c]);
-      dxr[c] = from_f32<T>(r * (dyw - s1 - xh * s2));
+    if (VEC) {
+      for (long c = threadIdx.x * 4; c < d; c += (long)blockDim.x * 4) {
+        PxV4<T> dyv4 = px_ld4(dyr, c), xv4 = px_ld4(xr, c),
+                wv4 = px_ld4(w, c);
+        PxV4<T> dxv;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          float xh = (to_f32<T>(xv4.v[k]) - m) * r;
+          float dyw = to_f32<T>(dyv4.v[k]) * to_f32<T>(wv4.v[k]);
+          dxv.v[k] = from_f32<T>(r * (dyw - s1 - xh * s2));
+        }
+        px_st4(dxr, c, dxv);
+      }
+    } else {
+      for (long c = threadIdx.x; c < d; c += blockDim.x) {
+        float xh = (to_f32<T>(xr[c]) - m) * r;
+        float dyw = to_f32<T>(dyr[c]) * to_f32<T>(w[c]);
+        dxr[c] = from_f32<T>(r * (dyw - s1 - xh * s2));
+      }
     }
   }
   __syncthreads();
@@ -216,17 +288,23 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
     auto dw32 = ws[0];
     auto db32 = ws[1];
     const size_t lds = 2 * (size_t)d * sizeof(float);
+    auto wc = w.contiguous();
+    const bool vec = (d % 4 == 0) && ln_aligned16(dy) &&
+                     ln_aligned16(x) && ln_aligned16(wc);
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::kBFloat16, at::kHalf, x.scalar_type(), "layernorm_bwd", [&] {
-          layernorm_bwd_lds_kernel<scalar_t>
-              <<<grid, PX_BLOCK, lds, stream>>>(
-                  reinterpret_cast<const scalar_t*>(dy.data_ptr()),
-                  reinterpret_cast<const scalar_t*>(x.data_ptr()),
-                  reinterpret_cast<const scalar_t*>(
-                      w.contiguous().data_ptr()),
-                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                  reinterpret_cast<scalar_t*>(dx.data_ptr()),
-                  dw32.data_ptr<float>(), db32.data_ptr<float>(), n, d);
+          auto launch = [&](auto vtag) {
+            layernorm_bwd_lds_kernel<scalar_t, decltype(vtag)::value>
+                <<<grid, PX_BLOCK, lds, stream>>>(
+                    reinterpret_cast<const scalar_t*>(dy.data_ptr()),
+                    reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                    reinterpret_cast<const scalar_t*>(wc.data_ptr()),
+                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                    reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                    dw32.data_ptr<float>(), db32.data_ptr<float>(), n, d);
+          };
+          if (vec) launch(std::true_type{});
+          else launch(std::false_type{});
           HIP_CHECK_KERNEL();
           fold_cast2_kernel<scalar_t><<<px_grid(d), PX_BLOCK, 0, stream>>>(
               dw32.data_ptr<float>(), db32.data_ptr<float>(),
