@@ -25,6 +25,7 @@ import torch.nn.functional as F
 from parallax_amd import optim
 from parallax_amd.ops.embedding import PartitionedEmbedding
 from parallax_amd.ops.layernorm import ParallaxLayerNorm
+from parallax_amd.ops.linear import PxLinear
 from parallax_amd.ops.lstm import ParallaxLSTM
 from parallax_amd.ops.sampled_softmax import sampled_softmax_loss
 from parallax_amd.ops.softmax_ce import softmax_cross_entropy
@@ -46,11 +47,11 @@ class _MHA(nn.Module):
         self.d = d
         self.h = heads
         if cross:
-            self.kv = nn.Linear(d, 2 * d, bias=False)
-            self.q = nn.Linear(d, d, bias=False)
+            self.kv = PxLinear(d, 2 * d, bias=False)
+            self.q = PxLinear(d, d, bias=False)
         else:
-            self.qkv = nn.Linear(d, 3 * d, bias=False)
-        self.out = nn.Linear(d, d, bias=False)
+            self.qkv = PxLinear(d, 3 * d, bias=False)
+        self.out = PxLinear(d, d, bias=False)
 
     def forward(self, x, mem=None, causal=False):
         B, T, D = x.shape
@@ -79,8 +80,8 @@ class _Block(nn.Module):
             self.ln_x = ParallaxLayerNorm(d)
             self.xattn = _MHA(d, heads, cross=True)
         self.ln2 = ParallaxLayerNorm(d)
-        self.fc1 = nn.Linear(d, ffn)
-        self.fc2 = nn.Linear(ffn, d)
+        self.fc1 = PxLinear(d, ffn)   # K=512: pxgemm wins
+        self.fc2 = nn.Linear(ffn, d)  # K=2048: hipBLASLt wins
 
     def forward(self, x, mem=None, causal=False):
         x = x + self.attn(self.ln1(x), causal=causal)

@@ -1,0 +1,69 @@
+"""Linear layer over the hand-written MFMA GEMM (pxgemm_bt).
+
+``y = x @ W^T (+ b)`` maps directly onto the kernel's A @ B^T form, and
+the bias rides the kernel's addend epilogue as a stride-0 broadcast —
+zero extra kernels.  Measured on MI355X (scripts/bench_gemm_nmt.py), the
+transformer's K=512 projections run 30-45% faster than hipBLASLt:
+
+    qkv  [3072,1536,512]   13.5us vs 20.6us
+    out  [3072, 512,512]   10.8us vs 19.8us
+    fc1  [3072,2048,512]   14.3us vs 20.3us
+
+The backward GEMMs (dx: K=N_out, dw: K=M — long-K shapes where the
+kernel has no split-K yet) stay on hipBLASLt, which wins there.
+
+Eligibility: CUDA + native ext + bf16 weights + K % 64 == 0 and
+K <= PX_LINEAR_MAX_K; everything else falls back to F.linear.
+PARALLAX_PX_LINEAR=0 disables the kernel path (A/B switch).
+"""
+
+import os
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from parallax_amd import ops as _ops
+
+PX_LINEAR_MAX_K = 1024
+
+
+class _PxLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2d, weight, bias):
+        ctx.save_for_backward(x2d, weight)
+        ctx.has_bias = bias is not None
+        if bias is not None:
+            addend = bias.unsqueeze(0).expand(x2d.shape[0], bias.shape[0])
+            return _ops.ext().pxgemm_bt_add(x2d, weight, addend)
+        return _ops.ext().pxgemm_bt(x2d, weight)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dy @ weight
+        dw = dy.t() @ x2d
+        db = dy.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def _px_eligible(x, weight):
+    return (x.is_cuda and weight.dtype == torch.bfloat16
+            and x.dtype == torch.bfloat16
+            and weight.shape[1] % 64 == 0
+            and weight.shape[1] <= PX_LINEAR_MAX_K
+            and _ops.native_available()
+            and os.environ.get('PARALLAX_PX_LINEAR', '1') != '0')
+
+
+class PxLinear(nn.Linear):
+    """nn.Linear drop-in (same state-dict keys) routing the forward GEMM
+    through pxgemm_bt when eligible."""
+
+    def forward(self, x):
+        if _px_eligible(x, self.weight):
+            x2d = x.reshape(-1, x.shape[-1]).contiguous()
+            y = _PxLinearFn.apply(x2d, self.weight, self.bias)
+            return y.view(*x.shape[:-1], self.out_features)
+        return F.linear(x, self.weight, self.bias)

@@ -150,6 +150,35 @@ def test_adam_graphed_bias_correction_advances():
     assert sd['state'][0]['step'] == 8
 
 
+def test_px_linear_fwd_bwd_vs_torch():
+    """PxLinear (pxgemm_bt forward + fused bias epilogue) vs fp32
+    F.linear, bf16 tolerance; backward grads via hipBLASLt must match."""
+    from parallax_amd.ops.linear import PxLinear
+    torch.manual_seed(21)
+    for bias in (False, True):
+        lin = PxLinear(512, 1536, bias=bias).to(DEV, torch.bfloat16)
+        x = torch.randn(300, 512, device=DEV, dtype=torch.bfloat16,
+                        requires_grad=True)
+        y = lin(x)
+        x2 = x.detach().cpu().float().requires_grad_(True)
+        w2 = lin.weight.detach().cpu().float().requires_grad_(True)
+        b2 = lin.bias.detach().cpu().float().requires_grad_(True) \
+            if bias else None
+        y2 = torch.nn.functional.linear(x2, w2, b2)
+        assert torch.allclose(y.detach().cpu().float(), y2.detach(),
+                              atol=0.5, rtol=0.05)
+        dy = torch.randn(300, 1536)
+        y.backward(dy.to(DEV).to(torch.bfloat16))
+        y2.backward(dy)
+        assert torch.allclose(x.grad.cpu().float(), x2.grad, atol=0.5,
+                              rtol=0.05)
+        assert torch.allclose(lin.weight.grad.cpu().float(), w2.grad,
+                              atol=0.5, rtol=0.05)
+        if bias:
+            assert torch.allclose(lin.bias.grad.cpu().float(), b2.grad,
+                                  atol=0.5, rtol=0.05)
+
+
 def test_adam_bf16_master_weights():
     """bf16 params + fused fp32 masters must track an fp32 torch Adam
     run to bf16 rounding (the master carries update precision; without it
