@@ -66,7 +66,8 @@ class _MHA(nn.Module):
             return t.view(B, n, self.h, D // self.h).transpose(1, 2)
 
         q, k, v = split(q, T), split(k, S), split(v, S)
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+        from parallax_amd.ops.attention import px_attention
+        o = px_attention(q, k, v, is_causal=causal)
         return self.out(o.transpose(1, 2).reshape(B, T, D))
 
 

@@ -87,6 +87,14 @@ at::Tensor pxconv3x3_fwd(const at::Tensor& xp, const at::Tensor& w2d,
                          long n_img, long oh, long ow, long cin,
                          long stride, long HP, long WP);
 
+// attention.hip
+at::Tensor px_attn_fwd(const at::Tensor& q, const at::Tensor& k,
+                       const at::Tensor& v, bool causal);
+std::vector<at::Tensor> px_attn_bwd(const at::Tensor& q,
+                                    const at::Tensor& k,
+                                    const at::Tensor& v,
+                                    const at::Tensor& dout, bool causal);
+
 // layernorm.hip
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
                                       const at::Tensor& w,
@@ -143,6 +151,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_rmsprop", &multi_tensor_rmsprop);
   m.def("multi_tensor_ema", &multi_tensor_ema);
   m.def("multi_tensor_zero", &multi_tensor_zero);
+  m.def("px_attn_fwd", &px_attn_fwd);
+  m.def("px_attn_bwd", &px_attn_bwd);
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
   m.def("multi_tensor_scale_by_norm", &multi_tensor_scale_by_norm);
   m.def("sparse_adagrad", &sparse_adagrad);

@@ -150,6 +150,55 @@ def test_adam_graphed_bias_correction_advances():
     assert sd['state'][0]['step'] == 8
 
 
+@pytest.mark.parametrize('causal', [False, True])
+def test_px_attention_fwd_bwd_vs_torch(causal):
+    """Fused short-seq attention vs fp32 torch SDPA (T=S=48, D=64, the
+    NMT shape), including the transposed-view layout _MHA produces."""
+    from parallax_amd.ops.attention import px_attention, _eligible
+    torch.manual_seed(31)
+    B, H, T, D = 4, 8, 48, 64
+    base = torch.randn(3, B, T, H, D)
+    q, k, v = [base[i].to(DEV).to(torch.bfloat16).transpose(1, 2)
+               .requires_grad_(True) for i in range(3)]
+    assert _eligible(q, k, v)
+    o = px_attention(q, k, v, is_causal=causal)
+    refs = [base[i].float().transpose(1, 2).requires_grad_(True)
+            for i in range(3)]
+    o2 = torch.nn.functional.scaled_dot_product_attention(
+        *refs, is_causal=causal)
+    assert torch.allclose(o.detach().cpu().float(), o2.detach(),
+                          atol=2e-2, rtol=5e-2)
+    do = torch.randn(B, H, T, D)
+    o.backward(do.to(DEV).to(torch.bfloat16))
+    o2.backward(do)
+    for got, ref, name in ((q, refs[0], 'dq'), (k, refs[1], 'dk'),
+                           (v, refs[2], 'dv')):
+        assert torch.allclose(got.grad.cpu().float(), ref.grad, atol=5e-2,
+                              rtol=5e-2), name
+
+
+def test_px_attention_cross_shape():
+    """Cross-attention: T != S."""
+    from parallax_amd.ops.attention import px_attention
+    torch.manual_seed(32)
+    q = torch.randn(2, 8, 40, 64, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(2, 8, 56, 64, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(2, 8, 56, 64, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = px_attention(q, k, v)
+    q2, k2, v2 = [t.detach().cpu().float().requires_grad_(True)
+                  for t in (q, k, v)]
+    o2 = torch.nn.functional.scaled_dot_product_attention(q2, k2, v2)
+    assert torch.allclose(o.detach().cpu().float(), o2.detach(),
+                          atol=2e-2, rtol=5e-2)
+    o.sum().backward()
+    o2.sum().backward()
+    assert torch.allclose(k.grad.cpu().float(), k2.grad, atol=0.1,
+                          rtol=5e-2)
+
+
 def test_px_linear_fwd_bwd_vs_torch():
     """PxLinear (pxgemm_bt forward + fused bias epilogue) vs fp32
     F.linear, bf16 tolerance; backward grads via hipBLASLt must match."""
