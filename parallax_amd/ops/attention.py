@@ -26,7 +26,8 @@ class _PxAttnFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v = ctx.saved_tensors
-        dq, dk, dv = _ops.ext().px_attn_bwd(q, k, v, dout, ctx.causal)
+        dq, dk, dv = _ops.ext().px_attn_bwd(q, k, v, dout.contiguous(),
+                                            ctx.causal)
         return dq, dk, dv, None
 
 
