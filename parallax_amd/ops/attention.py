@@ -5,7 +5,13 @@ and the P @ V product all live in LDS; backward recomputes P.  Eligible
 when T, S <= 64 and d_head == 64 (the NMT transformer's 48x48x64 shape,
 where library flash-attention kernels spend 17/28 us on ~2 us of HBM
 traffic); everything else falls back to torch SDPA.
-PARALLAX_PX_ATTN=0 disables (A/B switch).
+
+Measured (MI355X, scripts/bench_attn.py, B=64 H=8 T=48): fwd 26.9 us vs
+SDPA 33.7 us; fwd+bwd 83.7 vs 83.8 us — isolated it is a wash.  But
+END-TO-END NMT drops 308k -> 271k words/s with it on: the backward's
+60 KB-LDS workgroups block co-residency with the rest of the captured
+step.  OPT-IN via PARALLAX_PX_ATTN=1 until the bwd is split/tiled
+(TODO.md); default is torch SDPA.
 """
 
 import os
@@ -40,7 +46,7 @@ def _eligible(q, k, v):
             and all(t.stride(i) % 4 == 0
                     for t in (q, k, v) for i in (0, 1, 2))
             and _ops.native_available()
-            and os.environ.get('PARALLAX_PX_ATTN', '1') != '0')
+            and os.environ.get('PARALLAX_PX_ATTN', '0') == '1')
 
 
 def px_attention(q, k, v, is_causal=False):

@@ -151,9 +151,10 @@ def test_adam_graphed_bias_correction_advances():
 
 
 @pytest.mark.parametrize('causal', [False, True])
-def test_px_attention_fwd_bwd_vs_torch(causal):
+def test_px_attention_fwd_bwd_vs_torch(causal, monkeypatch):
     """Fused short-seq attention vs fp32 torch SDPA (T=S=48, D=64, the
     NMT shape), including the transposed-view layout _MHA produces."""
+    monkeypatch.setenv('PARALLAX_PX_ATTN', '1')  # opt-in kernel path
     from parallax_amd.ops.attention import px_attention, _eligible
     torch.manual_seed(31)
     B, H, T, D = 4, 8, 48, 64
@@ -177,8 +178,9 @@ def test_px_attention_fwd_bwd_vs_torch(causal):
                               rtol=5e-2), name
 
 
-def test_px_attention_cross_shape():
+def test_px_attention_cross_shape(monkeypatch):
     """Cross-attention: T != S."""
+    monkeypatch.setenv('PARALLAX_PX_ATTN', '1')
     from parallax_amd.ops.attention import px_attention
     torch.manual_seed(32)
     q = torch.randn(2, 8, 40, 64, device=DEV, dtype=torch.bfloat16,
