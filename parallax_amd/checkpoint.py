@@ -36,9 +36,25 @@ class CheckpointManager(object):
         if self.enabled and self.is_chief:
             os.makedirs(self.cfg.ckpt_dir, exist_ok=True)
 
+    def _sharded(self):
+        """True when any table uses shard-only storage
+        (replicate_variables=False): state_dict() is then a COLLECTIVE
+        and every rank holds rank-local optimizer state."""
+        return any(getattr(m, '_sharded', False)
+                   for m in self.model.modules())
+
+    def _rank_world(self):
+        import torch.distributed as dist
+        if dist.is_initialized():
+            return dist.get_rank(), dist.get_world_size()
+        return 0, 1
+
     # -- save ----------------------------------------------------------------
     def maybe_save(self, global_step):
-        if not (self.enabled and self.is_chief):
+        if not self.enabled:
+            return False
+        sharded = self._sharded()
+        if not (self.is_chief or sharded):
             return False
         due = False
         if (self.cfg.save_ckpt_steps
@@ -48,19 +64,50 @@ class CheckpointManager(object):
         if (self.cfg.save_ckpt_secs
                 and time.time() - self._last_save_time
                 >= self.cfg.save_ckpt_secs):
-            due = True
+            if sharded:
+                # wall-clock cadence diverges across ranks; in sharded
+                # mode the save is collective, so the chief's decision is
+                # broadcast
+                due = self._bcast_flag(due)
+            else:
+                due = True
         if due:
             self.save(global_step)
         return due
 
+    def _bcast_flag(self, flag):
+        import torch.distributed as dist
+        t = torch.tensor([1 if (flag and self.is_chief) else 0],
+                         dtype=torch.long)
+        dist.broadcast(t, src=0)
+        return bool(t.item())
+
     def save(self, global_step):
+        sharded = self._sharded()
+        rank, world = self._rank_world()
         path = os.path.join(self.cfg.ckpt_dir,
                             'model.ckpt-%d.pt' % global_step)
+        # model state_dict: collective in sharded mode (gathers shards);
+        # every rank builds it, only the chief writes it
+        model_sd = {k: v.detach().cpu()
+                    for k, v in self.model.state_dict().items()}
+        if sharded and self.optimizer is not None:
+            # per-row optimizer state is rank-local: each rank writes its
+            # own sidecar; restore requires the same world size
+            side = os.path.join(
+                self.cfg.ckpt_dir, 'model.ckpt-%d.optshard%d-of-%d.pt'
+                % (global_step, rank, world))
+            torch.save(self.optimizer.state_dict(), side + '.tmp')
+            os.replace(side + '.tmp', side)
+        if not self.is_chief:
+            self._last_save_time = time.time()
+            return
         payload = {
-            'model': {k: v.detach().cpu()
-                      for k, v in self.model.state_dict().items()},
-            'optimizer': self.optimizer.state_dict()
-            if self.optimizer is not None else None,
+            'model': model_sd,
+            'optimizer': (None if sharded else
+                          self.optimizer.state_dict()
+                          if self.optimizer is not None else None),
+            'sharded_world': world if sharded else None,
             'global_step': global_step,
         }
         tmp = path + '.tmp'
@@ -99,6 +146,20 @@ class CheckpointManager(object):
         self.model.load_state_dict(payload['model'])
         if self.optimizer is not None and payload.get('optimizer'):
             self.optimizer.load_state_dict(payload['optimizer'])
+        elif self.optimizer is not None and payload.get('sharded_world'):
+            rank, world = self._rank_world()
+            side = path.replace('.pt', '.optshard%d-of-%d.pt'
+                                % (rank, world))
+            if payload['sharded_world'] == world and os.path.isfile(side):
+                self.optimizer.load_state_dict(
+                    torch.load(side, map_location=map_location,
+                               weights_only=False))
+            else:
+                parallax_log.warning(
+                    'sharded checkpoint written at world=%s, running at '
+                    'world=%d: per-row optimizer state NOT restored '
+                    '(weights restored; accumulators start fresh)',
+                    payload['sharded_world'], world)
         parallax_log.info('restored checkpoint %s (global_step=%d)',
                           path, payload['global_step'])
         return int(payload['global_step'])

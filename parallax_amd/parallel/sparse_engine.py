@@ -22,10 +22,14 @@ Two modes, mirroring the reference's architecture split:
   every rank, so tables and optimizer state stay bit-identical with no
   mirror-refresh broadcast.  The table's PARTITION metadata (virtual
   shards) sets the granularity partition search varies and the
-  checkpoint-shard layout; with ``replicate_variables=False`` the
-  reference stored each shard only on its PS task — on a single xGMI node
-  that variant is strictly slower and is intentionally not implemented
-  (documented divergence).
+  checkpoint-shard layout.  ``replicate_variables=False`` (the
+  reference's shard-only PS placement) is ALSO supported for tables that
+  exceed the replication budget: each rank stores only partitions
+  k % world == rank, lookups fetch rows from their owners over an
+  all-to-all (embedding._ShardedLookup) and the grad push rides the
+  lookup's backward, so reduce_and_apply for those tables is purely
+  local.  On one xGMI node the replicated default is faster (no
+  per-lookup exchange); sharding is for capacity.
 
 Gradient capture: the engines intercept sparse COO grads from autograd via
 per-parameter hooks; the hook swallows the grad (replaces it with an empty
@@ -158,14 +162,38 @@ class SparsePSEngine(_SparseEngineBase):
                  local_aggregation=True, replicate_variables=True):
         super().__init__(comm, sparse_infos, optimizer, average_sparse)
         self.local_aggregation = local_aggregation
-        if not replicate_variables:
-            parallax_log.warning(
-                'replicate_variables=False (shard-only storage) is not '
-                'implemented on the single-node engine; using replicated '
-                'tables (see sparse_engine.py module docstring)')
+        self.sharded_params = set()
+        if not replicate_variables and comm.world_size > 1:
+            # shard-only storage (reference
+            # ps/between_graph_parallel.py: variables live only on their
+            # PS task): PartitionedEmbedding tables drop the rows they do
+            # not own; lookups fetch rows over an all-to-all
+            # (_ShardedLookup) and the grad push happens in its backward,
+            # so reduce_and_apply for these slots is purely local
+            for s in self.slots:
+                mod = getattr(s.info.param, '_parallax_module', None)
+                if mod is not None and hasattr(mod, 'to_sharded'):
+                    mod.to_sharded(comm)
+                    self.sharded_params.add(id(s.info.param))
+            if self.sharded_params:
+                parallax_log.info(
+                    'replicate_variables=False: %d table(s) sharded '
+                    'across %d ranks', len(self.sharded_params),
+                    comm.world_size)
+        elif not replicate_variables:
+            parallax_log.info('replicate_variables=False ignored at '
+                              'world_size=1')
 
     def reduce_and_apply(self):
         live = [s for s in self.slots if s.ids is not None]
+        # sharded tables: the cross-rank exchange already happened in the
+        # lookup's backward; ids are LOCAL rows — dedup locally and apply
+        for s in list(live):
+            if id(s.info.param) in self.sharded_params:
+                u_ids, u_rows, _ = unique_sum_rows(
+                    s.ids, s.rows, average=self.average_sparse)
+                self._apply_rows(s.info.param, u_ids, u_rows)
+                live.remove(s)
         if self.local_aggregation:
             # phase-split across tables: issue every table's sort/compare
             # first, THEN the host-syncing boundary extraction — each

@@ -105,6 +105,11 @@ class ParallaxSession(object):
         ``example_batch``: (tensors..., state_or_None) matching step_fn."""
         assert self.comm.device.type == 'cuda', 'cuda graphs need a GPU'
         g = self.graph
+        if any(getattr(m, '_sharded', False) for m in g.model.modules()):
+            raise RuntimeError(
+                'hipGraph capture is unavailable with shard-only tables '
+                '(replicate_variables=False): the lookup runs collectives '
+                'with data-dependent message sizes inside forward')
         if self.engine.dense_engine:
             self.engine.dense_engine.set_overlap(False)
 

@@ -331,6 +331,115 @@ def _body_lm1b_pure_bf16(rank, world):
     _body_lm1b_tiny_full_stack(rank, world, pure_bf16=True)
 
 
+def _run_lm1b_steps(cfg, batches, seed=0):
+    """Build tiny LM1B under cfg, run the given batches, return the final
+    full tables (collective in sharded mode) + loss trace."""
+    import parallax_amd
+    from parallax_amd.models import lm1b as M
+    torch.manual_seed(seed)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: M.build_lm1b_graph(batch_size=4, num_steps=3,
+                                   vocab_size=300, emb_dim=16,
+                                   state_size=32, proj_size=16,
+                                   num_sampled=32, min_partitions=4,
+                                   lr=0.3, ema_decay=None),
+        None, parallax_config=cfg)
+    losses = []
+    for b in batches:
+        out = sess.run((b[0], b[1], None))
+        losses.append(float(out['loss']))
+    m = sess.graph.model
+    emb = m.emb.full_weight().clone()
+    sw = m.softmax_w.full_weight().clone()
+    sess.close()
+    return emb, sw, losses
+
+
+def _body_sharded_tables_equivalence(rank, world):
+    """replicate_variables=False: shard-only storage must produce the SAME
+    trained tables as the replicated engine on identical data (the
+    reference's PS placement is a storage choice, not a semantics
+    change)."""
+    import torch.distributed as dist
+    from parallax_amd.config import ParallaxConfig
+
+    gen = torch.Generator().manual_seed(777)  # identical on every rank
+    batches = [(torch.randint(0, 300, (4, 3), generator=gen),
+                torch.randint(0, 300, (4, 3), generator=gen))
+               for _ in range(6)]
+
+    cfg_sh = ParallaxConfig(run_option='HYBRID')
+    cfg_sh.communication_config.ps_config.replicate_variables = False
+    emb_s, sw_s, losses_s = _run_lm1b_steps(cfg_sh, batches)
+    # full_weight() gathers identically on every rank
+    g = [torch.zeros_like(emb_s) for _ in range(world)]
+    dist.all_gather(g, emb_s)
+    assert torch.allclose(g[0], g[-1], atol=1e-6)
+
+    dist.destroy_process_group()
+    os.environ['MASTER_PORT'] = str(int(os.environ['MASTER_PORT']) + 1)
+    cfg_rep = ParallaxConfig(run_option='HYBRID')
+    emb_r, sw_r, losses_r = _run_lm1b_steps(cfg_rep, batches)
+
+    assert torch.allclose(emb_s, emb_r, atol=1e-4), \
+        (emb_s - emb_r).abs().max()
+    assert torch.allclose(sw_s, sw_r, atol=1e-4)
+    for a, b in zip(losses_s, losses_r):
+        assert abs(a - b) < 1e-3, (a, b)
+
+
+def _body_sharded_checkpoint(rank, world):
+    """Sharded-mode checkpoint: collective parts.k save (replicated
+    layout on disk), per-rank optimizer sidecars, restore resumes."""
+    import tempfile
+    import parallax_amd
+    import torch.distributed as dist
+    from parallax_amd.config import ParallaxConfig, CheckPointConfig
+    from parallax_amd.models import lm1b as M
+
+    tmp = os.path.join(tempfile.gettempdir(),
+                       'px_shard_ck_%s' % os.environ['MASTER_PORT'])
+    cfg = ParallaxConfig(
+        run_option='HYBRID',
+        ckpt_config=CheckPointConfig(ckpt_dir=tmp, save_ckpt_steps=2))
+    cfg.communication_config.ps_config.replicate_variables = False
+    kw = dict(batch_size=4, num_steps=3, vocab_size=300, emb_dim=16,
+              state_size=32, proj_size=16, num_sampled=32,
+              min_partitions=4, lr=0.3, ema_decay=None)
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: M.build_lm1b_graph(**kw), None, parallax_config=cfg)
+    for _ in range(2):
+        sess.run(M.synthetic_batch(4, 3, 300))
+    emb_saved = sess.graph.model.emb.full_weight().clone()
+    sess.close()
+    dist.barrier()
+    if rank == 0:
+        import glob as _g
+        assert os.path.isfile(os.path.join(tmp, 'model.ckpt-2.pt'))
+        sides = _g.glob(os.path.join(tmp, '*.optshard*-of-%d.pt' % world))
+        assert len(sides) == world, sides
+        payload = torch.load(os.path.join(tmp, 'model.ckpt-2.pt'),
+                             weights_only=False)
+        assert payload['sharded_world'] == world
+        assert 'emb.parts.0' in payload['model']
+    dist.destroy_process_group()
+
+    os.environ['MASTER_PORT'] = str(int(os.environ['MASTER_PORT']) + 1)
+    torch.manual_seed(99)  # different init; restore must overwrite
+    sess2, *_ = parallax_amd.parallel_run(
+        lambda: M.build_lm1b_graph(**kw), None, parallax_config=cfg)
+    assert sess2.global_step == 2
+    m2 = sess2.graph.model
+    assert torch.allclose(m2.emb.full_weight(), emb_saved, atol=1e-6)
+    # accumulators restored from this rank's sidecar
+    st = sess2.graph.optimizer.state.get(m2.emb.weight)
+    assert st is not None and 'sum' in st
+    out = sess2.run(M.synthetic_batch(4, 3, 300))
+    assert torch.isfinite(out['loss'])
+    sess2.close()
+
+
 # ---------------------------------------------------------------------------
 
 @pytest.mark.parametrize('body', [
@@ -352,6 +461,16 @@ def test_lm1b_full_stack_world4():
 
 def test_lm1b_pure_bf16_world2():
     _spawn(_body_lm1b_pure_bf16, get_free_port(), world=2, timeout=300)
+
+
+def test_sharded_tables_equivalence_world2():
+    _spawn(_body_sharded_tables_equivalence, get_free_port(), world=2,
+           timeout=300)
+
+
+def test_sharded_checkpoint_world2():
+    _spawn(_body_sharded_checkpoint, get_free_port(), world=2,
+           timeout=300)
 
 
 def test_average_sparse_world2():
