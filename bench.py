@@ -107,9 +107,10 @@ def main():
     else:
         from parallax_amd.models import resnet as M
         batch = args.batch_size or 64
-        build = lambda: M.build_resnet_graph(depth=50, batch_size=batch,
-                                             use_bf16=on_gpu,
-                                             channels_last=on_gpu)
+        build = lambda: M.build_resnet_graph(
+            depth=50, batch_size=batch, use_bf16=on_gpu,
+            channels_last=on_gpu,
+            pure_bf16=args.pure_bf16 and on_gpu)
         metric, unit = 'ResNet-50 images/sec', 'images/s'
         per_step_items = batch
         baseline = 7520.0

@@ -174,10 +174,21 @@ MODEL_REGISTRY = {
 
 def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
                     optimizer_name='momentum', lr=0.1, momentum=0.9,
-                    weight_decay=1e-4, use_bf16=True, channels_last=True):
+                    weight_decay=1e-4, use_bf16=True, channels_last=True,
+                    pure_bf16=False):
+    """pure_bf16: conv/linear weights live in bf16 (no per-step autocast
+    weight casts; fp32 update precision via the optimizers' fused master
+    weights); BatchNorm modules stay fp32 (weights + running stats — the
+    mixed bf16-input/fp32-param BN is the library's standard
+    mixed-precision path)."""
     model = MODEL_REGISTRY[model_name](num_classes)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
+    if pure_bf16:
+        model = model.to(torch.bfloat16)
+        for m in model.modules():
+            if isinstance(m, nn.modules.batchnorm._BatchNorm):
+                m.float()
     if optimizer_name == 'momentum':
         opt = optim.SGD(model.parameters(), lr=lr, momentum=momentum,
                         weight_decay=weight_decay)
@@ -195,7 +206,10 @@ def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
         x, y = batch
         if channels_last and x.dim() == 4:
             x = x.to(memory_format=torch.channels_last)
-        if use_bf16 and x.is_cuda:
+        if pure_bf16:
+            logits = m(x.to(torch.bfloat16))
+            loss = softmax_cross_entropy(logits.float(), y)
+        elif use_bf16 and x.is_cuda:
             with torch.autocast('cuda', dtype=torch.bfloat16):
                 logits = m(x)
             loss = softmax_cross_entropy(logits.float(), y)

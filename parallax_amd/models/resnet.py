@@ -138,7 +138,11 @@ class ResNet(nn.Module):
 
 def build_resnet_graph(depth=50, num_classes=1001, batch_size=64, lr=0.1,
                        momentum=0.9, weight_decay=1e-4, use_bf16=True,
-                       channels_last=True):
+                       channels_last=True, pure_bf16=False):
+    """pure_bf16: conv/FC weights live in bf16 (no per-step autocast
+    weight casts; fp32 update precision via the fused master-weight
+    optimizer path); BatchNorm stays fp32 (mixed bf16-input BN is the
+    library's standard mixed-precision path)."""
     model = ResNet(depth, num_classes)
     import os as _os
     if _os.environ.get('PARALLAX_NATIVE_CONV1X1') == '1' and channels_last:
@@ -149,15 +153,24 @@ def build_resnet_graph(depth=50, num_classes=1001, batch_size=64, lr=0.1,
         swap_conv3x3(model)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
+    if pure_bf16:
+        import torch.nn as _nn
+        model = model.to(torch.bfloat16)
+        for mod in model.modules():
+            if isinstance(mod, _nn.modules.batchnorm._BatchNorm):
+                mod.float()
     optimizer = optim.SGD(model.parameters(), lr=lr, momentum=momentum,
                           weight_decay=weight_decay)
-    use_amp = use_bf16
+    use_amp = use_bf16 and not pure_bf16
 
     def step_fn(m, batch):
         x, y = batch
         if channels_last and x.dim() == 4:
             x = x.to(memory_format=torch.channels_last)
-        if use_amp and x.is_cuda:
+        if pure_bf16:
+            logits = m(x.to(torch.bfloat16))
+            loss = softmax_cross_entropy(logits.float(), y)
+        elif use_amp and x.is_cuda:
             with torch.autocast('cuda', dtype=torch.bfloat16):
                 logits = m(x)
             loss = softmax_cross_entropy(logits.float(), y)

@@ -117,3 +117,37 @@ def test_lm1b_pure_bf16_learns(monkeypatch, free_port):
     sess.close()
     import torch.distributed as dist
     dist.destroy_process_group()
+
+
+def test_resnet_pure_bf16_structure_and_step(monkeypatch, free_port):
+    """pure_bf16 ResNet: conv/FC weights bf16, BatchNorm fp32 (weights,
+    stats); a training step runs and learns through the master-weight
+    momentum path."""
+    monkeypatch.setenv('WORLD_SIZE', '1')
+    monkeypatch.setenv('RANK', '0')
+    monkeypatch.setenv('MASTER_ADDR', '127.0.0.1')
+    monkeypatch.setenv('MASTER_PORT', str(free_port))
+    torch.manual_seed(0)
+    import torch.nn as nn
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: build_resnet_graph(depth=50, num_classes=10, lr=0.05,
+                                   use_bf16=False, channels_last=False,
+                                   pure_bf16=True),
+        None, parallax_config=ParallaxConfig(run_option='MPI'))
+    m = sess.graph.model
+    convs = [mod for mod in m.modules() if isinstance(mod, nn.Conv2d)]
+    bns = [mod for mod in m.modules()
+           if isinstance(mod, nn.modules.batchnorm._BatchNorm)]
+    assert convs and all(c.weight.dtype == torch.bfloat16 for c in convs)
+    assert bns and all(b.weight.dtype == torch.float32 for b in bns)
+    assert all(b.running_mean.dtype == torch.float32 for b in bns)
+    x = torch.randn(2, 3, 64, 64)
+    y = torch.randint(0, 10, (2,))
+    losses = [float(sess.run((x, y))['loss']) for _ in range(5)]
+    assert all(l == l for l in losses)
+    assert losses[-1] < losses[0]
+    st = sess.graph.optimizer.state[convs[0].weight]
+    assert st['master'].dtype == torch.float32
+    sess.close()
+    import torch.distributed as dist
+    dist.destroy_process_group()
