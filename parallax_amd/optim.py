@@ -27,6 +27,26 @@ def _use_native(params):
     return any(p.is_cuda for p in params) and _ops.native_available()
 
 
+def _dtype_groups(params, *lists):
+    """Split aligned per-param lists into same-dtype groups: the
+    multi-tensor kernels dispatch ONE element type per launch (a mixed
+    bf16-conv + fp32-BN model would otherwise reinterpret memory)."""
+    dtypes = {p.dtype for p in params}
+    if len(dtypes) == 1:
+        yield (params,) + lists
+        return
+    order = []
+    by_dt = {}
+    for i, p in enumerate(params):
+        if p.dtype not in by_dt:
+            by_dt[p.dtype] = []
+            order.append(p.dtype)
+        by_dt[p.dtype].append(i)
+    for dt in order:
+        ii = by_dt[dt]
+        yield tuple([lst[i] for i in ii] for lst in (params,) + lists)
+
+
 class ParallaxOptimizer(Optimizer):
     """Base: splits the fused dense path from the per-row sparse path."""
 
@@ -134,8 +154,10 @@ class SGD(ParallaxOptimizer):
         masters, has_m = self._ensure_masters(params)
         if mom == 0.0:
             if _use_native(params):
-                _ops.ext().multi_tensor_sgd(params, [g for g in grads],
-                                            lr, wd, masters)
+                for ps, gs, ms_ in _dtype_groups(params, list(grads),
+                                                 masters or params):
+                    _ops.ext().multi_tensor_sgd(
+                        ps, gs, lr, wd, ms_ if masters else [])
                 return
             for i, (p, g) in enumerate(zip(params, grads)):
                 w = masters[i] if has_m else p
@@ -154,9 +176,11 @@ class SGD(ParallaxOptimizer):
                     p, dtype=torch.float32)
             bufs.append(st['momentum_buffer'])
         if _use_native(params):
-            _ops.ext().multi_tensor_momentum(params, list(grads), bufs,
-                                             lr, mom, wd, nesterov,
-                                             masters)
+            for ps, gs, bs, ms_ in _dtype_groups(params, list(grads),
+                                                 bufs, masters or params):
+                _ops.ext().multi_tensor_momentum(
+                    ps, gs, bs, lr, mom, wd, nesterov,
+                    ms_ if masters else [])
             return
         for i, (p, g, buf) in enumerate(zip(params, grads, bufs)):
             w = masters[i] if has_m else p
@@ -205,8 +229,11 @@ class Adagrad(ParallaxOptimizer):
         accums = [self._accum(p, group) for p in params]
         masters, has_m = self._ensure_masters(params)
         if _use_native(params):
-            _ops.ext().multi_tensor_adagrad(params, list(grads), accums,
-                                            lr, eps, masters)
+            for ps, gs, acs, ms_ in _dtype_groups(params, list(grads),
+                                                  accums,
+                                                  masters or params):
+                _ops.ext().multi_tensor_adagrad(
+                    ps, gs, acs, lr, eps, ms_ if masters else [])
             return
         for i, (p, g, acc) in enumerate(zip(params, grads, accums)):
             w = masters[i] if has_m else p
@@ -296,9 +323,11 @@ class Adam(ParallaxOptimizer):
                 self._dev_steps[gi] = buf
             self._dev_step_params.update(params)
             buf.add_(1.0)
-            _ops.ext().multi_tensor_adam_dev(params, list(grads), ms, vs,
-                                             buf, lr, b1, b2, eps, wd,
-                                             masters)
+            for ps, gs, m_, v_, ms_ in _dtype_groups(
+                    params, list(grads), ms, vs, masters or params):
+                _ops.ext().multi_tensor_adam_dev(
+                    ps, gs, m_, v_, buf, lr, b1, b2, eps, wd,
+                    ms_ if masters else [])
             return
         for i, (p, g, m, v, t) in enumerate(zip(params, grads, ms, vs,
                                                 steps)):
@@ -360,8 +389,11 @@ class RMSProp(ParallaxOptimizer):
             bufs.append(st['momentum_buffer'])
         masters, has_m = self._ensure_masters(params)
         if _use_native(params):
-            _ops.ext().multi_tensor_rmsprop(params, list(grads), rms, bufs,
-                                            lr, decay, mom, eps, masters)
+            for ps, gs, rs, bs, ms_ in _dtype_groups(
+                    params, list(grads), rms, bufs, masters or params):
+                _ops.ext().multi_tensor_rmsprop(
+                    ps, gs, rs, bs, lr, decay, mom, eps,
+                    ms_ if masters else [])
             return
         for i, (p, g, sq, buf) in enumerate(zip(params, grads, rms, bufs)):
             w = masters[i] if has_m else p

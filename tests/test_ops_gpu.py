@@ -230,6 +230,34 @@ def test_px_linear_fwd_bwd_vs_torch():
                                   atol=0.5, rtol=0.05)
 
 
+def test_momentum_mixed_dtype_params():
+    """A mixed bf16+fp32 parameter list (pure-bf16 convs + fp32 BN) must
+    split into per-dtype multi-tensor launches — one launch would
+    reinterpret the fp32 tensors' memory as bf16."""
+    from parallax_amd import optim
+    torch.manual_seed(13)
+    pb = torch.nn.Parameter(torch.randn(500, device=DEV,
+                                        dtype=torch.bfloat16))
+    pf = torch.nn.Parameter(torch.randn(300, device=DEV))
+    ref_b = pb.detach().cpu().float().clone().requires_grad_(True)
+    ref_f = pf.detach().cpu().clone().requires_grad_(True)
+    o1 = optim.SGD([pb, pf], lr=0.1, momentum=0.9, weight_decay=1e-4)
+    o2 = torch.optim.SGD([ref_b, ref_f], lr=0.1, momentum=0.9,
+                         weight_decay=1e-4)
+    for _ in range(5):
+        gb, gf = torch.randn(500) * 0.1, torch.randn(300) * 0.1
+        pb.grad = gb.to(DEV).to(torch.bfloat16)
+        pf.grad = gf.to(DEV)
+        ref_b.grad, ref_f.grad = gb.clone(), gf.clone()
+        o1.step()
+        o2.step()
+    assert torch.allclose(pf.detach().cpu(), ref_f.detach(), atol=1e-5)
+    assert torch.allclose(pb.detach().cpu().float(), ref_b.detach(),
+                          atol=2e-2)
+    assert torch.isfinite(pb.detach().float()).all()
+    assert torch.isfinite(pf.detach()).all()
+
+
 def test_adam_bf16_master_weights():
     """bf16 params + fused fp32 masters must track an fp32 torch Adam
     run to bf16 rounding (the master carries update precision; without it
