@@ -156,15 +156,34 @@ class CommContext(object):
                 outs.append(recv[off:off + c].reshape((c,) + tail_shape))
                 off += c
             return outs
-        # gloo fallback: allgather everything, pick own column
-        all_lists = [None] * ws
-        # serialize shapes via all_gather_v per destination
-        received = []
+        # gloo fallback: pairwise isend/irecv (counts via one allgather,
+        # then point-to-point payloads — O(world) messages, not O(world)
+        # full allgathers)
+        my_counts = torch.tensor([t.shape[0] for t in tensors_per_rank],
+                                 dtype=torch.long)
+        all_counts = [torch.zeros_like(my_counts) for _ in range(ws)]
+        dist.all_gather(all_counts, my_counts)
+        tail_shape = tuple(tensors_per_rank[0].shape[1:])
+        dtype = tensors_per_rank[0].dtype
+        outs = [None] * ws
+        reqs = []
+        for src in range(ws):
+            if src == self.rank:
+                outs[src] = tensors_per_rank[src]
+                continue
+            n = int(all_counts[src][self.rank])
+            outs[src] = torch.empty((n,) + tail_shape, dtype=dtype)
+            if n > 0:
+                reqs.append(dist.irecv(outs[src], src=src))
         for dst in range(ws):
-            gathered = self.all_gather_v(tensors_per_rank[dst])
             if dst == self.rank:
-                received = gathered
-        return received
+                continue
+            t = tensors_per_rank[dst].contiguous()
+            if t.shape[0] > 0:
+                reqs.append(dist.isend(t, dst=dst))
+        for r in reqs:
+            r.wait()
+        return outs
 
     # -- a2a convenience for (ids, rows) pairs -------------------------------
     def exchange_rows(self, ids_per_rank, rows_per_rank):
