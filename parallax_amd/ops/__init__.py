@@ -83,3 +83,5 @@ from parallax_amd.ops.lstm import lstm_cell, ParallaxLSTM
 from parallax_amd.ops.sampled_softmax import sampled_softmax_loss
 from parallax_amd.ops.misc import (clip_by_global_norm, global_norm,
                                    ema_update)
+from parallax_amd.ops.linear import PxLinear
+from parallax_amd.ops.attention import px_attention

@@ -304,3 +304,45 @@ def test_optimizer_handles_bare_sparse_grad():
     assert not torch.allclose(before[2], after[2])
     assert not torch.allclose(before[7], after[7])
     assert torch.allclose(before[0], after[0])
+
+
+def test_dtype_groups_split():
+    """Mixed-dtype param lists split into per-dtype aligned groups
+    (first-appearance order preserved)."""
+    from parallax_amd.optim import _dtype_groups
+    pb = [torch.zeros(3, dtype=torch.bfloat16), torch.zeros(2),
+          torch.zeros(4, dtype=torch.bfloat16), torch.zeros(1)]
+    tags = ['b0', 'f0', 'b1', 'f1']
+    groups = list(_dtype_groups(pb, tags))
+    assert len(groups) == 2
+    (ps0, t0), (ps1, t1) = groups
+    assert t0 == ['b0', 'b1'] and all(
+        p.dtype == torch.bfloat16 for p in ps0)
+    assert t1 == ['f0', 'f1'] and all(
+        p.dtype == torch.float32 for p in ps1)
+    # uniform list passes through unchanged (no copies)
+    only = list(_dtype_groups(pb[:1], tags[:1]))
+    assert len(only) == 1 and only[0][0] is pb[:1] or \
+        only[0][0] == pb[:1]
+
+
+def test_px_linear_cpu_fallback_matches_nn_linear():
+    """On CPU (no kernel) PxLinear == nn.Linear exactly, including the
+    state-dict key layout."""
+    from parallax_amd.ops.linear import PxLinear
+    torch.manual_seed(5)
+    lin = PxLinear(16, 8)
+    ref = torch.nn.Linear(16, 8)
+    ref.load_state_dict(lin.state_dict())  # same keys
+    x = torch.randn(4, 16)
+    assert torch.equal(lin(x), ref(x))
+
+
+def test_px_attention_cpu_fallback_is_sdpa():
+    """Ineligible inputs (CPU) route to torch SDPA."""
+    from parallax_amd.ops.attention import px_attention, _eligible
+    q = torch.randn(2, 4, 8, 64)
+    assert not _eligible(q, q, q)
+    out = px_attention(q, q, q)
+    ref = torch.nn.functional.scaled_dot_product_attention(q, q, q)
+    assert torch.allclose(out, ref)
