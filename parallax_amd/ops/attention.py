@@ -6,12 +6,18 @@ when T, S <= 64 and d_head == 64 (the NMT transformer's 48x48x64 shape,
 where library flash-attention kernels spend 17/28 us on ~2 us of HBM
 traffic); everything else falls back to torch SDPA.
 
-Measured (MI355X, scripts/bench_attn.py, B=64 H=8 T=48): fwd 26.9 us vs
-SDPA 33.7 us; fwd+bwd 83.7 vs 83.8 us — isolated it is a wash.  But
-END-TO-END NMT drops 308k -> 271k words/s with it on: the backward's
-60 KB-LDS workgroups block co-residency with the rest of the captured
-step.  OPT-IN via PARALLAX_PX_ATTN=1 until the bwd is split/tiled
-(TODO.md); default is torch SDPA.
+Measured on MI355X: isolated wall-clock is at parity (fwd+bwd 83.7 vs
+83.8 us, scripts/bench_attn.py) but that comparison flatters us — the
+torch SDPA *call* carries ~15 us of dispatch overhead that disappears
+under hipGraph capture.  In-chain KERNEL time (rocprofv3, captured NMT
+step): ours 28.1 us fwd / 63.9 us bwd vs aotriton's 16.9 / 28.2 — ours
+loses ~1.0 ms/step e2e (308k -> 271k words/s).  Root cause: the
+per-(t,s) dot products re-read both operand rows from LDS for every
+MAC (~0.6 MB of LDS traffic per workgroup fwd, 3x that bwd, against a
+~65 TB/s aggregate LDS ceiling); aotriton's MFMA tiles amortize
+operands in registers.  The fix is a register-/MFMA-tiled rewrite
+(TODO.md), not a split.  OPT-IN via PARALLAX_PX_ATTN=1; default is
+torch SDPA.
 """
 
 import os
