@@ -109,7 +109,9 @@ __device__ __forceinline__ void att_scores_softmax(
 #pragma unroll
     for (int j = 0; j < ATT_CHUNK; ++j)
       kf[j] = (lane < S) ? bf2f(lk[lane * ATT_LD + c0 + j]) : 0.f;
-    for (int r = 0; r < rows; ++r) {
+#pragma unroll
+    for (int r = 0; r < ATT_RPW; ++r) {
+      if (r >= rows) break;
       const ushort_t* qr = &lq[(t0 + r) * ATT_LD + c0];
       float a = 0.f;
 #pragma unroll
@@ -118,7 +120,9 @@ __device__ __forceinline__ void att_scores_softmax(
     }
   }
   // per-row softmax across lanes (lane == s)
-  for (int r = 0; r < rows; ++r) {
+#pragma unroll
+  for (int r = 0; r < ATT_RPW; ++r) {
+    if (r >= rows) break;
     const int t = t0 + r;
     float sc = (lane < S && !(causal && lane > t))
                    ? acc[r] * scale
@@ -161,12 +165,16 @@ px_attn_fwd_kernel(const ushort_t* __restrict__ q,
   for (int r = 0; r < ATT_RPW; ++r) oacc[r] = 0.f;
   for (int s = 0; s < S; ++s) {
     const float vv = bf2f(lv[s * ATT_LD + lane]);
-    for (int r = 0; r < rows; ++r)
+#pragma unroll
+    for (int r = 0; r < ATT_RPW; ++r) {
+      if (r >= rows) break;
       oacc[r] += bf2f(lp[(t0 + r) * ATT_LD + s]) * vv;
+    }
   }
   ushort_t* og = o + (long)blockIdx.x * T * ATT_D;
-  for (int r = 0; r < rows; ++r)
-    og[(t0 + r) * ATT_D + lane] = f2bf(oacc[r]);
+#pragma unroll
+  for (int r = 0; r < ATT_RPW; ++r)
+    if (r < rows) og[(t0 + r) * ATT_D + lane] = f2bf(oacc[r]);
 }
 
 __global__ void __launch_bounds__(PX_BLOCK)
@@ -210,12 +218,16 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
     for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
     for (int t = 0; t < T; ++t) {
       const float dov = bf2f(ldo[t * ATT_LD + lane]);
-      for (int r = 0; r < srows; ++r)
+#pragma unroll
+      for (int r = 0; r < ATT_RPW; ++r) {
+        if (r >= srows) break;
         acc[r] += bf2f(lp[t * ATT_LD + (s0 + r)]) * dov;
+      }
     }
     ushort_t* dvg = dv + bh * (long)S * ATT_D;
-    for (int r = 0; r < srows; ++r)
-      dvg[(s0 + r) * ATT_D + lane] = f2bf(acc[r]);
+#pragma unroll
+    for (int r = 0; r < ATT_RPW; ++r)
+      if (r < srows) dvg[(s0 + r) * ATT_D + lane] = f2bf(acc[r]);
   }
   __syncthreads();  // dV read EVERY P row; dS below overwrites lp
   // dP[t][s] = dO[t] . V[s]; delta[t] = sum_s dP*P; dS overwrites lp.
@@ -229,7 +241,9 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
 #pragma unroll
       for (int j = 0; j < ATT_CHUNK; ++j)
         vf[j] = (lane < S) ? bf2f(lv[lane * ATT_LD + c0 + j]) : 0.f;
-      for (int r = 0; r < trows; ++r) {
+#pragma unroll
+      for (int r = 0; r < ATT_RPW; ++r) {
+        if (r >= trows) break;
         const ushort_t* dor = &ldo[(t0 + r) * ATT_LD + c0];
         float a = 0.f;
 #pragma unroll
@@ -237,7 +251,9 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
         acc[r] += a;
       }
     }
-    for (int r = 0; r < trows; ++r) {
+#pragma unroll
+    for (int r = 0; r < ATT_RPW; ++r) {
+      if (r >= trows) break;
       const int t = t0 + r;
       const float pv = (lane < S) ? bf2f(lp[t * ATT_LD + lane]) : 0.f;
       float delta = wave_reduce_sum(pv * acc[r]);
@@ -254,12 +270,16 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
     for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
     for (int s = 0; s < S; ++s) {
       const float kv = bf2f(lk[s * ATT_LD + lane]);
-      for (int r = 0; r < trows; ++r)
+#pragma unroll
+      for (int r = 0; r < ATT_RPW; ++r) {
+        if (r >= trows) break;
         acc[r] += bf2f(lp[(t0 + r) * ATT_LD + s]) * kv;
+      }
     }
     ushort_t* dqg = dq + bh * (long)T * ATT_D;
-    for (int r = 0; r < trows; ++r)
-      dqg[(t0 + r) * ATT_D + lane] = f2bf(acc[r]);
+#pragma unroll
+    for (int r = 0; r < ATT_RPW; ++r)
+      if (r < trows) dqg[(t0 + r) * ATT_D + lane] = f2bf(acc[r]);
   }
   // dK[s][d] = sum_t dS[t][s] Q[t][d]: lane owns d, wave owns s rows
   {
@@ -268,12 +288,16 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
     for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
     for (int t = 0; t < T; ++t) {
       const float qv = bf2f(lq[t * ATT_LD + lane]);
-      for (int r = 0; r < srows; ++r)
+#pragma unroll
+      for (int r = 0; r < ATT_RPW; ++r) {
+        if (r >= srows) break;
         acc[r] += bf2f(lp[t * ATT_LD + (s0 + r)]) * qv;
+      }
     }
     ushort_t* dkg = dk + bh * (long)S * ATT_D;
-    for (int r = 0; r < srows; ++r)
-      dkg[(s0 + r) * ATT_D + lane] = f2bf(acc[r]);
+#pragma unroll
+    for (int r = 0; r < ATT_RPW; ++r)
+      if (r < srows) dkg[(s0 + r) * ATT_D + lane] = f2bf(acc[r]);
   }
 }
 
