@@ -15,10 +15,8 @@
 // Backward recomputes P from q/k (cheaper than spilling it in fwd),
 // then:  dV = P^T dO;  dP = dO V^T;  dS = P o (dP - rowsum(dP o P));
 //        dQ = s * dS K;  dK = s * dS^T Q.
-// P/dS are kept in LDS as bf16; dP lives entirely in registers.
-// All matrix phases are register-tiled (see att_scores_softmax): each
-// wave owns a block of rows, each lane one column, so every LDS element
-// is read once per wave instead of once per MAC.
+// P is kept in LDS as bf16 (halves the footprint; matches bf16 training
+// precision), dP in fp32.
 
 #include <torch/extension.h>
 #include <ATen/ATen.h>
@@ -81,61 +79,6 @@ __device__ __forceinline__ float att_dot64(const ushort_t* a,
   return acc;
 }
 
-// ---------------------------------------------------------------------------
-// Register-tiled phases.  The naive per-(t,s) dot re-reads BOTH operand
-// rows from LDS for every MAC (~0.6 MB/WG) and is LDS-bandwidth-bound;
-// here each wave owns a contiguous block of rows and each LANE owns one
-// column (s or d), so every LDS element is read ONCE per wave (operands
-// amortized in registers) and row reductions happen across lanes.
-#define ATT_RPW (ATT_MAX / 4)  // max rows per wave (4 waves / block)
-#define ATT_CHUNK 16           // d-chunk cached in registers
-
-// scores + softmax -> P (bf16) in lp.  Wave w owns rows
-// [w*rpw, w*rpw+rows); lane owns column s = LANE.
-__device__ __forceinline__ void att_scores_softmax(
-    const ushort_t* lq, const ushort_t* lk, ushort_t* lp, int T, int S,
-    float scale, int causal) {
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
-  const int rpw = (T + 3) / 4;
-  const int t0 = wid * rpw;
-  const int rows = min(rpw, T - t0) > 0 ? min(rpw, T - t0) : 0;
-  float acc[ATT_RPW];
-#pragma unroll
-  for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
-  // k[s=lane] fragment per d-chunk, reused across the wave's rows
-  for (int c0 = 0; c0 < ATT_D; c0 += ATT_CHUNK) {
-    float kf[ATT_CHUNK];
-#pragma unroll
-    for (int j = 0; j < ATT_CHUNK; ++j)
-      kf[j] = (lane < S) ? bf2f(lk[lane * ATT_LD + c0 + j]) : 0.f;
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r) {
-      if (r >= rows) break;
-      const ushort_t* qr = &lq[(t0 + r) * ATT_LD + c0];
-      float a = 0.f;
-#pragma unroll
-      for (int j = 0; j < ATT_CHUNK; ++j) a += bf2f(qr[j]) * kf[j];
-      acc[r] += a;
-    }
-  }
-  // per-row softmax across lanes (lane == s)
-#pragma unroll
-  for (int r = 0; r < ATT_RPW; ++r) {
-    if (r >= rows) break;
-    const int t = t0 + r;
-    float sc = (lane < S && !(causal && lane > t))
-                   ? acc[r] * scale
-                   : -INFINITY;
-    float mx = wave_reduce_max(sc);
-    mx = __shfl(mx, 0, WAVE);
-    float e = (sc == -INFINITY) ? 0.f : __expf(sc - mx);
-    float sum = wave_reduce_sum(e);
-    sum = __shfl(sum, 0, WAVE);
-    if (lane < S) lp[t * ATT_LD + lane] = f2bf(e / sum);
-  }
-}
-
 __global__ void __launch_bounds__(PX_BLOCK)
 px_attn_fwd_kernel(const ushort_t* __restrict__ q,
                    const ushort_t* __restrict__ k,
@@ -146,35 +89,45 @@ px_attn_fwd_kernel(const ushort_t* __restrict__ q,
   __shared__ ushort_t lq[ATT_MAX * ATT_LD];
   __shared__ ushort_t lk[ATT_MAX * ATT_LD];
   __shared__ ushort_t lv[ATT_MAX * ATT_LD];
-  __shared__ ushort_t lp[ATT_MAX * ATT_LD];  // probabilities (bf16)
+  __shared__ float lp[ATT_MAX * ATT_LDF];  // scores -> probs (fp32)
   const long b = blockIdx.x / H, h = blockIdx.x % H;
   att_load_tile(q + b * qb + h * qh, qt, T, lq);
   att_load_tile(k + b * kb + h * kh, kt, S, lk);
   att_load_tile(v + b * vb + h * vh, vt, S, lv);
   __syncthreads();
-  att_scores_softmax(lq, lk, lp, T, S, scale, causal);
-  __syncthreads();
-  // O = P @ V: lane owns column d; each v element read once per wave
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
-  const int rpw = (T + 3) / 4;
-  const int t0 = wid * rpw;
-  const int rows = min(rpw, T - t0) > 0 ? min(rpw, T - t0) : 0;
-  float oacc[ATT_RPW];
-#pragma unroll
-  for (int r = 0; r < ATT_RPW; ++r) oacc[r] = 0.f;
-  for (int s = 0; s < S; ++s) {
-    const float vv = bf2f(lv[s * ATT_LD + lane]);
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r) {
-      if (r >= rows) break;
-      oacc[r] += bf2f(lp[(t0 + r) * ATT_LD + s]) * vv;
-    }
+  // scores
+  for (int i = threadIdx.x; i < T * S; i += blockDim.x) {
+    const int t = i / S, s = i % S;
+    lp[t * ATT_LDF + s] =
+        (causal && s > t)
+            ? -INFINITY
+            : att_dot64(&lq[t * ATT_LD], &lk[s * ATT_LD]) * scale;
   }
+  __syncthreads();
+  // softmax: one thread per row
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    float* row = &lp[t * ATT_LDF];
+    float mx = -INFINITY;
+    for (int s = 0; s < S; ++s) mx = fmaxf(mx, row[s]);
+    float sum = 0.f;
+    for (int s = 0; s < S; ++s) {
+      float e = __expf(row[s] - mx);
+      row[s] = e;
+      sum += e;
+    }
+    const float inv = 1.f / sum;
+    for (int s = 0; s < S; ++s) row[s] *= inv;
+  }
+  __syncthreads();
+  // O = P @ V  (o is a contiguous [B,H,T,64] allocation)
   ushort_t* og = o + (long)blockIdx.x * T * ATT_D;
-#pragma unroll
-  for (int r = 0; r < ATT_RPW; ++r)
-    if (r < rows) og[(t0 + r) * ATT_D + lane] = f2bf(oacc[r]);
+  for (int i = threadIdx.x; i < T * ATT_D; i += blockDim.x) {
+    const int t = i / ATT_D, d = i % ATT_D;
+    const float* pr = &lp[t * ATT_LDF];
+    float acc = 0.f;
+    for (int s = 0; s < S; ++s) acc += pr[s] * bf2f(lv[s * ATT_LD + d]);
+    og[t * ATT_D + d] = f2bf(acc);
+  }
 }
 
 __global__ void __launch_bounds__(PX_BLOCK)
@@ -193,7 +146,9 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
   __shared__ ushort_t lk[ATT_MAX * ATT_LD];
   __shared__ ushort_t lv[ATT_MAX * ATT_LD];
   __shared__ ushort_t ldo[ATT_MAX * ATT_LD];
-  __shared__ ushort_t lp[ATT_MAX * ATT_LD];  // P (bf16), then dS
+  __shared__ ushort_t lp[ATT_MAX * ATT_LD];   // P (bf16), then dS
+  __shared__ float ldp[ATT_MAX * ATT_LDF];    // dP (fp32)
+  __shared__ float ldelta[ATT_MAX];
   const long b = blockIdx.x / H, h = blockIdx.x % H;
   const long bh = blockIdx.x;
   att_load_tile(q + b * qb + h * qh, qt, T, lq);
@@ -201,103 +156,80 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
   att_load_tile(v + b * vb + h * vh, vt, S, lv);
   att_load_tile(dout + b * db + h * dh, dt, T, ldo);
   __syncthreads();
-  att_scores_softmax(lq, lk, lp, T, S, scale, causal);
-  __syncthreads();
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
-  const int rpw = (T + 3) / 4;       // q-side rows per wave
-  const int spw = (S + 3) / 4;       // k/v-side rows per wave
-  const int t0 = wid * rpw;
-  const int trows = min(rpw, T - t0) > 0 ? min(rpw, T - t0) : 0;
-  const int s0 = wid * spw;
-  const int srows = min(spw, S - s0) > 0 ? min(spw, S - s0) : 0;
-  // dV[s][d] = sum_t P[t][s] dO[t][d]: lane owns d, wave owns s rows
-  {
-    float acc[ATT_RPW];
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
-    for (int t = 0; t < T; ++t) {
-      const float dov = bf2f(ldo[t * ATT_LD + lane]);
-#pragma unroll
-      for (int r = 0; r < ATT_RPW; ++r) {
-        if (r >= srows) break;
-        acc[r] += bf2f(lp[t * ATT_LD + (s0 + r)]) * dov;
-      }
-    }
-    ushort_t* dvg = dv + bh * (long)S * ATT_D;
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r)
-      if (r < srows) dvg[(s0 + r) * ATT_D + lane] = f2bf(acc[r]);
-  }
-  __syncthreads();  // dV read EVERY P row; dS below overwrites lp
-  // dP[t][s] = dO[t] . V[s]; delta[t] = sum_s dP*P; dS overwrites lp.
-  // lane owns s; wave owns t rows (same mapping as the score phase)
-  {
-    float acc[ATT_RPW];
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
-    for (int c0 = 0; c0 < ATT_D; c0 += ATT_CHUNK) {
-      float vf[ATT_CHUNK];
-#pragma unroll
-      for (int j = 0; j < ATT_CHUNK; ++j)
-        vf[j] = (lane < S) ? bf2f(lv[lane * ATT_LD + c0 + j]) : 0.f;
-#pragma unroll
-      for (int r = 0; r < ATT_RPW; ++r) {
-        if (r >= trows) break;
-        const ushort_t* dor = &ldo[(t0 + r) * ATT_LD + c0];
-        float a = 0.f;
-#pragma unroll
-        for (int j = 0; j < ATT_CHUNK; ++j) a += bf2f(dor[j]) * vf[j];
-        acc[r] += a;
-      }
-    }
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r) {
-      if (r >= trows) break;
-      const int t = t0 + r;
-      const float pv = (lane < S) ? bf2f(lp[t * ATT_LD + lane]) : 0.f;
-      float delta = wave_reduce_sum(pv * acc[r]);
-      delta = __shfl(delta, 0, WAVE);
-      if (lane < S)
-        lp[t * ATT_LD + lane] = f2bf(pv * (acc[r] - delta) * scale);
-    }
+  // recompute scores into ldp (fp32 scratch)
+  for (int i = threadIdx.x; i < T * S; i += blockDim.x) {
+    const int t = i / S, s = i % S;
+    ldp[t * ATT_LDF + s] =
+        (causal && s > t)
+            ? -INFINITY
+            : att_dot64(&lq[t * ATT_LD], &lk[s * ATT_LD]) * scale;
   }
   __syncthreads();
-  // dQ[t][d] = sum_s dS[t][s] K[s][d]: lane owns d, wave owns t rows
-  {
-    float acc[ATT_RPW];
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
+  // softmax rows -> P (bf16 in lp)
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    float* row = &ldp[t * ATT_LDF];
+    float mx = -INFINITY;
+    for (int s = 0; s < S; ++s) mx = fmaxf(mx, row[s]);
+    float sum = 0.f;
     for (int s = 0; s < S; ++s) {
-      const float kv = bf2f(lk[s * ATT_LD + lane]);
-#pragma unroll
-      for (int r = 0; r < ATT_RPW; ++r) {
-        if (r >= trows) break;
-        acc[r] += bf2f(lp[(t0 + r) * ATT_LD + s]) * kv;
-      }
+      float e = __expf(row[s] - mx);
+      row[s] = e;
+      sum += e;
     }
-    ushort_t* dqg = dq + bh * (long)T * ATT_D;
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r)
-      if (r < trows) dqg[(t0 + r) * ATT_D + lane] = f2bf(acc[r]);
+    const float inv = 1.f / sum;
+    for (int s = 0; s < S; ++s)
+      lp[t * ATT_LD + s] = f2bf(row[s] * inv);
   }
-  // dK[s][d] = sum_t dS[t][s] Q[t][d]: lane owns d, wave owns s rows
-  {
-    float acc[ATT_RPW];
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r) acc[r] = 0.f;
-    for (int t = 0; t < T; ++t) {
-      const float qv = bf2f(lq[t * ATT_LD + lane]);
-#pragma unroll
-      for (int r = 0; r < ATT_RPW; ++r) {
-        if (r >= srows) break;
-        acc[r] += bf2f(lp[t * ATT_LD + (s0 + r)]) * qv;
-      }
-    }
-    ushort_t* dkg = dk + bh * (long)S * ATT_D;
-#pragma unroll
-    for (int r = 0; r < ATT_RPW; ++r)
-      if (r < srows) dkg[(s0 + r) * ATT_D + lane] = f2bf(acc[r]);
+  __syncthreads();
+  // dV[s][d] = sum_t P[t][s] * dO[t][d]   (outputs are contiguous
+  // [B,H,rows,64] allocations — fixed indexing)
+  ushort_t* dvg = dv + bh * (long)S * ATT_D;
+  for (int i = threadIdx.x; i < S * ATT_D; i += blockDim.x) {
+    const int s = i / ATT_D, d = i % ATT_D;
+    float acc = 0.f;
+    for (int t = 0; t < T; ++t)
+      acc += bf2f(lp[t * ATT_LD + s]) * bf2f(ldo[t * ATT_LD + d]);
+    dvg[s * ATT_D + d] = f2bf(acc);
+  }
+  // dP[t][s] = dO[t] . V[s]
+  for (int i = threadIdx.x; i < T * S; i += blockDim.x) {
+    const int t = i / S, s = i % S;
+    ldp[t * ATT_LDF + s] =
+        att_dot64(&ldo[t * ATT_LD], &lv[s * ATT_LD]);
+  }
+  __syncthreads();
+  // delta[t] = sum_s dP*P ; then dS (overwrites lp, scaled)
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    float acc = 0.f;
+    for (int s = 0; s < S; ++s)
+      acc += ldp[t * ATT_LDF + s] * bf2f(lp[t * ATT_LD + s]);
+    ldelta[t] = acc;
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < T * S; i += blockDim.x) {
+    const int t = i / S, s = i % S;
+    const float p = bf2f(lp[t * ATT_LD + s]);
+    lp[t * ATT_LD + s] =
+        f2bf(p * (ldp[t * ATT_LDF + s] - ldelta[t]) * scale);
+  }
+  __syncthreads();
+  // dQ[t][d] = sum_s dS[t][s] * K[s][d]
+  ushort_t* dqg = dq + bh * (long)T * ATT_D;
+  for (int i = threadIdx.x; i < T * ATT_D; i += blockDim.x) {
+    const int t = i / ATT_D, d = i % ATT_D;
+    float acc = 0.f;
+    for (int s = 0; s < S; ++s)
+      acc += bf2f(lp[t * ATT_LD + s]) * bf2f(lk[s * ATT_LD + d]);
+    dqg[t * ATT_D + d] = f2bf(acc);
+  }
+  // dK[s][d] = sum_t dS[t][s] * Q[t][d]
+  ushort_t* dkg = dk + bh * (long)S * ATT_D;
+  for (int i = threadIdx.x; i < S * ATT_D; i += blockDim.x) {
+    const int s = i / ATT_D, d = i % ATT_D;
+    float acc = 0.f;
+    for (int t = 0; t < T; ++t)
+      acc += bf2f(lp[t * ATT_LD + s]) * bf2f(lq[t * ATT_LD + d]);
+    dkg[s * ATT_D + d] = f2bf(acc);
   }
 }
 
