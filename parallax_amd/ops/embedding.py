@@ -130,15 +130,15 @@ class _ShardedLookup(torch.autograd.Function):
             out = torch.empty(flat_ids.shape[0], module.embedding_dim,
                               dtype=local_weight.dtype,
                               device=local_weight.device)
-            out[order] = torch.cat(got) if got else out
-        ctx.save_for_backward(order, counts, *recv_ids)
+            out[order] = torch.cat(got)
+        ctx.save_for_backward(order, *recv_ids)
         ctx.module = module
         ctx.counts_l = counts_l
         return out
 
     @staticmethod
     def backward(ctx, grad_out):
-        order, counts, *recv_ids = ctx.saved_tensors
+        order, *recv_ids = ctx.saved_tensors
         module = ctx.module
         comm = module._comm
         sorted_grads = grad_out.contiguous()[order]
