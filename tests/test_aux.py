@@ -135,3 +135,36 @@ def test_redirect_path(monkeypatch, free_port, tmp_path):
     subprocess.run([sys.executable, '-c', code], timeout=60)
     out = open(str(tmp_path / 'worker_3.out')).read()
     assert 'hello-redirect' in out
+
+
+def test_bench_contract_json_line(tmp_path):
+    """The driver depends on bench.py's output contract: one JSON line
+    with the documented keys, run via ``python bench.py`` with env
+    rendezvous — verify it end-to-end as a subprocess (tiny batch)."""
+    import json
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    for k in ('WORLD_SIZE', 'RANK', 'LOCAL_RANK', 'MASTER_ADDR',
+              'MASTER_PORT'):
+        env.pop(k, None)
+    from parallax_amd.lib import get_free_port
+    env['MASTER_PORT'] = str(get_free_port())
+    out = subprocess.run(
+        [sys.executable, 'bench.py', '--batch-size', '4', '--steps', '2',
+         '--warmup', '1', '--no-cuda-graph'],
+        capture_output=True, text=True, timeout=300, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith('{')][-1]
+    d = json.loads(line)
+    for key in ('metric', 'value', 'unit', 'n_gpus', 'steps', 'warmup',
+                'ms_per_step', 'higher_is_better', 'scaling',
+                'vs_baseline', 'dtype', 'data', 'config'):
+        assert key in d, key
+    assert d['metric'] == 'LM1B words/sec'
+    assert d['n_gpus'] == 1 and d['steps'] == 2 and d['warmup'] == 1
+    assert d['scaling'] == 'weak' and d['higher_is_better'] is True
+    assert d['data'] == 'synthetic'
+    assert d['config']['vocab'] == 793470
+    assert d['value'] > 0
