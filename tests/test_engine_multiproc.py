@@ -355,11 +355,12 @@ def _run_lm1b_steps(cfg, batches, seed=0):
     return emb, sw, losses
 
 
-def _body_sharded_tables_equivalence(rank, world):
+def _body_sharded_tables_equivalence(rank, world, run_option='HYBRID',
+                                     average_sparse=False):
     """replicate_variables=False: shard-only storage must produce the SAME
     trained tables as the replicated engine on identical data (the
     reference's PS placement is a storage choice, not a semantics
-    change)."""
+    change).  Checked for both sum and count-average sparse semantics."""
     import torch.distributed as dist
     from parallax_amd.config import ParallaxConfig
 
@@ -368,7 +369,8 @@ def _body_sharded_tables_equivalence(rank, world):
                 torch.randint(0, 300, (4, 3), generator=gen))
                for _ in range(6)]
 
-    cfg_sh = ParallaxConfig(run_option='HYBRID')
+    cfg_sh = ParallaxConfig(run_option=run_option,
+                            average_sparse=average_sparse)
     cfg_sh.communication_config.ps_config.replicate_variables = False
     emb_s, sw_s, losses_s = _run_lm1b_steps(cfg_sh, batches)
     # full_weight() gathers identically on every rank
@@ -378,7 +380,8 @@ def _body_sharded_tables_equivalence(rank, world):
 
     dist.destroy_process_group()
     os.environ['MASTER_PORT'] = str(int(os.environ['MASTER_PORT']) + 1)
-    cfg_rep = ParallaxConfig(run_option='HYBRID')
+    cfg_rep = ParallaxConfig(run_option=run_option,
+                             average_sparse=average_sparse)
     emb_r, sw_r, losses_r = _run_lm1b_steps(cfg_rep, batches)
 
     assert torch.allclose(emb_s, emb_r, atol=1e-4), \
@@ -386,6 +389,11 @@ def _body_sharded_tables_equivalence(rank, world):
     assert torch.allclose(sw_s, sw_r, atol=1e-4)
     for a, b in zip(losses_s, losses_r):
         assert abs(a - b) < 1e-3, (a, b)
+
+
+def _body_sharded_ps_average(rank, world):
+    _body_sharded_tables_equivalence(rank, world, run_option='PS',
+                                     average_sparse=True)
 
 
 def _body_sharded_checkpoint(rank, world):
@@ -470,6 +478,11 @@ def test_sharded_tables_equivalence_world2():
 
 def test_sharded_checkpoint_world2():
     _spawn(_body_sharded_checkpoint, get_free_port(), world=2,
+           timeout=300)
+
+
+def test_sharded_ps_average_world2():
+    _spawn(_body_sharded_ps_average, get_free_port(), world=2,
            timeout=300)
 
 
