@@ -61,16 +61,17 @@ class CheckpointManager(object):
                 and global_step % self.cfg.save_ckpt_steps == 0
                 and global_step > 0):
             due = True
-        if (self.cfg.save_ckpt_secs
-                and time.time() - self._last_save_time
-                >= self.cfg.save_ckpt_secs):
+        if self.cfg.save_ckpt_secs:
+            secs_due = (time.time() - self._last_save_time
+                        >= self.cfg.save_ckpt_secs)
             if sharded:
-                # wall-clock cadence diverges across ranks; in sharded
-                # mode the save is collective, so the chief's decision is
-                # broadcast
-                due = self._bcast_flag(due)
+                # wall-clock cadence diverges across ranks and the save
+                # is collective in sharded mode: EVERY rank must enter
+                # the broadcast every step (a rank-local condition would
+                # desynchronize the collective), the chief decides
+                due = self._bcast_flag(due or secs_due)
             else:
-                due = True
+                due = due or secs_due
         if due:
             self.save(global_step)
         return due
