@@ -375,7 +375,7 @@ struct ZeroOp {
 };
 
 struct ScaleOp {
-  static constexpr bool vec = false;  // single pass over one list; scalar
+  static constexpr bool vec = true;
   const float* norm;                  // device scalar
   float clip;
   template <typename T>
@@ -384,6 +384,16 @@ struct ScaleOp {
     float n = *norm;
     float scale = clip / fmaxf(n, clip);
     t[j] = from_f32<T>(to_f32<T>(t[j]) * scale);
+  }
+  template <typename T>
+  __device__ void apply4(const MTMeta& m, int ti, long j) const {
+    T* t = reinterpret_cast<T*>(m.a[ti]);
+    const float scale = clip / fmaxf(*norm, clip);
+    V4<T> v = ld4(t, j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      v.v[k] = from_f32<T>(to_f32<T>(v.v[k]) * scale);
+    st4(t, j, v);
   }
 };
 
@@ -618,6 +628,36 @@ __global__ void mt_sqsum_kernel(MTMeta meta, float* out) {
   if (threadIdx.x == 0) atomicAdd(out, acc);
 }
 
+// Vectorized variant: 4-element chunks (vcum index space) + scalar tails.
+template <typename T>
+__global__ void mt_sqsum_kernel_v4(MTMeta meta, float* out) {
+  __shared__ float scratch[PX_BLOCK / WAVE];
+  const long vtotal = meta.vcum[meta.n];
+  float acc = 0.f;
+  for (long vidx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+       vidx < vtotal; vidx += (long)gridDim.x * blockDim.x) {
+    const int ti = mt_find_in(meta.vcum, meta.n, vidx);
+    const long j = 4 * (vidx - meta.vcum[ti]);
+    V4<T> v = ld4(reinterpret_cast<const T*>(meta.a[ti]), j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float f = to_f32<T>(v.v[k]);
+      acc += f * f;
+    }
+  }
+  if (blockIdx.x == 0 && threadIdx.x < 4 * (unsigned)meta.n) {
+    const int ti = threadIdx.x >> 2;
+    const long numel = meta.cum[ti + 1] - meta.cum[ti];
+    const long j = (numel & ~3L) + (threadIdx.x & 3);
+    if (j < numel) {
+      float f = to_f32<T>(reinterpret_cast<const T*>(meta.a[ti])[j]);
+      acc += f * f;
+    }
+  }
+  acc = block_reduce_sum(acc, scratch);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
+}
+
 at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors) {
   TORCH_CHECK(!tensors.empty());
   check_contig(tensors, "mt_l2norm");
@@ -628,19 +668,30 @@ at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors) {
     MTMeta meta;
     meta.n = 0;
     meta.cum[0] = 0;
+    meta.vcum[0] = 0;
+    bool aligned = true;
     while (i < tensors.size() && meta.n < MT_MAX) {
       int k = meta.n;
       meta.a[k] = tensors[i].data_ptr();
       meta.cum[k + 1] = meta.cum[k] + tensors[i].numel();
+      meta.vcum[k + 1] = meta.vcum[k] + tensors[i].numel() / 4;
+      if (reinterpret_cast<uintptr_t>(meta.a[k]) & 15) aligned = false;
       meta.n++;
       i++;
     }
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::kBFloat16, at::kHalf, tensors[0].scalar_type(), "mt_l2norm",
         [&] {
-          mt_sqsum_kernel<scalar_t>
-              <<<px_grid(meta.cum[meta.n]), PX_BLOCK, 0, stream>>>(
-                  meta, out.data_ptr<float>());
+          if (aligned) {
+            const long vt = meta.vcum[meta.n];
+            mt_sqsum_kernel_v4<scalar_t>
+                <<<px_grid(vt > 0 ? vt : 1), PX_BLOCK, 0, stream>>>(
+                    meta, out.data_ptr<float>());
+          } else {
+            mt_sqsum_kernel<scalar_t>
+                <<<px_grid(meta.cum[meta.n]), PX_BLOCK, 0, stream>>>(
+                    meta, out.data_ptr<float>());
+          }
           HIP_CHECK_KERNEL();
         });
   }
