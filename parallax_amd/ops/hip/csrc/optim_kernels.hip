@@ -816,6 +816,43 @@ __global__ void mt_unpack_kernel(MTMeta meta, const TSrc* __restrict__ flat,
   }
 }
 
+// Vectorized variants: require every tensor's numel % 4 == 0 (so the
+// flat offsets stay 4-aligned) and 16B-aligned pointers — checked host
+// side; no tails by construction.
+template <typename TSrc, typename TDst>
+__global__ void mt_pack_kernel_v4(MTMeta meta, TDst* __restrict__ flat) {
+  const long vtotal = meta.vcum[meta.n];
+  for (long vidx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+       vidx < vtotal; vidx += (long)gridDim.x * blockDim.x) {
+    const int ti = mt_find_in(meta.vcum, meta.n, vidx);
+    const long j = 4 * (vidx - meta.vcum[ti]);
+    V4<TSrc> v = ld4(reinterpret_cast<const TSrc*>(meta.a[ti]), j);
+    V4<TDst> o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      o.v[k] = from_f32<TDst>(to_f32<TSrc>(v.v[k]));
+    st4(flat, meta.cum[ti] + j, o);
+  }
+}
+
+template <typename TSrc, typename TDst>
+__global__ void mt_unpack_kernel_v4(MTMeta meta,
+                                    const TSrc* __restrict__ flat,
+                                    float scale) {
+  const long vtotal = meta.vcum[meta.n];
+  for (long vidx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+       vidx < vtotal; vidx += (long)gridDim.x * blockDim.x) {
+    const int ti = mt_find_in(meta.vcum, meta.n, vidx);
+    const long j = 4 * (vidx - meta.vcum[ti]);
+    V4<TSrc> v = ld4(flat, meta.cum[ti] + j);
+    V4<TDst> o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      o.v[k] = from_f32<TDst>(to_f32<TSrc>(v.v[k]) * scale);
+    st4(reinterpret_cast<TDst*>(meta.a[ti]), j, o);
+  }
+}
+
 template <bool PACK>
 static void mt_pack_unpack(std::vector<at::Tensor>& tensors,
                            at::Tensor& flat, double scale) {
@@ -827,13 +864,20 @@ static void mt_pack_unpack(std::vector<at::Tensor>& tensors,
     MTMeta meta;
     meta.n = 0;
     meta.cum[0] = 0;
+    bool vec_ok = (base % 4 == 0);
+    meta.vcum[0] = 0;
     while (i < tensors.size() && meta.n < MT_MAX) {
       int k = meta.n;
       meta.a[k] = tensors[i].data_ptr();
       meta.cum[k + 1] = meta.cum[k] + tensors[i].numel();
+      meta.vcum[k + 1] = meta.vcum[k] + tensors[i].numel() / 4;
+      if (tensors[i].numel() % 4 != 0 ||
+          (reinterpret_cast<uintptr_t>(meta.a[k]) & 15))
+        vec_ok = false;
       meta.n++;
       i++;
     }
+    if (reinterpret_cast<uintptr_t>(flat.data_ptr()) & 15) vec_ok = false;
     const long total = meta.cum[meta.n];
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::kBFloat16, at::kHalf, tensors[0].scalar_type(), "mt_pack_t",
@@ -843,11 +887,24 @@ static void mt_pack_unpack(std::vector<at::Tensor>& tensors,
               at::kBFloat16, at::kHalf, flat.scalar_type(), "mt_pack_f",
               [&] {
                 using TF = scalar_t;
-                if (PACK) {
+                const long vt = meta.vcum[meta.n];
+                if (PACK && vec_ok) {
+                  mt_pack_kernel_v4<TP, TF>
+                      <<<px_grid(vt > 0 ? vt : 1), PX_BLOCK, 0, stream>>>(
+                          meta,
+                          reinterpret_cast<TF*>(flat.data_ptr()) + base);
+                } else if (PACK) {
                   mt_pack_kernel<TP, TF>
                       <<<px_grid(total), PX_BLOCK, 0, stream>>>(
                           meta,
                           reinterpret_cast<TF*>(flat.data_ptr()) + base);
+                } else if (vec_ok) {
+                  mt_unpack_kernel_v4<TF, TP>
+                      <<<px_grid(vt > 0 ? vt : 1), PX_BLOCK, 0, stream>>>(
+                          meta,
+                          reinterpret_cast<const TF*>(flat.data_ptr())
+                              + base,
+                          (float)scale);
                 } else {
                   mt_unpack_kernel<TF, TP>
                       <<<px_grid(total), PX_BLOCK, 0, stream>>>(
