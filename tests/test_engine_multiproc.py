@@ -258,6 +258,15 @@ def _body_comm_collectives(rank, world):
     multi = comm.all_gather_v_multi([t, t + 1.0])
     assert torch.allclose(torch.cat(multi[1][0:1])[0],
                           torch.cat(multi[0][0:1])[0] + 1.0)
+    # all_to_all_v with ZERO-length messages (rank 0 sends nothing to
+    # anyone) — the isend/irecv fallback must not post empty sends
+    zsend = [torch.zeros((0 if rank == 0 else 2, 3)) + rank
+             for _ in range(world)]
+    zrecv = comm.all_to_all_v(zsend)
+    for r in range(world):
+        assert zrecv[r].shape[0] == (0 if r == 0 else 2)
+        if zrecv[r].shape[0]:
+            assert torch.all(zrecv[r] == r)
 
 
 def _body_average_sparse(rank, world):
@@ -396,6 +405,37 @@ def _body_sharded_ps_average(rank, world):
                                      average_sparse=True)
 
 
+def _body_sharded_uneven_partitions(rank, world):
+    """to_sharded with a partition count NOT divisible by world (5 parts,
+    world 2 -> rank 0 owns {0,2,4}, rank 1 owns {1,3}): the shard slicing,
+    local indexing and full_weight reassembly must stay exact."""
+    import torch.distributed as dist
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.ops.embedding import PartitionedEmbedding
+    from parallax_amd.partitions import FixedSizePartitioner
+
+    comm = CommContext()
+    torch.manual_seed(7)  # same init everywhere (no broadcast needed)
+    emb = PartitionedEmbedding(103, 8, FixedSizePartitioner(5))
+    ref = emb.weight.detach().clone()
+    emb.to_sharded(comm)
+    owned_rows = sum(emb.sizes[k] for k in emb._owned)
+    assert emb.weight.shape == (owned_rows, 8)
+    # local_index maps a global id of an owned partition to its local row
+    for k in emb._owned:
+        gid = torch.tensor([emb.offsets[k]])
+        lid = emb.local_index(gid)
+        assert torch.equal(emb.weight.detach()[lid], ref[gid])
+    # collective reassembly returns the exact original table
+    full = emb.full_weight()
+    assert torch.equal(full, ref)
+    # a sharded forward fetches the right rows
+    ids = torch.tensor([[0, 50, 102, 17]])
+    out = emb(ids)
+    assert torch.allclose(out[0], ref[ids[0]])
+    dist.barrier()
+
+
 def _body_sharded_checkpoint(rank, world):
     """Sharded-mode checkpoint: collective parts.k save (replicated
     layout on disk), per-rank optimizer sidecars, restore resumes."""
@@ -484,6 +524,11 @@ def test_sharded_checkpoint_world2():
 def test_sharded_ps_average_world2():
     _spawn(_body_sharded_ps_average, get_free_port(), world=2,
            timeout=300)
+
+
+def test_sharded_uneven_partitions_world2():
+    _spawn(_body_sharded_uneven_partitions, get_free_port(), world=2,
+           timeout=180)
 
 
 def test_average_sparse_world2():
