@@ -75,9 +75,9 @@ class CommContext(object):
         hybrid/runner.py:142-149)."""
         if self.world_size == 1:
             return
-        for name in sorted(dict(named_params)):
-            t = dict(named_params)[name]
-            dist.broadcast(t.detach(), src=src)
+        by_name = dict(named_params)
+        for name in sorted(by_name):
+            dist.broadcast(by_name[name].detach(), src=src)
 
     # -- variable-length collectives -----------------------------------------
     def all_gather_v(self, tensor):
