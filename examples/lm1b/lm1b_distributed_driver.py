@@ -36,6 +36,8 @@ def main():
     ap.add_argument('--profile_dir', default=None)
     ap.add_argument('--profile_worker', type=int, default=0)
     ap.add_argument('--cuda_graph', action='store_true', default=True)
+    ap.add_argument('--pure_bf16', action='store_true', default=False,
+                    help='bf16 LSTM body + fused fp32 master weights')
     ap.add_argument('--resource_info', default=os.path.join(
         os.path.dirname(os.path.abspath(__file__)), 'resource_info'))
     args = ap.parse_args()
@@ -52,7 +54,8 @@ def main():
         lambda: lm1b.build_lm1b_graph(batch_size=args.batch_size,
                                       num_steps=args.num_steps,
                                       vocab_size=args.vocab_size,
-                                      lr=args.learning_rate),
+                                      lr=args.learning_rate,
+                                      pure_bf16=args.pure_bf16),
         args.resource_info, parallax_config=config)
     device = sess.comm.device
     if args.cuda_graph and device.type == 'cuda' \
