@@ -266,3 +266,47 @@ def test_checkpoint_roundtrip(monkeypatch, free_port, tmp_path):
         lambda: build_simple_graph(lr=0.05), None, parallax_config=cfg)
     assert sess2.global_step == 5
     sess2.close()
+
+
+def test_checkpoint_pure_bf16_masters_roundtrip(monkeypatch, free_port,
+                                                tmp_path):
+    """Pure-bf16 checkpoint: the fp32 masters ride the optimizer payload
+    and must restore at fp32 (torch's default load would round them to
+    the param dtype)."""
+    _worker_env(monkeypatch, free_port)
+    torch.manual_seed(0)
+    from parallax_amd.models.lm1b import build_lm1b_graph, synthetic_batch
+    ck = CheckPointConfig(ckpt_dir=str(tmp_path / 'ck'), save_ckpt_steps=3)
+    cfg = ParallaxConfig(run_option='HYBRID', ckpt_config=ck)
+    kw = dict(batch_size=2, num_steps=2, vocab_size=100, emb_dim=8,
+              state_size=16, proj_size=8, num_sampled=16,
+              min_partitions=4, pure_bf16=True, ema_decay=None)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: build_lm1b_graph(**kw), None, parallax_config=cfg)
+    for _ in range(3):
+        sess.run(synthetic_batch(2, 2, 100))
+    m = sess.graph.model
+    master = sess.graph.optimizer.state[m.lstm.w]['master'].clone()
+    w_bf16 = m.lstm.w.detach().clone()
+    sess.close()
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+    monkeypatch.setenv('MASTER_PORT', str(free_port + 1))
+    torch.manual_seed(99)
+    sess2, *_ = parallax_amd.parallel_run(
+        lambda: build_lm1b_graph(**kw), None, parallax_config=cfg)
+    assert sess2.global_step == 3
+    m2 = sess2.graph.model
+    assert m2.lstm.w.dtype == torch.bfloat16
+    assert torch.equal(m2.lstm.w.detach(), w_bf16)
+    master2 = sess2.graph.optimizer.state[m2.lstm.w]['master']
+    assert master2.dtype == torch.float32
+    assert torch.equal(master2, master)
+    # the bf16 param is exactly the rounded master after restore + a step
+    out = sess2.run(synthetic_batch(2, 2, 100))
+    assert torch.isfinite(out['loss'])
+    assert torch.equal(m2.lstm.w.detach(),
+                       sess2.graph.optimizer.state[m2.lstm.w]['master']
+                       .to(torch.bfloat16))
+    sess2.close()
