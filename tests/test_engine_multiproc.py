@@ -531,5 +531,72 @@ def test_sharded_uneven_partitions_world2():
            timeout=180)
 
 
+def _body_sharded_save_for_crossmode(rank, world):
+    """Save a sharded checkpoint + the expected full table for the
+    cross-mode restore check in the parent process."""
+    import parallax_amd
+    import torch.distributed as dist
+    from parallax_amd.config import ParallaxConfig, CheckPointConfig
+    from parallax_amd.models import lm1b as M
+    tmp = os.environ['PX_TEST_CKPT_DIR']
+    cfg = ParallaxConfig(
+        run_option='HYBRID',
+        ckpt_config=CheckPointConfig(ckpt_dir=tmp, save_ckpt_steps=2))
+    cfg.communication_config.ps_config.replicate_variables = False
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: M.build_lm1b_graph(
+            batch_size=4, num_steps=3, vocab_size=300, emb_dim=16,
+            state_size=32, proj_size=16, num_sampled=32, min_partitions=4,
+            lr=0.3, ema_decay=None),
+        None, parallax_config=cfg)
+    for _ in range(2):
+        sess.run(M.synthetic_batch(4, 3, 300))
+    full = sess.graph.model.emb.full_weight()
+    if rank == 0:
+        torch.save(full, os.path.join(tmp, 'expected_emb.pt'))
+    sess.close()
+    dist.barrier()
+
+
+def test_sharded_checkpoint_restores_replicated(monkeypatch, tmp_path):
+    """A shard-only run's checkpoint keeps the replicated parts.k layout,
+    so it restores into a plain (replicated, world-1) session."""
+    import parallax_amd
+    from parallax_amd.config import ParallaxConfig, CheckPointConfig
+    from parallax_amd.models import lm1b as M
+    ckdir = str(tmp_path / 'ck')
+    os.environ['PX_TEST_CKPT_DIR'] = ckdir
+    try:
+        _spawn(_body_sharded_save_for_crossmode, get_free_port(), world=2,
+               timeout=300)
+    finally:
+        os.environ.pop('PX_TEST_CKPT_DIR', None)
+    expected = torch.load(os.path.join(ckdir, 'expected_emb.pt'),
+                          weights_only=False)
+
+    monkeypatch.setenv('WORLD_SIZE', '1')
+    monkeypatch.setenv('RANK', '0')
+    monkeypatch.setenv('MASTER_ADDR', '127.0.0.1')
+    monkeypatch.setenv('MASTER_PORT', str(get_free_port()))
+    cfg = ParallaxConfig(
+        run_option='HYBRID',
+        ckpt_config=CheckPointConfig(ckpt_dir=ckdir, save_ckpt_steps=100))
+    torch.manual_seed(99)  # different init; restore must overwrite
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: M.build_lm1b_graph(
+            batch_size=4, num_steps=3, vocab_size=300, emb_dim=16,
+            state_size=32, proj_size=16, num_sampled=32, min_partitions=4,
+            lr=0.3, ema_decay=None),
+        None, parallax_config=cfg)
+    assert sess.global_step == 2
+    assert torch.allclose(sess.graph.model.emb.full_weight(), expected)
+    out = sess.run(M.synthetic_batch(4, 3, 300))
+    assert torch.isfinite(out['loss'])
+    sess.close()
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
 def test_average_sparse_world2():
     _spawn(_body_average_sparse, get_free_port())
