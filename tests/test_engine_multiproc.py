@@ -267,6 +267,11 @@ def _body_comm_collectives(rank, world):
         assert zrecv[r].shape[0] == (0 if r == 0 else 2)
         if zrecv[r].shape[0]:
             assert torch.all(zrecv[r] == r)
+    # all_gather_v where one rank contributes nothing
+    e = torch.ones(0 if rank == 0 else 5, 2)
+    eouts = comm.all_gather_v(e)
+    assert [o.shape[0] for o in eouts] == [0 if r == 0 else 5
+                                           for r in range(world)]
 
 
 def _body_average_sparse(rank, world):

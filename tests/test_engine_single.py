@@ -310,3 +310,24 @@ def test_checkpoint_pure_bf16_masters_roundtrip(monkeypatch, free_port,
                        sess2.graph.optimizer.state[m2.lstm.w]['master']
                        .to(torch.bfloat16))
     sess2.close()
+
+
+def test_dense_engine_overlap_toggle(monkeypatch, free_port):
+    """set_overlap(False/True) swaps hook-driven launches on and off (the
+    hipGraph-capture path turns overlap off because collective launches
+    move after replay)."""
+    _worker_env(monkeypatch, free_port)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: build_simple_graph(), None,
+        parallax_config=ParallaxConfig(run_option='MPI'))
+    eng = sess.engine.dense_engine
+    assert eng is not None
+    # world_size == 1: overlap stays off regardless (nothing to launch)
+    assert eng.overlap is False
+    eng.set_overlap(True)
+    assert eng.overlap is False and eng._hooks == []
+    # bucket bookkeeping is deterministic: every param assigned once
+    n_params = sum(len(b.params) for b in eng._buckets)
+    assert n_params == len(eng.infos)
+    assert set(eng._param_bucket) == {id(gi.param) for gi in eng.infos}
+    sess.close()
