@@ -78,8 +78,12 @@ class CheckpointManager(object):
 
     def _bcast_flag(self, flag):
         import torch.distributed as dist
+        # NCCL/RCCL (all GPU runs) broadcasts device tensors only; gloo
+        # wants CPU.  A CPU flag under nccl raises at dist.broadcast.
+        dev = (torch.device('cuda', torch.cuda.current_device())
+               if dist.get_backend() == 'nccl' else torch.device('cpu'))
         t = torch.tensor([1 if (flag and self.is_chief) else 0],
-                         dtype=torch.long)
+                         dtype=torch.long, device=dev)
         dist.broadcast(t, src=0)
         return bool(t.item())
 

@@ -38,6 +38,16 @@ class PrefetchLoader(object):
             return {k: mv(v) for k, v in batch.items()}
         return mv(batch)
 
+    def _record(self, batch, stream):
+        if isinstance(batch, (tuple, list)):
+            for b in batch:
+                self._record(b, stream)
+        elif isinstance(batch, dict):
+            for b in batch.values():
+                self._record(b, stream)
+        elif isinstance(batch, torch.Tensor) and batch.device.type == 'cuda':
+            batch.record_stream(stream)
+
     def __iter__(self):
         if self.stream is None:
             for batch in self.loader:
@@ -67,6 +77,12 @@ class PrefetchLoader(object):
                     next_event.record(self.stream)
             # compute stream waits for the copy of the CURRENT batch only
             ev.wait(torch.cuda.current_stream())
+            # Tie each yielded tensor's allocation to the consumer stream:
+            # without record_stream the caching allocator may hand the
+            # memory back to the NEXT side-stream prefetch copy while
+            # compute-stream kernels still read this batch (the classic
+            # apex/torchvision prefetcher hazard).
+            self._record(batch, torch.cuda.current_stream())
             yield batch
             if host is None:
                 return

@@ -14,13 +14,27 @@ import torch
 from parallax_amd import ops as _ops
 
 
+def _dtype_groups(tensors):
+    """Group tensors by dtype, preserving order.  The multi-tensor kernels
+    dispatch on tensors[0].scalar_type() and assume a uniform list — a
+    mixed bf16+fp32 list would reinterpret memory."""
+    groups = {}
+    for t in tensors:
+        groups.setdefault(t.dtype, []).append(t)
+    return list(groups.values())
+
+
 def global_norm(tensors):
     """sqrt(sum ||t||^2) over a tensor list (fp32 accumulation)."""
     tensors = [t for t in tensors if t is not None]
     if not tensors:
         return torch.tensor(0.0)
     if _ops.use_native(*tensors):
-        return _ops.ext().multi_tensor_l2norm(list(tensors))
+        sq = None
+        for grp in _dtype_groups(tensors):
+            n = _ops.ext().multi_tensor_l2norm(grp)
+            sq = n * n if sq is None else sq + n * n
+        return sq.sqrt()
     acc = torch.zeros((), dtype=torch.float32, device=tensors[0].device)
     for t in tensors:
         acc = acc + t.detach().float().pow(2).sum()
@@ -33,8 +47,9 @@ def clip_by_global_norm(tensors, clip_norm):
     tensors = [t for t in tensors if t is not None]
     norm = global_norm(tensors)
     if _ops.use_native(*tensors):
-        _ops.ext().multi_tensor_scale_by_norm(list(tensors), norm,
-                                              float(clip_norm))
+        for grp in _dtype_groups(tensors):
+            _ops.ext().multi_tensor_scale_by_norm(grp, norm,
+                                                  float(clip_norm))
         return norm
     scale = clip_norm / torch.clamp(norm, min=clip_norm)
     for t in tensors:

@@ -49,6 +49,8 @@ class DenseAllReduceEngine(object):
         self._hooks = []
         self._buckets = []
         self._param_bucket = {}
+        self._next_launch = 0
+        self._warned_unused = False
         self._build_buckets(bucket_bytes)
         if self.overlap:
             self._register_hooks()
@@ -90,14 +92,51 @@ class DenseAllReduceEngine(object):
         for b in self._buckets:
             b.pending = len(b.params)
             b.work = None
+        self._next_launch = 0
 
     def _on_grad_ready(self, p):
         bi, slot = self._param_bucket[id(p)]
         b = self._buckets[bi]
         b.pending -= 1
-        if b.pending == 0:
+        # Launch strictly in bucket-index order: bucket i fires only once
+        # buckets 0..i-1 are in flight.  This keeps the collective SEQUENCE
+        # identical on every rank even when a rank has a conditionally-
+        # unused parameter and launches the affected bucket late (from
+        # _finalize_launches) — NCCL/RCCL matches collectives by issue
+        # order, so out-of-order launches across ranks would pair wrong
+        # buckets (contract N6, deterministic ordering).
+        while (self._next_launch < len(self._buckets)
+               and self._buckets[self._next_launch].pending == 0):
+            nb = self._buckets[self._next_launch]
+            self._pack(nb)
+            self._launch(nb)
+            self._next_launch += 1
+
+    def _finalize_launches(self):
+        """Launch every bucket not yet in flight, zero-filling grads of
+        parameters that produced none this step (DDP find_unused_parameters
+        semantics: an unused param contributes zeros to the average; the
+        reference TF graph always computes every registered gradient, so
+        this path exists only for eager models with conditional branches)."""
+        if self.comm.world_size == 1:
+            return
+        for bi in range(self._next_launch, len(self._buckets)):
+            b = self._buckets[bi]
+            missing = [p for p in b.params if p.grad is None]
+            if missing:
+                if not self._warned_unused:
+                    import parallax_amd.lib as _lib
+                    _lib.parallax_log.warning(
+                        'dense all-reduce: %d parameter(s) received no '
+                        'gradient this step; contributing zeros '
+                        '(DDP-style unused-parameter handling)',
+                        len(missing))
+                    self._warned_unused = True
+                for p in missing:
+                    p.grad = torch.zeros_like(p)
             self._pack(b)
             self._launch(b)
+        self._next_launch = len(self._buckets)
 
     def _pack(self, b):
         """Copy the bucket's gradients into its flat fusion buffer — ONE
@@ -145,32 +184,18 @@ class DenseAllReduceEngine(object):
         """Pack + launch every bucket's all-reduce (async, comm stream)
         without waiting — lets finish_step overlap the sparse exchange
         with the dense collectives."""
-        if self.comm.world_size == 1 or self.overlap:
-            return  # overlap mode already launched from the hooks
-        for b in self._buckets:
-            if any(p.grad is not None for p in b.params):
-                self._pack(b)
-                self._launch(b)
+        if self.comm.world_size == 1:
+            return
+        # overlap mode already launched complete buckets from the hooks;
+        # either way, launch whatever remains (zero-filling unused params)
+        self._finalize_launches()
 
     def wait_and_unpack(self):
         if self.comm.world_size == 1:
             return
+        self._finalize_launches()
         denom = float(self.comm.world_size) if self.average else 1.0
         for b in self._buckets:
-            if self.overlap and b.pending > 0 \
-                    and any(p.grad is not None for p in b.params):
-                # a dense param produced no grad this step: the bucket
-                # never became ready and collective order would diverge
-                # across ranks — fail loudly (reference TF graphs always
-                # compute every registered gradient; so must step_fn)
-                missing = [i for i, p in enumerate(b.params)
-                           if p.grad is None]
-                raise RuntimeError(
-                    'dense all-reduce bucket incomplete: %d parameter(s) '
-                    'received no gradient this step (every dense '
-                    'parameter must be used in every step, or be marked '
-                    'requires_grad=False / excluded from the optimizer); '
-                    'missing slots %s' % (len(missing), missing[:8]))
             if b.flat is None:
                 continue
             if b.work is not None:

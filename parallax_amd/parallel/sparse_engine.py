@@ -113,38 +113,54 @@ class _SparseEngineBase(object):
             rows = rows.squeeze(1)
         self.optimizer.sparse_row_step(param, ids, rows)
 
+    # How ``average_sparse`` divides: the PS engine count-averages by
+    # per-id occurrence (reference take_indexed_slices_grad
+    # SPARSE_AVERAGE_BY_COUNTER, graph_transform_lib.py:385-390); the MPI
+    # engine divides by WORLD SIZE (Horovod hvd.allreduce(IndexedSlices,
+    # average=True), mpi/graph_transform.py:51-53) — the two reference
+    # paths genuinely differ, so the engines do too.
+    _average_mode = 'count'
+
     def _exchange_and_apply(self, slot, u_ids, u_rows, u_counts):
         """Allgather (ids, rows[, counts]) across ranks, dedup the union
-        (optionally count-averaged) and apply on every rank.  Row payloads
-        travel in ``comm_dtype`` (default bf16 — halves the xGMI bytes; the
-        reduction itself stays fp32 in the segment-sum kernel)."""
+        (optionally averaged, see _average_mode) and apply on every rank.
+        Row payloads travel in ``comm_dtype`` (default bf16 — halves the
+        xGMI bytes; the reduction itself stays fp32 in the segment-sum
+        kernel)."""
+        count_avg = self.average_sparse and self._average_mode == 'count'
         if self.comm.world_size > 1:
             send_rows = u_rows.to(self.comm_dtype) \
                 if self.comm_dtype is not None else u_rows
             tensors = [u_ids, send_rows]
-            if self.average_sparse:
+            if count_avg:
                 tensors.append(u_counts)
             gathered = self.comm.all_gather_v_multi(tensors)
             all_ids = torch.cat(gathered[0])
             all_rows = torch.cat(gathered[1]).to(u_rows.dtype)
-            if self.average_sparse:
+            if count_avg:
                 all_counts = torch.cat(gathered[2])
         else:
             all_ids, all_rows = u_ids, u_rows
-            if self.average_sparse:
+            if count_avg:
                 all_counts = u_counts
         g_ids, g_rows, _ = unique_sum_rows(all_ids, all_rows)
-        if self.average_sparse:
+        if count_avg:
             # total occurrence count per unique id (counts travel with the
             # rows so pre-deduplicated rows keep their multiplicity)
             _, tot, _ = unique_sum_rows(
                 all_ids, all_counts.to(torch.float32).unsqueeze(1))
             g_rows = g_rows / tot.to(g_rows.dtype)
+        elif self.average_sparse:
+            g_rows = g_rows / float(self.comm.world_size)
         self._apply_rows(slot.info.param, g_ids, g_rows)
 
 
 class SparseAllGatherEngine(_SparseEngineBase):
-    """MPI-mode sparse path: raw allgather (Horovod semantics), dedup after."""
+    """MPI-mode sparse path: raw allgather (Horovod semantics), dedup after.
+    ``average_sparse`` divides by world size — hvd.allreduce(IndexedSlices,
+    average=True) semantics (mpi/graph_transform.py:51-53)."""
+
+    _average_mode = 'world'
 
     def reduce_and_apply(self):
         for s in self.slots:

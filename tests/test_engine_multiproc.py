@@ -605,3 +605,104 @@ def test_sharded_checkpoint_restores_replicated(monkeypatch, tmp_path):
 
 def test_average_sparse_world2():
     _spawn(_body_average_sparse, get_free_port())
+
+
+def _body_unused_dense_param(rank, world):
+    """A conditionally-unused dense parameter must not wedge or corrupt
+    the bucketed all-reduce (DDP-style zero-contribution semantics; the
+    reference never hits this — TF graphs compute every registered grad —
+    so this is robustness for eager user models with branches)."""
+    import torch.nn as nn
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.dense_engine import DenseAllReduceEngine
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    torch.manual_seed(321)
+    model = nn.Module()
+    model.a = nn.Linear(4, 4)
+    model.b = nn.Linear(4, 4)   # used only on rank 0 / only on odd steps
+    model.c = nn.Linear(4, 2)
+    infos = classify_params(model)
+    eng = DenseAllReduceEngine(comm, infos, bucket_bytes=64,
+                               overlap=True)
+    for step in range(3):
+        eng.prepare_step()
+        model.zero_grad(set_to_none=True)
+        torch.manual_seed(1000 * step + rank)
+        x = torch.randn(5, 4)
+        h = model.a(x)
+        use_b = (rank == 0) if step == 0 else (step % 2 == 1)
+        if use_b:
+            h = model.b(h)
+        loss = model.c(h).pow(2).sum()
+        loss.backward()
+        eng.reduce_gradients()
+        # every param now has a grad, identical across ranks
+        for p in model.parameters():
+            assert p.grad is not None
+            got = [torch.zeros_like(p.grad) for _ in range(world)]
+            torch.distributed.all_gather(got, p.grad)
+            assert torch.allclose(got[0], got[1], atol=1e-6)
+        # recompute the expected average by hand on every rank
+        ref = nn.Module()
+        ref.a, ref.b, ref.c = (nn.Linear(4, 4), nn.Linear(4, 4),
+                               nn.Linear(4, 2))
+        ref.load_state_dict(model.state_dict())
+        expected = {n: torch.zeros_like(p)
+                    for n, p in ref.named_parameters()}
+        for r in range(world):
+            ref.zero_grad()
+            torch.manual_seed(1000 * step + r)
+            xr = torch.randn(5, 4)
+            hr = ref.a(xr)
+            if (r == 0) if step == 0 else (step % 2 == 1):
+                hr = ref.b(hr)
+            ref.c(hr).pow(2).sum().backward()
+            for n, p in ref.named_parameters():
+                if p.grad is not None:
+                    expected[n] += p.grad / world
+        for (n, p) in model.named_parameters():
+            assert torch.allclose(p.grad, expected[n], atol=1e-5), \
+                (n, (p.grad - expected[n]).abs().max())
+    eng.close()
+
+
+def test_unused_dense_param_world2():
+    _spawn(_body_unused_dense_param, get_free_port())
+
+
+def _body_mpi_average_world(rank, world):
+    """MPI mode + average_sparse: divide by WORLD SIZE (Horovod
+    hvd.allreduce(IndexedSlices, average=True) semantics,
+    mpi/graph_transform.py:51-53) — not by occurrence count."""
+    import torch.nn as nn
+    from parallax_amd import optim
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.sparse_engine import SparseAllGatherEngine
+    from parallax_amd.ops.embedding import ParallaxEmbedding
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    torch.manual_seed(77)
+    model = nn.Module()
+    model.emb = ParallaxEmbedding(20, 4)
+    opt = optim.SGD(model.parameters(), lr=1.0)
+    eng = SparseAllGatherEngine(comm, classify_params(model), opt,
+                                average_sparse=True)
+    eng.prepare_step()
+    # rank 0 uses row 5 three times, rank 1 once -> sum = 4*ones,
+    # Horovod average divides by world (2) -> grad = 2*ones -> delta = -2
+    n = 3 if rank == 0 else 1
+    ids = torch.full((n,), 5, dtype=torch.long)
+    before = model.emb.weight[5].detach().clone()
+    model.emb(ids).sum().backward()
+    eng.reduce_and_apply()
+    after = model.emb.weight[5].detach()
+    assert torch.allclose(after - before, -2.0 * torch.ones(4),
+                          atol=1e-5), (after - before)
+    eng.close()
+
+
+def test_mpi_average_sparse_world2():
+    _spawn(_body_mpi_average_world, get_free_port())
