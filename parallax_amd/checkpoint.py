@@ -49,6 +49,31 @@ class CheckpointManager(object):
             return dist.get_rank(), dist.get_world_size()
         return 0, 1
 
+    def _opt_param_index(self):
+        """param -> flat index in the optimizer state_dict numbering."""
+        params = [p for g in self.optimizer.param_groups
+                  for p in g['params']]
+        return {id(p): i for i, p in enumerate(params)}
+
+    def _sharded_modules(self):
+        return [m for m in self.model.modules()
+                if getattr(m, '_sharded', False)]
+
+    def _shard_layout(self):
+        """Manifest: optimizer-state index -> this rank's shard layout
+        (partition offsets/sizes are global; 'owned' is rank-local)."""
+        idx_of = self._opt_param_index()
+        layout = {}
+        for m in self._sharded_modules():
+            idx = idx_of.get(id(m.weight))
+            if idx is None:
+                continue
+            layout[idx] = {'offsets': list(m.offsets),
+                           'sizes': list(m.sizes),
+                           'owned': list(m._owned),
+                           'rows': m.num_embeddings}
+        return layout
+
     # -- save ----------------------------------------------------------------
     def maybe_save(self, global_step):
         if not self.enabled:
@@ -98,11 +123,14 @@ class CheckpointManager(object):
                     for k, v in self.model.state_dict().items()}
         if sharded and self.optimizer is not None:
             # per-row optimizer state is rank-local: each rank writes its
-            # own sidecar; restore requires the same world size
+            # own sidecar WITH a shard-layout manifest so a restore at a
+            # DIFFERENT world size can reassemble the full per-row state
+            # and re-slice it for the new placement
             side = os.path.join(
                 self.cfg.ckpt_dir, 'model.ckpt-%d.optshard%d-of-%d.pt'
                 % (global_step, rank, world))
-            torch.save(self.optimizer.state_dict(), side + '.tmp')
+            torch.save({'opt': self.optimizer.state_dict(),
+                        'layout': self._shard_layout()}, side + '.tmp')
             os.replace(side + '.tmp', side)
         if not self.is_chief:
             self._last_save_time = time.time()
@@ -153,18 +181,85 @@ class CheckpointManager(object):
             self.optimizer.load_state_dict(payload['optimizer'])
         elif self.optimizer is not None and payload.get('sharded_world'):
             rank, world = self._rank_world()
+            old_world = payload['sharded_world']
             side = path.replace('.pt', '.optshard%d-of-%d.pt'
                                 % (rank, world))
-            if payload['sharded_world'] == world and os.path.isfile(side):
-                self.optimizer.load_state_dict(
-                    torch.load(side, map_location=map_location,
-                               weights_only=False))
-            else:
+            if old_world == world and os.path.isfile(side):
+                sd = torch.load(side, map_location=map_location,
+                                weights_only=False)
+                if isinstance(sd, dict) and 'opt' in sd:
+                    sd = sd['opt']
+                self.optimizer.load_state_dict(sd)
+            elif not self._reshard_restore(path, old_world, map_location):
                 parallax_log.warning(
                     'sharded checkpoint written at world=%s, running at '
                     'world=%d: per-row optimizer state NOT restored '
                     '(weights restored; accumulators start fresh)',
-                    payload['sharded_world'], world)
+                    old_world, world)
         parallax_log.info('restored checkpoint %s (global_step=%d)',
                           path, payload['global_step'])
         return int(payload['global_step'])
+
+    def _reshard_restore(self, path, old_world, map_location):
+        """Rebuild this rank's optimizer state from a sharded checkpoint
+        written at a DIFFERENT world size: assemble the full per-row
+        state from every old rank's sidecar (each carries a layout
+        manifest), then slice it for the CURRENT placement.  Dense /
+        replicated state is identical across ranks, so old rank 0's copy
+        is taken verbatim.  Returns True on success."""
+        import copy
+        sides = [path.replace('.pt', '.optshard%d-of-%d.pt' % (r, old_world))
+                 for r in range(old_world)]
+        if not all(os.path.isfile(s) for s in sides):
+            return False
+        payloads = [torch.load(s, map_location=map_location,
+                               weights_only=False) for s in sides]
+        if not all(isinstance(p, dict) and 'opt' in p and 'layout' in p
+                   for p in payloads):
+            return False  # pre-manifest sidecar format
+        base = copy.deepcopy(payloads[0]['opt'])
+        idx_of = self._opt_param_index()
+        mod_by_idx = {idx_of[id(m.weight)]: m
+                      for m in self._sharded_modules()
+                      if id(m.weight) in idx_of}
+        for idx, meta0 in payloads[0]['layout'].items():
+            offsets, sizes = meta0['offsets'], meta0['sizes']
+            nrows = meta0['rows']
+            # assemble the full [V, ...] tensor per row-shaped state key
+            full = {}
+            for pl in payloads:
+                st_r = pl['opt'].get('state', {}).get(idx)
+                if st_r is None:
+                    continue
+                owned_r = pl['layout'][idx]['owned']
+                local_rows = sum(sizes[k] for k in owned_r)
+                local_off = 0
+                for k in owned_r:
+                    off, size = offsets[k], sizes[k]
+                    for key, t in st_r.items():
+                        if (torch.is_tensor(t) and t.dim() >= 1
+                                and t.shape[0] == local_rows):
+                            dst = full.get(key)
+                            if dst is None:
+                                dst = torch.empty(
+                                    (nrows,) + tuple(t.shape[1:]),
+                                    dtype=t.dtype)
+                                full[key] = dst
+                            dst[off:off + size] = \
+                                t[local_off:local_off + size]
+                    local_off += size
+            mod = mod_by_idx.get(idx)
+            st = base.setdefault('state', {}).setdefault(idx, {})
+            for key, t in full.items():
+                if mod is not None and mod._sharded:
+                    pieces = [t[offsets[k]:offsets[k] + sizes[k]]
+                              for k in mod._owned]
+                    st[key] = (torch.cat(pieces) if pieces else
+                               t.new_zeros((0,) + tuple(t.shape[1:])))
+                else:  # restoring into a replicated/world-1 session
+                    st[key] = t
+        self.optimizer.load_state_dict(base)
+        parallax_log.info(
+            'resharded optimizer sidecars: world %d -> %d',
+            old_world, self._rank_world()[1])
+        return True

@@ -220,16 +220,24 @@ class PartitionedEmbedding(nn.Module):
         return torch.searchsorted(offs, flat_ids, right=True) - 1
 
     # -- shard-only storage (PSConfig.replicate_variables=False) ------------
-    def to_sharded(self, comm):
-        """Drop every row this rank does not own: rank r keeps partitions
-        k % world == r.  Must run AFTER the rank-0 broadcast (all ranks
-        hold the same full table) and BEFORE the first step."""
+    def to_sharded(self, comm, part_to_rank=None):
+        """Drop every row this rank does not own.  ``part_to_rank`` maps
+        partition k -> owning rank; the default is round-robin k % world.
+        The sparse engine passes a byte-size greedy least-loaded placement
+        (reference GreedyLoadBalancingStrategy,
+        ps/between_graph_parallel.py:30-70).  Must run AFTER the rank-0
+        broadcast (all ranks hold the same full table) and BEFORE the
+        first step; the placement must be computed identically on every
+        rank (it is deterministic in registration order)."""
         if self._sharded:
             return
         world, rank = comm.world_size, comm.rank
         dev = self.weight.device
+        if part_to_rank is None:
+            part_to_rank = [k % world for k in range(self.num_partitions)]
+        assert len(part_to_rank) == self.num_partitions
         owned = [k for k in range(self.num_partitions)
-                 if k % world == rank]
+                 if part_to_rank[k] == rank]
         pieces, local_base, local_off = [], [-1] * self.num_partitions, 0
         for k in owned:
             off, size = self.offsets[k], self.sizes[k]
@@ -242,13 +250,16 @@ class PartitionedEmbedding(nn.Module):
         self._sharded = True
         self._comm = comm
         self._owned = owned
+        self._part_to_rank = list(part_to_rank)
+        self._part_rank_t = torch.tensor(part_to_rank, dtype=torch.long,
+                                         device=dev)
         self._local_rows = local.shape[0]
         self._local_base_t = torch.tensor(local_base, dtype=torch.long,
                                           device=dev)
 
     def owner_of(self, flat_ids):
-        """Owning rank for each id (partition k lives on rank k % world)."""
-        return self.shard_of(flat_ids) % self._comm.world_size
+        """Owning rank for each id (partition -> rank placement map)."""
+        return self._part_rank_t[self.shard_of(flat_ids)]
 
     def local_index(self, flat_ids):
         """Global id -> row index in THIS rank's local shard storage
@@ -275,7 +286,7 @@ class PartitionedEmbedding(nn.Module):
         for r in range(world):
             local_off = 0
             for k in range(self.num_partitions):
-                if k % world != r:
+                if self._part_to_rank[k] != r:
                     continue
                 off, size = self.offsets[k], self.sizes[k]
                 full[off:off + size] = gathered[r][local_off:local_off

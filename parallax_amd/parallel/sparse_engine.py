@@ -185,12 +185,29 @@ class SparsePSEngine(_SparseEngineBase):
             # PS task): PartitionedEmbedding tables drop the rows they do
             # not own; lookups fetch rows over an all-to-all
             # (_ShardedLookup) and the grad push happens in its backward,
-            # so reduce_and_apply for these slots is purely local
+            # so reduce_and_apply for these slots is purely local.
+            # Placement = byte-size greedy least-loaded across ALL tables
+            # (reference GreedyLoadBalancingStrategy,
+            # ps/between_graph_parallel.py:30-70), deterministic in slot
+            # registration order so every rank computes the same map.
+            mods, seen = [], set()
             for s in self.slots:
                 mod = getattr(s.info.param, '_parallax_module', None)
-                if mod is not None and hasattr(mod, 'to_sharded'):
-                    mod.to_sharded(comm)
-                    self.sharded_params.add(id(s.info.param))
+                if (mod is not None and hasattr(mod, 'to_sharded')
+                        and id(mod) not in seen):
+                    seen.add(id(mod))
+                    mods.append((mod, s.info.param))
+            loads = [0] * comm.world_size
+            for mod, param in mods:
+                row_bytes = mod.embedding_dim * param.element_size()
+                part_to_rank = []
+                for size in mod.sizes:
+                    r = min(range(comm.world_size),
+                            key=lambda i: (loads[i], i))
+                    part_to_rank.append(r)
+                    loads[r] += size * row_bytes
+                mod.to_sharded(comm, part_to_rank)
+                self.sharded_params.add(id(param))
             if self.sharded_params:
                 parallax_log.info(
                     'replicate_variables=False: %d table(s) sharded '

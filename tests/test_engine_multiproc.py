@@ -558,8 +558,20 @@ def _body_sharded_save_for_crossmode(rank, world):
     for _ in range(2):
         sess.run(M.synthetic_batch(4, 3, 300))
     full = sess.graph.model.emb.full_weight()
+    # expected FULL Adagrad accumulator (each row owned by exactly one
+    # rank; zeros elsewhere + all_reduce SUM assembles it)
+    mod = sess.graph.model.emb
+    st = sess.graph.optimizer.state[mod.weight]['sum']
+    full_sum = torch.zeros(mod.num_embeddings, mod.embedding_dim)
+    local_off = 0
+    for k in mod._owned:
+        off, size = mod.offsets[k], mod.sizes[k]
+        full_sum[off:off + size] = st[local_off:local_off + size]
+        local_off += size
+    dist.all_reduce(full_sum)
     if rank == 0:
         torch.save(full, os.path.join(tmp, 'expected_emb.pt'))
+        torch.save(full_sum, os.path.join(tmp, 'expected_sum.pt'))
     sess.close()
     dist.barrier()
 
@@ -596,6 +608,14 @@ def test_sharded_checkpoint_restores_replicated(monkeypatch, tmp_path):
         None, parallax_config=cfg)
     assert sess.global_step == 2
     assert torch.allclose(sess.graph.model.emb.full_weight(), expected)
+    # per-row optimizer state RESHARDED from the world-2 sidecars into
+    # this replicated world-1 session (layout manifests in the sidecars)
+    expected_sum = torch.load(os.path.join(ckdir, 'expected_sum.pt'),
+                              weights_only=False)
+    st = sess.graph.optimizer.state.get(sess.graph.model.emb.weight)
+    assert st is not None and 'sum' in st, 'optimizer state not resharded'
+    assert st['sum'].shape == expected_sum.shape
+    assert torch.allclose(st['sum'], expected_sum, atol=1e-5)
     out = sess.run(M.synthetic_batch(4, 3, 300))
     assert torch.isfinite(out['loss'])
     sess.close()
