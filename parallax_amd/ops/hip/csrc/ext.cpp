@@ -49,9 +49,10 @@ at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors);
 void multi_tensor_scale_by_norm(std::vector<at::Tensor> tensors,
                                 at::Tensor norm, double clip);
 void sparse_adagrad(at::Tensor param, at::Tensor accum, at::Tensor ids,
-                    at::Tensor grad_rows, double lr, double eps);
+                    at::Tensor grad_rows, double lr, double eps,
+                    c10::optional<at::Tensor> nvalid);
 void sparse_sgd(at::Tensor param, at::Tensor ids, at::Tensor grad_rows,
-                double lr);
+                double lr, c10::optional<at::Tensor> nvalid);
 void multi_tensor_pack(std::vector<at::Tensor> tensors, at::Tensor flat);
 void multi_tensor_unpack(std::vector<at::Tensor> tensors, at::Tensor flat,
                          double scale);
@@ -155,8 +156,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("px_attn_bwd", &px_attn_bwd);
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
   m.def("multi_tensor_scale_by_norm", &multi_tensor_scale_by_norm);
-  m.def("sparse_adagrad", &sparse_adagrad);
-  m.def("sparse_sgd", &sparse_sgd);
+  m.def("sparse_adagrad", &sparse_adagrad, py::arg("param"),
+        py::arg("accum"), py::arg("ids"), py::arg("grad_rows"),
+        py::arg("lr"), py::arg("eps"), py::arg("nvalid") = py::none());
+  m.def("sparse_sgd", &sparse_sgd, py::arg("param"), py::arg("ids"),
+        py::arg("grad_rows"), py::arg("lr"),
+        py::arg("nvalid") = py::none());
   m.def("multi_tensor_pack", &multi_tensor_pack);
   m.def("multi_tensor_unpack", &multi_tensor_unpack);
   m.def("batchnorm_fwd", &batchnorm_fwd);

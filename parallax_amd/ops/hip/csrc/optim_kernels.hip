@@ -708,16 +708,24 @@ void multi_tensor_scale_by_norm(std::vector<at::Tensor> tensors,
 
 // --- sparse row optimizers (SparseApplyAdagrad etc.) ------------------------
 
+// ``nvalid``: optional 0-dim int64 DEVICE scalar bounding the number of
+// real rows — the sync-free padded dedup (ops/sparse.py
+// unique_sum_rows_padded) emits fixed-shape id/row buffers whose tail is
+// (id 0, zero rows); masking here avoids the tail's read-modify-write of
+// row 0 racing a real id-0 update.  Each thread loads it once.
 template <typename T>
 __global__ void sparse_adagrad_kernel(T* __restrict__ p,
                                       float* __restrict__ acc,
                                       const int64_t* __restrict__ ids,
                                       const T* __restrict__ grows, long u,
-                                      long d, float lr, float eps) {
+                                      long d, float lr, float eps,
+                                      const int64_t* __restrict__ nvalid) {
+  const long nv = nvalid ? *nvalid : u;
   const long total = u * d;
   for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
     const long r = idx / d;
+    if (r >= nv) continue;
     const long c = idx - r * d;
     const long row = ids[r];
     float gv = to_f32<T>(grows[r * d + c]);
@@ -729,13 +737,19 @@ __global__ void sparse_adagrad_kernel(T* __restrict__ p,
 }
 
 void sparse_adagrad(at::Tensor param, at::Tensor accum, at::Tensor ids,
-                    at::Tensor grad_rows, double lr, double eps) {
+                    at::Tensor grad_rows, double lr, double eps,
+                    c10::optional<at::Tensor> nvalid) {
   TORCH_CHECK(param.is_cuda() && param.is_contiguous());
   TORCH_CHECK(accum.scalar_type() == at::kFloat);
   const long u = ids.numel();
   if (u == 0) return;
   const long d = param.dim() == 1 ? 1 : param.size(1);
   auto g = grad_rows.contiguous();
+  const int64_t* nv = nullptr;
+  if (nvalid.has_value()) {
+    TORCH_CHECK(nvalid->scalar_type() == at::kLong && nvalid->is_cuda());
+    nv = nvalid->data_ptr<int64_t>();
+  }
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, param.scalar_type(), "sparse_adagrad", [&] {
@@ -744,7 +758,7 @@ void sparse_adagrad(at::Tensor param, at::Tensor accum, at::Tensor ids,
                 reinterpret_cast<scalar_t*>(param.data_ptr()),
                 accum.data_ptr<float>(), ids.data_ptr<int64_t>(),
                 reinterpret_cast<const scalar_t*>(g.data_ptr()), u, d,
-                (float)lr, (float)eps);
+                (float)lr, (float)eps, nv);
         HIP_CHECK_KERNEL();
       });
 }
@@ -753,11 +767,14 @@ template <typename T>
 __global__ void sparse_sgd_kernel(T* __restrict__ p,
                                   const int64_t* __restrict__ ids,
                                   const T* __restrict__ grows, long u, long d,
-                                  float lr) {
+                                  float lr,
+                                  const int64_t* __restrict__ nvalid) {
+  const long nv = nvalid ? *nvalid : u;
   const long total = u * d;
   for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
     const long r = idx / d;
+    if (r >= nv) continue;
     const long c = idx - r * d;
     const long row = ids[r];
     p[row * d + c] = from_f32<T>(to_f32<T>(p[row * d + c]) -
@@ -766,12 +783,17 @@ __global__ void sparse_sgd_kernel(T* __restrict__ p,
 }
 
 void sparse_sgd(at::Tensor param, at::Tensor ids, at::Tensor grad_rows,
-                double lr) {
+                double lr, c10::optional<at::Tensor> nvalid) {
   TORCH_CHECK(param.is_cuda() && param.is_contiguous());
   const long u = ids.numel();
   if (u == 0) return;
   const long d = param.dim() == 1 ? 1 : param.size(1);
   auto g = grad_rows.contiguous();
+  const int64_t* nv = nullptr;
+  if (nvalid.has_value()) {
+    TORCH_CHECK(nvalid->scalar_type() == at::kLong && nvalid->is_cuda());
+    nv = nvalid->data_ptr<int64_t>();
+  }
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, param.scalar_type(), "sparse_sgd", [&] {
@@ -780,7 +802,7 @@ void sparse_sgd(at::Tensor param, at::Tensor ids, at::Tensor grad_rows,
                 reinterpret_cast<scalar_t*>(param.data_ptr()),
                 ids.data_ptr<int64_t>(),
                 reinterpret_cast<const scalar_t*>(g.data_ptr()), u, d,
-                (float)lr);
+                (float)lr, nv);
         HIP_CHECK_KERNEL();
       });
 }

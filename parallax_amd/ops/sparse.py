@@ -83,6 +83,58 @@ def unique_sum_rows(indices, values, average=False, presorted=None):
     return unique_vals, (out.squeeze(1) if squeeze else out), counts
 
 
+def unique_sum_rows_padded(indices, values, average=False, presorted=None,
+                           in_counts=None):
+    """Sync-free fixed-shape dedup: returns (ids[n], rows[n, d],
+    counts[n], nvalid) where n == indices.numel() and only the first
+    ``nvalid`` (a 0-dim DEVICE tensor) entries are real; the tail is
+    (id 0, zero rows, zero counts).  Zero-rows applied to row 0 are
+    no-ops for sum-style row updates, and consumers that cannot tolerate
+    them mask with ``nvalid`` (the native sparse optimizer kernels do).
+
+    Unlike unique_sum_rows there is NO host synchronization (no dynamic
+    unique count), so the whole sparse tail is hipGraph-capturable and
+    costs no ~70 us/table stall (round-1 TODO item 3).
+
+    ``in_counts``: pre-aggregated per-entry multiplicities (float) whose
+    segment SUMS become the output counts (the second dedup of a padded
+    exchange); default = occurrence counting.
+    """
+    assert indices.dim() == 1
+    squeeze = False
+    if values.dim() == 1:
+        values = values.unsqueeze(1)
+        squeeze = True
+    n = indices.numel()
+    dev = indices.device
+    if n == 0:
+        out = values.new_zeros((0, values.shape[1]))
+        counts = torch.zeros(0, dtype=torch.float32, device=dev)
+        nvalid = torch.zeros((), dtype=torch.long, device=dev)
+        return indices, out.squeeze(1) if squeeze else out, counts, nvalid
+    sorted_vals, order, mask = (presorted if presorted is not None
+                                else _sorted_segments_presort(indices))
+    seg_id = torch.cumsum(mask, 0) - 1          # segment of each sorted pos
+    nvalid = seg_id[-1] + 1
+    uniq = torch.zeros(n, dtype=indices.dtype, device=dev)
+    uniq.scatter_(0, seg_id, sorted_vals)
+    gathered = values.index_select(0, order).to(torch.float32)
+    out = torch.zeros((n, values.shape[1]), dtype=torch.float32, device=dev)
+    out.index_add_(0, seg_id, gathered)
+    counts = torch.zeros(n, dtype=torch.float32, device=dev)
+    if in_counts is None:
+        counts.scatter_add_(0, seg_id,
+                            torch.ones(n, dtype=torch.float32, device=dev))
+    else:
+        counts.scatter_add_(0, seg_id,
+                            in_counts.index_select(0, order)
+                            .to(torch.float32))
+    if average:
+        out = out / counts.clamp(min=1.0).unsqueeze(1)
+    out = out.to(values.dtype)
+    return uniq, (out.squeeze(1) if squeeze else out), counts, nvalid
+
+
 def segment_sum_rows(values, order, seg_starts, average=False):
     """Row segment-sum given precomputed sort ``order`` and segment
     boundaries (see _sorted_segments).  out[s] = sum_k values[order[k]]

@@ -53,6 +53,15 @@ class ParallaxOptimizer(Optimizer):
     def supports_sparse(self):
         return True
 
+    def supports_padded_sparse(self):
+        """True when sparse_row_step accepts fixed-shape padded inputs
+        (tail = id 0 / zero rows, bounded by a DEVICE nvalid scalar) —
+        the sync-free sparse path.  Requires either pure sum-style index
+        updates (safe under zero-duplicates) or an nvalid-masked native
+        kernel; Adam/RMSProp's index_copy_ row state is not safe under
+        duplicate tail ids, so they opt out."""
+        return False
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = None
@@ -193,14 +202,20 @@ class SGD(ParallaxOptimizer):
             if w is not p:
                 p.copy_(w)
 
+    def supports_padded_sparse(self):
+        return self.defaults.get('momentum', 0.0) == 0.0
+
     @torch.no_grad()
-    def sparse_row_step(self, param, ids, grad_rows, group=None):
+    def sparse_row_step(self, param, ids, grad_rows, group=None,
+                        nvalid=None):
         group = group or self.group_of(param)
         lr = group['lr']
         if _use_native([param]):
             _ops.ext().sparse_sgd(param, ids.to(torch.int64),
-                                  grad_rows.contiguous(), lr)
+                                  grad_rows.contiguous(), lr,
+                                  nvalid=nvalid)
             return
+        # index_add is a pure sum: padded zero-rows are no-ops, no mask
         param.index_add_(0, ids, grad_rows.to(param.dtype), alpha=-lr)
 
 
@@ -243,15 +258,22 @@ class Adagrad(ParallaxOptimizer):
             if w is not p:
                 p.copy_(w)
 
+    def supports_padded_sparse(self):
+        return True
+
     @torch.no_grad()
-    def sparse_row_step(self, param, ids, grad_rows, group=None):
+    def sparse_row_step(self, param, ids, grad_rows, group=None,
+                        nvalid=None):
         group = group or self.group_of(param)
         lr, eps = group['lr'], group['eps']
         acc = self._accum(param, group)
         if _use_native([param]):
             _ops.ext().sparse_adagrad(param, acc, ids.to(torch.int64),
-                                      grad_rows.contiguous(), lr, eps)
+                                      grad_rows.contiguous(), lr, eps,
+                                      nvalid=nvalid)
             return
+        # sum-style index ops: padded zero-rows add zero everywhere (the
+        # duplicate-id-0 tail is benign without masking on this path)
         g32 = grad_rows.float()
         acc.index_add_(0, ids, g32 * g32)
         denom = acc.index_select(0, ids).sqrt() + eps
@@ -346,9 +368,11 @@ class Adam(ParallaxOptimizer):
                 p.copy_(w)
 
     @torch.no_grad()
-    def sparse_row_step(self, param, ids, grad_rows, group=None):
+    def sparse_row_step(self, param, ids, grad_rows, group=None,
+                        nvalid=None):
         """Lazy-Adam style: moments updated only on touched rows (the TF
         sparse Adam apply)."""
+        assert nvalid is None, 'Adam opts out of padded sparse inputs'
         group = group or self.group_of(param)
         lr, (b1, b2), eps = group['lr'], group['betas'], group['eps']
         st = self._state(param)
@@ -406,7 +430,9 @@ class RMSProp(ParallaxOptimizer):
                 p.copy_(w)
 
     @torch.no_grad()
-    def sparse_row_step(self, param, ids, grad_rows, group=None):
+    def sparse_row_step(self, param, ids, grad_rows, group=None,
+                        nvalid=None):
+        assert nvalid is None, 'RMSProp opts out of padded sparse inputs'
         group = group or self.group_of(param)
         lr, decay = group['lr'], group['decay']
         mom, eps = group['momentum'], group['eps']

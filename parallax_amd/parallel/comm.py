@@ -127,6 +127,28 @@ class CommContext(object):
                              for r in range(self.world_size)])
         return outs_all
 
+    def all_gather_fixed(self, tensor):
+        """Equal-shape allgather, concatenated along dim 0 — the
+        sync-free exchange of the padded sparse path (every rank
+        contributes exactly the same count, so no counts exchange and no
+        padding).  Returns a [world * n0, ...] tensor."""
+        if self.world_size == 1:
+            return tensor
+        t = tensor.contiguous()
+        out = t.new_empty((self.world_size * t.shape[0],)
+                          + tuple(t.shape[1:]))
+        if self.backend == 'nccl':
+            dist.all_gather_into_tensor(out, t)
+        else:
+            chunks = list(out.chunk(self.world_size, dim=0))
+            dist.all_gather(chunks, t)
+            for r, c in enumerate(chunks):  # gloo may return new tensors
+                start = r * t.shape[0]
+                if not c.data_ptr() == out[start:start
+                                           + t.shape[0]].data_ptr():
+                    out[start:start + t.shape[0]] = c
+        return out
+
     def all_to_all_v(self, tensors_per_rank):
         """Send tensors_per_rank[r] to rank r; receive one tensor from each
         rank (contract N7 PS transport).  Works on nccl (all_to_all) and

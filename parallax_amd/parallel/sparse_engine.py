@@ -37,11 +37,15 @@ sparse tensor) so neither autograd accumulation nor the optimizer's own
 sparse path sees it — the engine applies the update itself.
 """
 
+import os
+
 import torch
 import torch.distributed as dist
 
 from parallax_amd.lib import parallax_log
-from parallax_amd.ops.sparse import unique_sum_rows
+from parallax_amd.ops.sparse import (unique_sum_rows,
+                                     unique_sum_rows_padded,
+                                     _sorted_segments_presort)
 
 
 class _SparseGradSlot(object):
@@ -97,21 +101,81 @@ class _SparseEngineBase(object):
         for slot in self.slots:
             h = slot.info.param.register_hook(_capture_hook(slot))
             self._hooks.append(h)
+        # sync-free padded path (fixed shapes, device-side valid count):
+        # removes every per-table host stall and makes the sparse tail
+        # hipGraph-capturable.  GPU + padded-capable optimizer only; at
+        # world>1 additionally requires every rank to contribute the SAME
+        # id count each step (verified ONCE, below).
+        env = os.environ.get('PARALLAX_PADDED_SPARSE', '1')
+        self._padded = ((comm.device.type == 'cuda' or env == 'force')
+                        and env != '0'
+                        and getattr(optimizer, 'supports_padded_sparse',
+                                    lambda: False)())
+        self._counts_static = None  # None = unverified; True/False after
 
     def prepare_step(self):
         for s in self.slots:
             s.ids = None
             s.rows = None
 
+    def capture_safe(self):
+        """True when reduce_and_apply is hipGraph-capturable: the padded
+        sync-free path with no collectives (world 1)."""
+        return self._padded and self.comm.world_size == 1
+
+    def _verify_static_counts(self, live):
+        """One-time check (one sync) that every rank sees identical
+        per-slot id counts — the requirement for the fixed-shape padded
+        exchange.  Models with data-dependent lookup counts (bucketed
+        batching) fall back to the dynamic-size path permanently."""
+        if self._counts_static is not None:
+            return self._counts_static
+        counts = torch.tensor([s.ids.numel() for s in live]
+                              + [len(live)],
+                              dtype=torch.long, device=self.comm.device)
+        lo = counts.clone()
+        hi = counts.clone()
+        dist.all_reduce(lo, op=dist.ReduceOp.MIN)
+        dist.all_reduce(hi, op=dist.ReduceOp.MAX)
+        self._counts_static = bool(torch.equal(lo, hi))
+        if not self._counts_static:
+            parallax_log.info(
+                'sparse id counts differ across ranks: using the '
+                'dynamic-size exchange (padded path disabled)')
+        return self._counts_static
+
     def close(self):
         for h in self._hooks:
             h.remove()
         self._hooks = []
 
-    def _apply_rows(self, param, ids, rows):
+    def _apply_rows(self, param, ids, rows, nvalid=None):
         if param.dim() == 1 and rows.dim() == 2:
             rows = rows.squeeze(1)
-        self.optimizer.sparse_row_step(param, ids, rows)
+        self.optimizer.sparse_row_step(param, ids, rows, nvalid=nvalid)
+
+    def _padded_exchange_and_apply(self, slot, u_ids, u_rows, u_counts,
+                                   count_avg):
+        """Fixed-shape exchange: every rank contributes exactly
+        len(u_ids) entries (tail = id 0 / zero rows), so the allgather
+        needs no counts exchange, no padding to max and no host syncs.
+        ``u_counts`` carries per-id multiplicities (None = raw,
+        multiplicity 1 each)."""
+        send_rows = u_rows.to(self.comm_dtype) \
+            if self.comm_dtype is not None else u_rows
+        all_ids = self.comm.all_gather_fixed(u_ids)
+        all_rows = self.comm.all_gather_fixed(
+            send_rows.contiguous()).to(u_rows.dtype)
+        in_counts = None
+        if count_avg:
+            in_counts = self.comm.all_gather_fixed(
+                u_counts if u_counts is not None
+                else torch.ones_like(u_ids, dtype=torch.float32))
+        g_ids, g_rows, g_counts, nvalid = unique_sum_rows_padded(
+            all_ids, all_rows, average=count_avg, in_counts=in_counts)
+        if self.average_sparse and not count_avg:  # Horovod world-average
+            g_rows = g_rows / float(self.comm.world_size)
+        self._apply_rows(slot.info.param, g_ids, g_rows, nvalid=nvalid)
 
     # How ``average_sparse`` divides: the PS engine count-averages by
     # per-id occurrence (reference take_indexed_slices_grad
@@ -163,9 +227,24 @@ class SparseAllGatherEngine(_SparseEngineBase):
     _average_mode = 'world'
 
     def reduce_and_apply(self):
-        for s in self.slots:
-            if s.ids is None:
-                continue
+        live = [s for s in self.slots if s.ids is not None]
+        if self._padded and (self.comm.world_size == 1
+                             or self._verify_static_counts(live)):
+            for s in live:
+                if self.comm.world_size == 1:
+                    g_ids, g_rows, _, nvalid = unique_sum_rows_padded(
+                        s.ids, s.rows)
+                    if self.average_sparse:
+                        pass  # /world_size == /1
+                    self._apply_rows(s.info.param, g_ids, g_rows,
+                                     nvalid=nvalid)
+                else:
+                    # raw (no local agg — Horovod semantics); the raw
+                    # buffers are already fixed-shape per rank
+                    self._padded_exchange_and_apply(
+                        s, s.ids, s.rows, None, count_avg=False)
+            return
+        for s in live:
             counts = torch.ones_like(s.ids)
             self._exchange_and_apply(s, s.ids, s.rows, counts)
 
@@ -223,15 +302,42 @@ class SparsePSEngine(_SparseEngineBase):
         # lookup's backward; ids are LOCAL rows — dedup locally and apply
         for s in list(live):
             if id(s.info.param) in self.sharded_params:
-                u_ids, u_rows, _ = unique_sum_rows(
-                    s.ids, s.rows, average=self.average_sparse)
-                self._apply_rows(s.info.param, u_ids, u_rows)
+                if self._padded:
+                    u_ids, u_rows, _, nvalid = unique_sum_rows_padded(
+                        s.ids, s.rows, average=self.average_sparse)
+                    self._apply_rows(s.info.param, u_ids, u_rows,
+                                     nvalid=nvalid)
+                else:
+                    u_ids, u_rows, _ = unique_sum_rows(
+                        s.ids, s.rows, average=self.average_sparse)
+                    self._apply_rows(s.info.param, u_ids, u_rows)
                 live.remove(s)
+        # -- sync-free padded path (fixed shapes; hipGraph-capturable at
+        #    world 1; at world>1 one fixed-size allgather, no counts
+        #    exchange, no padding-to-max) -------------------------------
+        if self._padded and self.local_aggregation \
+                and (self.comm.world_size == 1
+                     or self._verify_static_counts(live)):
+            pre = [_sorted_segments_presort(s.ids) for s in live]
+            for s, p in zip(live, pre):
+                if self.comm.world_size == 1:
+                    u_ids, u_rows, _, nvalid = unique_sum_rows_padded(
+                        s.ids, s.rows, average=self.average_sparse,
+                        presorted=p)
+                    self._apply_rows(s.info.param, u_ids, u_rows,
+                                     nvalid=nvalid)
+                else:
+                    u_ids, u_rows, u_counts, _ = unique_sum_rows_padded(
+                        s.ids, s.rows, presorted=p)
+                    self._padded_exchange_and_apply(
+                        s, u_ids, u_rows, u_counts,
+                        count_avg=self.average_sparse)
+            return
+        # -- dynamic-size path ------------------------------------------
         if self.local_aggregation:
             # phase-split across tables: issue every table's sort/compare
             # first, THEN the host-syncing boundary extraction — each
             # sync then overlaps the other tables' queued GPU work
-            from parallax_amd.ops.sparse import _sorted_segments_presort
             pre = [_sorted_segments_presort(s.ids) for s in live]
         for i, s in enumerate(live):
             if self.local_aggregation:

@@ -188,6 +188,12 @@ class ParallaxSession(object):
         self._opt_graph_steps += 1
         if self._opt_graph_steps < 3:
             return False
+        # the sync-free padded sparse tail (sort + fixed-shape dedup +
+        # masked row update, no collectives at world 1) is captured INTO
+        # the optimizer graph: sparse dedup/apply then costs one replay
+        # instead of ~15 eager launches per table per step
+        sp = self.engine.sparse_engine
+        self._sparse_in_opt_graph = self._opt_graph_owns_sparse()
         try:
             # the warm call below EXECUTES this step's real update; the
             # capture that follows only records — so return False this
@@ -195,19 +201,39 @@ class ParallaxSession(object):
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
+                if self._sparse_in_opt_graph:
+                    sp.reduce_and_apply()
                 self.graph.optimizer.step()
             torch.cuda.current_stream().wait_stream(side)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
+                if self._sparse_in_opt_graph:
+                    sp.reduce_and_apply()
                 self.graph.optimizer.step()
             self._opt_graph = g
             self._opt_graph_state = 'just_captured'
-            parallax_log.info('optimizer-step hipGraph captured')
+            parallax_log.info('optimizer-step hipGraph captured%s',
+                              ' (sparse tail included)'
+                              if self._sparse_in_opt_graph else '')
         except Exception as e:
             parallax_log.warning('optimizer graph capture failed (%r); '
                                  'staying eager', e)
             self._opt_graph_state = 'disabled'
+            self._sparse_in_opt_graph = False
         return False
+
+    def _opt_graph_owns_sparse(self):
+        """Will the optimizer graph (its warm/capture call this step, or
+        its replay) run the sparse tail — so the eager call must skip it?"""
+        if self._opt_graph_state == 'disabled':
+            return False
+        sp = self.engine.sparse_engine
+        if sp is None or not sp.capture_safe() \
+                or os.environ.get('PARALLAX_SPARSE_GRAPH', '1') == '0':
+            return False
+        if self._opt_graph_state in ('ready', 'just_captured'):
+            return getattr(self, '_sparse_in_opt_graph', False)
+        return self._opt_graph_steps + 1 >= 3  # this step captures
 
     def _run_step_graphed(self, batch):
         # copy dynamic inputs into the static buffers (state feeds back
@@ -220,7 +246,7 @@ class ParallaxSession(object):
         eng = self.engine
         if eng.dense_engine:
             eng.dense_engine.launch_async()
-        if eng.sparse_engine:
+        if eng.sparse_engine and not self._opt_graph_owns_sparse():
             eng.sparse_engine.reduce_and_apply()
         if eng.dense_engine:
             eng.dense_engine.wait_and_unpack()

@@ -726,3 +726,68 @@ def _body_mpi_average_world(rank, world):
 
 def test_mpi_average_sparse_world2():
     _spawn(_body_mpi_average_world, get_free_port())
+
+
+def _body_padded_exchange_equivalence(rank, world):
+    """World-2 padded (fixed-shape, sync-free) sparse exchange ==
+    dynamic-size exchange, for PS (count-average) and MPI (world-average)
+    engines."""
+    import torch.nn as nn
+    from parallax_amd import optim
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.sparse_engine import (SparsePSEngine,
+                                                     SparseAllGatherEngine)
+    from parallax_amd.ops.embedding import ParallaxEmbedding
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    for cls, avg in [(SparsePSEngine, True), (SparsePSEngine, False),
+                     (SparseAllGatherEngine, True)]:
+        results = {}
+        for mode in ('force', '0'):
+            os.environ['PARALLAX_PADDED_SPARSE'] = mode
+            try:
+                torch.manual_seed(13)  # same init both modes and ranks
+                model = nn.Module()
+                model.emb = ParallaxEmbedding(50, 4)
+                opt = optim.Adagrad(model.parameters(), lr=0.5)
+                eng = cls(comm, classify_params(model), opt,
+                          average_sparse=avg)
+                eng.prepare_step()
+                # same COUNT per rank (static shapes), different ids
+                ids = torch.tensor([1 + rank, 5, 5, 30 + 2 * rank])
+                model.emb(ids).sum().backward()
+                eng.reduce_and_apply()
+                results[mode] = model.emb.weight.detach().clone()
+                if mode == 'force':
+                    assert eng._padded and eng._counts_static is True
+                eng.close()
+            finally:
+                os.environ.pop('PARALLAX_PADDED_SPARSE', None)
+        assert torch.allclose(results['force'], results['0'], atol=1e-5), \
+            (cls.__name__, avg,
+             (results['force'] - results['0']).abs().max())
+    # dynamic fallback: DIFFERENT per-rank counts must be detected once
+    # and the engine must still produce identical tables on both ranks
+    os.environ['PARALLAX_PADDED_SPARSE'] = 'force'
+    try:
+        torch.manual_seed(29)
+        model = nn.Module()
+        model.emb = ParallaxEmbedding(50, 4)
+        opt = optim.Adagrad(model.parameters(), lr=0.5)
+        eng = SparsePSEngine(comm, classify_params(model), opt)
+        eng.prepare_step()
+        ids = torch.tensor([2, 3, 4]) if rank == 0 else torch.tensor([7])
+        model.emb(ids).sum().backward()
+        eng.reduce_and_apply()
+        assert eng._counts_static is False
+        got = [torch.zeros_like(model.emb.weight) for _ in range(world)]
+        torch.distributed.all_gather(got, model.emb.weight.detach())
+        assert torch.allclose(got[0], got[1], atol=1e-6)
+        eng.close()
+    finally:
+        os.environ.pop('PARALLAX_PADDED_SPARSE', None)
+
+
+def test_padded_exchange_world2():
+    _spawn(_body_padded_exchange_equivalence, get_free_port())

@@ -48,6 +48,81 @@ def test_unique_sum_rows_empty():
     assert u.numel() == 0 and s.shape == (0, 4)
 
 
+def test_unique_sum_rows_padded_matches_exact():
+    """The sync-free fixed-shape dedup must agree with unique_sum_rows on
+    the first nvalid entries and emit (id 0, zero rows, zero counts)
+    tails."""
+    from parallax_amd.ops.sparse import unique_sum_rows_padded
+    torch.manual_seed(3)
+    for n, hi in [(1, 1), (7, 3), (64, 10), (200, 500)]:
+        ids = torch.randint(0, hi, (n,))
+        vals = torch.randn(n, 5)
+        u, s, c = unique_sum_rows(ids, vals)
+        pu, ps, pc, nv = unique_sum_rows_padded(ids, vals)
+        nv = int(nv)
+        assert nv == u.numel()
+        assert torch.equal(pu[:nv], u)
+        assert torch.allclose(ps[:nv], s, atol=1e-5)
+        assert torch.equal(pc[:nv].long(), c)
+        assert pu.numel() == n and ps.shape[0] == n
+        assert (pu[nv:] == 0).all()
+        assert (ps[nv:] == 0).all() and (pc[nv:] == 0).all()
+
+
+def test_unique_sum_rows_padded_average_and_counts():
+    from parallax_amd.ops.sparse import unique_sum_rows_padded
+    ids = torch.tensor([0, 0, 5, 5, 5])
+    vals = torch.tensor([[2.0], [4.0], [9.0], [3.0], [0.0]])
+    pu, ps, pc, nv = unique_sum_rows_padded(ids, vals, average=True)
+    assert int(nv) == 2
+    assert ps[:2].squeeze(1).tolist() == [3.0, 4.0]
+    # in_counts: pre-aggregated multiplicities are SUMMED (the second
+    # dedup of a padded exchange)
+    in_c = torch.tensor([3.0, 1.0, 2.0, 2.0, 2.0])
+    pu, ps, pc, nv = unique_sum_rows_padded(ids, vals, average=True,
+                                            in_counts=in_c)
+    assert pc[:2].tolist() == [4.0, 6.0]
+    assert ps[:2].squeeze(1).tolist() == [1.5, 2.0]
+
+
+def test_padded_sparse_world1_engine_equivalence():
+    """PS engine world-1: padded (forced) vs dynamic path produce the
+    same table after a step."""
+    import os as _os
+    import torch.nn as nn
+    from parallax_amd.parallel.sparse_engine import SparsePSEngine
+    from parallax_amd.ops.embedding import ParallaxEmbedding
+    from parallax_amd.tagging import classify_params
+
+    class _FakeComm:
+        world_size = 1
+        rank = 0
+        device = torch.device('cpu')
+        backend = 'gloo'
+
+    results = {}
+    for mode in ('force', '0'):
+        _os.environ['PARALLAX_PADDED_SPARSE'] = mode
+        try:
+            torch.manual_seed(11)
+            model = nn.Module()
+            model.emb = ParallaxEmbedding(40, 6)
+            opt = optim.Adagrad(model.parameters(), lr=0.7)
+            eng = SparsePSEngine(_FakeComm(), classify_params(model), opt,
+                                 average_sparse=True,
+                                 local_aggregation=True)
+            assert eng._padded == (mode == 'force')
+            eng.prepare_step()
+            ids = torch.tensor([0, 3, 3, 0, 17, 3])
+            model.emb(ids).sum().backward()
+            eng.reduce_and_apply()
+            results[mode] = model.emb.weight.detach().clone()
+            eng.close()
+        finally:
+            _os.environ.pop('PARALLAX_PADDED_SPARSE', None)
+    assert torch.allclose(results['force'], results['0'], atol=1e-6)
+
+
 def test_rows_gather():
     w = torch.randn(10, 4)
     ids = torch.tensor([[1, 2], [9, 1]])
