@@ -522,8 +522,14 @@ def test_sharded_tables_equivalence_world2():
 
 
 def test_sharded_checkpoint_world2():
-    _spawn(_body_sharded_checkpoint, get_free_port(), world=2,
-           timeout=300)
+    import shutil
+    import tempfile
+    port = get_free_port()
+    # the body keys its ckpt dir by MASTER_PORT; free ports recycle, so a
+    # stale dir from an earlier test would be restored as a checkpoint
+    stale = os.path.join(tempfile.gettempdir(), 'px_shard_ck_%d' % port)
+    shutil.rmtree(stale, ignore_errors=True)
+    _spawn(_body_sharded_checkpoint, port, world=2, timeout=300)
 
 
 def test_sharded_ps_average_world2():
