@@ -134,6 +134,18 @@ void lstm_gates_fwd_out(const at::Tensor& gates, const at::Tensor& c_prev,
                         double forget_bias, at::Tensor h_out,
                         at::Tensor c_out, at::Tensor i_s, at::Tensor g_t,
                         at::Tensor f_s, at::Tensor o_s);
+// lstm_fused.hip
+void lstm_step_fwd_fused(const at::Tensor& hp, const at::Tensor& whT,
+                         const at::Tensor& gx_t, const at::Tensor& c_prev,
+                         double forget_bias, at::Tensor s_out,
+                         at::Tensor c_out, at::Tensor i_s, at::Tensor g_t,
+                         at::Tensor f_s, at::Tensor o_s);
+void lstm_step_bwd_fused(const at::Tensor& dhp, const at::Tensor& wproj,
+                         const at::Tensor& dc_in, const at::Tensor& c_prev,
+                         const at::Tensor& c_out, const at::Tensor& i_s,
+                         const at::Tensor& g_t, const at::Tensor& f_s,
+                         const at::Tensor& o_s, at::Tensor dgates,
+                         at::Tensor dc_prev);
 void lstm_gates_bwd_out(const at::Tensor& dh, const at::Tensor& dc_in,
                         const at::Tensor& c_prev, const at::Tensor& c_out,
                         const at::Tensor& i_s, const at::Tensor& g_t,
@@ -184,4 +196,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_gates_bwd", &lstm_gates_bwd);
   m.def("lstm_gates_fwd_out", &lstm_gates_fwd_out);
   m.def("lstm_gates_bwd_out", &lstm_gates_bwd_out);
+  m.def("lstm_step_fwd_fused", &lstm_step_fwd_fused,
+        "fused LSTM step: recurrent gates GEMM + gate pointwise (MFMA)");
+  m.def("lstm_step_bwd_fused", &lstm_step_bwd_fused,
+        "fused LSTM step backward: ds GEMM + gate backward (MFMA)");
 }

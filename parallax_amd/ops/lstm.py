@@ -103,25 +103,43 @@ class _LSTMSeqFn(torch.autograd.Function):
                            device=gx.device)
         c_all[0] = c0.to(gx.dtype)
         hp_all[0] = h0.to(gx.dtype)
+        import os as _os
+        # round-2 default: the fused step kernel (gates GEMM + pointwise
+        # in ONE launch, gate-grouped columns so the pointwise is
+        # register-local in the MFMA C fragments) — the [B,4H] gates
+        # tensor never exists in memory and the dependent chain loses a
+        # launch per step (lstm_fused.hip)
+        use_fused = (gx.dtype == torch.bfloat16 and B % 64 == 0
+                     and B <= 128 and H % 32 == 0 and P % 64 == 0
+                     and hasattr(ext, 'lstm_step_fwd_fused')
+                     and _os.environ.get('PARALLAX_FUSED_LSTM',
+                                         '1') != '0')
         # pxgemm beats hipBLASLt 2.4x on the recurrent gates shape in
         # ISOLATION (7.7 vs 18.8 us) but measured ~4%% SLOWER end-to-end
         # inside the captured step (A/B fwd=0/bwd=0 756.7k vs fwd=1/bwd=1
         # 727.4k words/s) — dependent-chain pipelining favors the library
         # kernels here, so the substitution is opt-in
         # (PARALLAX_PX_LSTM_FWD/BWD=1)
-        import os as _os
-        use_px = (gx.dtype == torch.bfloat16 and B % 128 == 0
+        use_px = (not use_fused and gx.dtype == torch.bfloat16
+                  and B % 128 == 0
                   and (4 * H) % 128 == 0 and P % 64 == 0
                   and _os.environ.get('PARALLAX_PX_LSTM_FWD', '0') == '1')
-        w_hT = w_h.t().contiguous() if use_px else None
+        w_hT = w_h.t().contiguous() if (use_px or use_fused) else None
+        ctx.used_fused = use_fused
         for t in range(T):
-            if use_px:
-                gates = ext.pxgemm_bt_add(hp_all[t], w_hT, gx[:, t])
+            if use_fused:
+                ext.lstm_step_fwd_fused(
+                    hp_all[t], w_hT, gx[:, t], c_all[t],
+                    float(forget_bias), s_all[t], c_all[t + 1],
+                    acts[0, t], acts[1, t], acts[2, t], acts[3, t])
             else:
-                gates = torch.addmm(gx[:, t], hp_all[t], w_h)
-            ext.lstm_gates_fwd_out(gates, c_all[t], float(forget_bias),
-                                   s_all[t], c_all[t + 1], acts[0, t],
-                                   acts[1, t], acts[2, t], acts[3, t])
+                if use_px:
+                    gates = ext.pxgemm_bt_add(hp_all[t], w_hT, gx[:, t])
+                else:
+                    gates = torch.addmm(gx[:, t], hp_all[t], w_h)
+                ext.lstm_gates_fwd_out(gates, c_all[t], float(forget_bias),
+                                       s_all[t], c_all[t + 1], acts[0, t],
+                                       acts[1, t], acts[2, t], acts[3, t])
             torch.mm(s_all[t], w_proj, out=hp_all[t + 1])
         ctx.save_for_backward(hp_all, s_all, c_all, acts, w_h, w_proj)
         ctx.fbias = forget_bias
@@ -141,18 +159,25 @@ class _LSTMSeqFn(torch.autograd.Function):
         dc_a = torch.empty(B, H, dtype=w_h.dtype, device=w_h.device)
         dc_b = torch.empty(B, H, dtype=w_h.dtype, device=w_h.device)
         w_h_t = w_h.t().contiguous()
-        w_proj_t = w_proj.t().contiguous()
         if d_cT is None:
             dc_a.zero_()
         else:
             dc_a.copy_(d_cT)
         dc, dc_next = dc_a, dc_b
+        import os as _os
+        # fused bwd step: ds GEMM + gate backward in one launch
+        # (w_proj's natural [H, P] layout IS the transposed operand)
+        use_fused = (getattr(ctx, 'used_fused', False)
+                     and hasattr(ext, 'lstm_step_bwd_fused')
+                     and _os.environ.get('PARALLAX_FUSED_LSTM',
+                                         '1') != '0')
         # ds = dhp @ w_proj^T maps DIRECTLY onto pxgemm_bt (w_proj is
         # [H, P] row-major = the transposed operand): 7.6 vs 19.3 us
-        import os as _os
-        use_px = (w_h.dtype == torch.bfloat16 and B % 128 == 0
+        use_px = (not use_fused and w_h.dtype == torch.bfloat16
+                  and B % 128 == 0
                   and H % 128 == 0 and P % 64 == 0
                   and _os.environ.get('PARALLAX_PX_LSTM_BWD', '0') == '1')
+        w_proj_t = None if use_fused else w_proj.t().contiguous()
         for t in range(T - 1, -1, -1):
             if t == T - 1:
                 dHP[t].copy_(d_outs[:, t])
@@ -162,13 +187,19 @@ class _LSTMSeqFn(torch.autograd.Function):
                 # (hipBLASLt picks a worse kernel on the strided-C path)
                 torch.add(d_outs[:, t].to(w_h.dtype),
                           torch.mm(dG[t + 1], w_h_t), out=dHP[t])
-            if use_px:
-                ds = ext.pxgemm_bt(dHP[t], w_proj)
+            if use_fused:
+                ext.lstm_step_bwd_fused(
+                    dHP[t], w_proj, dc, c_all[t], c_all[t + 1],
+                    acts[0, t], acts[1, t], acts[2, t], acts[3, t],
+                    dG[t], dc_next)
             else:
-                torch.mm(dHP[t], w_proj_t, out=ds)
-            ext.lstm_gates_bwd_out(ds, dc, c_all[t], c_all[t + 1],
-                                   acts[0, t], acts[1, t], acts[2, t],
-                                   acts[3, t], dG[t], dc_next)
+                if use_px:
+                    ds = ext.pxgemm_bt(dHP[t], w_proj)
+                else:
+                    torch.mm(dHP[t], w_proj_t, out=ds)
+                ext.lstm_gates_bwd_out(ds, dc, c_all[t], c_all[t + 1],
+                                       acts[0, t], acts[1, t], acts[2, t],
+                                       acts[3, t], dG[t], dc_next)
             dc, dc_next = dc_next, dc
         # batched weight grads: one GEMM each over all timesteps
         dG_flat = dG.reshape(T * B, 4 * H)

@@ -451,6 +451,61 @@ def test_lstm_gates_fwd_bwd_vs_ref():
     assert torch.allclose(c0.grad.cpu(), c0c.grad, atol=1e-5)
 
 
+def test_lstm_fused_step_matches_unfused():
+    """The fused step kernel (gates GEMM + pointwise in one launch,
+    lstm_fused.hip) must match the unfused bf16 path (addmm +
+    lstm_gates kernels) and the fp32 CPU reference."""
+    import os as _os
+    from parallax_amd.ops.lstm import ParallaxLSTM
+    torch.manual_seed(15)
+    B, T, I, H, P = 128, 5, 64, 128, 64
+    m = ParallaxLSTM(I, H, proj_size=P, dtype=torch.bfloat16).to(DEV)
+    m_cpu = ParallaxLSTM(I, H, proj_size=P)
+    m_cpu.load_state_dict({k: v.float().cpu()
+                           for k, v in m.state_dict().items()})
+    xs = (torch.randn(B, T, I) * 0.5)
+    h0 = torch.randn(B, P) * 0.5
+    c0 = torch.randn(B, H) * 0.5
+    dloss = torch.randn(B, T, P)
+
+    def run(fused):
+        _os.environ['PARALLAX_FUSED_LSTM'] = '1' if fused else '0'
+        try:
+            for p in m.parameters():
+                p.grad = None
+            out, (h, c) = m(xs.bfloat16().to(DEV),
+                            (h0.bfloat16().to(DEV), c0.bfloat16().to(DEV)))
+            out.backward(dloss.bfloat16().to(DEV))
+            return (out.detach().float().cpu(), c.detach().float().cpu(),
+                    {n: p.grad.float().cpu().clone()
+                     for n, p in m.named_parameters() if p.grad is not None})
+        finally:
+            _os.environ.pop('PARALLAX_FUSED_LSTM', None)
+
+    out_f, c_f, g_f = run(True)
+    out_u, c_u, g_u = run(False)
+    assert torch.allclose(out_f, out_u, atol=3e-2, rtol=3e-2), \
+        (out_f - out_u).abs().max()
+    assert torch.allclose(c_f, c_u, atol=3e-2, rtol=3e-2)
+    for n in g_u:
+        assert torch.allclose(g_f[n], g_u[n], atol=0.08, rtol=0.08), \
+            (n, (g_f[n] - g_u[n]).abs().max())
+
+    # fp32 CPU ground truth (loose: bf16 body)
+    for p in m_cpu.parameters():
+        p.grad = None
+    out_c, (h_c, c_c) = m_cpu(xs, (h0, c0))
+    out_c.backward(dloss)
+    assert torch.allclose(out_f, out_c.detach(), atol=0.1, rtol=0.1), \
+        (out_f - out_c.detach()).abs().max()
+    gc = dict(m_cpu.named_parameters())
+    for n in g_f:
+        ref = gc[n].grad
+        denom = ref.abs().max().clamp(min=1.0)
+        assert ((g_f[n] - ref).abs().max() / denom) < 0.12, \
+            (n, ((g_f[n] - ref).abs().max() / denom))
+
+
 def test_sampled_ce_fused_vs_composed():
     """Fused sampled-softmax CE kernel vs the composed fp32 reference
     (add corrections, cat, cross_entropy) — values and all four grads."""
