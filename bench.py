@@ -136,7 +136,7 @@ def main():
                        'global_batch': batch * n, 'seq_len': seq,
                        'num_sampled': M.NUM_SAMPLED,
                        'parallelism': _parallelism_label(args.run_option, n)}
-        dtype = 'bf16' if on_gpu else 'fp32'
+        dtype = 'bf16' if (on_gpu and args.pure_bf16) else 'fp32'
     elif args.model == 'nmt':
         from parallax_amd.models import nmt as M
         batch = args.batch_size or 64
@@ -149,7 +149,7 @@ def main():
         config_json = {'model': 'nmt-transformer', 'global_batch': batch * n,
                        'src_len': 48, 'tgt_len': tgt_len,
                        'parallelism': _parallelism_label(args.run_option, n)}
-        dtype = 'bf16' if on_gpu else 'fp32'
+        dtype = 'bf16' if (on_gpu and args.pure_bf16) else 'fp32'
     else:
         from parallax_amd.models import resnet as M
         batch = args.batch_size or 64

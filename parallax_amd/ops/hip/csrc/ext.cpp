@@ -82,6 +82,11 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
 
 // gemm.hip
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
+at::Tensor pxgemm8_bt(const at::Tensor& a, const at::Tensor& bt);
+at::Tensor pxgemm8_bt_v(const at::Tensor& a, const at::Tensor& bt,
+                        long variant);
+at::Tensor pxgemm8_bt_add(const at::Tensor& a, const at::Tensor& bt,
+                          const c10::optional<at::Tensor>& addend);
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
                          const c10::optional<at::Tensor>& addend);
 at::Tensor pxconv3x3_fwd(const at::Tensor& xp, const at::Tensor& w2d,
@@ -180,6 +185,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("gru_gates_fwd", &gru_gates_fwd);
   m.def("gru_gates_bwd", &gru_gates_bwd);
+  m.def("pxgemm8_bt_v", &pxgemm8_bt_v,
+        "pxgemm8_bt with explicit schedule-variant selection (A/B)");
+  m.def("pxgemm8_bt", &pxgemm8_bt,
+        "bf16 MFMA GEMM C=A@B^T, 256^2 8-phase deep-pipelined schedule");
+  m.def("pxgemm8_bt_add", &pxgemm8_bt_add,
+        "pxgemm8_bt with fused bf16 addend epilogue");
   m.def("pxgemm_bt", &pxgemm_bt,
         "hand-written MFMA bf16 GEMM C = A @ B^T");
   m.def("pxgemm_bt_add", &pxgemm_bt_add,

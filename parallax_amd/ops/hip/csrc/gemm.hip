@@ -185,6 +185,409 @@ void pxgemm_bt_kernel(const ushort* __restrict__ A,
   }
 }
 
+// ---------------------------------------------------------------------------
+// 8-phase 256x256 deep-pipelined MFMA GEMM (guide cdna_hip_programming.md
+// §5 "The 256² 8-phase template"): C[M,N] = A[M,K] @ B[N,K]^T, bf16.
+//
+//   * 256x256 tile, BK=64, 8 waves (2M x 4N), per-wave output 128x64
+//     (acc[8][4] f32x4 = 128 VGPRs);
+//   * LDS 128 KiB: 2 buffers x (A 32 KiB + B 32 KiB), each operand tile
+//     [2 k-half][256 rows][32 k] bf16 with a 16-B-slot XOR swizzle
+//     q ^= (-(row>>2))&3 (verified conflict-free against the
+//     ds_read_b128 4x16 lane groups; involution, so the same XOR is the
+//     glds source permutation — rule 21);
+//   * 4 phases per K-tile: (k-half, n-half) quadrants of 16 MFMAs each;
+//     every phase issues 2 global_load_lds_dwordx4 staging the NEXT
+//     K-tile (each wave stages its own 2 A + 2 B 1-KiB pieces per
+//     k-half), so a full K-tile is always in flight;
+//   * counted s_waitcnt vmcnt(4) only at phases 2 and 4 of each K-tile
+//     (never 0 in the main loop — T4), raw s_barrier (never
+//     __syncthreads: an in-flight glds would be drained — §5
+//     "Pipelining across barriers"), s_setprio(1) around the MFMA
+//     cluster (T5), XCD-aware bijective block remap (T1).
+// ---------------------------------------------------------------------------
+
+#define P8_BM 256
+#define P8_BN 256
+#define P8_THREADS 512
+
+// LDS geometry: buffer b at b*65536; A at +0, B at +32768;
+// k-half kh at +kh*16384; row r occupies 64 B; 16-B slot q at
+// (q ^ ((-(r>>2))&3)) * 16.
+#define P8_LDSA(b) (smem + (b) * 65536)
+#define P8_LDSB(b) (smem + (b) * 65536 + 32768)
+
+template <int V>
+__global__ __launch_bounds__(P8_THREADS, 2)
+void pxgemm8_bt_kernel(const ushort* __restrict__ A,
+                       const ushort* __restrict__ B,
+                       ushort* __restrict__ C, long M, long N, long K,
+                       int nbm, int nbn, const ushort* __restrict__ Cadd,
+                       long add_rstride) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  int bid = blockIdx.x;
+  {
+    const int nwg = nbm * nbn;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, pos = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const long bm = (long)(bid / nbn) * P8_BM;
+  const long bn = (long)(bid % nbn) * P8_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;        // 0..7
+  const int wm = (wave >> 2) * 128; // 2 M groups
+  const int wn = (wave & 3) * 64;   // 4 N groups
+  const int fr = lane & 15;         // fragment row within 16
+  const int fq = lane >> 4;         // fragment 16-B slot (k 8-group)
+
+  f32x4 acc[8][4] = {};
+
+  // ---- staging constants (per lane) -----------------------------------
+  // Each wave stages, per k-half, its two 1-KiB pieces of A (LDS rows
+  // pi*16..pi*16+15, pi in {wave, wave+8}) and the same two of B.
+  // lane covers LDS row pi*16 + (lane>>2), slot (lane&3); the glds
+  // source column carries the inverse (== same) swizzle.
+  const int srow = lane >> 2;                       // 0..15 within piece
+  const int sq = (lane & 3) ^ ((-(lane >> 4)) & 3); // swizzled src slot
+  // global rows for piece 0 (pi=wave) and piece 1 (pi=wave+8):
+  long arow0 = bm + wave * 16 + srow;
+  long arow1 = arow0 + 128;
+  long brow0 = bn + wave * 16 + srow;
+  long brow1 = brow0 + 128;
+  if (arow0 >= M) arow0 = M - 1;  // clamped edge reads; stale fragments
+  if (arow1 >= M) arow1 = M - 1;  // are discarded by the guarded stores
+  if (brow0 >= N) brow0 = N - 1;
+  if (brow1 >= N) brow1 = N - 1;
+  // bake per-lane base pointers once: the per-stage address is then a
+  // single add of the wave-uniform k offset (no 64-bit mult in the loop)
+  const ushort* aptr0 = A + arow0 * K + sq * 8;
+  const ushort* aptr1 = A + arow1 * K + sq * 8;
+  const ushort* bptr0 = B + brow0 * K + sq * 8;
+  const ushort* bptr1 = B + brow1 * K + sq * 8;
+  // LDS byte destinations (lane-linear within the piece):
+  const int a_dst0 = wave * 1024 + lane * 16;          // + kh*16384
+  const int a_dst1 = (wave + 8) * 1024 + lane * 16;
+  const int ktiles = (int)(K / BK);
+
+  // one piece: operand base, global row, k element offset, LDS dest
+#define P8_STAGE(gptr, kelem, buf, lds_off, bsel)                          \
+  __builtin_amdgcn_global_load_lds(                                        \
+      (const __attribute__((address_space(1))) void*)((gptr) + (kelem)),   \
+      (__attribute__((address_space(3))) void*)(                           \
+          (bsel ? P8_LDSB(buf) : P8_LDSA(buf)) + (lds_off)),               \
+      16, 0, 0)
+
+  // stage phase ph (0..3) of K-tile kt into buffer (kt&1)
+#define P8_STAGE_PHASE(kt, ph)                                             \
+  do {                                                                     \
+    const int _b = (kt) & 1;                                               \
+    const long _kb = (long)(kt) * BK + ((ph) >> 1) * 32;                   \
+    const int _kh_off = ((ph) >> 1) * 16384;                               \
+    if (((ph) & 1) == 0) {                                                 \
+      P8_STAGE(aptr0, _kb, _b, _kh_off + a_dst0, 0);                       \
+      P8_STAGE(bptr0, _kb, _b, _kh_off + a_dst0, 1);                       \
+    } else {                                                               \
+      P8_STAGE(aptr1, _kb, _b, _kh_off + a_dst1, 0);                       \
+      P8_STAGE(bptr1, _kb, _b, _kh_off + a_dst1, 1);                       \
+    }                                                                      \
+  } while (0)
+
+  // ---- fragment read addressing ---------------------------------------
+  // a_frag[i]: LDS row wm + i*16 + fr of k-half kh, slot fq ^ f(row);
+  // f(row) = (-(row>>2))&3 and (wm+i*16)>>2 is 0 mod 4, so the XOR is
+  // the per-lane constant (-(fr>>2))&3.
+  const int rswz = ((-(fr >> 2)) & 3) ^ fq;  // swizzled slot for reads
+
+#define P8_AFRAG(kh, i)                                                    \
+  (*(const bf16x8*)(P8_LDSA(cur) + (kh) * 16384 +                          \
+                    (wm + (i) * 16 + fr) * 64 + rswz * 16))
+#define P8_BFRAG(kh, nh, jj)                                               \
+  (*(const bf16x8*)(P8_LDSB(cur) + (kh) * 16384 +                          \
+                    (wn + (nh) * 32 + (jj) * 16 + fr) * 64 + rswz * 16))
+
+  // prologue: stage all of K-tile 0, publish its first k-half
+  P8_STAGE_PHASE(0, 0);
+  P8_STAGE_PHASE(0, 1);
+  P8_STAGE_PHASE(0, 2);
+  P8_STAGE_PHASE(0, 3);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int cur = 0;
+  // ---- main loop: one iteration = one K-tile = 4 phases ---------------
+  // phase (kh, nh): ds_read frags -> stage next tile's phase -> [vmcnt]
+  // -> barrier -> MFMA quadrant. vmcnt(4) at phases 1 and 3 publishes
+  // the k-half the NEXT phase pair will read.
+#define P8_PHASE(kh, nh, DO_STAGE, kt, WAITN)                              \
+  do {                                                                     \
+    bf16x8 b_frag[2];                                                      \
+    if ((nh) == 0) {                                                       \
+      _Pragma("unroll") for (int i = 0; i < 8; ++i)                        \
+          a_frag[i] = P8_AFRAG(kh, i);                                     \
+    }                                                                      \
+    b_frag[0] = P8_BFRAG(kh, nh, 0);                                       \
+    b_frag[1] = P8_BFRAG(kh, nh, 1);                                       \
+    if (DO_STAGE) P8_STAGE_PHASE((kt) + 1, (kh) * 2 + (nh));               \
+    if ((WAITN) == 0) {                                                    \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                     \
+    } else if ((WAITN) > 0) {                                              \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                     \
+    }                                                                      \
+    __builtin_amdgcn_s_barrier();                                          \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                     \
+    __builtin_amdgcn_sched_barrier(0);                                     \
+    __builtin_amdgcn_s_setprio(1);                                         \
+    _Pragma("unroll") for (int i = 0; i < 8; ++i) {                        \
+      acc[i][(nh) * 2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(          \
+          a_frag[i], b_frag[0], acc[i][(nh) * 2], 0, 0, 0);                \
+      acc[i][(nh) * 2 + 1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(      \
+          a_frag[i], b_frag[1], acc[i][(nh) * 2 + 1], 0, 0, 0);            \
+    }                                                                      \
+    __builtin_amdgcn_s_setprio(0);                                         \
+    __builtin_amdgcn_s_barrier();                                          \
+  } while (0)
+
+  // merged variant: one barrier pair per k-half (12 ds_reads + 4 glds +
+  // 32 MFMAs per phase, 4 barriers per K-tile instead of 8)
+#define P8_PHASE_M(kh, DO_STAGE, kt, WAITN)                                \
+  do {                                                                     \
+    bf16x8 b_frag[4];                                                      \
+    _Pragma("unroll") for (int i = 0; i < 8; ++i)                          \
+        a_frag[i] = P8_AFRAG(kh, i);                                       \
+    b_frag[0] = P8_BFRAG(kh, 0, 0);                                        \
+    b_frag[1] = P8_BFRAG(kh, 0, 1);                                        \
+    b_frag[2] = P8_BFRAG(kh, 1, 0);                                        \
+    b_frag[3] = P8_BFRAG(kh, 1, 1);                                        \
+    if (DO_STAGE) {                                                        \
+      P8_STAGE_PHASE((kt) + 1, (kh) * 2);                                  \
+      P8_STAGE_PHASE((kt) + 1, (kh) * 2 + 1);                              \
+    }                                                                      \
+    if ((WAITN) == 0) {                                                    \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                     \
+    } else if ((WAITN) > 0) {                                              \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                     \
+    }                                                                      \
+    __builtin_amdgcn_s_barrier();                                          \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                     \
+    __builtin_amdgcn_sched_barrier(0);                                     \
+    __builtin_amdgcn_s_setprio(1);                                         \
+    _Pragma("unroll") for (int i = 0; i < 8; ++i) {                        \
+      _Pragma("unroll") for (int j = 0; j < 4; ++j) {                      \
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(               \
+            a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);                     \
+      }                                                                    \
+    }                                                                      \
+    __builtin_amdgcn_s_setprio(0);                                         \
+    __builtin_amdgcn_s_barrier();                                          \
+  } while (0)
+
+  bf16x8 a_frag[8];
+  if (V == 2) {
+    // ---- software-pipelined fragment reads (one phase ahead) --------
+    // Phase quadrant = (k-half, m-half): B fragments are read once per
+    // k-half (4 regs per set), A alternates between two 4-frag banks
+    // with a static period-2 pattern, so every phase's MFMA window
+    // contains only the 16 MFMAs plus the issue of the NEXT phase's
+    // ds_reads (their latency hides under the MFMAs).  The
+    // barrier-to-barrier critical path is the MFMA cluster itself.
+    bf16x8 a0[4], a1[4], b0[4], b1[4];
+#define P8_RD_A4(bank, kh, ilo)                                            \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
+        bank[i] = P8_AFRAG(kh, (ilo) + i)
+#define P8_RD_B4(bank, kh)                                                 \
+    _Pragma("unroll") for (int j = 0; j < 4; ++j)                          \
+        bank[j] = P8_BFRAG(kh, (j) >> 1, (j) & 1)
+#define P8_MM4(abank, bbank, mh)                                           \
+    __builtin_amdgcn_s_setprio(1);                                         \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+      _Pragma("unroll") for (int j = 0; j < 4; ++j) {                      \
+        acc[(mh) * 4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+            abank[i], bbank[j], acc[(mh) * 4 + i][j], 0, 0, 0);            \
+      }                                                                    \
+    }                                                                      \
+    __builtin_amdgcn_s_setprio(0)
+
+    // prologue reads: tile-0 k-half-0 first A bank + B set
+    P8_RD_A4(a0, 0, 0);
+    P8_RD_B4(b0, 0);
+    for (int kt = 0; kt < ktiles; ++kt) {
+      const bool last = (kt == ktiles - 1);
+      // P1 (kh0, mh0): consume a0,b0; read A(kh0, i4..7) -> a1
+      if (!last) P8_STAGE_PHASE(kt + 1, 0);
+      __builtin_amdgcn_s_barrier();
+      P8_RD_A4(a1, 0, 4);
+      P8_MM4(a0, b0, 0);
+      __builtin_amdgcn_s_barrier();
+      // P2 (kh0, mh1): publish kh1; read A(kh1, i0..3) -> a0, B(kh1) -> b1
+      if (!last) {
+        P8_STAGE_PHASE(kt + 1, 1);
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      P8_RD_A4(a0, 1, 0);
+      P8_RD_B4(b1, 1);
+      P8_MM4(a1, b0, 1);
+      __builtin_amdgcn_s_barrier();
+      // P3 (kh1, mh0): consume a0,b1; read A(kh1, i4..7) -> a1
+      if (!last) P8_STAGE_PHASE(kt + 1, 2);
+      __builtin_amdgcn_s_barrier();
+      P8_RD_A4(a1, 1, 4);
+      P8_MM4(a0, b1, 0);
+      __builtin_amdgcn_s_barrier();
+      // P4 (kh1, mh1): publish next tile kh0; read its first A bank + B
+      if (!last) {
+        P8_STAGE_PHASE(kt + 1, 3);
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      cur ^= 1;   // reads below target the next tile's buffer
+      if (!last) {
+        P8_RD_A4(a0, 0, 0);
+        P8_RD_B4(b0, 0);
+      }
+      P8_MM4(a1, b1, 1);
+      __builtin_amdgcn_s_barrier();
+    }
+#undef P8_RD_A4
+#undef P8_RD_B4
+#undef P8_MM4
+  } else if (V == 1) {
+    for (int kt = 0; kt < ktiles - 1; ++kt) {
+      P8_PHASE_M(0, true, kt, 4);
+      P8_PHASE_M(1, true, kt, 4);
+      cur ^= 1;
+    }
+    const int kt = ktiles - 1;
+    P8_PHASE_M(0, false, kt, 0);
+    P8_PHASE_M(1, false, kt, -1);
+  } else {
+    for (int kt = 0; kt < ktiles - 1; ++kt) {
+      P8_PHASE(0, 0, true, kt, -1);
+      P8_PHASE(0, 1, true, kt, 4);   // publish next pair's k-half
+      P8_PHASE(1, 0, true, kt, -1);
+      P8_PHASE(1, 1, true, kt, 4);   // publish next K-tile's k-half 0
+      cur ^= 1;
+    }
+    // final K-tile: nothing left to stage; drain what remains
+    const int kt = ktiles - 1;
+    P8_PHASE(0, 0, false, kt, -1);
+    P8_PHASE(0, 1, false, kt, 0);  // k-half 1 fully landed
+    P8_PHASE(1, 0, false, kt, -1);
+    P8_PHASE(1, 1, false, kt, -1);
+  }
+#undef P8_PHASE
+#undef P8_PHASE_M
+#undef P8_STAGE_PHASE
+#undef P8_STAGE
+#undef P8_AFRAG
+#undef P8_BFRAG
+
+  // ---- epilogue --------------------------------------------------------
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = bm + wm + i * 16 + crow0 + r;
+        const long col = bn + wn + j * 16 + fr;
+        if (row >= M || col >= N) continue;
+        float fv = acc[i][j][r];
+        if (Cadd != nullptr) {
+          __hip_bfloat16 av = *reinterpret_cast<const __hip_bfloat16*>(
+              &Cadd[row * add_rstride + col]);
+          fv += __bfloat162float(av);
+        }
+        __hip_bfloat16 v = __float2bfloat16(fv);
+        C[row * N + col] = *reinterpret_cast<ushort*>(&v);
+      }
+    }
+  }
+}
+
+at::Tensor pxgemm8_bt_addv(const at::Tensor& a, const at::Tensor& bt,
+                           const c10::optional<at::Tensor>& addend,
+                           long variant) {
+  TORCH_CHECK(a.is_cuda() && bt.is_cuda(), "pxgemm8_bt: GPU only");
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 &&
+              bt.scalar_type() == at::kBFloat16, "pxgemm8_bt: bf16 only");
+  TORCH_CHECK(a.dim() == 2 && bt.dim() == 2 && a.is_contiguous() &&
+              bt.is_contiguous());
+  const long M = a.size(0), K = a.size(1), N = bt.size(0);
+  TORCH_CHECK(bt.size(1) == K, "pxgemm8_bt: K mismatch");
+  TORCH_CHECK(K % BK == 0, "pxgemm8_bt needs K % 64 == 0");
+  TORCH_CHECK(M >= 1 && N >= 16, "pxgemm8_bt: degenerate shape");
+  auto c = at::empty({M, N}, a.options());
+  const int nbm = (int)((M + P8_BM - 1) / P8_BM);
+  const int nbn = (int)((N + P8_BN - 1) / P8_BN);
+  const int smem = 128 * 1024;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&pxgemm8_bt_kernel<0>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, smem);
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&pxgemm8_bt_kernel<1>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, smem);
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&pxgemm8_bt_kernel<2>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, smem);
+  const ushort* add_ptr = nullptr;
+  long add_rs = 0;
+  if (addend.has_value()) {
+    const auto& ad = addend.value();
+    TORCH_CHECK(ad.scalar_type() == at::kBFloat16 && ad.dim() == 2 &&
+                ad.size(0) == M && ad.size(1) == N && ad.stride(1) == 1,
+                "pxgemm8_bt: bad addend");
+    add_ptr = reinterpret_cast<const ushort*>(ad.data_ptr());
+    add_rs = ad.stride(0);
+  }
+#define P8_LAUNCH(VV)                                                     \
+  pxgemm8_bt_kernel<VV><<<nbm * nbn, P8_THREADS, smem, stream>>>(          \
+      reinterpret_cast<const ushort*>(a.data_ptr()),                       \
+      reinterpret_cast<const ushort*>(bt.data_ptr()),                      \
+      reinterpret_cast<ushort*>(c.data_ptr()), M, N, K, nbm, nbn,          \
+      add_ptr, add_rs)
+  if (variant == 1) {
+    P8_LAUNCH(1);
+  } else if (variant == 2) {
+    P8_LAUNCH(2);
+  } else {
+    P8_LAUNCH(0);
+  }
+#undef P8_LAUNCH
+  HIP_CHECK_KERNEL();
+  return c;
+}
+
+static long p8_default_variant() {
+  static long v = [] {
+    const char* e = getenv("PARALLAX_GEMM8_VARIANT");
+    return e ? atol(e) : 0L;
+  }();
+  return v;
+}
+
+at::Tensor pxgemm8_bt_add(const at::Tensor& a, const at::Tensor& bt,
+                          const c10::optional<at::Tensor>& addend) {
+  return pxgemm8_bt_addv(a, bt, addend, p8_default_variant());
+}
+
+at::Tensor pxgemm8_bt(const at::Tensor& a, const at::Tensor& bt) {
+  return pxgemm8_bt_addv(a, bt, c10::nullopt, p8_default_variant());
+}
+
+at::Tensor pxgemm8_bt_v(const at::Tensor& a, const at::Tensor& bt,
+                        long variant) {
+  return pxgemm8_bt_addv(a, bt, c10::nullopt, variant);
+}
+
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
                          const c10::optional<at::Tensor>& addend) {
   TORCH_CHECK(a.is_cuda() && bt.is_cuda(), "pxgemm_bt: GPU only");

@@ -29,9 +29,16 @@ def check(M, N, K):
     ci = e.pxgemm_bt(ai, b)
     ref_i = ai.float() @ b.float().t()
     ok_t = torch.allclose(ci.float(), ref_i, atol=1e-2)
-    print('M=%5d N=%5d K=%5d  max_abs_err=%.4f rel=%.5f identityA_ok=%s'
-          % (M, N, K, err, rel, ok_t), flush=True)
-    return rel < 0.02 and ok_t
+    rel8 = 0.0
+    ok8 = True
+    if hasattr(e, 'pxgemm8_bt'):
+        c8 = e.pxgemm8_bt(a, b)
+        rel8 = float((c8.float() - ref).abs().max() /
+                     ref.abs().max().clamp_min(1e-6))
+        ok8 = torch.allclose(e.pxgemm8_bt(ai, b).float(), ref_i, atol=1e-2)
+    print('M=%5d N=%5d K=%5d  2ph rel=%.5f idA=%s   8ph rel=%.5f idA=%s'
+          % (M, N, K, rel, ok_t, rel8, ok8), flush=True)
+    return rel < 0.02 and ok_t and rel8 < 0.02 and ok8
 
 
 def bench(M, N, K, iters=30):
@@ -49,11 +56,13 @@ def bench(M, N, K, iters=30):
         return (time.perf_counter() - t0) / iters
 
     t_ours = t(lambda: e.pxgemm_bt(a, b))
+    t_p8 = t(lambda: e.pxgemm8_bt(a, b)) if hasattr(e, 'pxgemm8_bt') else 0
     t_blas = t(lambda: a @ b.t())
     fl = 2.0 * M * N * K
-    print('M=%5d N=%5d K=%5d  ours %7.1f us (%6.1f TF)   '
-          'hipblaslt %7.1f us (%6.1f TF)'
+    print('M=%5d N=%5d K=%5d  2ph %7.1f us (%6.1f TF)  8ph %7.1f us '
+          '(%6.1f TF)  hipblaslt %7.1f us (%6.1f TF)'
           % (M, N, K, t_ours * 1e6, fl / t_ours / 1e12,
+             t_p8 * 1e6, fl / max(t_p8, 1e-12) / 1e12,
              t_blas * 1e6, fl / t_blas / 1e12), flush=True)
 
 

@@ -604,6 +604,37 @@ def test_pxgemm_bt_vs_matmul():
                               atol=1e-2)
 
 
+def test_pxgemm8_bt_vs_matmul():
+    """8-phase 256^2 deep-pipelined MFMA GEMM vs hipBLASLt, incl. the
+    transpose-detecting identity-A check (guide G9) and edge tiles."""
+    e = _ext()
+    if not hasattr(e, 'pxgemm8_bt'):
+        pytest.skip('pxgemm8 not built')
+    torch.manual_seed(52)
+    for (M, N, K) in ((256, 256, 64), (512, 512, 256), (300, 700, 128),
+                      (1024, 2048, 512), (2048, 1024, 1024)):
+        a = torch.rand(M, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
+        b = torch.rand(N, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
+        c = e.pxgemm8_bt(a, b)
+        ref = a.float() @ b.float().t()
+        rel = (c.float() - ref).abs().max() / ref.abs().max()
+        assert float(rel) < 0.02, (M, N, K, float(rel))
+        ai = torch.zeros(M, K, device=DEV, dtype=torch.bfloat16)
+        d = min(M, K)
+        ai[torch.arange(d), torch.arange(d)] = 1.0
+        ci = e.pxgemm8_bt(ai, b)
+        assert torch.allclose(ci.float(), ai.float() @ b.float().t(),
+                              atol=1e-2)
+    # fused addend epilogue
+    M, N, K = 512, 512, 128
+    a = torch.rand(M, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
+    b = torch.rand(N, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
+    add = torch.rand(M, N, device=DEV, dtype=torch.bfloat16) * 2 - 1
+    c = e.pxgemm8_bt_add(a, b, add)
+    ref = a.float() @ b.float().t() + add.float()
+    assert float((c.float() - ref).abs().max()) < 0.25
+
+
 def test_conv1x1_nhwc_vs_torch():
     """MFMA 1x1 NHWC conv fwd/bwd vs nn.functional.conv2d."""
     from parallax_amd.ops.conv import conv1x1_nhwc
