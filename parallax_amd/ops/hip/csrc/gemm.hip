@@ -217,6 +217,40 @@ void pxgemm_bt_kernel(const ushort* __restrict__ A,
 #define P8_LDSA(b) (smem + (b) * 65536)
 #define P8_LDSB(b) (smem + (b) * 65536 + 32768)
 
+// Two-level XCD-aware super-tile block remap.  The plain per-XCD stripe
+// remap (T1) makes the ~256 concurrent blocks an 8-row x nbn stripe:
+// at 8192^3 that is 128 MB of distinct B panels live at once, nothing
+// fits in L2 or L3, and the kernel devolves to streaming 8 GB from HBM
+// (measured: 1.24 ms = bytes / 6.3 TB/s exactly).  Here each XCD owns
+// ONE n-column at a time (its B panel lives in its private L2) and the
+// concurrent super-tile is SM m-rows x 8 n-cols, whose A panels are
+// shared across XCDs through L3.  Requires nbn % 8 == 0; callers fall
+// back to the stripe remap otherwise.
+__device__ __forceinline__ void p8_supertile_remap(
+    int bid, int nbm, int nbn, int* bm_out, int* bn_out) {
+  const int SM = 16;                  // super-tile height (m-tiles)
+  const int xcd = bid % 8;
+  const int pos = bid / 8;            // sequential index within the XCD
+  const int ncol = nbn / 8;           // n-columns owned by each XCD
+  const int full_sr = nbm / SM;
+  const int per_sr = SM * ncol;
+  int sr, c, m_in;
+  if (pos < full_sr * per_sr) {
+    sr = pos / per_sr;
+    const int w = pos - sr * per_sr;
+    c = w / SM;
+    m_in = w - c * SM;
+  } else {                            // ragged last super-row
+    const int rem_m = nbm - full_sr * SM;
+    const int w = pos - full_sr * per_sr;
+    c = w / rem_m;
+    m_in = w - c * rem_m;
+    sr = full_sr;
+  }
+  *bm_out = sr * SM + m_in;
+  *bn_out = c * 8 + xcd;
+}
+
 template <int V>
 __global__ __launch_bounds__(P8_THREADS, 2)
 void pxgemm8_bt_kernel(const ushort* __restrict__ A,
@@ -226,15 +260,20 @@ void pxgemm8_bt_kernel(const ushort* __restrict__ A,
                        long add_rstride) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
-  int bid = blockIdx.x;
-  {
+  int bmt, bnt;
+  if (nbn % 8 == 0 && nbm * nbn >= 64) {
+    p8_supertile_remap(blockIdx.x, nbm, nbn, &bmt, &bnt);
+  } else {
+    int bid = blockIdx.x;
     const int nwg = nbm * nbn;
     const int q = nwg / 8, r = nwg % 8;
     const int xcd = bid % 8, pos = bid / 8;
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    bmt = bid / nbn;
+    bnt = bid % nbn;
   }
-  const long bm = (long)(bid / nbn) * P8_BM;
-  const long bn = (long)(bid % nbn) * P8_BN;
+  const long bm = (long)bmt * P8_BM;
+  const long bn = (long)bnt * P8_BN;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -569,7 +608,7 @@ at::Tensor pxgemm8_bt_addv(const at::Tensor& a, const at::Tensor& bt,
 static long p8_default_variant() {
   static long v = [] {
     const char* e = getenv("PARALLAX_GEMM8_VARIANT");
-    return e ? atol(e) : 0L;
+    return e ? atol(e) : 2L;  // measured best (ab_gemm8_v3.log)
   }();
   return v;
 }
