@@ -139,6 +139,50 @@ class NMTTransformer(nn.Module):
                                       num_classes=self.tgt_vocab)
         return losses.mean()
 
+    @torch.no_grad()
+    def encode(self, src):
+        scale = math.sqrt(self.d_model)
+        body = self.pos.dtype
+        e = self.src_emb(src).to(body) * scale + self.pos[:src.shape[1]]
+        for blk in self.enc:
+            e = blk(e)
+        return e
+
+    @torch.no_grad()
+    def greedy_decode(self, src, sos_id=1, eos_id=2, max_len=64):
+        """Inference-time greedy decoding over the FULL tied softmax
+        (sampled softmax is train-only; eval scores every class —
+        reference nmt/model.py infer path / evaluation_utils BLEU).
+        Returns a list of id-lists (without SOS, truncated at EOS)."""
+        mem = self.encode(src)
+        B = src.shape[0]
+        W = self.tgt_emb.full_weight().to(mem.dtype)   # [V, D]
+        ys = torch.full((B, 1), sos_id, dtype=torch.long,
+                        device=src.device)
+        done = torch.zeros(B, dtype=torch.bool, device=src.device)
+        scale = math.sqrt(self.d_model)
+        for _ in range(max_len):
+            d = self.tgt_emb(ys).to(mem.dtype) * scale \
+                + self.pos[:ys.shape[1]]
+            for blk in self.dec:
+                d = blk(d, mem=mem, causal=True)
+            d = self.ln_f(d)[:, -1]                    # [B, D]
+            nxt = (d @ W.t()).argmax(dim=-1)           # [B]
+            nxt = torch.where(done, torch.full_like(nxt, eos_id), nxt)
+            ys = torch.cat([ys, nxt[:, None]], dim=1)
+            done |= nxt == eos_id
+            if bool(done.all()):
+                break
+        outs = []
+        for row in ys[:, 1:].tolist():
+            hyp = []
+            for t in row:
+                if t == eos_id:
+                    break
+                hyp.append(t)
+            outs.append(hyp)
+        return outs
+
 
 class GNMTAttention(nn.Module):
     """Reference-architecture seq2seq: LSTM encoder, LSTM decoder with

@@ -1,0 +1,146 @@
+"""NMT example pipeline tests (reference examples/nmt test surface:
+vocab/BPE handling, bucketed iterator semantics, BLEU, greedy decode —
+model_test.py + iterator_utils/vocab_utils/evaluation_utils)."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, os.path.join(REPO, 'examples', 'nmt'))
+
+from utils import evaluation_utils, iterator_utils, vocab_utils  # noqa: E402
+
+
+def test_vocab_roundtrip(tmp_path):
+    v = vocab_utils.Vocab(['hello', 'world', 'foo'])
+    assert v.lookup('<unk>') == 0
+    assert v.lookup('<s>') == 1
+    assert v.lookup('</s>') == 2
+    assert v.lookup('hello') == 3
+    assert v.lookup('MISSING') == vocab_utils.UNK_ID
+    p = tmp_path / 'vocab.txt'
+    v.save(str(p))
+    v2 = vocab_utils.Vocab.load(str(p))
+    assert v2.itos == v.itos
+    ids = v.encode('hello world zzz'.split())
+    assert ids == [3, 4, 0]
+    assert v.decode(ids + [vocab_utils.EOS_ID, 5]) == \
+        ['hello', 'world', '<unk>']
+
+
+def test_build_vocab_freq_order(tmp_path):
+    p = tmp_path / 'c.txt'
+    p.write_text('a a a b b c\n' 'a b\n')
+    v = vocab_utils.build_vocab([str(p)], max_size=5)
+    # 3 specials + 2 most frequent
+    assert len(v) == 5
+    assert v.itos[3:] == ['a', 'b']
+
+
+def test_bpe_learn_apply_restore(tmp_path):
+    p = tmp_path / 'c.txt'
+    p.write_text('lower lower lowest newer newer new\n' * 3)
+    merges = vocab_utils.learn_bpe([str(p)], 40)
+    toks = vocab_utils.apply_bpe('lower newest unseen'.split(), merges)
+    # every token splits into >= 1 piece; continuation marked with @@
+    assert vocab_utils.restore_bpe(toks) == ['lower', 'newest', 'unseen']
+    # frequent whole words merge back into single symbols
+    toks_l = vocab_utils.apply_bpe(['lower'], merges)
+    assert vocab_utils.restore_bpe(toks_l) == ['lower']
+
+
+def _toy_examples(n=64, seed=0):
+    import random
+    rng = random.Random(seed)
+    exs = []
+    for _ in range(n):
+        sl = rng.randint(1, 20)
+        tl = rng.randint(1, 20)
+        exs.append(([rng.randint(3, 50) for _ in range(sl)],
+                    [rng.randint(3, 50) for _ in range(tl)]))
+    return exs
+
+
+def test_bucketed_batches_semantics():
+    exs = _toy_examples(200)
+    seen = []
+    for batch in iterator_utils.bucketed_batches(exs, 8, bucket_width=4,
+                                                 shuffle=True, seed=1):
+        src, tgt_in, tgt_out, src_len, tgt_len = batch
+        bs = src.shape[0]
+        assert bs <= 8
+        # bucketing: src lengths within one batch differ < 2*bucket_width
+        lens = src_len.tolist()
+        if bs == 8:   # full buckets are width-homogeneous
+            assert max(lens) // 4 == min(lens) // 4
+        # teacher forcing: tgt_in starts with SOS, tgt_out ends with EOS
+        assert (tgt_in[:, 0] == vocab_utils.SOS_ID).all()
+        for i in range(bs):
+            t = int(tgt_len[i])
+            assert int(tgt_out[i, t - 1]) == vocab_utils.EOS_ID
+            assert torch.equal(tgt_in[i, 1:t], tgt_out[i, :t - 1])
+        seen.extend((tuple(src[i, :src_len[i]].tolist()))
+                    for i in range(bs))
+    assert len(seen) == 200   # nothing dropped
+
+
+def test_bucketed_batches_sharding_disjoint_and_complete():
+    exs = _toy_examples(100)
+    got = []
+    for shard in range(4):
+        for batch in iterator_utils.bucketed_batches(
+                exs, 8, shuffle=False, shard_id=shard, num_shards=4):
+            src, _ti, _to, sl, _tl = batch
+            got.extend(tuple(src[i, :sl[i]].tolist())
+                       for i in range(src.shape[0]))
+    assert len(got) == 100   # i % num_shards partition covers everything
+
+
+def test_make_examples_filters():
+    sv = vocab_utils.Vocab(['a', 'b'])
+    tv = vocab_utils.Vocab(['x'])
+    exs = iterator_utils.make_examples(
+        ['a b', '', 'a a a a'], ['x', 'x', 'x'], sv, tv, src_max_len=3)
+    assert len(exs) == 1
+    assert exs[0][0] == [3, 4]
+
+
+def test_corpus_bleu_known_values():
+    ref = ['the cat sat on the mat'.split()]
+    assert evaluation_utils.corpus_bleu(ref, ref) == pytest.approx(1.0)
+    hyp = ['the cat sat on mat'.split()]
+    b = evaluation_utils.corpus_bleu(ref, hyp)
+    assert 0.0 < b < 1.0
+    # brevity penalty: shorter hypothesis scores below a same-precision
+    # longer one
+    assert evaluation_utils.corpus_bleu(ref, [['the']], smooth=True) < 0.2
+    assert evaluation_utils.corpus_bleu(ref, [['zzz'] * 6]) == 0.0
+
+
+def test_greedy_decode_shapes_and_eos():
+    from parallax_amd.models.nmt import NMTTransformer
+    torch.manual_seed(0)
+    m = NMTTransformer(src_vocab=64, tgt_vocab=64, d_model=32, heads=2,
+                       ffn=64, enc_layers=1, dec_layers=1,
+                       min_partitions=2, max_len=32)
+    m.eval()
+    src = torch.randint(3, 64, (3, 7))
+    outs = m.greedy_decode(src, max_len=9)
+    assert len(outs) == 3
+    for row in outs:
+        assert len(row) <= 9
+        assert all(0 <= t < 64 for t in row)
+        assert vocab_utils.EOS_ID not in row   # truncated at EOS
+
+
+def test_eval_driver_smoke():
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, 'examples', 'nmt',
+                                      'nmt_eval.py')],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert 'BLEU=' in out.stdout
