@@ -8,6 +8,17 @@ over xGMI.  The public surface mirrors the reference
 ``get_partitioner``, ``shard``, ``log`` and the config-class tree.
 """
 
+import os as _os
+
+# Ship the exhaustively-tuned MIOpen user DB (miopen_udb/, produced by
+# MIOPEN_FIND_ENFORCE=4 on an MI355X: ResNet-50 4.9k -> 5.36k img/s) with
+# the tree, and point MIOpen at it before its first use.  An explicit
+# MIOPEN_USER_DB_PATH in the environment wins.
+_udb = _os.path.join(_os.path.dirname(_os.path.dirname(
+    _os.path.abspath(__file__))), 'miopen_udb')
+if _os.path.isdir(_udb):
+    _os.environ.setdefault('MIOPEN_USER_DB_PATH', _udb)
+
 from parallax_amd.partitions import get_partitioner
 from parallax_amd.runner import parallel_run
 from parallax_amd import shard
