@@ -25,9 +25,9 @@ _LAYERS = {50: (3, 4, 6, 3), 101: (3, 4, 23, 3), 152: (3, 8, 36, 3)}
 
 
 def _bn(c, bn_momentum, relu_after=False):
-    """nn.BatchNorm2d, or the hand-written NHWC BN with the following
-    ReLU fused when PARALLAX_NATIVE_BN=1 (ops/batchnorm.py)."""
-    if _os.environ.get('PARALLAX_NATIVE_BN') == '1':
+    """Hand-written NHWC BN with the following ReLU fused (default;
+    ops/batchnorm.py), or nn.BatchNorm2d with PARALLAX_NATIVE_BN=0."""
+    if _bn_fused():
         from parallax_amd.ops.batchnorm import ParallaxBatchNorm2d
         return ParallaxBatchNorm2d(c, momentum=bn_momentum,
                                    relu=relu_after)
@@ -35,7 +35,10 @@ def _bn(c, bn_momentum, relu_after=False):
 
 
 def _bn_fused():
-    return _os.environ.get('PARALLAX_NATIVE_BN') == '1'
+    # default ON: the fused NHWC BN(+ReLU)(+residual add) kernels remove
+    # the MIOpen BN + elementwise glue that measured ~50% of the tuned
+    # ResNet step (gpurun_out/prof_rn50, round 2)
+    return _os.environ.get('PARALLAX_NATIVE_BN', '1') == '1'
 
 
 class Bottleneck(nn.Module):
@@ -50,7 +53,10 @@ class Bottleneck(nn.Module):
                                bias=False)
         self.bn2 = _bn(width, bn_momentum, relu_after=True)
         self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
-        self.bn3 = _bn(cout, bn_momentum)
+        # fused mode: bn3 also folds the shortcut add + final ReLU
+        self.bn3 = _bn(cout, bn_momentum, relu_after=True) \
+            if _bn_fused() else _bn(cout, bn_momentum)
+        self.fused_bn3 = _bn_fused()
         self.relu = nn.ReLU(inplace=True)
         # with fused BN+ReLU the separate activation after bn1/bn2 is a
         # no-op; keep it as Identity to preserve the module structure
@@ -66,6 +72,8 @@ class Bottleneck(nn.Module):
         identity = x if self.downsample is None else self.downsample(x)
         out = self.act(self.bn1(self.conv1(x)))
         out = self.act(self.bn2(self.conv2(out)))
+        if self.fused_bn3:
+            return self.bn3(self.conv3(out), identity)
         out = self.bn3(self.conv3(out))
         return self.relu(out + identity)
 
@@ -109,7 +117,9 @@ class ResNet(nn.Module):
         self.version = version
         layers = _LAYERS[depth]
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64, momentum=bn_momentum)
+        self.bn1 = _bn(64, bn_momentum, relu_after=True) \
+            if version == 1 else nn.BatchNorm2d(64, momentum=bn_momentum)
+        self.stem_fused = _bn_fused() and version == 1
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         block = Bottleneck if version == 1 else BottleneckV2
@@ -128,7 +138,10 @@ class ResNet(nn.Module):
         self.fc = nn.Linear(512 * block.expansion, num_classes)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.bn1(self.conv1(x))
+        if not self.stem_fused:
+            x = self.relu(x)
+        x = self.maxpool(x)
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         if self.version == 2:
             x = self.relu(self.bn_final(x))
@@ -155,9 +168,11 @@ def build_resnet_graph(depth=50, num_classes=1001, batch_size=64, lr=0.1,
         model = model.to(memory_format=torch.channels_last)
     if pure_bf16:
         import torch.nn as _nn
+        from parallax_amd.ops.batchnorm import ParallaxBatchNorm2d
         model = model.to(torch.bfloat16)
         for mod in model.modules():
-            if isinstance(mod, _nn.modules.batchnorm._BatchNorm):
+            if isinstance(mod, (_nn.modules.batchnorm._BatchNorm,
+                                ParallaxBatchNorm2d)):
                 mod.float()
     optimizer = optim.SGD(model.parameters(), lr=lr, momentum=momentum,
                           weight_decay=weight_decay)

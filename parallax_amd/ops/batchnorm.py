@@ -18,22 +18,31 @@ from parallax_amd import ops as _ops
 
 
 class _BNFn(torch.autograd.Function):
+    """Fused BN(+residual add)(+ReLU).  The saved OUTPUT y provides the
+    ReLU mask in backward (with the add fused the mask depends on the
+    post-add value); the identity branch's gradient dy' = dy * (y > 0)
+    comes out of the same dx kernel pass."""
+
     @staticmethod
     def forward(ctx, x2d, weight, bias, run_mean, run_var, momentum, eps,
-                relu):
+                relu, addend):
         y, mean, invstd = _ops.ext().batchnorm_fwd(
-            x2d, weight, bias, run_mean, run_var, momentum, eps, relu)
-        ctx.save_for_backward(x2d, mean, invstd, weight, bias)
+            x2d, weight, bias, run_mean, run_var, momentum, eps, relu,
+            addend)
+        ctx.save_for_backward(x2d, y, mean, invstd, weight)
         ctx.relu = relu
+        ctx.has_add = addend is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x2d, mean, invstd, weight, bias = ctx.saved_tensors
-        dx, dw, db = _ops.ext().batchnorm_bwd(dy, x2d, mean, invstd,
-                                              weight, bias, ctx.relu)
-        return (dx, dw.to(weight.dtype), db.to(bias.dtype), None, None,
-                None, None, None)
+        x2d, y, mean, invstd, weight = ctx.saved_tensors
+        out = _ops.ext().batchnorm_bwd(dy, x2d, y, mean, invstd, weight,
+                                       ctx.relu, ctx.has_add)
+        dx, dw, db = out[0], out[1], out[2]
+        dxid = out[3] if ctx.has_add else None
+        return (dx, dw.to(weight.dtype), db.to(weight.dtype), None, None,
+                None, None, None, dxid)
 
 
 class ParallaxBatchNorm2d(nn.Module):
@@ -66,19 +75,28 @@ class ParallaxBatchNorm2d(nn.Module):
     def _native_ok(self, x):
         c = self.num_features
         return (x.is_cuda and _ops.native_available() and self.training
-                and (c & (c - 1)) == 0 and 64 <= c <= 2048)
+                and x.dtype in (torch.bfloat16, torch.float16)
+                and c % 8 == 0 and c <= 2048)
 
-    def forward(self, x):
+    def forward(self, x, addend=None):
+        """addend: optional residual tensor fused as
+        y = [relu](bn(x) + addend) — the bottleneck's bn3 + shortcut."""
         N, C, H, W = x.shape
         if self._native_ok(x):
             self.num_batches_tracked += 1
             x2d = (x.permute(0, 2, 3, 1).reshape(N * H * W, C)
                    .contiguous())
+            a2d = None
+            if addend is not None:
+                a2d = (addend.permute(0, 2, 3, 1).reshape(N * H * W, C)
+                       .to(x.dtype).contiguous())
             y = _BNFn.apply(x2d, self.weight, self.bias,
                             self.running_mean, self.running_var,
-                            self.momentum, self.eps, self.relu)
+                            self.momentum, self.eps, self.relu, a2d)
             return y.view(N, H, W, C).permute(0, 3, 1, 2)
         y = torch.nn.functional.batch_norm(
             x, self.running_mean, self.running_var, self.weight,
             self.bias, self.training, self.momentum, self.eps)
+        if addend is not None:
+            y = y + addend
         return torch.relu(y) if self.relu else y

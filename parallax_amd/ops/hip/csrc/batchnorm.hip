@@ -1,15 +1,28 @@
-// NHWC BatchNorm2d training fwd/bwd with optional fused ReLU
-// (SURVEY §2.4 worklist: "BatchNorm fwd/bwd (decay .999) + ReLU").
+// NHWC BatchNorm2d training fwd/bwd with fused ReLU and fused residual
+// add (SURVEY §2.4 worklist: "BatchNorm fwd/bwd (decay .999) + ReLU";
+// reference convnet_builder.py:428-501 conv+BN+ReLU chains and the
+// ResNet bottleneck's bn3 + shortcut-add + ReLU).
 //
-// Layout: channels_last activations viewed [M = N*H*W, C] with C
-// contiguous; C is a power of two (ResNet: 64..2048).
-// Stats reduction: each 256-thread block sweeps a row slab; a thread owns
-// channels {tid mod C} (C <= 256) or {tid, tid+256, ...} (C > 256), so
-// per-element accumulation is PRIVATE registers; one global atomicAdd per
-// (block, owned channel) merges the partials — no LDS traffic, no
-// per-element atomics.  Backward reuses the structure for the two
-// per-channel sums; the fused-ReLU mask is recomputed from
-// (xhat*gamma+beta > 0), so nothing extra is saved.
+// These are pure HBM-streaming kernels; the design rules are guide G13
+// (bf16 loads vectorized 16 B/lane — scalar bf16 is ~2x slower) and a
+// FIXED channel-group-per-thread mapping so per-channel accumulators
+// and scale/shift constants live in registers, never LDS:
+//
+//   * activations are channels_last views [M = N*H*W, C], C % 8 == 0,
+//     C <= 2048;
+//   * a thread owns ONE group of 8 consecutive channels; threads_per_row
+//     = C/8; a 256-thread block covers floor(256/tpr) rows per sweep and
+//     grid-strides over rows — every load is a 16-byte short8;
+//   * stats reduce: per-thread fp32 acc[8] -> LDS combine -> one global
+//     atomicAdd per (block, channel);
+//   * finalize precomputes scale = gamma*invstd and shift = beta -
+//     mean*scale, so the normalize pass is one fma per element (+ the
+//     optional residual addend read and ReLU) — y = relu(x*scale + shift
+//     [+ id]);
+//   * backward reads the SAVED OUTPUT y for the ReLU mask (with the add
+//     fused, the mask depends on the post-add value, which cannot be
+//     recomputed from x alone) and optionally writes the identity
+//     branch's gradient dy' = dy * (y > 0) in the same pass.
 
 #include <torch/extension.h>
 #include <ATen/ATen.h>
@@ -17,89 +30,88 @@
 
 #include "common.h"
 
-#define BN_MAXCPT 8  // max channels per thread (C <= 2048 at 256 threads)
+typedef __attribute__((ext_vector_type(8))) short short8v;
 
-template <typename T, bool XHAT>
-__global__ __launch_bounds__(PX_BLOCK)
-void bn_colsum_kernel(const T* __restrict__ x,
-                                 const T* __restrict__ dy,
-                                 const float* __restrict__ mean,
-                                 const float* __restrict__ invstd,
-                                 const float* __restrict__ gamma,
-                                 const float* __restrict__ beta,
-                                 float* __restrict__ out1,
-                                 float* __restrict__ out2, long M, int C,
-                                 bool relu) {
-  // XHAT=false: out1 = sum(x), out2 = sum(x^2)            (forward stats)
-  // XHAT=true:  out1 = sum(dy_eff), out2 = sum(dy_eff * xhat)   (backward)
+template <typename T>
+__device__ __forceinline__ float bn_elt(const short8v v, int k);
+template <>
+__device__ __forceinline__ float bn_elt<at::BFloat16>(const short8v v,
+                                                      int k) {
+  union { short s; __hip_bfloat16 b; } u;
+  u.s = v[k];
+  return __bfloat162float(u.b);
+}
+template <>
+__device__ __forceinline__ float bn_elt<at::Half>(const short8v v, int k) {
+  union { short s; __half h; } u;
+  u.s = v[k];
+  return __half2float(u.h);
+}
+
+template <typename T>
+__device__ __forceinline__ short bn_pack(float f);
+template <>
+__device__ __forceinline__ short bn_pack<at::BFloat16>(float f) {
+  union { short s; __hip_bfloat16 b; } u;
+  u.b = __float2bfloat16(f);
+  return u.s;
+}
+template <>
+__device__ __forceinline__ short bn_pack<at::Half>(float f) {
+  union { short s; __half h; } u;
+  u.h = __float2half(f);
+  return u.s;
+}
+
+// ---- forward stats: s1[c] = sum x, s2[c] = sum x^2 ----------------------
+template <typename T>
+__global__ __launch_bounds__(256)
+void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ s1,
+                     float* __restrict__ s2, long M, int C) {
+  const int tpr = C >> 3;                    // thread-groups per row
+  const int rpb = 256 / tpr;                 // rows per block sweep
   const int tid = threadIdx.x;
-  const int cpt = C > (int)blockDim.x ? C / (int)blockDim.x : 1;
-  int own_c[BN_MAXCPT];
-  float acc1[BN_MAXCPT], acc2[BN_MAXCPT];
+  const int rsub = tid / tpr;                // row slot within the block
+  const int cg = (tid - rsub * tpr) << 3;    // first owned channel
+  float a1[8] = {}, a2[8] = {};
+  if (rsub < rpb) {
+    const long rstep = (long)gridDim.x * rpb;
+    for (long r = (long)blockIdx.x * rpb + rsub; r < M; r += rstep) {
+      const short8v v = *(const short8v*)(x + r * C + cg);
 #pragma unroll
-  for (int k = 0; k < BN_MAXCPT; ++k) {
-    own_c[k] = (C > (int)blockDim.x) ? (tid + k * (int)blockDim.x)
-                                     : (tid & (C - 1));
-    acc1[k] = 0.f;
-    acc2[k] = 0.f;
-  }
-  const long total = M * (long)C;
-  for (long idx = (long)blockIdx.x * blockDim.x + tid; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    const int c_here = (int)(idx & (C - 1));
-    // which of this thread's owned slots is c_here?  With the mapping
-    // above and blockDim=256, idx stride keeps c_here == own_c[k] for
-    // k = (idx/blockDim) % cpt when C > blockDim, and own_c[0] always
-    // when C <= blockDim (stride % C == 0).
-    const int k = (C > (int)blockDim.x)
-                      ? (int)((idx / blockDim.x) % cpt)
-                      : 0;
-    if (XHAT) {
-      const float xh = (to_f32<T>(x[idx]) - mean[c_here]) *
-                       invstd[c_here];
-      float dyv = to_f32<T>(dy[idx]);
-      if (relu) {
-        const float y = xh * gamma[c_here] + beta[c_here];
-        dyv = y > 0.f ? dyv : 0.f;
+      for (int k = 0; k < 8; ++k) {
+        const float f = bn_elt<T>(v, k);
+        a1[k] += f;
+        a2[k] += f * f;
       }
-      acc1[k] += dyv;
-      acc2[k] += dyv * xh;
-    } else {
-      const float v = to_f32<T>(x[idx]);
-      acc1[k] += v;
-      acc2[k] += v * v;
     }
   }
-  // block-level LDS accumulation first: ONE global atomic per
-  // (block, channel), not per thread — 512 blocks x C addresses instead
-  // of 131k threads hammering C cells
-  __shared__ float lsum1[256 * BN_MAXCPT];
-  __shared__ float lsum2[256 * BN_MAXCPT];
-  for (int c = tid; c < C; c += (int)blockDim.x) {
-    lsum1[c] = 0.f;
-    lsum2[c] = 0.f;
-  }
+  __shared__ float ls[2 * 2048];
+  for (int c = tid; c < 2 * C; c += 256) ls[c] = 0.f;
   __syncthreads();
+  if (rsub < rpb) {
 #pragma unroll
-  for (int k = 0; k < BN_MAXCPT; ++k) {
-    if (k < cpt) {
-      atomicAdd(&lsum1[own_c[k]], acc1[k]);
-      atomicAdd(&lsum2[own_c[k]], acc2[k]);
+    for (int k = 0; k < 8; ++k) {
+      atomicAdd(&ls[cg + k], a1[k]);
+      atomicAdd(&ls[C + cg + k], a2[k]);
     }
   }
   __syncthreads();
-  for (int c = tid; c < C; c += (int)blockDim.x) {
-    if (lsum1[c] != 0.f || lsum2[c] != 0.f) {
-      atomicAdd(&out1[c], lsum1[c]);
-      atomicAdd(&out2[c], lsum2[c]);
-    }
+  for (int c = tid; c < C; c += 256) {
+    atomicAdd(&s1[c], ls[c]);
+    atomicAdd(&s2[c], ls[C + c]);
   }
 }
 
+// ---- finalize: mean/invstd, running stats, scale/shift ------------------
 __global__ void bn_finalize_kernel(const float* __restrict__ s1,
                                    const float* __restrict__ s2,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ beta,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
+                                   float* __restrict__ scale,
+                                   float* __restrict__ shift,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var, long M,
                                    int C, float eps, float momentum) {
@@ -108,8 +120,12 @@ __global__ void bn_finalize_kernel(const float* __restrict__ s1,
   const float m = s1[c] / (float)M;
   float var = s2[c] / (float)M - m * m;
   var = fmaxf(var, 0.f);
+  const float is = rsqrtf(var + eps);
   mean[c] = m;
-  invstd[c] = rsqrtf(var + eps);
+  invstd[c] = is;
+  const float sc = gamma[c] * is;
+  scale[c] = sc;
+  shift[c] = beta[c] - m * sc;
   if (running_mean != nullptr) {
     running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
     const float unbiased = var * (float)M / (float)(M > 1 ? M - 1 : 1);
@@ -118,49 +134,153 @@ __global__ void bn_finalize_kernel(const float* __restrict__ s1,
   }
 }
 
-template <typename T>
-__global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
-                               const float* __restrict__ mean,
-                               const float* __restrict__ invstd,
-                               const float* __restrict__ gamma,
-                               const float* __restrict__ beta, long M,
-                               int C, bool relu) {
-  const long total = M * (long)C;
-  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    const int c = (int)(idx & (C - 1));
-    float v = (to_f32<T>(x[idx]) - mean[c]) * invstd[c] * gamma[c] +
-              beta[c];
-    if (relu) v = fmaxf(v, 0.f);
-    y[idx] = from_f32<T>(v);
+// ---- normalize (+ residual add) (+ relu) --------------------------------
+template <typename T, bool ADD, bool RELU>
+__global__ __launch_bounds__(256)
+void bn_norm_kernel(const T* __restrict__ x, const T* __restrict__ addend,
+                    T* __restrict__ y, const float* __restrict__ scale,
+                    const float* __restrict__ shift, long M, int C) {
+  const int tpr = C >> 3;
+  const int rpb = 256 / tpr;
+  const int tid = threadIdx.x;
+  const int rsub = tid / tpr;
+  const int cg = (tid - rsub * tpr) << 3;
+  if (rsub >= rpb) return;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    sc[k] = scale[cg + k];
+    sh[k] = shift[cg + k];
+  }
+  const long rstep = (long)gridDim.x * rpb;
+  for (long r = (long)blockIdx.x * rpb + rsub; r < M; r += rstep) {
+    const long base = r * C + cg;
+    const short8v v = *(const short8v*)(x + base);
+    short8v a;
+    if (ADD) a = *(const short8v*)(addend + base);
+    short8v o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float f = fmaf(bn_elt<T>(v, k), sc[k], sh[k]);
+      if (ADD) f += bn_elt<T>(a, k);
+      if (RELU) f = fmaxf(f, 0.f);
+      o[k] = bn_pack<T>(f);
+    }
+    *(short8v*)(y + base) = o;
   }
 }
 
-template <typename T>
-__global__ void bn_dx_kernel(const T* __restrict__ x,
-                             const T* __restrict__ dy, T* __restrict__ dx,
-                             const float* __restrict__ mean,
-                             const float* __restrict__ invstd,
-                             const float* __restrict__ gamma,
-                             const float* __restrict__ beta,
-                             const float* __restrict__ sum_dy,
-                             const float* __restrict__ sum_dy_xhat, long M,
-                             int C, bool relu) {
-  const long total = M * (long)C;
-  const float invM = 1.f / (float)M;
-  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    const int c = (int)(idx & (C - 1));
-    const float xh = (to_f32<T>(x[idx]) - mean[c]) * invstd[c];
-    float dyv = to_f32<T>(dy[idx]);
-    if (relu) {
-      const float yv = xh * gamma[c] + beta[c];
-      dyv = yv > 0.f ? dyv : 0.f;
+// ---- backward stats: sum dy', sum dy'*xhat ------------------------------
+// dy' = dy * (y > 0) when RELU else dy; xhat = (x - mean) * invstd.
+template <typename T, bool RELU>
+__global__ __launch_bounds__(256)
+void bn_bstats_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                      const T* __restrict__ y,
+                      const float* __restrict__ mean,
+                      const float* __restrict__ invstd,
+                      float* __restrict__ s1, float* __restrict__ s2,
+                      long M, int C) {
+  const int tpr = C >> 3;
+  const int rpb = 256 / tpr;
+  const int tid = threadIdx.x;
+  const int rsub = tid / tpr;
+  const int cg = (tid - rsub * tpr) << 3;
+  float a1[8] = {}, a2[8] = {};
+  if (rsub < rpb) {
+    float mn[8], is[8];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      mn[k] = mean[cg + k];
+      is[k] = invstd[cg + k];
     }
-    const float g = gamma[c] * invstd[c];
-    dx[idx] = from_f32<T>(
-        g * (dyv - (sum_dy[c] + xh * sum_dy_xhat[c]) * invM));
+    const long rstep = (long)gridDim.x * rpb;
+    for (long r = (long)blockIdx.x * rpb + rsub; r < M; r += rstep) {
+      const long base = r * C + cg;
+      const short8v xv = *(const short8v*)(x + base);
+      const short8v dv = *(const short8v*)(dy + base);
+      short8v yv;
+      if (RELU) yv = *(const short8v*)(y + base);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float d = bn_elt<T>(dv, k);
+        if (RELU && !(bn_elt<T>(yv, k) > 0.f)) d = 0.f;
+        a1[k] += d;
+        a2[k] += d * (bn_elt<T>(xv, k) - mn[k]) * is[k];
+      }
+    }
   }
+  __shared__ float ls[2 * 2048];
+  for (int c = tid; c < 2 * C; c += 256) ls[c] = 0.f;
+  __syncthreads();
+  if (rsub < rpb) {
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      atomicAdd(&ls[cg + k], a1[k]);
+      atomicAdd(&ls[C + cg + k], a2[k]);
+    }
+  }
+  __syncthreads();
+  for (int c = tid; c < C; c += 256) {
+    atomicAdd(&s1[c], ls[c]);
+    atomicAdd(&s2[c], ls[C + c]);
+  }
+}
+
+// ---- backward dx (+ identity-branch grad) -------------------------------
+// dx = gamma*invstd * (dy' - sum_dy'/M - xhat * sum_dy_xhat/M)
+// dxid = dy'   (gradient through the fused residual add)
+template <typename T, bool RELU, bool WANT_DID>
+__global__ __launch_bounds__(256)
+void bn_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                  const T* __restrict__ y, T* __restrict__ dx,
+                  T* __restrict__ dxid, const float* __restrict__ mean,
+                  const float* __restrict__ invstd,
+                  const float* __restrict__ gamma,
+                  const float* __restrict__ sum_dy,
+                  const float* __restrict__ sum_dy_xhat, long M, int C) {
+  const int tpr = C >> 3;
+  const int rpb = 256 / tpr;
+  const int tid = threadIdx.x;
+  const int rsub = tid / tpr;
+  const int cg = (tid - rsub * tpr) << 3;
+  if (rsub >= rpb) return;
+  const float invM = 1.f / (float)M;
+  float mn[8], is[8], g[8], k1[8], k2[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    mn[k] = mean[cg + k];
+    is[k] = invstd[cg + k];
+    g[k] = gamma[cg + k] * is[k];
+    k1[k] = sum_dy[cg + k] * invM;
+    k2[k] = sum_dy_xhat[cg + k] * invM;
+  }
+  const long rstep = (long)gridDim.x * rpb;
+  for (long r = (long)blockIdx.x * rpb + rsub; r < M; r += rstep) {
+    const long base = r * C + cg;
+    const short8v xv = *(const short8v*)(x + base);
+    const short8v dv = *(const short8v*)(dy + base);
+    short8v yv;
+    if (RELU) yv = *(const short8v*)(y + base);
+    short8v o, oid;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float d = bn_elt<T>(dv, k);
+      if (RELU && !(bn_elt<T>(yv, k) > 0.f)) d = 0.f;
+      const float xh = (bn_elt<T>(xv, k) - mn[k]) * is[k];
+      o[k] = bn_pack<T>(g[k] * (d - k1[k] - xh * k2[k]));
+      if (WANT_DID) oid[k] = bn_pack<T>(d);
+    }
+    *(short8v*)(dx + base) = o;
+    if (WANT_DID) *(short8v*)(dxid + base) = oid;
+  }
+}
+
+static int bn_grid(long M, int C) {
+  const int rpb = 256 / (C >> 3);
+  long blocks = (M + rpb - 1) / rpb;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
 }
 
 std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
@@ -169,84 +289,152 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
                                       c10::optional<at::Tensor> run_mean,
                                       c10::optional<at::Tensor> run_var,
                                       double momentum, double eps,
-                                      bool relu) {
+                                      bool relu,
+                                      c10::optional<at::Tensor> addend) {
   TORCH_CHECK(x2d.is_cuda() && x2d.dim() == 2 && x2d.is_contiguous());
   const long M = x2d.size(0);
   const int C = (int)x2d.size(1);
-  TORCH_CHECK((C & (C - 1)) == 0 && C <= 256 * BN_MAXCPT,
-              "batchnorm: C must be a power of two <= 2048");
+  TORCH_CHECK(C % 8 == 0 && C <= 2048 && (C >> 3) <= 256,
+              "batchnorm: need C % 8 == 0 and C <= 2048");
+  TORCH_CHECK(x2d.scalar_type() == at::kBFloat16 ||
+              x2d.scalar_type() == at::kHalf,
+              "batchnorm: bf16/fp16 activations only");
   auto f32 = x2d.options().dtype(at::kFloat);
   auto s1 = at::zeros({C}, f32);
   auto s2 = at::zeros({C}, f32);
   auto mean = at::empty({C}, f32);
   auto invstd = at::empty({C}, f32);
+  auto scale = at::empty({C}, f32);
+  auto shift = at::empty({C}, f32);
   auto y = at::empty_like(x2d);
   auto g32 = gamma.to(at::kFloat).contiguous();
   auto b32 = beta.to(at::kFloat).contiguous();
+  const bool add = addend.has_value();
+  if (add) {
+    TORCH_CHECK(addend->is_contiguous() &&
+                addend->sizes() == x2d.sizes() &&
+                addend->scalar_type() == x2d.scalar_type());
+  }
   auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = bn_grid(M, C);
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x2d.scalar_type(), "bn_fwd", [&] {
-        bn_colsum_kernel<scalar_t, false>
-            <<<px_grid(M * C, PX_BLOCK, 512), PX_BLOCK, 0, stream>>>(
-                reinterpret_cast<const scalar_t*>(x2d.data_ptr()), nullptr,
-                nullptr, nullptr, nullptr, nullptr, s1.data_ptr<float>(),
-                s2.data_ptr<float>(), M, C, false);
-        HIP_CHECK_KERNEL();
-        bn_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
-            s1.data_ptr<float>(), s2.data_ptr<float>(),
-            mean.data_ptr<float>(), invstd.data_ptr<float>(),
-            run_mean.has_value() ? run_mean->data_ptr<float>() : nullptr,
-            run_var.has_value() ? run_var->data_ptr<float>() : nullptr, M,
-            C, (float)eps, (float)momentum);
-        HIP_CHECK_KERNEL();
-        bn_norm_kernel<scalar_t><<<px_grid(M * C), PX_BLOCK, 0, stream>>>(
-            reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
-            reinterpret_cast<scalar_t*>(y.data_ptr()),
-            mean.data_ptr<float>(), invstd.data_ptr<float>(),
-            g32.data_ptr<float>(), b32.data_ptr<float>(), M, C, relu);
-        HIP_CHECK_KERNEL();
+        if constexpr (std::is_same<scalar_t, float>::value ||
+                      std::is_same<scalar_t, double>::value) {
+          TORCH_CHECK(false, "batchnorm: bf16/fp16 only");
+        } else {
+          bn_stats_kernel<scalar_t><<<grid, 256, 0, stream>>>(
+              reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
+              s1.data_ptr<float>(), s2.data_ptr<float>(), M, C);
+          HIP_CHECK_KERNEL();
+          bn_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
+              s1.data_ptr<float>(), s2.data_ptr<float>(),
+              g32.data_ptr<float>(), b32.data_ptr<float>(),
+              mean.data_ptr<float>(), invstd.data_ptr<float>(),
+              scale.data_ptr<float>(), shift.data_ptr<float>(),
+              run_mean.has_value() ? run_mean->data_ptr<float>() : nullptr,
+              run_var.has_value() ? run_var->data_ptr<float>() : nullptr,
+              M, C, (float)eps, (float)momentum);
+          HIP_CHECK_KERNEL();
+          const scalar_t* ap = add
+              ? reinterpret_cast<const scalar_t*>(addend->data_ptr())
+              : nullptr;
+          const scalar_t* xp =
+              reinterpret_cast<const scalar_t*>(x2d.data_ptr());
+          scalar_t* yp = reinterpret_cast<scalar_t*>(y.data_ptr());
+          float* scp = scale.data_ptr<float>();
+          float* shp = shift.data_ptr<float>();
+          if (add && relu) {
+            bn_norm_kernel<scalar_t, true, true><<<grid, 256, 0, stream>>>(
+                xp, ap, yp, scp, shp, M, C);
+          } else if (add) {
+            bn_norm_kernel<scalar_t, true, false><<<grid, 256, 0,
+                                                    stream>>>(
+                xp, ap, yp, scp, shp, M, C);
+          } else if (relu) {
+            bn_norm_kernel<scalar_t, false, true><<<grid, 256, 0,
+                                                    stream>>>(
+                xp, ap, yp, scp, shp, M, C);
+          } else {
+            bn_norm_kernel<scalar_t, false, false><<<grid, 256, 0,
+                                                     stream>>>(
+                xp, ap, yp, scp, shp, M, C);
+          }
+          HIP_CHECK_KERNEL();
+        }
       });
   return {y, mean, invstd};
 }
 
 std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
                                       const at::Tensor& x2d,
+                                      const at::Tensor& y2d,
                                       const at::Tensor& mean,
                                       const at::Tensor& invstd,
-                                      const at::Tensor& gamma,
-                                      const at::Tensor& beta, bool relu) {
+                                      const at::Tensor& gamma, bool relu,
+                                      bool want_dxid) {
   const long M = x2d.size(0);
   const int C = (int)x2d.size(1);
   auto f32 = x2d.options().dtype(at::kFloat);
   auto sum_dy = at::zeros({C}, f32);
   auto sum_dy_xhat = at::zeros({C}, f32);
   auto dx = at::empty_like(x2d);
+  auto dxid = want_dxid ? at::empty_like(x2d) : at::Tensor();
   auto g32 = gamma.to(at::kFloat).contiguous();
-  auto b32 = beta.to(at::kFloat).contiguous();
   auto dyc = dy2d.contiguous();
   auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = bn_grid(M, C);
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x2d.scalar_type(), "bn_bwd", [&] {
-        bn_colsum_kernel<scalar_t, true>
-            <<<px_grid(M * C, PX_BLOCK, 512), PX_BLOCK, 0, stream>>>(
-                reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
-                reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
-                mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                g32.data_ptr<float>(), b32.data_ptr<float>(),
-                sum_dy.data_ptr<float>(), sum_dy_xhat.data_ptr<float>(),
-                M, C, relu);
-        HIP_CHECK_KERNEL();
-        bn_dx_kernel<scalar_t><<<px_grid(M * C), PX_BLOCK, 0, stream>>>(
-            reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
-            reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
-            reinterpret_cast<scalar_t*>(dx.data_ptr()),
-            mean.data_ptr<float>(), invstd.data_ptr<float>(),
-            g32.data_ptr<float>(), b32.data_ptr<float>(),
-            sum_dy.data_ptr<float>(), sum_dy_xhat.data_ptr<float>(), M, C,
-            relu);
-        HIP_CHECK_KERNEL();
+        if constexpr (std::is_same<scalar_t, float>::value ||
+                      std::is_same<scalar_t, double>::value) {
+          TORCH_CHECK(false, "batchnorm: bf16/fp16 only");
+        } else {
+          const scalar_t* xp =
+              reinterpret_cast<const scalar_t*>(x2d.data_ptr());
+          const scalar_t* dp =
+              reinterpret_cast<const scalar_t*>(dyc.data_ptr());
+          const scalar_t* yp =
+              reinterpret_cast<const scalar_t*>(y2d.data_ptr());
+          if (relu) {
+            bn_bstats_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+                xp, dp, yp, mean.data_ptr<float>(),
+                invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                sum_dy_xhat.data_ptr<float>(), M, C);
+          } else {
+            bn_bstats_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+                xp, dp, yp, mean.data_ptr<float>(),
+                invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                sum_dy_xhat.data_ptr<float>(), M, C);
+          }
+          HIP_CHECK_KERNEL();
+          scalar_t* didp = want_dxid
+              ? reinterpret_cast<scalar_t*>(dxid.data_ptr()) : nullptr;
+          scalar_t* dxp = reinterpret_cast<scalar_t*>(dx.data_ptr());
+          float* mp = mean.data_ptr<float>();
+          float* ip = invstd.data_ptr<float>();
+          float* gp = g32.data_ptr<float>();
+          float* s1p = sum_dy.data_ptr<float>();
+          float* s2p = sum_dy_xhat.data_ptr<float>();
+          if (relu && want_dxid) {
+            bn_dx_kernel<scalar_t, true, true><<<grid, 256, 0, stream>>>(
+                xp, dp, yp, dxp, didp, mp, ip, gp, s1p, s2p, M, C);
+          } else if (relu) {
+            bn_dx_kernel<scalar_t, true, false><<<grid, 256, 0, stream>>>(
+                xp, dp, yp, dxp, didp, mp, ip, gp, s1p, s2p, M, C);
+          } else if (want_dxid) {
+            bn_dx_kernel<scalar_t, false, true><<<grid, 256, 0, stream>>>(
+                xp, dp, yp, dxp, didp, mp, ip, gp, s1p, s2p, M, C);
+          } else {
+            bn_dx_kernel<scalar_t, false, false><<<grid, 256, 0,
+                                                   stream>>>(
+                xp, dp, yp, dxp, didp, mp, ip, gp, s1p, s2p, M, C);
+          }
+          HIP_CHECK_KERNEL();
+        }
       });
   // dgamma = sum_dy_xhat, dbeta = sum_dy (per-channel sums of the
-  // (relu-masked) upstream grad)
+  // relu-masked upstream grad)
+  if (want_dxid) return {dx, sum_dy_xhat, sum_dy, dxid};
   return {dx, sum_dy_xhat, sum_dy};
 }

@@ -72,13 +72,15 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
                                       c10::optional<at::Tensor> run_mean,
                                       c10::optional<at::Tensor> run_var,
                                       double momentum, double eps,
-                                      bool relu);
+                                      bool relu,
+                                      c10::optional<at::Tensor> addend);
 std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
                                       const at::Tensor& x2d,
+                                      const at::Tensor& y2d,
                                       const at::Tensor& mean,
                                       const at::Tensor& invstd,
-                                      const at::Tensor& gamma,
-                                      const at::Tensor& beta, bool relu);
+                                      const at::Tensor& gamma, bool relu,
+                                      bool want_dxid);
 
 // gemm.hip
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);

@@ -803,35 +803,66 @@ def test_conv3x3_nhwc_vs_torch():
 
 
 def test_batchnorm_nhwc_vs_torch():
-    """Hand-written NHWC BN (train) fwd/bwd + running stats + fused ReLU
-    vs nn.BatchNorm2d."""
+    """Hand-written NHWC bf16 BN (train) fwd/bwd + running stats + fused
+    ReLU vs an fp32 torch reference on the same bf16 input."""
     from parallax_amd.ops.batchnorm import ParallaxBatchNorm2d
     torch.manual_seed(62)
     for relu in (False, True):
-        N, C, H, W = 4, 64, 9, 9
-        ours = ParallaxBatchNorm2d(C, momentum=0.1, relu=relu).to(DEV)
-        ref = torch.nn.BatchNorm2d(C, momentum=0.1).to(DEV)
-        x1 = (torch.randn(N, C, H, W, device=DEV)
-              .to(memory_format=torch.channels_last).requires_grad_(True))
-        x2 = x1.detach().clone().requires_grad_(True)
-        y1 = ours(x1)
-        y2 = ref(x2)
-        if relu:
-            y2 = torch.relu(y2)
-        assert torch.allclose(y1, y2, atol=1e-4), \
-            (y1 - y2).abs().max()
-        assert torch.allclose(ours.running_mean, ref.running_mean,
-                              atol=1e-4)
-        assert torch.allclose(ours.running_var, ref.running_var,
-                              atol=1e-3)
-        dy = torch.randn_like(y2)
-        y1.backward(dy)
-        y2.backward(dy)
-        assert torch.allclose(x1.grad, x2.grad, atol=1e-4), \
-            (x1.grad - x2.grad).abs().max()
-        assert torch.allclose(ours.weight.grad, ref.weight.grad,
-                              atol=1e-3)
-        assert torch.allclose(ours.bias.grad, ref.bias.grad, atol=1e-3)
+        for C in (64, 192, 256):
+            N, H, W = 4, C, 9, 9 if C == 64 else 5
+            N, H, W = 4, 9, 5
+            ours = ParallaxBatchNorm2d(C, momentum=0.1, relu=relu).to(DEV)
+            ref = torch.nn.BatchNorm2d(C, momentum=0.1).to(DEV)
+            x1 = (torch.randn(N, C, H, W, device=DEV)
+                  .to(torch.bfloat16)
+                  .to(memory_format=torch.channels_last)
+                  .requires_grad_(True))
+            x2 = (x1.detach().float().clone().requires_grad_(True))
+            y1 = ours(x1)
+            y2 = ref(x2)
+            if relu:
+                y2 = torch.relu(y2)
+            assert torch.allclose(y1.float(), y2, atol=5e-2), \
+                (C, relu, (y1.float() - y2).abs().max())
+            assert torch.allclose(ours.running_mean, ref.running_mean,
+                                  atol=1e-2)
+            assert torch.allclose(ours.running_var, ref.running_var,
+                                  atol=1e-2)
+            dy = torch.randn_like(y2)
+            y1.backward(dy.to(torch.bfloat16))
+            y2.backward(dy)
+            assert torch.allclose(x1.grad.float(), x2.grad, atol=5e-2), \
+                (C, relu, (x1.grad.float() - x2.grad).abs().max())
+            assert torch.allclose(ours.weight.grad, ref.weight.grad,
+                                  atol=0.3, rtol=3e-2)
+            assert torch.allclose(ours.bias.grad, ref.bias.grad,
+                                  atol=0.3, rtol=3e-2)
+
+
+def test_batchnorm_fused_residual_add():
+    """bn3 + shortcut-add + ReLU fusion: y = relu(bn(x) + id); backward
+    produces the identity branch's gradient dy * (y > 0)."""
+    from parallax_amd.ops.batchnorm import ParallaxBatchNorm2d
+    torch.manual_seed(63)
+    N, C, H, W = 4, 128, 7, 7
+    ours = ParallaxBatchNorm2d(C, momentum=0.1, relu=True).to(DEV)
+    ref = torch.nn.BatchNorm2d(C, momentum=0.1).to(DEV)
+    x1 = (torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16)
+          .to(memory_format=torch.channels_last).requires_grad_(True))
+    id1 = (torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16)
+           .to(memory_format=torch.channels_last).requires_grad_(True))
+    x2 = x1.detach().float().clone().requires_grad_(True)
+    id2 = id1.detach().float().clone().requires_grad_(True)
+    y1 = ours(x1, addend=id1)
+    y2 = torch.relu(ref(x2) + id2)
+    assert torch.allclose(y1.float(), y2, atol=5e-2), \
+        (y1.float() - y2).abs().max()
+    dy = torch.randn_like(y2)
+    y1.backward(dy.to(torch.bfloat16))
+    y2.backward(dy)
+    assert torch.allclose(x1.grad.float(), x2.grad, atol=5e-2)
+    assert torch.allclose(id1.grad.float(), id2.grad, atol=2e-2), \
+        (id1.grad.float() - id2.grad).abs().max()
 
 
 def test_batchnorm_small_batch_bf16():
