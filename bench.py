@@ -185,10 +185,14 @@ def main():
         out = sess.run(make_batch(state))
         state = out.get('state')
 
-    # resnet is NOT captured: MIOpen picks graph-safe conv algorithms
-    # under capture that measure ~20% slower than its tuned eager path
-    # (4.06k vs 5.06k img/s) — launch overhead there is negligible anyway.
-    if args.cuda_graph and on_gpu and args.model in ('lm1b', 'nmt'):
+    # resnet capture is opt-in (PARALLAX_RN_GRAPH=1): round-1 measured
+    # MIOpen picking slower graph-safe conv algorithms under capture
+    # (4.06k vs 5.06k img/s); with the shipped exhaustive-find DB the
+    # tuned algorithms may now be capture-eligible — A/B before shipping.
+    capture_ok = args.model in ('lm1b', 'nmt') or (
+        args.model == 'resnet50'
+        and os.environ.get('PARALLAX_RN_GRAPH') == '1')
+    if args.cuda_graph and on_gpu and capture_ok:
         try:
             if args.model == 'lm1b':
                 state0 = M.make_init_state(batch, device=device,

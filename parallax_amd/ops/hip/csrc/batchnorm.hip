@@ -97,15 +97,19 @@ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ s1,
     }
   }
   __syncthreads();
+  // per-block partials, plain stores (no global atomics, nothing to
+  // pre-zero): part[block][0..C) = sums, part[block][C..2C) = sumsq
+  float* mine = s1 + (long)blockIdx.x * 2 * C;
   for (int c = tid; c < C; c += 256) {
-    atomicAdd(&s1[c], ls[c]);
-    atomicAdd(&s2[c], ls[C + c]);
+    mine[c] = ls[c];
+    mine[C + c] = ls[C + c];
   }
+  (void)s2;
 }
 
 // ---- finalize: mean/invstd, running stats, scale/shift ------------------
-__global__ void bn_finalize_kernel(const float* __restrict__ s1,
-                                   const float* __restrict__ s2,
+__global__ void bn_finalize_kernel(const float* __restrict__ part,
+                                   int nblk,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
                                    float* __restrict__ mean,
@@ -117,8 +121,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ s1,
                                    int C, float eps, float momentum) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  const float m = s1[c] / (float)M;
-  float var = s2[c] / (float)M - m * m;
+  float sum = 0.f, sumsq = 0.f;
+  for (int b = 0; b < nblk; ++b) {
+    sum += part[(long)b * 2 * C + c];
+    sumsq += part[(long)b * 2 * C + C + c];
+  }
+  const float m = sum / (float)M;
+  float var = sumsq / (float)M - m * m;
   var = fmaxf(var, 0.f);
   const float is = rsqrtf(var + eps);
   mean[c] = m;
@@ -220,10 +229,28 @@ void bn_bstats_kernel(const T* __restrict__ x, const T* __restrict__ dy,
     }
   }
   __syncthreads();
+  // per-block partials, plain stores (nothing to pre-zero)
+  float* mine = s1 + (long)blockIdx.x * 2 * C;
   for (int c = tid; c < C; c += 256) {
-    atomicAdd(&s1[c], ls[c]);
-    atomicAdd(&s2[c], ls[C + c]);
+    mine[c] = ls[c];
+    mine[C + c] = ls[C + c];
   }
+  (void)s2;
+}
+
+// reduce backward partials -> sum_dy / sum_dy_xhat
+__global__ void bn_breduce_kernel(const float* __restrict__ part, int nblk,
+                                  float* __restrict__ sum_dy,
+                                  float* __restrict__ sum_dy_xhat, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s1 = 0.f, s2 = 0.f;
+  for (int b = 0; b < nblk; ++b) {
+    s1 += part[(long)b * 2 * C + c];
+    s2 += part[(long)b * 2 * C + C + c];
+  }
+  sum_dy[c] = s1;
+  sum_dy_xhat[c] = s2;
 }
 
 // ---- backward dx (+ identity-branch grad) -------------------------------
@@ -278,8 +305,8 @@ void bn_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
 static int bn_grid(long M, int C) {
   const int rpb = 256 / (C >> 3);
   long blocks = (M + rpb - 1) / rpb;
-  if (blocks > 2048) blocks = 2048;
-  if (blocks < 1) blocks = 1;
+  if (blocks > 512) blocks = 512;  // bounds the partial buffer; 2 blocks
+  if (blocks < 1) blocks = 1;      // per CU saturate a streaming kernel
   return (int)blocks;
 }
 
@@ -300,12 +327,17 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
               x2d.scalar_type() == at::kHalf,
               "batchnorm: bf16/fp16 activations only");
   auto f32 = x2d.options().dtype(at::kFloat);
-  auto s1 = at::zeros({C}, f32);
-  auto s2 = at::zeros({C}, f32);
-  auto mean = at::empty({C}, f32);
-  auto invstd = at::empty({C}, f32);
-  auto scale = at::empty({C}, f32);
-  auto shift = at::empty({C}, f32);
+  const int grid = bn_grid(M, C);
+  // one workspace: [grid][2C] partials + mean/invstd/scale/shift
+  auto ws = at::empty({(long)grid * 2 * C + 4 * C}, f32);
+  float* wsp = ws.data_ptr<float>();
+  float* partp = wsp;
+  float* meanp = wsp + (long)grid * 2 * C;
+  float* invstdp = meanp + C;
+  float* scp = invstdp + C;
+  float* shp = scp + C;
+  auto mean = ws.narrow(0, (long)grid * 2 * C, C);
+  auto invstd = ws.narrow(0, (long)grid * 2 * C + C, C);
   auto y = at::empty_like(x2d);
   auto g32 = gamma.to(at::kFloat).contiguous();
   auto b32 = beta.to(at::kFloat).contiguous();
@@ -316,7 +348,6 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
                 addend->scalar_type() == x2d.scalar_type());
   }
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int grid = bn_grid(M, C);
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x2d.scalar_type(), "bn_fwd", [&] {
         if constexpr (std::is_same<scalar_t, float>::value ||
@@ -325,13 +356,11 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
         } else {
           bn_stats_kernel<scalar_t><<<grid, 256, 0, stream>>>(
               reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
-              s1.data_ptr<float>(), s2.data_ptr<float>(), M, C);
+              partp, nullptr, M, C);
           HIP_CHECK_KERNEL();
           bn_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
-              s1.data_ptr<float>(), s2.data_ptr<float>(),
-              g32.data_ptr<float>(), b32.data_ptr<float>(),
-              mean.data_ptr<float>(), invstd.data_ptr<float>(),
-              scale.data_ptr<float>(), shift.data_ptr<float>(),
+              partp, grid, g32.data_ptr<float>(), b32.data_ptr<float>(),
+              meanp, invstdp, scp, shp,
               run_mean.has_value() ? run_mean->data_ptr<float>() : nullptr,
               run_var.has_value() ? run_var->data_ptr<float>() : nullptr,
               M, C, (float)eps, (float)momentum);
@@ -342,8 +371,6 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
           const scalar_t* xp =
               reinterpret_cast<const scalar_t*>(x2d.data_ptr());
           scalar_t* yp = reinterpret_cast<scalar_t*>(y.data_ptr());
-          float* scp = scale.data_ptr<float>();
-          float* shp = shift.data_ptr<float>();
           if (add && relu) {
             bn_norm_kernel<scalar_t, true, true><<<grid, 256, 0, stream>>>(
                 xp, ap, yp, scp, shp, M, C);
@@ -376,14 +403,18 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
   const long M = x2d.size(0);
   const int C = (int)x2d.size(1);
   auto f32 = x2d.options().dtype(at::kFloat);
-  auto sum_dy = at::zeros({C}, f32);
-  auto sum_dy_xhat = at::zeros({C}, f32);
+  const int grid = bn_grid(M, C);
+  auto ws = at::empty({(long)grid * 2 * C + 2 * C}, f32);
+  float* partp = ws.data_ptr<float>();
+  float* sdyp = partp + (long)grid * 2 * C;
+  float* sdxp = sdyp + C;
+  auto sum_dy = ws.narrow(0, (long)grid * 2 * C, C);
+  auto sum_dy_xhat = ws.narrow(0, (long)grid * 2 * C + C, C);
   auto dx = at::empty_like(x2d);
   auto dxid = want_dxid ? at::empty_like(x2d) : at::Tensor();
   auto g32 = gamma.to(at::kFloat).contiguous();
   auto dyc = dy2d.contiguous();
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int grid = bn_grid(M, C);
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, x2d.scalar_type(), "bn_bwd", [&] {
         if constexpr (std::is_same<scalar_t, float>::value ||
@@ -399,14 +430,15 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
           if (relu) {
             bn_bstats_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
                 xp, dp, yp, mean.data_ptr<float>(),
-                invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
-                sum_dy_xhat.data_ptr<float>(), M, C);
+                invstd.data_ptr<float>(), partp, nullptr, M, C);
           } else {
             bn_bstats_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
                 xp, dp, yp, mean.data_ptr<float>(),
-                invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
-                sum_dy_xhat.data_ptr<float>(), M, C);
+                invstd.data_ptr<float>(), partp, nullptr, M, C);
           }
+          HIP_CHECK_KERNEL();
+          bn_breduce_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
+              partp, grid, sdyp, sdxp, C);
           HIP_CHECK_KERNEL();
           scalar_t* didp = want_dxid
               ? reinterpret_cast<scalar_t*>(dxid.data_ptr()) : nullptr;
@@ -414,8 +446,8 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
           float* mp = mean.data_ptr<float>();
           float* ip = invstd.data_ptr<float>();
           float* gp = g32.data_ptr<float>();
-          float* s1p = sum_dy.data_ptr<float>();
-          float* s2p = sum_dy_xhat.data_ptr<float>();
+          float* s1p = sdyp;
+          float* s2p = sdxp;
           if (relu && want_dxid) {
             bn_dx_kernel<scalar_t, true, true><<<grid, 256, 0, stream>>>(
                 xp, dp, yp, dxp, didp, mp, ip, gp, s1p, s2p, M, C);
