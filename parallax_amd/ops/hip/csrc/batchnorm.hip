@@ -108,7 +108,11 @@ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ s1,
 }
 
 // ---- finalize: mean/invstd, running stats, scale/shift ------------------
-__global__ void bn_finalize_kernel(const float* __restrict__ part,
+// one WAVE per channel (the partials are L2-hot from the stats kernel;
+// 64 lanes x few loads each + shuffle reduce, instead of one thread
+// looping over every block's partial)
+__global__ __launch_bounds__(256)
+void bn_finalize_kernel(const float* __restrict__ part,
                                    int nblk,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
@@ -119,13 +123,20 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var, long M,
                                    int C, float eps, float momentum) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (c >= C) return;
+  const int lane = threadIdx.x & 63;
   float sum = 0.f, sumsq = 0.f;
-  for (int b = 0; b < nblk; ++b) {
+  for (int b = lane; b < nblk; b += 64) {
     sum += part[(long)b * 2 * C + c];
     sumsq += part[(long)b * 2 * C + C + c];
   }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    sum += __shfl_down(sum, off, 64);
+    sumsq += __shfl_down(sumsq, off, 64);
+  }
+  if (lane != 0) return;
   const float m = sum / (float)M;
   float var = sumsq / (float)M - m * m;
   var = fmaxf(var, 0.f);
@@ -239,16 +250,24 @@ void bn_bstats_kernel(const T* __restrict__ x, const T* __restrict__ dy,
 }
 
 // reduce backward partials -> sum_dy / sum_dy_xhat
-__global__ void bn_breduce_kernel(const float* __restrict__ part, int nblk,
+__global__ __launch_bounds__(256)
+void bn_breduce_kernel(const float* __restrict__ part, int nblk,
                                   float* __restrict__ sum_dy,
                                   float* __restrict__ sum_dy_xhat, int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (c >= C) return;
+  const int lane = threadIdx.x & 63;
   float s1 = 0.f, s2 = 0.f;
-  for (int b = 0; b < nblk; ++b) {
+  for (int b = lane; b < nblk; b += 64) {
     s1 += part[(long)b * 2 * C + c];
     s2 += part[(long)b * 2 * C + C + c];
   }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s1 += __shfl_down(s1, off, 64);
+    s2 += __shfl_down(s2, off, 64);
+  }
+  if (lane != 0) return;
   sum_dy[c] = s1;
   sum_dy_xhat[c] = s2;
 }
@@ -358,7 +377,7 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
               reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
               partp, nullptr, M, C);
           HIP_CHECK_KERNEL();
-          bn_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
+          bn_finalize_kernel<<<(C + 3) / 4, 256, 0, stream>>>(
               partp, grid, g32.data_ptr<float>(), b32.data_ptr<float>(),
               meanp, invstdp, scp, shp,
               run_mean.has_value() ? run_mean->data_ptr<float>() : nullptr,
@@ -437,7 +456,7 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
                 invstd.data_ptr<float>(), partp, nullptr, M, C);
           }
           HIP_CHECK_KERNEL();
-          bn_breduce_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
+          bn_breduce_kernel<<<(C + 3) / 4, 256, 0, stream>>>(
               partp, grid, sdyp, sdxp, C);
           HIP_CHECK_KERNEL();
           scalar_t* didp = want_dxid
