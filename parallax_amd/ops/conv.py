@@ -27,20 +27,40 @@ def _pad_rows(x, mult):
 
 
 class _Conv1x1Fn(torch.autograd.Function):
-    @staticmethod
-    def forward(ctx, x2d, weight):
-        # x2d [M, Cin] bf16, weight [Cout, Cin] bf16; the kernel handles
-        # arbitrary M/N with clamped edge-tile loads + guarded stores, so
-        # no padding copies
-        ctx.save_for_backward(x2d, weight)
-        return _ops.ext().pxgemm_bt(x2d.contiguous(), weight.contiguous())
+    """fwd + dgrad on the 8-phase MFMA GEMM; wgrad through MIOpen's wrw
+    igemm (hipBLASLt's TN split-K GEMM on this shape picks a ~5x-slower
+    schedule — gpurun_out/pc11 profile)."""
 
     @staticmethod
-    def backward(ctx, dy):
-        x2d, weight = ctx.saved_tensors
-        dx = dy @ weight            # [M, Cout] @ [Cout, Cin]
-        dw = dy.t().conj() @ x2d    # [Cout, M] @ [M, Cin]
-        return dx, dw
+    def forward(ctx, x, x2d, weight4):
+        ctx.save_for_backward(x, weight4)
+        e = _ops.ext()
+        w2d = weight4.reshape(weight4.shape[0], weight4.shape[1])
+        x2d = x2d.contiguous()
+        w = w2d.contiguous()
+        if hasattr(e, 'pxgemm8_bt') and x2d.shape[0] >= 512:
+            return e.pxgemm8_bt(x2d, w)
+        return e.pxgemm_bt(x2d, w)
+
+    @staticmethod
+    def backward(ctx, dy2d):
+        x, weight4 = ctx.saved_tensors
+        e = _ops.ext()
+        dy2d = dy2d.contiguous()
+        w2d = weight4.reshape(weight4.shape[0], weight4.shape[1])
+        # dgrad on the MFMA kernel: dy @ W == dy @ (W^T)^T
+        if (dy2d.shape[1] % 64 == 0 and hasattr(e, 'pxgemm8_bt')
+                and dy2d.shape[0] >= 512):
+            dx2d = e.pxgemm8_bt(dy2d, w2d.t().contiguous())
+        else:
+            dx2d = dy2d @ w2d
+        N, C, H, W = x.shape
+        cout = weight4.shape[0]
+        dy4 = (dy2d.view(N, H, W, cout).permute(0, 3, 1, 2))
+        _, dw, _ = torch.ops.aten.convolution_backward(
+            dy4, x, weight4.to(x.dtype), None, [1, 1], [0, 0], [1, 1],
+            False, [0, 0], 1, [False, True, False])
+        return None, dx2d, dw.to(weight4.dtype)
 
 
 def conv1x1_nhwc(x, weight):
@@ -52,7 +72,7 @@ def conv1x1_nhwc(x, weight):
     w2d = weight.reshape(cout, C)
     if x.is_cuda and _ops.native_available() \
             and x.dtype == torch.bfloat16 and C % 64 == 0:
-        out = _Conv1x1Fn.apply(x2d.contiguous(), w2d.to(x.dtype))
+        out = _Conv1x1Fn.apply(x, x2d, weight.to(x.dtype))
     else:
         out = x2d @ w2d.to(x2d.dtype).t()
     return out.view(N, H, W, cout).permute(0, 3, 1, 2)
