@@ -16,7 +16,9 @@ per-(t,s) dot products re-read both operand rows from LDS for every
 MAC (~0.6 MB of LDS traffic per workgroup fwd, 3x that bwd, against a
 ~65 TB/s aggregate LDS ceiling); aotriton's MFMA tiles amortize
 operands in registers.  The fix is a register-/MFMA-tiled rewrite
-(TODO.md), not a split.  OPT-IN via PARALLAX_PX_ATTN=1; default is
+(TODO.md), not a split.  DEFAULT-ON since the round-2 MFMA rewrite
+(fwd 6.8 us / bwd 19.3 us vs SDPA-aotriton 33/63 at the NMT shape —
+gpurun_out/attn_ab_mfma.log); PARALLAX_PX_ATTN=0 restores SDPA; was
 torch SDPA.
 """
 
@@ -52,7 +54,7 @@ def _eligible(q, k, v):
             and all(t.stride(i) % 4 == 0
                     for t in (q, k, v) for i in (0, 1, 2))
             and _ops.native_available()
-            and os.environ.get('PARALLAX_PX_ATTN', '0') == '1')
+            and os.environ.get('PARALLAX_PX_ATTN', '1') == '1')
 
 
 def px_attention(q, k, v, is_causal=False):

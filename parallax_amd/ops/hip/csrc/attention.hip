@@ -233,6 +233,331 @@ px_attn_bwd_kernel(const ushort_t* __restrict__ q,
   }
 }
 
+
+// ===========================================================================
+// MFMA variants (round 2): the short-seq shape on the matrix cores.
+// One workgroup = 4 waves = one (b, h); every GEMM is
+// mfma_f32_16x16x32_bf16 over zero-padded 64x64 LDS tiles; fragments are
+// read as X[row][k-contiguous] / Y'[col][k-contiguous], so every
+// reduction over a NON-contiguous axis gets a transposed twin image
+// written once at load (or after the producing stage).  Row-wise softmax
+// reductions are 4-step __shfl_xor within each 16-lane group (the C
+// fragment holds col = lane&15, row = (lane>>4)*4 + reg).
+// ===========================================================================
+
+typedef __attribute__((ext_vector_type(8))) short att_bf16x8;
+typedef __attribute__((ext_vector_type(4))) float att_f32x4;
+
+// load a 16-byte (8-elem) A/B fragment from an LDS tile with row stride
+// ATT_LD (136 B: 8-aligned, not 16) as two b64 reads
+__device__ __forceinline__ att_bf16x8 att_frag(const ushort_t* lds,
+                                               int row, int k0) {
+  const ushort_t* p = lds + row * ATT_LD + k0;
+  att_bf16x8 f;
+  *reinterpret_cast<uint2*>(&f) = *reinterpret_cast<const uint2*>(p);
+  *(reinterpret_cast<uint2*>(&f) + 1) =
+      *reinterpret_cast<const uint2*>(p + 4);
+  return f;
+}
+
+// load an [R,64] global tile into lds ([64][ATT_LD], rows >= R zeroed)
+// and, when ldsT != nullptr, its transpose into ldsT[64][ATT_LD]
+__device__ __forceinline__ void att_load_tile_t(
+    const ushort_t* __restrict__ g, long rs, int rows, ushort_t* lds,
+    ushort_t* ldsT) {
+  for (int i = threadIdx.x; i < 64 * (ATT_D / 4); i += blockDim.x) {
+    const int r = i / (ATT_D / 4);
+    const int c = (i % (ATT_D / 4)) * 4;
+    uint2 v;
+    if (r < rows) {
+      v = *reinterpret_cast<const uint2*>(&g[r * rs + c]);
+    } else {
+      v.x = 0u;
+      v.y = 0u;
+    }
+    *reinterpret_cast<uint2*>(&lds[r * ATT_LD + c]) = v;
+    if (ldsT != nullptr) {
+      const ushort_t* e = reinterpret_cast<const ushort_t*>(&v);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) ldsT[(c + j) * ATT_LD + r] = e[j];
+    }
+  }
+}
+
+// row-wise reduce over the 16 lanes of a C-fragment lane group
+__device__ __forceinline__ float att_gmax(float v) {
+#pragma unroll
+  for (int w = 1; w < 16; w <<= 1) v = fmaxf(v, __shfl_xor(v, w, 64));
+  return v;
+}
+__device__ __forceinline__ float att_gsum(float v) {
+#pragma unroll
+  for (int w = 1; w < 16; w <<= 1) v += __shfl_xor(v, w, 64);
+  return v;
+}
+
+#define ATT_TILE_B (64 * ATT_LD)  // elems per LDS tile
+
+// compute one 16x64 score row-block: acc[j] += A_rows(w16+fr) @ B_rows
+__device__ __forceinline__ void att_qkt(const ushort_t* A,
+                                        const ushort_t* B, int arow0,
+                                        att_f32x4* acc) {
+  const int lane = threadIdx.x & 63;
+  const int fr = lane & 15;
+  const int ko = (lane >> 4) * 8;
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    const att_bf16x8 a = att_frag(A, arow0 + fr, kk * 32 + ko);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const att_bf16x8 b = att_frag(B, j * 16 + fr, kk * 32 + ko);
+      acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[j], 0,
+                                                       0, 0);
+    }
+  }
+}
+
+// write a C-layout fragment block (rows w16+crow0+r, cols j*16+cl) of
+// bf16 values into an LDS tile, optionally also its transpose
+__device__ __forceinline__ void att_cwrite(ushort_t* lds, ushort_t* ldsT,
+                                           int row0, const float* vals4,
+                                           int j) {
+  const int lane = threadIdx.x & 63;
+  const int cl = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const ushort_t bv = f2bf(vals4[r]);
+    const int row = row0 + crow0 + r;
+    const int col = j * 16 + cl;
+    lds[row * ATT_LD + col] = bv;
+    if (ldsT != nullptr) ldsT[col * ATT_LD + row] = bv;
+  }
+}
+
+__global__ __launch_bounds__(256)
+void px_attn_fwd_mfma_kernel(
+    const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
+    const ushort_t* __restrict__ v, ushort_t* __restrict__ out, int B,
+    int H, int T, int S, float scale, int causal, long qb, long qh,
+    long qt, long kb, long kh, long kt, long vb, long vh, long vt) {
+  __shared__ ushort_t lq[ATT_TILE_B];
+  __shared__ ushort_t lk[ATT_TILE_B];
+  __shared__ ushort_t lvT[ATT_TILE_B];
+  __shared__ ushort_t lp[ATT_TILE_B];
+  const int bh = blockIdx.x;
+  const int b = bh / H, h = bh % H;
+  att_load_tile_t(q + b * qb + h * qh, qt, T, lq, nullptr);
+  // K plain for QK^T; V transposed only: load V rows into lvT transpose
+  for (int i = threadIdx.x; i < 64 * (ATT_D / 4); i += blockDim.x) {
+    const int r = i / (ATT_D / 4);
+    const int c = (i % (ATT_D / 4)) * 4;
+    uint2 kv, vv;
+    if (r < S) {
+      kv = *reinterpret_cast<const uint2*>(&k[b * kb + h * kh + r * kt
+                                              + c]);
+      vv = *reinterpret_cast<const uint2*>(&v[b * vb + h * vh + r * vt
+                                              + c]);
+    } else {
+      kv.x = kv.y = 0u;
+      vv.x = vv.y = 0u;
+    }
+    *reinterpret_cast<uint2*>(&lk[r * ATT_LD + c]) = kv;
+    const ushort_t* e = reinterpret_cast<const ushort_t*>(&vv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) lvT[(c + j) * ATT_LD + r] = e[j];
+  }
+  __syncthreads();
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int cl = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+  const int row0 = wave * 16;
+
+  att_f32x4 sc[4] = {};
+  att_qkt(lq, lk, row0, sc);
+
+  // mask + softmax (rows row0+crow0+r, cols j*16+cl)
+  float p[4][4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = row0 + crow0 + r;
+    float mx = -1e30f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float sv = sc[j][r] * scale;
+      const int col = j * 16 + cl;
+      if (col >= S || (causal && col > row)) sv = -1e30f;
+      p[j][r] = sv;
+      mx = fmaxf(mx, sv);
+    }
+    mx = att_gmax(mx);
+    float sum = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float e = p[j][r] > -1e29f ? __expf(p[j][r] - mx) : 0.f;
+      p[j][r] = e;
+      sum += e;
+    }
+    sum = att_gsum(sum);
+    const float inv = sum > 0.f ? 1.f / sum : 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) p[j][r] *= inv;
+  }
+  // P -> LDS (plain layout only)
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    float col4[4] = {p[j][0], p[j][1], p[j][2], p[j][3]};
+    att_cwrite(lp, nullptr, row0, col4, j);
+  }
+  __syncthreads();
+
+  // O = P @ V  (B' = V^T image)
+  att_f32x4 oc[4] = {};
+  att_qkt(lp, lvT, row0, oc);
+
+  // store (out is [B,H,T,64] contiguous)
+  ushort_t* ob = out + ((long)bh) * T * ATT_D;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = row0 + crow0 + r;
+      if (row < T) ob[row * ATT_D + j * 16 + cl] = f2bf(oc[j][r]);
+    }
+  }
+}
+
+__global__ __launch_bounds__(256)
+void px_attn_bwd_mfma_kernel(
+    const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
+    const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
+    ushort_t* __restrict__ dq, ushort_t* __restrict__ dk,
+    ushort_t* __restrict__ dv, int B, int H, int T, int S, float scale,
+    int causal, long qb, long qh, long qt, long kb, long kh, long kt,
+    long vb, long vh, long vt, long db, long dh, long dt) {
+  extern __shared__ __attribute__((aligned(16))) char attsm[];
+  ushort_t* lq = reinterpret_cast<ushort_t*>(attsm);
+  ushort_t* lqT = lq + ATT_TILE_B;
+  ushort_t* lk = lqT + ATT_TILE_B;
+  ushort_t* lkT = lk + ATT_TILE_B;
+  ushort_t* lv = lkT + ATT_TILE_B;
+  ushort_t* ldo = lv + ATT_TILE_B;
+  ushort_t* ldoT = ldo + ATT_TILE_B;
+  ushort_t* lpT = ldoT + ATT_TILE_B;
+  ushort_t* lds_ = lpT + ATT_TILE_B;     // dS plain
+  ushort_t* ldsT = lds_ + ATT_TILE_B;    // dS transposed
+  const int bh = blockIdx.x;
+  const int b = bh / H, h = bh % H;
+  att_load_tile_t(q + b * qb + h * qh, qt, T, lq, lqT);
+  att_load_tile_t(k + b * kb + h * kh, kt, S, lk, lkT);
+  att_load_tile_t(v + b * vb + h * vh, vt, S, lv, nullptr);
+  att_load_tile_t(dout + b * db + h * dh, dt, T, ldo, ldoT);
+  __syncthreads();
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int cl = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+  const int row0 = wave * 16;
+
+  // recompute P (rows = t) and dP, both in C-fragment layout
+  att_f32x4 sc[4] = {};
+  att_qkt(lq, lk, row0, sc);
+  float p[4][4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = row0 + crow0 + r;
+    float mx = -1e30f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float sv = sc[j][r] * scale;
+      const int col = j * 16 + cl;
+      if (col >= S || (causal && col > row)) sv = -1e30f;
+      p[j][r] = sv;
+      mx = fmaxf(mx, sv);
+    }
+    mx = att_gmax(mx);
+    float sum = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float e = p[j][r] > -1e29f ? __expf(p[j][r] - mx) : 0.f;
+      p[j][r] = e;
+      sum += e;
+    }
+    sum = att_gsum(sum);
+    const float inv = sum > 0.f ? 1.f / sum : 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) p[j][r] *= inv;
+  }
+  // write P^T (needed by dV); P itself stays in registers for dS
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = row0 + crow0 + r;
+      const int col = j * 16 + cl;
+      lpT[col * ATT_LD + row] = f2bf(p[j][r]);
+    }
+  }
+  // dP = dO @ V^T   (A' = dO, B' = V, both plain k(d)-major)
+  att_f32x4 dpc[4] = {};
+  att_qkt(ldo, lv, row0, dpc);
+  // dS = P o (dP - rowsum(dP o P)) * scale
+  float dsv[4][4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    float rs = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) rs += dpc[j][r] * p[j][r];
+    rs = att_gsum(rs);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      dsv[j][r] = p[j][r] * (dpc[j][r] - rs) * scale;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = row0 + crow0 + r;
+      const int col = j * 16 + cl;
+      const ushort_t bv = f2bf(dsv[j][r]);
+      lds_[row * ATT_LD + col] = bv;
+      ldsT[col * ATT_LD + row] = bv;
+    }
+  }
+  __syncthreads();
+
+  // dQ = dS @ K        (A' = dS,  B' = K^T)
+  att_f32x4 dqc[4] = {};
+  att_qkt(lds_, lkT, row0, dqc);
+  // dK = dS^T @ Q      (A' = dS^T, B' = Q^T)
+  att_f32x4 dkc[4] = {};
+  att_qkt(ldsT, lqT, row0, dkc);
+  // dV = P^T @ dO      (A' = P^T,  B' = dO^T)
+  att_f32x4 dvc[4] = {};
+  att_qkt(lpT, ldoT, row0, dvc);
+
+  ushort_t* dqb = dq + ((long)bh) * T * ATT_D;
+  ushort_t* dkb = dk + ((long)bh) * S * ATT_D;
+  ushort_t* dvb = dv + ((long)bh) * S * ATT_D;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = row0 + crow0 + r;
+      const int col = j * 16 + cl;
+      if (row < T) dqb[row * ATT_D + col] = f2bf(dqc[j][r]);
+      if (row < S) {
+        dkb[row * ATT_D + col] = f2bf(dkc[j][r]);
+        dvb[row * ATT_D + col] = f2bf(dvc[j][r]);
+      }
+    }
+  }
+}
+
 static void att_check(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16 &&
                   t.dim() == 4 && t.stride(3) == 1 && t.size(3) == ATT_D,
@@ -255,6 +580,22 @@ at::Tensor px_attn_fwd(const at::Tensor& q, const at::Tensor& k,
   auto o = at::empty({B, H, T, ATT_D}, q.options());
   const float scale = 1.0f / sqrtf((float)ATT_D);
   auto stream = at::cuda::getCurrentCUDAStream();
+  static const bool use_mfma = [] {
+    const char* e = getenv("PARALLAX_ATTN_MFMA");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (use_mfma) {
+    px_attn_fwd_mfma_kernel<<<(int)(B * H), 256, 0, stream>>>(
+        reinterpret_cast<const ushort_t*>(q.data_ptr()),
+        reinterpret_cast<const ushort_t*>(k.data_ptr()),
+        reinterpret_cast<const ushort_t*>(v.data_ptr()),
+        reinterpret_cast<ushort_t*>(o.data_ptr()), (int)B, (int)H,
+        (int)T, (int)S, scale, causal ? 1 : 0, q.stride(0), q.stride(1),
+        q.stride(2), k.stride(0), k.stride(1), k.stride(2), v.stride(0),
+        v.stride(1), v.stride(2));
+    HIP_CHECK_KERNEL();
+    return o;
+  }
   px_attn_fwd_kernel<<<(int)(B * H), PX_BLOCK, 0, stream>>>(
       reinterpret_cast<const ushort_t*>(q.data_ptr()),
       reinterpret_cast<const ushort_t*>(k.data_ptr()),
@@ -283,6 +624,30 @@ std::vector<at::Tensor> px_attn_bwd(const at::Tensor& q,
   auto dv = at::empty({B, H, S, ATT_D}, q.options());
   const float scale = 1.0f / sqrtf((float)ATT_D);
   auto stream = at::cuda::getCurrentCUDAStream();
+  static const bool use_mfma = [] {
+    const char* e = getenv("PARALLAX_ATTN_MFMA");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (use_mfma) {
+    const int smem = 10 * ATT_TILE_B * 2;
+    hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&px_attn_bwd_mfma_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, smem);
+    px_attn_bwd_mfma_kernel<<<(int)(B * H), 256, smem, stream>>>(
+        reinterpret_cast<const ushort_t*>(q.data_ptr()),
+        reinterpret_cast<const ushort_t*>(k.data_ptr()),
+        reinterpret_cast<const ushort_t*>(v.data_ptr()),
+        reinterpret_cast<const ushort_t*>(dc.data_ptr()),
+        reinterpret_cast<ushort_t*>(dq.data_ptr()),
+        reinterpret_cast<ushort_t*>(dk.data_ptr()),
+        reinterpret_cast<ushort_t*>(dv.data_ptr()), (int)B, (int)H,
+        (int)T, (int)S, scale, causal ? 1 : 0, q.stride(0), q.stride(1),
+        q.stride(2), k.stride(0), k.stride(1), k.stride(2), v.stride(0),
+        v.stride(1), v.stride(2), dc.stride(0), dc.stride(1),
+        dc.stride(2));
+    HIP_CHECK_KERNEL();
+    return {dq, dk, dv};
+  }
   px_attn_bwd_kernel<<<(int)(B * H), PX_BLOCK, 0, stream>>>(
       reinterpret_cast<const ushort_t*>(q.data_ptr()),
       reinterpret_cast<const ushort_t*>(k.data_ptr()),
