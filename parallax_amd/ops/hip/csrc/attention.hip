@@ -340,7 +340,8 @@ void px_attn_fwd_mfma_kernel(
     const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
     const ushort_t* __restrict__ v, ushort_t* __restrict__ out, int B,
     int H, int T, int S, float scale, int causal, long qb, long qh,
-    long qt, long kb, long kh, long kt, long vb, long vh, long vt) {
+    long qt, long kb, long kh, long kt, long vb, long vh, long vt,
+    long ob_, long oh_, long ot_) {
   __shared__ ushort_t lq[ATT_TILE_B];
   __shared__ ushort_t lk[ATT_TILE_B];
   __shared__ ushort_t lvT[ATT_TILE_B];
@@ -417,14 +418,15 @@ void px_attn_fwd_mfma_kernel(
   att_f32x4 oc[4] = {};
   att_qkt(lp, lvT, row0, oc);
 
-  // store (out is [B,H,T,64] contiguous)
-  ushort_t* ob = out + ((long)bh) * T * ATT_D;
+  // strided store (the host allocates out [B,T,H,D]-permuted so the
+  // model's transpose(1,2).reshape afterwards is a free view)
+  ushort_t* ob = out + b * ob_ + h * oh_;
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = row0 + crow0 + r;
-      if (row < T) ob[row * ATT_D + j * 16 + cl] = f2bf(oc[j][r]);
+      if (row < T) ob[row * ot_ + j * 16 + cl] = f2bf(oc[j][r]);
     }
   }
 }
@@ -540,19 +542,22 @@ void px_attn_bwd_mfma_kernel(
   att_f32x4 dvc[4] = {};
   att_qkt(lpT, ldoT, row0, dvc);
 
-  ushort_t* dqb = dq + ((long)bh) * T * ATT_D;
-  ushort_t* dkb = dk + ((long)bh) * S * ATT_D;
-  ushort_t* dvb = dv + ((long)bh) * S * ATT_D;
+  // grads are [B,T|S,H,64] contiguous ([B,H,*,64] permuted views on the
+  // host): element (b,h,row,col) at ((b*rows + row)*H + h)*64 + col
+  const long grt = (long)H * ATT_D;
+  ushort_t* dqb = dq + ((long)b * T) * grt + (long)h * ATT_D;
+  ushort_t* dkb = dk + ((long)b * S) * grt + (long)h * ATT_D;
+  ushort_t* dvb = dv + ((long)b * S) * grt + (long)h * ATT_D;
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = row0 + crow0 + r;
       const int col = j * 16 + cl;
-      if (row < T) dqb[row * ATT_D + col] = f2bf(dqc[j][r]);
+      if (row < T) dqb[row * grt + col] = f2bf(dqc[j][r]);
       if (row < S) {
-        dkb[row * ATT_D + col] = f2bf(dkc[j][r]);
-        dvb[row * ATT_D + col] = f2bf(dvc[j][r]);
+        dkb[row * grt + col] = f2bf(dkc[j][r]);
+        dvb[row * grt + col] = f2bf(dvc[j][r]);
       }
     }
   }
@@ -585,16 +590,21 @@ at::Tensor px_attn_fwd(const at::Tensor& q, const at::Tensor& k,
     return e == nullptr || e[0] != '0';
   }();
   if (use_mfma) {
+    // allocate [B,T,H,D] and return the [B,H,T,D] permutation: the
+    // model's transpose(1,2).reshape(B,T,H*D) is then a free view
+    auto ob = at::empty({B, T, H, ATT_D}, q.options());
+    auto ov = ob.permute({0, 2, 1, 3});
     px_attn_fwd_mfma_kernel<<<(int)(B * H), 256, 0, stream>>>(
         reinterpret_cast<const ushort_t*>(q.data_ptr()),
         reinterpret_cast<const ushort_t*>(k.data_ptr()),
         reinterpret_cast<const ushort_t*>(v.data_ptr()),
-        reinterpret_cast<ushort_t*>(o.data_ptr()), (int)B, (int)H,
+        reinterpret_cast<ushort_t*>(ob.data_ptr()), (int)B, (int)H,
         (int)T, (int)S, scale, causal ? 1 : 0, q.stride(0), q.stride(1),
         q.stride(2), k.stride(0), k.stride(1), k.stride(2), v.stride(0),
-        v.stride(1), v.stride(2));
+        v.stride(1), v.stride(2), ov.stride(0), ov.stride(1),
+        ov.stride(2));
     HIP_CHECK_KERNEL();
-    return o;
+    return ov;
   }
   px_attn_fwd_kernel<<<(int)(B * H), PX_BLOCK, 0, stream>>>(
       reinterpret_cast<const ushort_t*>(q.data_ptr()),
@@ -618,10 +628,17 @@ std::vector<at::Tensor> px_attn_bwd(const at::Tensor& q,
   att_check(dout, "dout");
   const long B = q.size(0), H = q.size(1), T = q.size(2), S = k.size(2);
   TORCH_CHECK(T <= ATT_MAX && S <= ATT_MAX, "px_attn: T,S must be <= 64");
-  auto dc = dout.contiguous();
-  auto dq = at::empty({B, H, T, ATT_D}, q.options());
-  auto dk = at::empty({B, H, S, ATT_D}, q.options());
-  auto dv = at::empty({B, H, S, ATT_D}, q.options());
+  const bool d_ok = dout.stride(3) == 1 && dout.stride(0) % 4 == 0 &&
+                    dout.stride(1) % 4 == 0 && dout.stride(2) % 4 == 0;
+  auto dc = d_ok ? dout : dout.contiguous();
+  // grads as [B,T,H,D] permutations: the qkv-chunk cat in the caller's
+  // backward then reads them contiguously
+  auto dqb_ = at::empty({B, T, H, ATT_D}, q.options());
+  auto dkb_ = at::empty({B, S, H, ATT_D}, q.options());
+  auto dvb_ = at::empty({B, S, H, ATT_D}, q.options());
+  auto dq = dqb_.permute({0, 2, 1, 3});
+  auto dk = dkb_.permute({0, 2, 1, 3});
+  auto dv = dvb_.permute({0, 2, 1, 3});
   const float scale = 1.0f / sqrtf((float)ATT_D);
   auto stream = at::cuda::getCurrentCUDAStream();
   static const bool use_mfma = [] {
@@ -638,9 +655,9 @@ std::vector<at::Tensor> px_attn_bwd(const at::Tensor& q,
         reinterpret_cast<const ushort_t*>(k.data_ptr()),
         reinterpret_cast<const ushort_t*>(v.data_ptr()),
         reinterpret_cast<const ushort_t*>(dc.data_ptr()),
-        reinterpret_cast<ushort_t*>(dq.data_ptr()),
-        reinterpret_cast<ushort_t*>(dk.data_ptr()),
-        reinterpret_cast<ushort_t*>(dv.data_ptr()), (int)B, (int)H,
+        reinterpret_cast<ushort_t*>(dqb_.data_ptr()),
+        reinterpret_cast<ushort_t*>(dkb_.data_ptr()),
+        reinterpret_cast<ushort_t*>(dvb_.data_ptr()), (int)B, (int)H,
         (int)T, (int)S, scale, causal ? 1 : 0, q.stride(0), q.stride(1),
         q.stride(2), k.stride(0), k.stride(1), k.stride(2), v.stride(0),
         v.stride(1), v.stride(2), dc.stride(0), dc.stride(1),
