@@ -448,8 +448,11 @@ void px_attn_bwd_mfma_kernel(
   ushort_t* ldo = lv + ATT_TILE_B;
   ushort_t* ldoT = ldo + ATT_TILE_B;
   ushort_t* lpT = ldoT + ATT_TILE_B;
-  ushort_t* lds_ = lpT + ATT_TILE_B;     // dS plain
-  ushort_t* ldsT = lds_ + ATT_TILE_B;    // dS transposed
+  // dS images ALIAS lv/ldo (both dead once dP is computed; a barrier
+  // separates the last dP read from the first dS write) — 8 tiles of
+  // LDS instead of 10 keeps 2 workgroups per CU resident
+  ushort_t* lds_ = ldo;                  // dS plain
+  ushort_t* ldsT = lv;                   // dS transposed
   const int bh = blockIdx.x;
   const int b = bh / H, h = bh % H;
   att_load_tile_t(q + b * qb + h * qh, qt, T, lq, lqT);
@@ -506,6 +509,7 @@ void px_attn_bwd_mfma_kernel(
   // dP = dO @ V^T   (A' = dO, B' = V, both plain k(d)-major)
   att_f32x4 dpc[4] = {};
   att_qkt(ldo, lv, row0, dpc);
+  __syncthreads();  // every wave done reading lv/ldo before dS aliases
   // dS = P o (dP - rowsum(dP o P)) * scale
   float dsv[4][4];
 #pragma unroll
@@ -646,7 +650,7 @@ std::vector<at::Tensor> px_attn_bwd(const at::Tensor& q,
     return e == nullptr || e[0] != '0';
   }();
   if (use_mfma) {
-    const int smem = 10 * ATT_TILE_B * 2;
+    const int smem = 8 * ATT_TILE_B * 2;
     hipFuncSetAttribute(
         reinterpret_cast<const void*>(&px_attn_bwd_mfma_kernel),
         hipFuncAttributeMaxDynamicSharedMemorySize, smem);
