@@ -91,6 +91,9 @@ at::Tensor pxgemm8_bt_add(const at::Tensor& a, const at::Tensor& bt,
                           const c10::optional<at::Tensor>& addend);
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
                          const c10::optional<at::Tensor>& addend);
+void px_skinny_bt(const at::Tensor& a, const at::Tensor& bt,
+                  at::Tensor ws, at::Tensor out,
+                  const c10::optional<at::Tensor>& addend);
 at::Tensor pxconv3x3_fwd(const at::Tensor& xp, const at::Tensor& w2d,
                          long n_img, long oh, long ow, long cin,
                          long stride, long HP, long WP);
@@ -187,6 +190,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("gru_gates_fwd", &gru_gates_fwd);
   m.def("gru_gates_bwd", &gru_gates_bwd);
+  m.def("px_skinny_bt", &px_skinny_bt,
+        "split-K skinny GEMM (M<=128) with fp32-atomic accumulation + "
+        "fused cast/addend/re-zero epilogue (the LSTM chain launches)");
   m.def("pxgemm8_bt_v", &pxgemm8_bt_v,
         "pxgemm8_bt with explicit schedule-variant selection (A/B)");
   m.def("pxgemm8_bt", &pxgemm8_bt,

@@ -126,6 +126,17 @@ class _LSTMSeqFn(torch.autograd.Function):
                   and _os.environ.get('PARALLAX_PX_LSTM_FWD', '0') == '1')
         w_hT = w_h.t().contiguous() if (use_px or use_fused) else None
         ctx.used_fused = use_fused
+        # per-timestep projection on the split-K skinny GEMM (the chain's
+        # dependent [B,P] = [B,H] @ [H,P] launches: hipBLASLt ~19 us,
+        # px_skinny ~5 us + cast)
+        use_skinny = (gx.dtype == torch.bfloat16 and B <= 128
+                      and P % 64 == 0 and H % 64 == 0
+                      and hasattr(ext, 'px_skinny_bt')
+                      and _os.environ.get('PARALLAX_SKINNY', '1') == '1')
+        if use_skinny:
+            w_projT = w_proj.t().contiguous()
+            ws_p = torch.zeros(B, P, dtype=torch.float32,
+                               device=gx.device)
         for t in range(T):
             if use_fused:
                 ext.lstm_step_fwd_fused(
@@ -140,7 +151,11 @@ class _LSTMSeqFn(torch.autograd.Function):
                 ext.lstm_gates_fwd_out(gates, c_all[t], float(forget_bias),
                                        s_all[t], c_all[t + 1], acts[0, t],
                                        acts[1, t], acts[2, t], acts[3, t])
-            torch.mm(s_all[t], w_proj, out=hp_all[t + 1])
+            if use_skinny:
+                ext.px_skinny_bt(s_all[t], w_projT, ws_p, hp_all[t + 1],
+                                 None)
+            else:
+                torch.mm(s_all[t], w_proj, out=hp_all[t + 1])
         ctx.save_for_backward(hp_all, s_all, c_all, acts, w_h, w_proj)
         ctx.fbias = forget_bias
         # outs view [B, T, P]: non-contiguous transpose of hp_all[1:]; the
@@ -178,9 +193,23 @@ class _LSTMSeqFn(torch.autograd.Function):
                   and H % 128 == 0 and P % 64 == 0
                   and _os.environ.get('PARALLAX_PX_LSTM_BWD', '0') == '1')
         w_proj_t = None if use_fused else w_proj.t().contiguous()
+        # the recurrent carry dhp = dG @ w_h^T + upstream slice on the
+        # split-K skinny GEMM with the add fused into its cast epilogue
+        use_skinny = (w_h.dtype == torch.bfloat16
+                      and d_outs.dtype == torch.bfloat16 and B <= 128
+                      and P % 64 == 0 and (4 * H) % 64 == 0
+                      and d_outs.stride(2) == 1
+                      and hasattr(ext, 'px_skinny_bt')
+                      and _os.environ.get('PARALLAX_SKINNY', '1') == '1')
+        if use_skinny:
+            ws_c = torch.zeros(B, P, dtype=torch.float32,
+                               device=w_h.device)
         for t in range(T - 1, -1, -1):
             if t == T - 1:
                 dHP[t].copy_(d_outs[:, t])
+            elif use_skinny:
+                ext.px_skinny_bt(dG[t + 1], w_h, ws_c, dHP[t],
+                                 d_outs[:, t])
             else:
                 # dhp = upstream slice + recurrent carry; NOTE an addmm
                 # (beta=1) fusion of the add measured ~2% SLOWER overall

@@ -635,6 +635,34 @@ def test_pxgemm8_bt_vs_matmul():
     assert float((c.float() - ref).abs().max()) < 0.25
 
 
+def test_px_skinny_bt_vs_matmul():
+    """Split-K skinny GEMM + fused cast/addend/re-zero epilogue vs
+    fp32 matmul (the LSTM chain shapes)."""
+    e = _ext()
+    if not hasattr(e, 'px_skinny_bt'):
+        pytest.skip('px_skinny not built')
+    torch.manual_seed(53)
+    for (M, N, K) in ((128, 512, 2048), (128, 512, 8192), (64, 128, 64),
+                      (100, 192, 256)):
+        a = torch.rand(M, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
+        b = torch.rand(N, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
+        ws = torch.zeros(M, N, dtype=torch.float32, device=DEV)
+        out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+        e.px_skinny_bt(a, b, ws, out, None)
+        ref = a.float() @ b.float().t()
+        rel = (out.float() - ref).abs().max() / ref.abs().max()
+        assert float(rel) < 0.02, (M, N, K, float(rel))
+        assert float(ws.abs().max()) == 0.0          # re-zeroed
+        # addend + reuse of the zeroed workspace
+        add = torch.rand(M, 2 * N, device=DEV,
+                         dtype=torch.bfloat16)[:, :N] * 2 - 1
+        e.px_skinny_bt(a, b, ws, out, add.contiguous()
+                       if add.stride(1) != 1 else add)
+        ref2 = ref + add.float()
+        rel2 = (out.float() - ref2).abs().max() / ref2.abs().max().clamp_min(1e-3)
+        assert float(rel2) < 0.05, (M, N, K, float(rel2))
+
+
 def test_conv1x1_nhwc_vs_torch():
     """MFMA 1x1 NHWC conv fwd/bwd vs nn.functional.conv2d."""
     from parallax_amd.ops.conv import conv1x1_nhwc
