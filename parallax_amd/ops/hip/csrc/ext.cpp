@@ -92,7 +92,7 @@ at::Tensor pxgemm8_bt_add(const at::Tensor& a, const at::Tensor& bt,
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
                          const c10::optional<at::Tensor>& addend);
 void px_skinny_bt(const at::Tensor& a, const at::Tensor& bt,
-                  at::Tensor ws, at::Tensor out,
+                  at::Tensor ws, at::Tensor cnt, at::Tensor out,
                   const c10::optional<at::Tensor>& addend);
 at::Tensor pxconv3x3_fwd(const at::Tensor& xp, const at::Tensor& w2d,
                          long n_img, long oh, long ow, long cin,

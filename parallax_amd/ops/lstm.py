@@ -137,6 +137,8 @@ class _LSTMSeqFn(torch.autograd.Function):
             w_projT = w_proj.t().contiguous()
             ws_p = torch.zeros(B, P, dtype=torch.float32,
                                device=gx.device)
+            cnt_p = torch.zeros(P // 64, dtype=torch.int32,
+                                device=gx.device)
         for t in range(T):
             if use_fused:
                 ext.lstm_step_fwd_fused(
@@ -152,8 +154,8 @@ class _LSTMSeqFn(torch.autograd.Function):
                                        s_all[t], c_all[t + 1], acts[0, t],
                                        acts[1, t], acts[2, t], acts[3, t])
             if use_skinny:
-                ext.px_skinny_bt(s_all[t], w_projT, ws_p, hp_all[t + 1],
-                                 None)
+                ext.px_skinny_bt(s_all[t], w_projT, ws_p, cnt_p,
+                                 hp_all[t + 1], None)
             else:
                 torch.mm(s_all[t], w_proj, out=hp_all[t + 1])
         ctx.save_for_backward(hp_all, s_all, c_all, acts, w_h, w_proj)
@@ -204,11 +206,13 @@ class _LSTMSeqFn(torch.autograd.Function):
         if use_skinny:
             ws_c = torch.zeros(B, P, dtype=torch.float32,
                                device=w_h.device)
+            cnt_c = torch.zeros(P // 64, dtype=torch.int32,
+                                device=w_h.device)
         for t in range(T - 1, -1, -1):
             if t == T - 1:
                 dHP[t].copy_(d_outs[:, t])
             elif use_skinny:
-                ext.px_skinny_bt(dG[t + 1], w_h, ws_c, dHP[t],
+                ext.px_skinny_bt(dG[t + 1], w_h, ws_c, cnt_c, dHP[t],
                                  d_outs[:, t])
             else:
                 # dhp = upstream slice + recurrent carry; NOTE an addmm

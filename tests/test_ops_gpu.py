@@ -647,8 +647,9 @@ def test_px_skinny_bt_vs_matmul():
         a = torch.rand(M, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
         b = torch.rand(N, K, device=DEV, dtype=torch.bfloat16) * 2 - 1
         ws = torch.zeros(M, N, dtype=torch.float32, device=DEV)
+        cnt = torch.zeros(N // 64, dtype=torch.int32, device=DEV)
         out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
-        e.px_skinny_bt(a, b, ws, out, None)
+        e.px_skinny_bt(a, b, ws, cnt, out, None)
         ref = a.float() @ b.float().t()
         rel = (out.float() - ref).abs().max() / ref.abs().max()
         assert float(rel) < 0.02, (M, N, K, float(rel))
@@ -656,7 +657,8 @@ def test_px_skinny_bt_vs_matmul():
         # addend + reuse of the zeroed workspace
         add = torch.rand(M, 2 * N, device=DEV,
                          dtype=torch.bfloat16)[:, :N] * 2 - 1
-        e.px_skinny_bt(a, b, ws, out, add.contiguous()
+        assert int(cnt.abs().max()) == 0             # tickets reset
+        e.px_skinny_bt(a, b, ws, cnt, out, add.contiguous()
                        if add.stride(1) != 1 else add)
         ref2 = ref + add.float()
         rel2 = (out.float() - ref2).abs().max() / ref2.abs().max().clamp_min(1e-3)
