@@ -71,6 +71,36 @@ def _expected_count(ids, num_classes, num_sampled):
     return probs * num_sampled
 
 
+class _PxBtFn(torch.autograd.Function):
+    """inputs @ w^T on the 8-phase MFMA GEMM (w is [S, D] row-major =
+    the kernel's natural B operand); backward through library GEMMs
+    (dw is a transposed-A split-K shape)."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        ctx.save_for_backward(a, b)
+        from parallax_amd import ops as _o
+        return _o.ext().pxgemm8_bt(a.contiguous(), b.contiguous())
+
+    @staticmethod
+    def backward(ctx, g):
+        a, b = ctx.saved_tensors
+        return g @ b, g.transpose(0, 1) @ a
+
+
+def _candidates_matmul(inputs, sampled_w):
+    import os as _os
+    from parallax_amd import ops as _o
+    if (inputs.is_cuda and inputs.dtype == torch.bfloat16
+            and sampled_w.dtype == torch.bfloat16
+            and inputs.shape[1] % 64 == 0 and sampled_w.shape[0] % 16 == 0
+            and _o.native_available()
+            and hasattr(_o.ext(), 'pxgemm8_bt')
+            and _os.environ.get('PARALLAX_PX_CANDIDATES', '1') == '1'):
+        return _PxBtFn.apply(inputs, sampled_w)
+    return inputs @ sampled_w.t()
+
+
 def sampled_softmax_loss(weight, bias, labels, inputs, num_sampled,
                          num_classes, generator=None,
                          remove_accidental_hits=False):
@@ -92,7 +122,7 @@ def sampled_softmax_loss(weight, bias, labels, inputs, num_sampled,
     true_w = rows(weight, labels).to(inputs.dtype)    # [N, D]
     sampled_w = rows(weight, sampled_ids).to(inputs.dtype)  # [S, D]
     true_logits = (inputs * true_w).sum(-1)           # [N]
-    sampled_logits = inputs @ sampled_w.t()           # [N, S]
+    sampled_logits = _candidates_matmul(inputs, sampled_w)  # [N, S]
 
     # per-row / per-column logit corrections: + bias - log(expected_count)
     # (TF sampled-softmax correction); tiny [N] / [S] vectors
