@@ -121,7 +121,11 @@ class ResNet(nn.Module):
             if version == 1 else nn.BatchNorm2d(64, momentum=bn_momentum)
         self.stem_fused = _bn_fused() and version == 1
         self.relu = nn.ReLU(inplace=True)
-        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        if _os.environ.get('PARALLAX_NATIVE_POOL', '1') == '1':
+            from parallax_amd.ops.pool import ParallaxMaxPool3x3s2
+            self.maxpool = ParallaxMaxPool3x3s2()
+        else:
+            self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         block = Bottleneck if version == 1 else BottleneckV2
         cin, stages = 64, []
         for i, (width, n) in enumerate(zip((64, 128, 256, 512), layers)):
@@ -134,7 +138,13 @@ class ResNet(nn.Module):
         self.layer1, self.layer2, self.layer3, self.layer4 = stages
         if version == 2:
             self.bn_final = nn.BatchNorm2d(cin, momentum=bn_momentum)
-        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        if _os.environ.get('PARALLAX_NATIVE_POOL', '1') == '1':
+            from parallax_amd.ops.pool import ParallaxGlobalAvgPool
+            self.avgpool = ParallaxGlobalAvgPool()
+            self._flat_pool = True
+        else:
+            self.avgpool = nn.AdaptiveAvgPool2d(1)
+            self._flat_pool = False
         self.fc = nn.Linear(512 * block.expansion, num_classes)
 
     def forward(self, x):
@@ -145,8 +155,10 @@ class ResNet(nn.Module):
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         if self.version == 2:
             x = self.relu(self.bn_final(x))
-        x = self.avgpool(x).flatten(1)
-        return self.fc(x)
+        x = self.avgpool(x)
+        if not self._flat_pool:
+            x = x.flatten(1)
+        return self.fc(x.to(self.fc.weight.dtype))
 
 
 def build_resnet_graph(depth=50, num_classes=1001, batch_size=64, lr=0.1,

@@ -82,6 +82,14 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
                                       const at::Tensor& gamma, bool relu,
                                       bool want_dxid);
 
+// pool.hip
+std::vector<at::Tensor> maxpool3x3s2_fwd(const at::Tensor& x2d, long N,
+                                         long H, long W, long C);
+at::Tensor maxpool3x3s2_bwd(const at::Tensor& dy, const at::Tensor& arg,
+                            long N, long H, long W, long C);
+at::Tensor gavgpool_fwd(const at::Tensor& x2d, long N, long HW, long C);
+at::Tensor gavgpool_bwd(const at::Tensor& dy, long N, long HW, long C);
+
 // gemm.hip
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
 at::Tensor pxgemm8_bt(const at::Tensor& a, const at::Tensor& bt);
@@ -190,6 +198,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("gru_gates_fwd", &gru_gates_fwd);
   m.def("gru_gates_bwd", &gru_gates_bwd);
+  m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
+  m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
+  m.def("gavgpool_fwd", &gavgpool_fwd);
+  m.def("gavgpool_bwd", &gavgpool_bwd);
   m.def("px_skinny_bt", &px_skinny_bt,
         "split-K skinny GEMM (M<=128) with fp32-atomic accumulation + "
         "fused cast/addend/re-zero epilogue (the LSTM chain launches)");

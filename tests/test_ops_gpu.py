@@ -635,6 +635,38 @@ def test_pxgemm8_bt_vs_matmul():
     assert float((c.float() - ref).abs().max()) < 0.25
 
 
+def test_pool_nhwc_vs_torch():
+    """NHWC max-pool 3x3/s2/p1 and global avg pool fwd/bwd vs torch."""
+    from parallax_amd.ops.pool import (ParallaxMaxPool3x3s2,
+                                       ParallaxGlobalAvgPool)
+    torch.manual_seed(64)
+    for (N, C, H, W) in ((2, 64, 112, 112), (3, 16, 9, 7)):
+        x1 = (torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16)
+              .to(memory_format=torch.channels_last).requires_grad_(True))
+        x2 = x1.detach().float().clone().requires_grad_(True)
+        y1 = ParallaxMaxPool3x3s2()(x1)
+        y2 = torch.nn.functional.max_pool2d(x2, 3, 2, 1)
+        assert y1.shape == y2.shape
+        assert torch.allclose(y1.float(), y2, atol=1e-2)
+        dy = torch.randn_like(y2)
+        y1.backward(dy.to(torch.bfloat16))
+        y2.backward(dy)
+        # dy is bf16-rounded per element before the (<=4-way) gather sum
+        assert torch.allclose(x1.grad.float(), x2.grad, atol=6e-2), \
+            (N, C, H, W, (x1.grad.float() - x2.grad).abs().max())
+    # global avg pool
+    x1 = (torch.randn(4, 512, 7, 7, device=DEV).to(torch.bfloat16)
+          .to(memory_format=torch.channels_last).requires_grad_(True))
+    x2 = x1.detach().float().clone().requires_grad_(True)
+    y1 = ParallaxGlobalAvgPool()(x1)
+    y2 = torch.nn.functional.adaptive_avg_pool2d(x2, 1).flatten(1)
+    assert torch.allclose(y1, y2, atol=1e-2)
+    g = torch.randn_like(y2)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.allclose(x1.grad.float(), x2.grad, atol=1e-2)
+
+
 def test_px_skinny_bt_vs_matmul():
     """Split-K skinny GEMM + fused cast/addend/re-zero epilogue vs
     fp32 matmul (the LSTM chain shapes)."""
