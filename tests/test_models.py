@@ -136,8 +136,10 @@ def test_resnet_pure_bf16_structure_and_step(monkeypatch, free_port):
         None, parallax_config=ParallaxConfig(run_option='MPI'))
     m = sess.graph.model
     convs = [mod for mod in m.modules() if isinstance(mod, nn.Conv2d)]
+    from parallax_amd.ops.batchnorm import ParallaxBatchNorm2d
     bns = [mod for mod in m.modules()
-           if isinstance(mod, nn.modules.batchnorm._BatchNorm)]
+           if isinstance(mod, (nn.modules.batchnorm._BatchNorm,
+                               ParallaxBatchNorm2d))]
     assert convs and all(c.weight.dtype == torch.bfloat16 for c in convs)
     assert bns and all(b.weight.dtype == torch.float32 for b in bns)
     assert all(b.running_mean.dtype == torch.float32 for b in bns)
