@@ -52,7 +52,11 @@ def parse_args():
                    action='store_true',
                    help='nmt: bf16 body + fused fp32 master weights')
     p.add_argument('--no-pure-bf16', dest='pure_bf16',
-                   action='store_false')
+                   action='store_false',
+                   help='drop to the autocast-bf16 body (fp32 weights)')
+    p.add_argument('--fp32', action='store_true',
+                   help='full fp32 computation (no autocast either) — '
+                        'the reference-precision parity arm')
     p.add_argument('--companion', dest='companion', default=True,
                    action='store_true',
                    help='single-GPU lm1b runs also measure the resnet50 '
@@ -121,6 +125,10 @@ def main():
     cfg.communication_config.dense_comm_dtype = 'bf16'
 
     on_gpu = torch.cuda.is_available()
+    if args.fp32:
+        args.pure_bf16 = False
+    precision = ('fp32' if args.fp32 or not on_gpu else
+                 ('pure_bf16' if args.pure_bf16 else 'autocast_bf16'))
     if args.model == 'lm1b':
         from parallax_amd.models import lm1b as M
         batch = args.batch_size or M.BATCH_SIZE          # 128
@@ -128,6 +136,7 @@ def main():
         vocab = M.VOCAB_SIZE                             # 793470
         build = lambda: M.build_lm1b_graph(
             batch_size=batch, num_steps=seq, min_partitions=32,
+            use_bf16=not args.fp32,
             pure_bf16=args.pure_bf16 and on_gpu)
         metric, unit = 'LM1B words/sec', 'words/s'
         per_step_items = batch * seq                     # words per rank
@@ -135,26 +144,29 @@ def main():
         config_json = {'model': 'lm1b', 'vocab': vocab,
                        'global_batch': batch * n, 'seq_len': seq,
                        'num_sampled': M.NUM_SAMPLED,
-                       'parallelism': _parallelism_label(args.run_option, n)}
-        dtype = 'bf16' if (on_gpu and args.pure_bf16) else 'fp32'
+                       'parallelism': _parallelism_label(args.run_option, n),
+                       'precision': precision}
+        dtype = 'fp32' if precision == 'fp32' else 'bf16'
     elif args.model == 'nmt':
         from parallax_amd.models import nmt as M
         batch = args.batch_size or 64
         tgt_len = 48
         build = lambda: M.build_nmt_graph(
-            arch='transformer', pure_bf16=args.pure_bf16 and on_gpu)
+            arch='transformer', use_bf16=not args.fp32,
+            pure_bf16=args.pure_bf16 and on_gpu)
         metric, unit = 'NMT tgt words/sec', 'words/s'
         per_step_items = batch * tgt_len
         baseline = None  # no published reference number
         config_json = {'model': 'nmt-transformer', 'global_batch': batch * n,
                        'src_len': 48, 'tgt_len': tgt_len,
-                       'parallelism': _parallelism_label(args.run_option, n)}
-        dtype = 'bf16' if (on_gpu and args.pure_bf16) else 'fp32'
+                       'parallelism': _parallelism_label(args.run_option, n),
+                       'precision': precision}
+        dtype = 'fp32' if precision == 'fp32' else 'bf16'
     else:
         from parallax_amd.models import resnet as M
         batch = args.batch_size or 64
         build = lambda: M.build_resnet_graph(
-            depth=50, batch_size=batch, use_bf16=on_gpu,
+            depth=50, batch_size=batch, use_bf16=on_gpu and not args.fp32,
             channels_last=on_gpu,
             pure_bf16=args.pure_bf16 and on_gpu)
         metric, unit = 'ResNet-50 images/sec', 'images/s'
@@ -163,8 +175,9 @@ def main():
         config_json = {'model': 'resnet50', 'global_batch': batch * n,
                        'image_size': 224,
                        'parallelism': _parallelism_label(args.run_option, n,
-                                                         dense_only=True)}
-        dtype = 'bf16' if on_gpu else 'fp32'
+                                                         dense_only=True),
+                       'precision': precision}
+        dtype = 'fp32' if precision == 'fp32' else 'bf16'
 
     sess, num_workers, worker_id, _ = parallax_amd.parallel_run(
         build, None, parallax_config=cfg)
