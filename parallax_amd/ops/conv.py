@@ -26,6 +26,19 @@ def _pad_rows(x, mult):
     return torch.nn.functional.pad(x, (0, 0, 0, pad)), m
 
 
+def _best_gemm(e, a, b):
+    """Tile-family dispatch: the 256x64 tall tile for M-huge/N-small
+    (fills the chip where the wide tiles waste columns, measured
+    profiles/tall_ab_r2.log), 256^2 8-phase otherwise."""
+    M, N = a.shape[0], b.shape[0]
+    if (hasattr(e, 'px_tall_bt') and M >= 8 * N
+            and (M // 256) * ((N + 63) // 64) >= 256):
+        return e.px_tall_bt(a, b)
+    if hasattr(e, 'pxgemm8_bt') and M >= 512:
+        return e.pxgemm8_bt(a, b)
+    return e.pxgemm_bt(a, b)
+
+
 class _Conv1x1Fn(torch.autograd.Function):
     """fwd + dgrad on the 8-phase MFMA GEMM; wgrad through MIOpen's wrw
     igemm (hipBLASLt's TN split-K GEMM on this shape picks a ~5x-slower
@@ -38,9 +51,7 @@ class _Conv1x1Fn(torch.autograd.Function):
         w2d = weight4.reshape(weight4.shape[0], weight4.shape[1])
         x2d = x2d.contiguous()
         w = w2d.contiguous()
-        if hasattr(e, 'pxgemm8_bt') and x2d.shape[0] >= 512:
-            return e.pxgemm8_bt(x2d, w)
-        return e.pxgemm_bt(x2d, w)
+        return _best_gemm(e, x2d, w)
 
     @staticmethod
     def backward(ctx, dy2d):
@@ -51,7 +62,7 @@ class _Conv1x1Fn(torch.autograd.Function):
         # dgrad on the MFMA kernel: dy @ W == dy @ (W^T)^T
         if (dy2d.shape[1] % 64 == 0 and hasattr(e, 'pxgemm8_bt')
                 and dy2d.shape[0] >= 512):
-            dx2d = e.pxgemm8_bt(dy2d, w2d.t().contiguous())
+            dx2d = _best_gemm(e, dy2d, w2d.t().contiguous())
         else:
             dx2d = dy2d @ w2d
         N, C, H, W = x.shape
@@ -115,9 +126,27 @@ def swap_conv1x1(module):
     return module
 
 
+def _tall_grid(M, cout):
+    return (M // 256) * ((cout + 63) // 64)
+
+
+def _conv3x3_gemm(xp, w2d, N, oh, ow, C, stride, HP, WP):
+    """Pick the tall-tile conv kernel when its grid can occupy the chip
+    (>= 96 blocks at 2 blocks/CU), the 128^2 kernel otherwise."""
+    e = _ops.ext()
+    M = N * oh * ow
+    cout = w2d.shape[0]
+    if (hasattr(e, 'px_tall_conv3x3') and _tall_grid(M, cout) >= 96):
+        return e.px_tall_conv3x3(xp, w2d, N, oh, ow, C, stride, HP, WP)
+    return e.pxconv3x3_fwd(xp, w2d, N, oh, ow, C, stride, HP, WP)
+
+
 class _Conv3x3Fn(torch.autograd.Function):
-    """Forward: implicit-GEMM MFMA kernel over a host-padded NHWC input.
-    Backward: library dgrad/wgrad (aten.convolution_backward -> MIOpen)."""
+    """Forward: implicit-GEMM MFMA kernel over a host-padded NHWC input
+    (tall 256x64 tile when the grid fills the chip).  Backward: stride-1
+    dgrad is the SAME implicit GEMM over pad(dy) with the
+    rotated/transposed weight; wgrad (and stride-2 dgrad) go through
+    MIOpen (aten.convolution_backward with a masked output)."""
 
     @staticmethod
     def forward(ctx, x, weight, stride):
@@ -132,17 +161,40 @@ class _Conv3x3Fn(torch.autograd.Function):
         xp = torch.nn.functional.pad(x_nhwc, (0, 0, 1, 1, 1, 1))
         w2d = (weight.permute(0, 2, 3, 1).reshape(cout, 9 * C)
                .to(x.dtype).contiguous())
-        out = _ops.ext().pxconv3x3_fwd(xp.contiguous(), w2d, N, oh, ow, C,
-                                       stride, H + 2, W + 2)
+        out = _conv3x3_gemm(xp.contiguous(), w2d, N, oh, ow, C,
+                            stride, H + 2, W + 2)
         return out.view(N, oh, ow, cout).permute(0, 3, 1, 2)
 
     @staticmethod
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         s = ctx.stride
-        dx, dw, _ = torch.ops.aten.convolution_backward(
-            dy.to(x.dtype), x, weight.to(x.dtype), None, [s, s], [1, 1],
-            [1, 1], False, [0, 0], 1, [True, True, False])
+        N, C, H, W = x.shape
+        cout = weight.shape[0]
+        native_dgrad = (s == 1 and cout >= 64
+                        and (cout & (cout - 1)) == 0
+                        and _tall_grid(N * H * W, C) >= 96
+                        and ctx.needs_input_grad[0])
+        if native_dgrad:
+            # dx = conv3x3(pad(dy), Wrot) with
+            # Wrot[ci][dy][dx][co] = w[co][2-dy][2-dx][ci]
+            dyc = dy.to(x.dtype)
+            dy_nhwc = dyc.permute(0, 2, 3, 1)
+            dyp = torch.nn.functional.pad(dy_nhwc, (0, 0, 1, 1, 1, 1))
+            wrot = (weight.to(x.dtype).flip(2, 3)
+                    .permute(1, 2, 3, 0).reshape(C, 9 * cout)
+                    .contiguous())
+            dx2 = _conv3x3_gemm(dyp.contiguous(), wrot, N, H, W, cout,
+                                1, H + 2, W + 2)
+            dx = (dx2.view(N, H, W, C).permute(0, 3, 1, 2))
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dyc, x, weight.to(x.dtype), None, [s, s], [1, 1],
+                [1, 1], False, [0, 0], 1, [False, True, False])
+        else:
+            dx, dw, _ = torch.ops.aten.convolution_backward(
+                dy.to(x.dtype), x, weight.to(x.dtype), None, [s, s],
+                [1, 1], [1, 1], False, [0, 0], 1,
+                [ctx.needs_input_grad[0], True, False])
         return dx, dw.to(weight.dtype), None
 
 
@@ -150,9 +202,13 @@ def conv3x3_nhwc(x, weight, stride=1):
     """3x3 pad-1 NHWC convolution; forward on the hand-written MFMA
     implicit-GEMM kernel when eligible (bf16, Cin power-of-two >= 64)."""
     C = x.shape[1]
+    N, _, H, W = x.shape
+    oh = (H + 2 - 3) // stride + 1
+    ow = (W + 2 - 3) // stride + 1
     if (x.is_cuda and _ops.native_available()
             and x.dtype == torch.bfloat16 and C >= 64
-            and (C & (C - 1)) == 0):
+            and (C & (C - 1)) == 0
+            and _tall_grid(N * oh * ow, weight.shape[0]) >= 96):
         return _Conv3x3Fn.apply(x, weight, stride)
     return torch.nn.functional.conv2d(x, weight.to(x.dtype),
                                       stride=stride, padding=1)

@@ -99,6 +99,10 @@ at::Tensor pxgemm8_bt_add(const at::Tensor& a, const at::Tensor& bt,
                           const c10::optional<at::Tensor>& addend);
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
                          const c10::optional<at::Tensor>& addend);
+at::Tensor px_tall_bt(const at::Tensor& a, const at::Tensor& bt);
+at::Tensor px_tall_conv3x3(const at::Tensor& xp, const at::Tensor& w2d,
+                           long n_img, long oh, long ow, long cin,
+                           long stride, long HP, long WP);
 void px_skinny_bt(const at::Tensor& a, const at::Tensor& bt,
                   at::Tensor ws, at::Tensor cnt, at::Tensor out,
                   const c10::optional<at::Tensor>& addend);
@@ -202,6 +206,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("gavgpool_fwd", &gavgpool_fwd);
   m.def("gavgpool_bwd", &gavgpool_bwd);
+  m.def("px_tall_conv3x3", &px_tall_conv3x3,
+        "implicit-GEMM 3x3 NHWC conv fwd (and stride-1 dgrad via "
+        "rotated weights) on the tall-tile structure");
+  m.def("px_tall_bt", &px_tall_bt,
+        "256x64-tile MFMA GEMM for M-huge/N-small (conv-as-GEMM) shapes");
   m.def("px_skinny_bt", &px_skinny_bt,
         "split-K skinny GEMM (M<=128) with fp32-atomic accumulation + "
         "fused cast/addend/re-zero epilogue (the LSTM chain launches)");

@@ -1101,3 +1101,336 @@ void px_skinny_bt(const at::Tensor& a, const at::Tensor& bt,
       cnt.data_ptr<int>(), add_ptr, add_rs, M, N, K, spl, kslice);
   HIP_CHECK_KERNEL();
 }
+
+// ---------------------------------------------------------------------------
+// Tall-tile MFMA GEMM: 256x64 output tile, 8 waves (8M x 1N), BK=64,
+// double-buffered glds, 80 KiB LDS -> 2 blocks/CU (the co-resident
+// block's MFMAs cover this block's staging phases).  For the
+// M-huge / N-small shapes (ResNet conv-as-GEMM: N = Cout in {64..512})
+// where the 128/256-wide tiles compute mostly-discarded columns.
+// ---------------------------------------------------------------------------
+
+#define TL_BM 256
+#define TL_THREADS 512
+
+__global__ __launch_bounds__(TL_THREADS, 2)
+void px_tall_kernel(const ushort* __restrict__ A,
+                    const ushort* __restrict__ B,
+                    ushort* __restrict__ C, long M, long N, long K,
+                    int nbm, int nbn) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [2 buf][A 32 KiB | B 8 KiB]
+#define TL_LDSA(b) (smem + (b) * 40960)
+#define TL_LDSB(b) (smem + (b) * 40960 + 32768)
+  int bid = blockIdx.x;
+  {
+    const int nwg = nbm * nbn;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, pos = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const long bm = (long)(bid / nbn) * TL_BM;
+  const long bn = (long)(bid % nbn) * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave * 32;
+  const int fr = lane & 15;
+  const int ko = (lane >> 4) * 8;
+  f32x4 acc[2][4] = {};
+
+  // staging: A = 4 rounds of 512x16B, B = 1 round (512 x 16 B = the
+  // whole 8 KiB tile); swizzle ((row&7)<<4)
+  int pa_row[4], pa_kb[4];
+  long a_src[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int p = (tid + i * TL_THREADS) * 16;
+    pa_row[i] = p / 128;
+    pa_kb[i] = (p % 128) ^ ((pa_row[i] & 7) << 4);
+    long ar = bm + pa_row[i];
+    a_src[i] = ar < M ? ar : M - 1;
+  }
+  int pb_row, pb_kb;
+  long b_src;
+  {
+    const int p = tid * 16;
+    pb_row = p / 128;
+    pb_kb = (p % 128) ^ ((pb_row & 7) << 4);
+    long br = bn + pb_row;
+    b_src = br < N ? br : N - 1;
+  }
+  const int ktiles = (int)(K / 64);
+
+#define TL_STAGE(buf, kt)                                                  \
+  do {                                                                     \
+    const long kb = (long)(kt) * 64;                                       \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) void*)(                 \
+              A + a_src[i] * K + kb + (pa_kb[i] >> 1)),                    \
+          (__attribute__((address_space(3))) void*)(TL_LDSA(buf) +         \
+                                                    (tid + i * 512) * 16), \
+          16, 0, 0);                                                       \
+    }                                                                      \
+    __builtin_amdgcn_global_load_lds(                                      \
+        (const __attribute__((address_space(1))) void*)(                   \
+            B + b_src * K + kb + (pb_kb >> 1)),                            \
+        (__attribute__((address_space(3))) void*)(TL_LDSB(buf) +           \
+                                                  tid * 16),               \
+        16, 0, 0);                                                         \
+  } while (0)
+
+  TL_STAGE(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  int cur = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    if (kt + 1 < ktiles) TL_STAGE(cur ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 a_frag[2], b_frag[4];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int ar = wm + i * 16 + fr;
+        a_frag[i] = *(const bf16x8*)(TL_LDSA(cur) +
+                                     swz(ar, (kk * 32 + ko) * 2));
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int br = j * 16 + fr;
+        b_frag[j] = *(const bf16x8*)(TL_LDSB(cur) +
+                                     swz(br, (kk * 32 + ko) * 2));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = bm + wm + i * 16 + crow0 + r;
+        const long col = bn + j * 16 + fr;
+        if (row >= M || col >= N) continue;
+        __hip_bfloat16 v = __float2bfloat16(acc[i][j][r]);
+        C[row * N + col] = *reinterpret_cast<ushort*>(&v);
+      }
+    }
+  }
+#undef TL_STAGE
+#undef TL_LDSA
+#undef TL_LDSB
+}
+
+at::Tensor px_tall_bt(const at::Tensor& a, const at::Tensor& bt) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16 &&
+              bt.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(a.dim() == 2 && bt.dim() == 2 && a.is_contiguous() &&
+              bt.is_contiguous());
+  const long M = a.size(0), K = a.size(1), N = bt.size(0);
+  TORCH_CHECK(bt.size(1) == K && K % 64 == 0 && N >= 16);
+  auto c = at::empty({M, N}, a.options());
+  const int nbm = (int)((M + TL_BM - 1) / TL_BM);
+  const int nbn = (int)((N + 63) / 64);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&px_tall_kernel),
+      hipFuncAttributeMaxDynamicSharedMemorySize, 81920);
+  px_tall_kernel<<<nbm * nbn, TL_THREADS, 81920, stream>>>(
+      reinterpret_cast<const ushort*>(a.data_ptr()),
+      reinterpret_cast<const ushort*>(bt.data_ptr()),
+      reinterpret_cast<ushort*>(c.data_ptr()), M, N, K, nbm, nbn);
+  HIP_CHECK_KERNEL();
+  return c;
+}
+
+// ---------------------------------------------------------------------------
+// 3x3 NHWC conv forward as implicit GEMM on the tall-tile structure:
+// A = im2col of the HOST-PADDED input (halo materialized once per call;
+// glds cannot predicate), B = weight [Cout, 9*Cin] k-major.  Stride 1/2.
+// The SAME kernel computes stride-1 dgrad when fed pad(dy) and the
+// rotated/transposed weight [Cin, 9*Cout] (host prep, tiny).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(TL_THREADS, 2)
+void px_tall_conv3x3_kernel(const ushort* __restrict__ XP,
+                            const ushort* __restrict__ B,
+                            ushort* __restrict__ C, long M, long N,
+                            long K, int nbm, int nbn, int OH, int OW,
+                            int WP, int HP, int Cin, int log2cin,
+                            int stride) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+#define TL_LDSA(b) (smem + (b) * 40960)
+#define TL_LDSB(b) (smem + (b) * 40960 + 32768)
+  int bid = blockIdx.x;
+  {
+    const int nwg = nbm * nbn;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, pos = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const long bm = (long)(bid / nbn) * TL_BM;
+  const long bn = (long)(bid % nbn) * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave * 32;
+  const int fr = lane & 15;
+  const int ko = (lane >> 4) * 8;
+  f32x4 acc[2][4] = {};
+
+  int pa_kb[4];
+  long a_base[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int p = (tid + i * TL_THREADS) * 16;
+    const int prow = p / 128;
+    pa_kb[i] = (p % 128) ^ ((prow & 7) << 4);
+    long ar = bm + prow;
+    if (ar >= M) ar = M - 1;
+    const long n_img = ar / ((long)OH * OW);
+    const long rem = ar - n_img * ((long)OH * OW);
+    const int oy = (int)(rem / OW);
+    const int ox = (int)(rem - (long)oy * OW);
+    a_base[i] = (((n_img * (long)HP) + (long)oy * stride) * WP +
+                 (long)ox * stride) * Cin;
+  }
+  int pb_row, pb_kb;
+  long b_src;
+  {
+    const int p = tid * 16;
+    pb_row = p / 128;
+    pb_kb = (p % 128) ^ ((pb_row & 7) << 4);
+    long br = bn + pb_row;
+    b_src = br < N ? br : N - 1;
+  }
+  const int ktiles = (int)(K / 64);
+
+#define TLC_STAGE(buf, kt)                                                 \
+  do {                                                                     \
+    const int kb = (kt) * 64;                                              \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+      const int ka = kb + (pa_kb[i] >> 1);                                 \
+      const int dydx = ka >> log2cin;                                      \
+      const int ci = ka - (dydx << log2cin);                               \
+      const int dy = dydx >= 6 ? 2 : (dydx >= 3 ? 1 : 0);                  \
+      const int dx = dydx - 3 * dy;                                        \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) void*)(                 \
+              XP + a_base[i] + ((long)dy * WP + dx) * Cin + ci),           \
+          (__attribute__((address_space(3))) void*)(TL_LDSA(buf) +         \
+                                                    (tid + i * 512) * 16), \
+          16, 0, 0);                                                       \
+    }                                                                      \
+    __builtin_amdgcn_global_load_lds(                                      \
+        (const __attribute__((address_space(1))) void*)(                   \
+            B + b_src * K + kb + (pb_kb >> 1)),                            \
+        (__attribute__((address_space(3))) void*)(TL_LDSB(buf) +           \
+                                                  tid * 16),               \
+        16, 0, 0);                                                         \
+  } while (0)
+
+  TLC_STAGE(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  int cur = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    if (kt + 1 < ktiles) TLC_STAGE(cur ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 a_frag[2], b_frag[4];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int ar = wm + i * 16 + fr;
+        a_frag[i] = *(const bf16x8*)(TL_LDSA(cur) +
+                                     swz(ar, (kk * 32 + ko) * 2));
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int br = j * 16 + fr;
+        b_frag[j] = *(const bf16x8*)(TL_LDSB(cur) +
+                                     swz(br, (kk * 32 + ko) * 2));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = bm + wm + i * 16 + crow0 + r;
+        const long col = bn + j * 16 + fr;
+        if (row >= M || col >= N) continue;
+        __hip_bfloat16 v = __float2bfloat16(acc[i][j][r]);
+        C[row * N + col] = *reinterpret_cast<ushort*>(&v);
+      }
+    }
+  }
+#undef TLC_STAGE
+#undef TL_LDSA
+#undef TL_LDSB
+}
+
+at::Tensor px_tall_conv3x3(const at::Tensor& xp, const at::Tensor& w2d,
+                           long n_img, long oh, long ow, long cin,
+                           long stride, long HP, long WP) {
+  TORCH_CHECK(xp.is_cuda() && w2d.is_cuda());
+  TORCH_CHECK(xp.scalar_type() == at::kBFloat16 &&
+              w2d.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(xp.is_contiguous() && w2d.is_contiguous());
+  TORCH_CHECK((cin & (cin - 1)) == 0 && cin >= 64,
+              "px_tall_conv3x3: Cin must be a power of two >= 64");
+  const long cout = w2d.size(0);
+  const long K = w2d.size(1);
+  TORCH_CHECK(K == 9 * cin && K % 64 == 0);
+  const long M = n_img * oh * ow;
+  TORCH_CHECK((oh - 1) * stride + 3 <= HP && (ow - 1) * stride + 3 <= WP);
+  TORCH_CHECK(xp.numel() == n_img * HP * WP * cin);
+  int log2cin = 0;
+  while ((1 << log2cin) < cin) log2cin++;
+  auto c = at::empty({M, cout}, xp.options());
+  const int nbm = (int)((M + TL_BM - 1) / TL_BM);
+  const int nbn = (int)((cout + 63) / 64);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&px_tall_conv3x3_kernel),
+      hipFuncAttributeMaxDynamicSharedMemorySize, 81920);
+  px_tall_conv3x3_kernel<<<nbm * nbn, TL_THREADS, 81920, stream>>>(
+      reinterpret_cast<const ushort*>(xp.data_ptr()),
+      reinterpret_cast<const ushort*>(w2d.data_ptr()),
+      reinterpret_cast<ushort*>(c.data_ptr()), M, cout, K, nbm, nbn,
+      (int)oh, (int)ow, (int)WP, (int)HP, (int)cin, log2cin,
+      (int)stride);
+  HIP_CHECK_KERNEL();
+  return c;
+}
