@@ -90,6 +90,9 @@ at::Tensor maxpool3x3s2_bwd(const at::Tensor& dy, const at::Tensor& arg,
 at::Tensor gavgpool_fwd(const at::Tensor& x2d, long N, long HW, long C);
 at::Tensor gavgpool_bwd(const at::Tensor& dy, long N, long HW, long C);
 
+// reduce.hip
+at::Tensor col_sum(const at::Tensor& x);
+
 // gemm.hip
 at::Tensor pxgemm_bt(const at::Tensor& a, const at::Tensor& bt);
 at::Tensor pxgemm8_bt(const at::Tensor& a, const at::Tensor& bt);
@@ -206,6 +209,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("gavgpool_fwd", &gavgpool_fwd);
   m.def("gavgpool_bwd", &gavgpool_bwd);
+  m.def("col_sum", &col_sum,
+        "fp32 column sum of a bf16 [M, N] matrix (bias-grad shape)");
   m.def("px_tall_conv3x3", &px_tall_conv3x3,
         "implicit-GEMM 3x3 NHWC conv fwd (and stride-1 dgrad via "
         "rotated weights) on the tall-tile structure");

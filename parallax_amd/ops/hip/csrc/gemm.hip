@@ -1077,8 +1077,16 @@ void px_skinny_bt(const at::Tensor& a, const at::Tensor& bt,
   TORCH_CHECK(out.scalar_type() == at::kBFloat16 && out.numel() == M * N
               && out.is_contiguous() && (N % 4) == 0);
   const int ntiles = (int)(N / 64);
-  int spl = (int)(256 / ntiles);
   const int kt64 = (int)(K / 64);
+  // deep-K split measured SLOWER (859k vs 900k lm1b words/s,
+  // gpurun_out/lm_deep*) — the extra atomic traffic and cast-tile
+  // contention beat the shorter serial k-loop; keep the toggle for
+  // future shapes but default off
+  static const bool deep = [] {
+    const char* e = getenv("PARALLAX_SKINNY_DEEP");
+    return e != nullptr && e[0] == '1';
+  }();
+  int spl = (int)(((deep && kt64 >= 64) ? 512 : 256) / ntiles);
   if (spl > kt64) spl = kt64;
   if (spl < 1) spl = 1;
   const int kslice = (int)(((kt64 + spl - 1) / spl) * 64);

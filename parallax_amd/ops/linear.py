@@ -44,7 +44,15 @@ class _PxLinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dy @ weight
         dw = dy.t() @ x2d
-        db = dy.sum(0) if ctx.has_bias else None
+        if ctx.has_bias:
+            e = _ops.ext()
+            if (hasattr(e, 'col_sum') and dy.dtype == torch.bfloat16
+                    and dy.shape[1] % 8 == 0 and dy.shape[1] <= 2048):
+                db = e.col_sum(dy)
+            else:
+                db = dy.sum(0)
+        else:
+            db = None
         return dx, dw, db
 
 

@@ -635,6 +635,20 @@ def test_pxgemm8_bt_vs_matmul():
     assert float((c.float() - ref).abs().max()) < 0.25
 
 
+def test_col_sum_vs_torch():
+    e = _ext()
+    if not hasattr(e, 'col_sum'):
+        pytest.skip('col_sum not built')
+    torch.manual_seed(65)
+    for (M, N) in ((3072, 512), (2560, 2048), (100, 8), (7, 64)):
+        x = torch.randn(M, N, device=DEV).to(torch.bfloat16)
+        out = e.col_sum(x)
+        ref = x.float().sum(0)
+        assert out.dtype == torch.float32
+        assert torch.allclose(out, ref, atol=0.5, rtol=1e-3), \
+            (M, N, (out - ref).abs().max())
+
+
 def test_pool_nhwc_vs_torch():
     """NHWC max-pool 3x3/s2/p1 and global avg pool fwd/bwd vs torch."""
     from parallax_amd.ops.pool import (ParallaxMaxPool3x3s2,
