@@ -32,7 +32,9 @@ def run(N,C,H,W,add=False,iters=30):
         torch.cuda.synchronize()
         out[name]=(time.perf_counter()-t0)/iters*1e6
     mb=N*C*H*W*2/1e6
-    print('N%d C%4d %3dx%-3d add=%d  ours %8.1f us  miopen %8.1f us  (%.0f MB)'%(N,C,H,W,add,out['ours'],out['miopen'],mb),flush=True)
+    print('N%d C%4d %3dx%-3d add=%d  ours %8.1f us  miopen %8.1f us'
+          '  (%.0f MB)' % (N, C, H, W, add, out['ours'], out['miopen'],
+                           mb), flush=True)
 
 if __name__=='__main__':
     for (C,HW) in ((64,112),(256,56),(64,56),(512,28),(128,28),(1024,14),(256,14),(2048,7),(512,7)):
