@@ -329,6 +329,21 @@ static int bn_grid(long M, int C) {
   return (int)blocks;
 }
 
+// stats/bstats grid: ALSO proportional to the tensor, so the per-block
+// partial rows (written by the reduce, re-read by finalize) stay <= a
+// few % of the input bytes — a fixed 512-block grid put a ~15 us floor
+// under every small-layer BN (partials 512 x 2C fp32 vs a 3 MB input).
+static int bn_grid_stats(long M, int C) {
+  long blocks = M / 32;
+  if (blocks > 512) blocks = 512;
+  if (blocks < 16) blocks = 16;
+  const int rpb = 256 / (C >> 3);
+  long max_useful = (M + rpb - 1) / rpb;
+  if (blocks > max_useful) blocks = max_useful;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
 std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
                                       const at::Tensor& gamma,
                                       const at::Tensor& beta,
@@ -347,16 +362,17 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
               "batchnorm: bf16/fp16 activations only");
   auto f32 = x2d.options().dtype(at::kFloat);
   const int grid = bn_grid(M, C);
-  // one workspace: [grid][2C] partials + mean/invstd/scale/shift
-  auto ws = at::empty({(long)grid * 2 * C + 4 * C}, f32);
+  const int sgrid = bn_grid_stats(M, C);
+  // one workspace: [sgrid][2C] partials + mean/invstd/scale/shift
+  auto ws = at::empty({(long)sgrid * 2 * C + 4 * C}, f32);
   float* wsp = ws.data_ptr<float>();
   float* partp = wsp;
-  float* meanp = wsp + (long)grid * 2 * C;
+  float* meanp = wsp + (long)sgrid * 2 * C;
   float* invstdp = meanp + C;
   float* scp = invstdp + C;
   float* shp = scp + C;
-  auto mean = ws.narrow(0, (long)grid * 2 * C, C);
-  auto invstd = ws.narrow(0, (long)grid * 2 * C + C, C);
+  auto mean = ws.narrow(0, (long)sgrid * 2 * C, C);
+  auto invstd = ws.narrow(0, (long)sgrid * 2 * C + C, C);
   auto y = at::empty_like(x2d);
   auto g32 = gamma.to(at::kFloat).contiguous();
   auto b32 = beta.to(at::kFloat).contiguous();
@@ -373,12 +389,12 @@ std::vector<at::Tensor> batchnorm_fwd(const at::Tensor& x2d,
                       std::is_same<scalar_t, double>::value) {
           TORCH_CHECK(false, "batchnorm: bf16/fp16 only");
         } else {
-          bn_stats_kernel<scalar_t><<<grid, 256, 0, stream>>>(
+          bn_stats_kernel<scalar_t><<<sgrid, 256, 0, stream>>>(
               reinterpret_cast<const scalar_t*>(x2d.data_ptr()),
               partp, nullptr, M, C);
           HIP_CHECK_KERNEL();
           bn_finalize_kernel<<<(C + 3) / 4, 256, 0, stream>>>(
-              partp, grid, g32.data_ptr<float>(), b32.data_ptr<float>(),
+              partp, sgrid, g32.data_ptr<float>(), b32.data_ptr<float>(),
               meanp, invstdp, scp, shp,
               run_mean.has_value() ? run_mean->data_ptr<float>() : nullptr,
               run_var.has_value() ? run_var->data_ptr<float>() : nullptr,
@@ -423,12 +439,13 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
   const int C = (int)x2d.size(1);
   auto f32 = x2d.options().dtype(at::kFloat);
   const int grid = bn_grid(M, C);
-  auto ws = at::empty({(long)grid * 2 * C + 2 * C}, f32);
+  const int sgrid = bn_grid_stats(M, C);
+  auto ws = at::empty({(long)sgrid * 2 * C + 2 * C}, f32);
   float* partp = ws.data_ptr<float>();
-  float* sdyp = partp + (long)grid * 2 * C;
+  float* sdyp = partp + (long)sgrid * 2 * C;
   float* sdxp = sdyp + C;
-  auto sum_dy = ws.narrow(0, (long)grid * 2 * C, C);
-  auto sum_dy_xhat = ws.narrow(0, (long)grid * 2 * C + C, C);
+  auto sum_dy = ws.narrow(0, (long)sgrid * 2 * C, C);
+  auto sum_dy_xhat = ws.narrow(0, (long)sgrid * 2 * C + C, C);
   auto dx = at::empty_like(x2d);
   auto dxid = want_dxid ? at::empty_like(x2d) : at::Tensor();
   auto g32 = gamma.to(at::kFloat).contiguous();
@@ -447,17 +464,17 @@ std::vector<at::Tensor> batchnorm_bwd(const at::Tensor& dy2d,
           const scalar_t* yp =
               reinterpret_cast<const scalar_t*>(y2d.data_ptr());
           if (relu) {
-            bn_bstats_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+            bn_bstats_kernel<scalar_t, true><<<sgrid, 256, 0, stream>>>(
                 xp, dp, yp, mean.data_ptr<float>(),
                 invstd.data_ptr<float>(), partp, nullptr, M, C);
           } else {
-            bn_bstats_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+            bn_bstats_kernel<scalar_t, false><<<sgrid, 256, 0, stream>>>(
                 xp, dp, yp, mean.data_ptr<float>(),
                 invstd.data_ptr<float>(), partp, nullptr, M, C);
           }
           HIP_CHECK_KERNEL();
           bn_breduce_kernel<<<(C + 3) / 4, 256, 0, stream>>>(
-              partp, grid, sdyp, sdxp, C);
+              partp, sgrid, sdyp, sdxp, C);
           HIP_CHECK_KERNEL();
           scalar_t* didp = want_dxid
               ? reinterpret_cast<scalar_t*>(dxid.data_ptr()) : nullptr;

@@ -1023,45 +1023,6 @@ void px_skinny_kernel(const ushort* __restrict__ A,
 #undef SK_LDSB
 }
 
-// out = bf16(WS) [+ addend]; WS = 0  (vectorized 4-wide)
-__global__ __launch_bounds__(256)
-void px_castzero_kernel(float* __restrict__ WS, ushort* __restrict__ out,
-                        const ushort* __restrict__ addend, long add_rs,
-                        long M, long N) {
-  const long total4 = M * N / 4;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total4;
-       i += (long)gridDim.x * blockDim.x) {
-    const long e = i * 4;
-    float4 w = *reinterpret_cast<float4*>(&WS[e]);
-    if (addend != nullptr) {
-      const long row = e / N, col = e - row * N;
-      const ushort* ap = addend + row * add_rs + col;
-      w.x += __bfloat162float(
-          *reinterpret_cast<const __hip_bfloat16*>(ap));
-      w.y += __bfloat162float(
-          *reinterpret_cast<const __hip_bfloat16*>(ap + 1));
-      w.z += __bfloat162float(
-          *reinterpret_cast<const __hip_bfloat16*>(ap + 2));
-      w.w += __bfloat162float(
-          *reinterpret_cast<const __hip_bfloat16*>(ap + 3));
-    }
-    ushort o[4];
-    __hip_bfloat16 b0 = __float2bfloat16(w.x);
-    __hip_bfloat16 b1 = __float2bfloat16(w.y);
-    __hip_bfloat16 b2 = __float2bfloat16(w.z);
-    __hip_bfloat16 b3 = __float2bfloat16(w.w);
-    o[0] = *reinterpret_cast<ushort*>(&b0);
-    o[1] = *reinterpret_cast<ushort*>(&b1);
-    o[2] = *reinterpret_cast<ushort*>(&b2);
-    o[3] = *reinterpret_cast<ushort*>(&b3);
-    *reinterpret_cast<uint2*>(&out[e]) =
-        *reinterpret_cast<uint2*>(&o[0]);
-    float4 z;
-    z.x = z.y = z.z = z.w = 0.f;
-    *reinterpret_cast<float4*>(&WS[e]) = z;
-  }
-}
-
 void px_skinny_bt(const at::Tensor& a, const at::Tensor& bt,
                   at::Tensor ws, at::Tensor cnt, at::Tensor out,
                   const c10::optional<at::Tensor>& addend) {
