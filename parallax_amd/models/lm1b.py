@@ -52,13 +52,23 @@ class LM1BModel(nn.Module):
         from parallax_amd.tagging import mark_sparse
         mark_sparse(self.softmax_b)
 
+    # reference keep_prob = 0.9 during training (language_model.py:24):
+    # dropout on the embedded inputs (:69) and on each timestep's
+    # projected output before the softmax (:87 — the RECURRENT h stays
+    # un-dropped, so one dropout over the stacked outputs is equivalent)
+    dropout_p = 0.1
+
     def forward(self, x, y, state=None):
         """x, y: [B, T] int64.  Returns (mean loss per word, final state)."""
         emb = self.emb(x)                      # [B, T, D]
         body = self.lstm.w.dtype               # bf16 in pure-bf16 mode
         if emb.dtype != body and not torch.is_autocast_enabled():
             emb = emb.to(body)
+        if self.training and self.dropout_p > 0:
+            emb = torch.nn.functional.dropout(emb, self.dropout_p)
         hs, state = self.lstm(emb, state)      # [B, T, P]
+        if self.training and self.dropout_p > 0:
+            hs = torch.nn.functional.dropout(hs, self.dropout_p)
         flat_h = hs.reshape(-1, hs.shape[-1])
         flat_y = y.reshape(-1)
         losses = sampled_softmax_loss(self.softmax_w, self.softmax_b,

@@ -72,8 +72,9 @@ class _MHA(nn.Module):
 
 
 class _Block(nn.Module):
-    def __init__(self, d, heads, ffn, cross=False):
+    def __init__(self, d, heads, ffn, cross=False, dropout=0.1):
         super().__init__()
+        self.p = dropout
         self.ln1 = ParallaxLayerNorm(d)
         self.attn = _MHA(d, heads)
         self.cross = cross
@@ -84,11 +85,18 @@ class _Block(nn.Module):
         self.fc1 = PxLinear(d, ffn)   # K=512: pxgemm wins
         self.fc2 = nn.Linear(ffn, d)  # K=2048: hipBLASLt wins
 
+    def _drop(self, t):
+        if self.training and self.p > 0:
+            return F.dropout(t, self.p)
+        return t
+
     def forward(self, x, mem=None, causal=False):
-        x = x + self.attn(self.ln1(x), causal=causal)
+        # residual dropout (the reference NMT's dropout=0.2-class
+        # regularization; standard transformer placement)
+        x = x + self._drop(self.attn(self.ln1(x), causal=causal))
         if self.cross:
-            x = x + self.xattn(self.ln_x(x), mem=mem)
-        x = x + self.fc2(F.gelu(self.fc1(self.ln2(x))))
+            x = x + self._drop(self.xattn(self.ln_x(x), mem=mem))
+        x = x + self._drop(self.fc2(F.gelu(self.fc1(self.ln2(x)))))
         return x
 
 
@@ -204,8 +212,13 @@ class GNMTAttention(nn.Module):
         self.units = units
         self.tgt_vocab = tgt_vocab
 
+    dropout_p = 0.2   # reference nmt hparams dropout default
+
     def forward(self, src, tgt_in, tgt_out):
-        enc_out, (h, c) = self.encoder(self.src_emb(src))
+        semb = self.src_emb(src)
+        if self.training and self.dropout_p > 0:
+            semb = torch.nn.functional.dropout(semb, self.dropout_p)
+        enc_out, (h, c) = self.encoder(semb)
         B, S, U = enc_out.shape
         T = tgt_in.shape[1]
         emb = self.tgt_emb(tgt_in)
