@@ -61,6 +61,10 @@ class ParallaxBatchNorm2d(nn.Module):
         self.register_buffer('running_var', torch.ones(num_features))
         self.register_buffer('num_batches_tracked',
                              torch.tensor(0, dtype=torch.long))
+        # forward() counts host-side (one CUDA add kernel per BN per
+        # step otherwise — 53 launches/step in ResNet-50); the pending
+        # count is flushed into the buffer at state_dict time
+        self._nbt_pending = 0
 
     @classmethod
     def from_bn(cls, bn, relu=False):
@@ -78,12 +82,25 @@ class ParallaxBatchNorm2d(nn.Module):
                 and x.dtype in (torch.bfloat16, torch.float16)
                 and c % 8 == 0 and c <= 2048)
 
+    def _flush_nbt(self):
+        if self._nbt_pending:
+            self.num_batches_tracked += self._nbt_pending
+            self._nbt_pending = 0
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        self._flush_nbt()
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+
+    def _load_from_state_dict(self, *args, **kw):
+        self._nbt_pending = 0
+        super()._load_from_state_dict(*args, **kw)
+
     def forward(self, x, addend=None):
         """addend: optional residual tensor fused as
         y = [relu](bn(x) + addend) — the bottleneck's bn3 + shortcut."""
         N, C, H, W = x.shape
         if self._native_ok(x):
-            self.num_batches_tracked += 1
+            self._nbt_pending += 1
             x2d = (x.permute(0, 2, 3, 1).reshape(N * H * W, C)
                    .contiguous())
             a2d = None
@@ -94,6 +111,8 @@ class ParallaxBatchNorm2d(nn.Module):
                             self.running_mean, self.running_var,
                             self.momentum, self.eps, self.relu, a2d)
             return y.view(N, H, W, C).permute(0, 3, 1, 2)
+        if self.training:
+            self._nbt_pending += 1
         y = torch.nn.functional.batch_norm(
             x, self.running_mean, self.running_var, self.weight,
             self.bias, self.training, self.momentum, self.eps)
