@@ -233,30 +233,38 @@ __global__ void layernorm_bwd_lds_kernel(
     }
   }
   __syncthreads();
-  float* dwb = dw32 + (blockIdx.x % LN_NBUF) * d;
-  float* dbb = db32 + (blockIdx.x % LN_NBUF) * d;
+  // per-block partial rows, plain stores (nothing pre-zeroed, no atomic
+  // contention; a wave-per-column reduce folds them + casts)
+  float* dwb = dw32 + (long)blockIdx.x * 2 * d;
   for (long c = threadIdx.x; c < d; c += blockDim.x) {
-    atomicAdd(&dwb[c], sdw[c]);
-    atomicAdd(&dbb[c], sdb[c]);
+    dwb[c] = sdw[c];
+    dwb[d + c] = sdb[c];
   }
+  (void)db32;
 }
 
-// Fold the LN_NBUF stripes and cast to the weight dtype in one launch.
+// Reduce the [nblk][2d] partials (one wave per output column) and cast
+// to the weight dtype in one launch.
 template <typename T>
-__global__ void fold_cast2_kernel(const float* __restrict__ a32,
-                                  const float* __restrict__ b32,
-                                  T* __restrict__ a, T* __restrict__ b,
-                                  long d) {
-  for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < d;
-       c += (long)gridDim.x * blockDim.x) {
-    float sa = 0.f, sb = 0.f;
-    for (int k = 0; k < LN_NBUF; ++k) {
-      sa += a32[k * d + c];
-      sb += b32[k * d + c];
-    }
-    a[c] = from_f32<T>(sa);
-    b[c] = from_f32<T>(sb);
+__global__ __launch_bounds__(256)
+void fold_cast2_kernel(const float* __restrict__ part, int nblk,
+                       T* __restrict__ a, T* __restrict__ b, long d) {
+  const long c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (c >= d) return;
+  const int lane = threadIdx.x & 63;
+  float sa = 0.f, sb = 0.f;
+  for (int k = lane; k < nblk; k += 64) {
+    sa += part[(long)k * 2 * d + c];
+    sb += part[(long)k * 2 * d + d + c];
   }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    sa += __shfl_down(sa, off, 64);
+    sb += __shfl_down(sb, off, 64);
+  }
+  if (lane != 0) return;
+  a[c] = from_f32<T>(sa);
+  b[c] = from_f32<T>(sb);
 }
 
 template <typename T>
@@ -282,11 +290,16 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
   auto db = at::empty_like(w);
   auto stream = at::cuda::getCurrentCUDAStream();
   if (d <= LN_LDS_MAX_D) {
-    const int grid = (int)(n < 512 ? (n > 0 ? n : 1) : 512);
-    // one allocation + one fill for both stripe buffers
-    auto ws = at::zeros({2, LN_NBUF * d}, x.options().dtype(at::kFloat));
-    auto dw32 = ws[0];
-    auto db32 = ws[1];
+    long g = n / 8;
+    if (g > 512) g = 512;
+    if (g < 16) g = 16;
+    if (g > n) g = n > 0 ? n : 1;
+    const int grid = (int)g;
+    // per-block partials, nothing to pre-zero
+    auto ws = at::empty({(long)grid * 2 * d},
+                        x.options().dtype(at::kFloat));
+    auto dw32 = ws;
+    auto db32 = ws;
     const size_t lds = 2 * (size_t)d * sizeof(float);
     auto wc = w.contiguous();
     const bool vec = (d % 4 == 0) && ln_aligned16(dy) &&
@@ -306,8 +319,9 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
           if (vec) launch(std::true_type{});
           else launch(std::false_type{});
           HIP_CHECK_KERNEL();
-          fold_cast2_kernel<scalar_t><<<px_grid(d), PX_BLOCK, 0, stream>>>(
-              dw32.data_ptr<float>(), db32.data_ptr<float>(),
+          fold_cast2_kernel<scalar_t>
+              <<<(int)((d + 3) / 4), 256, 0, stream>>>(
+              dw32.data_ptr<float>(), grid,
               reinterpret_cast<scalar_t*>(dw.data_ptr()),
               reinterpret_cast<scalar_t*>(db.data_ptr()), d);
           HIP_CHECK_KERNEL();
