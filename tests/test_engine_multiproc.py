@@ -797,3 +797,12 @@ def _body_padded_exchange_equivalence(rank, world):
 
 def test_padded_exchange_world2():
     _spawn(_body_padded_exchange_equivalence, get_free_port())
+
+
+def test_world8_full_stack(free_port):
+    """Scaling-path insurance at the driver's maximum world size: the
+    hybrid engine stack (dense buckets + sparse PS exchange + optimizer)
+    at world_size=8 on gloo, tiny model — exercises bucket membership,
+    padded-count verification and collective ordering at the exact rank
+    count of the round-end 8-GPU run."""
+    _spawn(_body_parallel_run_hybrid, free_port, world=8)
