@@ -425,7 +425,7 @@ void pxgemm8_bt_kernel(const ushort* __restrict__ A,
   } while (0)
 
   bf16x8 a_frag[8];
-  if (V == 2) {
+  if (V == 2 || V == 3) {
     // ---- software-pipelined fragment reads (one phase ahead) --------
     // Phase quadrant = (k-half, m-half): B fragments are read once per
     // k-half (4 regs per set), A alternates between two 4-frag banks
@@ -453,6 +453,48 @@ void pxgemm8_bt_kernel(const ushort* __restrict__ A,
     // prologue reads: tile-0 k-half-0 first A bank + B set
     P8_RD_A4(a0, 0, 0);
     P8_RD_B4(b0, 0);
+    if (V == 3) {
+      // single barrier per phase: the stage moves AFTER the barrier
+      // (so the previous phase's LDS reads are collectively retired
+      // before any wave's DMA can overwrite the old buffer), halving
+      // the barrier count; counted waits shrink to vmcnt(2)
+      for (int kt = 0; kt < ktiles; ++kt) {
+        const bool last = (kt == ktiles - 1);
+        // P1 (kh0, mh0)
+        __builtin_amdgcn_s_barrier();
+        if (!last) P8_STAGE_PHASE(kt + 1, 0);
+        P8_RD_A4(a1, 0, 4);
+        P8_MM4(a0, b0, 0);
+        // P2 (kh0, mh1): publish kh1
+        if (!last) {
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+        if (!last) P8_STAGE_PHASE(kt + 1, 1);
+        P8_RD_A4(a0, 1, 0);
+        P8_RD_B4(b1, 1);
+        P8_MM4(a1, b0, 1);
+        // P3 (kh1, mh0)
+        __builtin_amdgcn_s_barrier();
+        if (!last) P8_STAGE_PHASE(kt + 1, 2);
+        P8_RD_A4(a1, 1, 4);
+        P8_MM4(a0, b1, 0);
+        // P4 (kh1, mh1): publish next tile's kh0
+        if (!last) {
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+        if (!last) P8_STAGE_PHASE(kt + 1, 3);
+        cur ^= 1;
+        if (!last) {
+          P8_RD_A4(a0, 0, 0);
+          P8_RD_B4(b0, 0);
+        }
+        P8_MM4(a1, b1, 1);
+      }
+    } else {
     for (int kt = 0; kt < ktiles; ++kt) {
       const bool last = (kt == ktiles - 1);
       // P1 (kh0, mh0): consume a0,b0; read A(kh0, i4..7) -> a1
@@ -492,6 +534,7 @@ void pxgemm8_bt_kernel(const ushort* __restrict__ A,
       }
       P8_MM4(a1, b1, 1);
       __builtin_amdgcn_s_barrier();
+    }
     }
 #undef P8_RD_A4
 #undef P8_RD_B4
@@ -577,6 +620,9 @@ at::Tensor pxgemm8_bt_addv(const at::Tensor& a, const at::Tensor& bt,
   hipFuncSetAttribute(
       reinterpret_cast<const void*>(&pxgemm8_bt_kernel<2>),
       hipFuncAttributeMaxDynamicSharedMemorySize, smem);
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&pxgemm8_bt_kernel<3>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, smem);
   const ushort* add_ptr = nullptr;
   long add_rs = 0;
   if (addend.has_value()) {
@@ -597,6 +643,8 @@ at::Tensor pxgemm8_bt_addv(const at::Tensor& a, const at::Tensor& bt,
     P8_LAUNCH(1);
   } else if (variant == 2) {
     P8_LAUNCH(2);
+  } else if (variant == 3) {
+    P8_LAUNCH(3);
   } else {
     P8_LAUNCH(0);
   }

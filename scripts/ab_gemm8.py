@@ -24,6 +24,7 @@ def run(M, N, K, rounds=9, iters=20):
         'v0-8ph': lambda: e.pxgemm8_bt_v(a, b, 0),
         'v1-merged': lambda: e.pxgemm8_bt_v(a, b, 1),
         'v2-pipe': lambda: e.pxgemm8_bt_v(a, b, 2),
+        'v3-1bar': lambda: e.pxgemm8_bt_v(a, b, 3),
         'tall': lambda: e.px_tall_bt(a, b),
         'hipblaslt': lambda: a @ b.t(),
     }
