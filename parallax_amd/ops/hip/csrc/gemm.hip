@@ -656,7 +656,7 @@ at::Tensor pxgemm8_bt_addv(const at::Tensor& a, const at::Tensor& bt,
 static long p8_default_variant() {
   static long v = [] {
     const char* e = getenv("PARALLAX_GEMM8_VARIANT");
-    return e ? atol(e) : 2L;  // measured best (ab_gemm8_v3.log)
+    return e ? atol(e) : 3L;  // measured best (gpurun_out/ab_v3.log)
   }();
   return v;
 }

@@ -38,7 +38,7 @@ def test_fuzz_gemm_family():
             c = getattr(e, name)(a, b)
             rel = float((c.float() - ref).abs().max()) / den
             assert rel < 0.02, (name, M, N, K, rel)
-        for v in (0, 1, 2):
+        for v in (0, 1, 2, 3):
             c = e.pxgemm8_bt_v(a, b, v)
             rel = float((c.float() - ref).abs().max()) / den
             assert rel < 0.02, ('v%d' % v, M, N, K, rel)
