@@ -187,9 +187,22 @@ class _Conv3x3Fn(torch.autograd.Function):
             dx2 = _conv3x3_gemm(dyp.contiguous(), wrot, N, H, W, cout,
                                 1, H + 2, W + 2)
             dx = (dx2.view(N, H, W, C).permute(0, 3, 1, 2))
-            _, dw, _ = torch.ops.aten.convolution_backward(
-                dyc, x, weight.to(x.dtype), None, [s, s], [1, 1],
-                [1, 1], False, [0, 0], 1, [False, True, False])
+            e = _ops.ext()
+            if hasattr(e, 'px_wgrad3x3') and cout % 8 == 0:
+                # native wgrad: LDS-transposed implicit GEMM over the
+                # re-padded input; fp32 accumulate, one cast at the end
+                oh = ow = H  # stride-1 branch: output size == input
+                xp = torch.nn.functional.pad(
+                    x.permute(0, 2, 3, 1), (0, 0, 1, 1, 1, 1))
+                dy2 = (dy_nhwc.reshape(N * H * W, cout).contiguous())
+                ws = e.px_wgrad3x3(dy2, xp.contiguous(), N, oh, ow, C,
+                                   1, H + 2, W + 2)
+                dw = (ws.view(cout, 3, 3, C).permute(0, 3, 1, 2)
+                      .to(weight.dtype))
+            else:
+                _, dw, _ = torch.ops.aten.convolution_backward(
+                    dyc, x, weight.to(x.dtype), None, [s, s], [1, 1],
+                    [1, 1], False, [0, 0], 1, [False, True, False])
         else:
             dx, dw, _ = torch.ops.aten.convolution_backward(
                 dy.to(x.dtype), x, weight.to(x.dtype), None, [s, s],

@@ -103,6 +103,9 @@ at::Tensor pxgemm8_bt_add(const at::Tensor& a, const at::Tensor& bt,
 at::Tensor pxgemm_bt_add(const at::Tensor& a, const at::Tensor& bt,
                          const c10::optional<at::Tensor>& addend);
 at::Tensor px_tall_bt(const at::Tensor& a, const at::Tensor& bt);
+at::Tensor px_wgrad3x3(const at::Tensor& dy2d, const at::Tensor& xp,
+                       long n_img, long oh, long ow, long cin,
+                       long stride, long HP, long WP);
 at::Tensor px_tall_conv3x3(const at::Tensor& xp, const at::Tensor& w2d,
                            long n_img, long oh, long ow, long cin,
                            long stride, long HP, long WP);
@@ -211,6 +214,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gavgpool_bwd", &gavgpool_bwd);
   m.def("col_sum", &col_sum,
         "fp32 column sum of a bf16 [M, N] matrix (bias-grad shape)");
+  m.def("px_wgrad3x3", &px_wgrad3x3,
+        "implicit-GEMM 3x3 NHWC conv wgrad (LDS-transposed staging, "
+        "split-M fp32 atomic accumulation); returns fp32 [Cout, 9*Cin]");
   m.def("px_tall_conv3x3", &px_tall_conv3x3,
         "implicit-GEMM 3x3 NHWC conv fwd (and stride-1 dgrad via "
         "rotated weights) on the tall-tile structure");

@@ -1451,3 +1451,152 @@ at::Tensor px_tall_conv3x3(const at::Tensor& xp, const at::Tensor& w2d,
   HIP_CHECK_KERNEL();
   return c;
 }
+
+// ---------------------------------------------------------------------------
+// 3x3 NHWC conv WGRAD as implicit GEMM with LDS-transposed staging:
+// dW[co][kc] += sum_m dy[m][co] * im2col[m][kc]  (m = N*OH*OW rows).
+// Both operands are m-major in memory but the MFMA fragments need
+// k(=m)-contiguous reads, so each 64-row m-tile is staged TRANSPOSED
+// into LDS (scatter b16 writes, padded 144-B rows -> 2-way banks, the
+// attention-loader pattern).  The m axis splits over blocks; partial
+// tiles accumulate with fp32 global atomics into a zeroed workspace
+// (the caller casts once).  Each 64-wide kc tile lies inside ONE of the
+// nine taps (Cin % 64 == 0), so its im2col rows are contiguous slices
+// of the padded input.
+// ---------------------------------------------------------------------------
+
+#define WG_LDP 72   // padded LDS row stride (elements)
+
+__global__ __launch_bounds__(256, 4)
+void px_wgrad3x3_kernel(const ushort* __restrict__ DY,  // [M, Cout]
+                        const ushort* __restrict__ XP,  // padded NHWC
+                        float* __restrict__ WS,         // [Cout, 9*Cin]
+                        long M, long Cout, long K9, int OH, int OW,
+                        int WP, int HP, int Cin, int log2cin, int stride,
+                        int nco, int nkc, int splitm, int mtiles) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  ushort* dyT = reinterpret_cast<ushort*>(smem);            // [64][WG_LDP]
+  ushort* xT = dyT + 64 * WG_LDP;                           // [64][WG_LDP]
+  const int bid = blockIdx.x;
+  const int co0 = (bid % nco) * 64;
+  const int kc0 = ((bid / nco) % nkc) * 64;
+  const int ms = bid / (nco * nkc);          // m-slice index
+  const int tps = (mtiles + splitm - 1) / splitm;
+  const int mt0 = ms * tps;
+  int mt1 = mt0 + tps;
+  if (mt1 > mtiles) mt1 = mtiles;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int fr = lane & 15;
+  const int ko = (lane >> 4) * 8;
+  // wave owns co rows wave*16..wave*16+15 x all 64 kc
+  f32x4 acc[4] = {};
+
+  // kc tile -> tap + channel base (whole tile inside one tap)
+  const int dydx = kc0 >> log2cin;
+  const int tdy = dydx >= 6 ? 2 : (dydx >= 3 ? 1 : 0);
+  const int tdx = dydx - 3 * tdy;
+  const int ci0 = kc0 - (dydx << log2cin);
+
+  for (int mt = mt0; mt < mt1; ++mt) {
+    const long m0 = (long)mt * 64;
+    // stage both tiles transposed: 2 rounds x 256 threads x 8 elems
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      const int idx = tid + r * 256;
+      const int mm = idx & 63;
+      const int cg = (idx >> 6) * 8;
+      const long m = m0 + mm;
+      uint2 dv, xv;
+      if (m < M) {
+        dv = *reinterpret_cast<const uint2*>(&DY[m * Cout + co0 + cg]);
+        const long n_img = m / ((long)OH * OW);
+        const long rem = m - n_img * ((long)OH * OW);
+        const int oy = (int)(rem / OW);
+        const int ox = (int)(rem - (long)oy * OW);
+        const ushort* src = XP +
+            (((n_img * HP) + (long)oy * stride + tdy) * WP +
+             (long)ox * stride + tdx) * Cin + ci0 + cg;
+        xv = *reinterpret_cast<const uint2*>(src);
+      } else {
+        dv.x = dv.y = 0u;
+        xv.x = xv.y = 0u;
+      }
+      const ushort* de = reinterpret_cast<const ushort*>(&dv);
+      const ushort* xe = reinterpret_cast<const ushort*>(&xv);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        dyT[(cg + j) * WG_LDP + mm] = de[j];
+        xT[(cg + j) * WG_LDP + mm] = xe[j];
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 a_frag, b_frag[4];
+      a_frag = *(const bf16x8*)(&dyT[(wave * 16 + fr) * WG_LDP +
+                                     kk * 32 + ko]);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        b_frag[j] = *(const bf16x8*)(&xT[(j * 16 + fr) * WG_LDP +
+                                         kk * 32 + ko]);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag[j], acc[j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long co = co0 + wave * 16 + crow0 + r;
+      const long kc = kc0 + j * 16 + fr;
+      if (co >= Cout) continue;
+      atomicAdd(&WS[co * K9 + kc], acc[j][r]);
+    }
+  }
+}
+
+at::Tensor px_wgrad3x3(const at::Tensor& dy2d, const at::Tensor& xp,
+                       long n_img, long oh, long ow, long cin,
+                       long stride, long HP, long WP) {
+  TORCH_CHECK(dy2d.is_cuda() && xp.is_cuda());
+  TORCH_CHECK(dy2d.scalar_type() == at::kBFloat16 &&
+              xp.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dy2d.dim() == 2 && dy2d.is_contiguous() &&
+              xp.is_contiguous());
+  const long M = n_img * oh * ow;
+  const long Cout = dy2d.size(1);
+  TORCH_CHECK(dy2d.size(0) == M && Cout % 8 == 0);
+  TORCH_CHECK((cin & (cin - 1)) == 0 && cin >= 64);
+  TORCH_CHECK(xp.numel() == n_img * HP * WP * cin);
+  const long K9 = 9 * cin;
+  int log2cin = 0;
+  while ((1 << log2cin) < cin) log2cin++;
+  auto ws = at::zeros({Cout, K9},
+                      dy2d.options().dtype(at::kFloat));
+  const int nco = (int)((Cout + 63) / 64);
+  const int nkc = (int)(K9 / 64);
+  const int mtiles = (int)((M + 63) / 64);
+  int splitm = (int)(384 / (nco * nkc));
+  if (splitm > mtiles) splitm = mtiles;
+  if (splitm < 1) splitm = 1;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int smem = 2 * 64 * WG_LDP * 2;
+  px_wgrad3x3_kernel<<<nco * nkc * splitm, 256, smem, stream>>>(
+      reinterpret_cast<const ushort*>(dy2d.data_ptr()),
+      reinterpret_cast<const ushort*>(xp.data_ptr()),
+      ws.data_ptr<float>(), M, Cout, K9, (int)oh, (int)ow, (int)WP,
+      (int)HP, (int)cin, log2cin, (int)stride, nco, nkc, splitm, mtiles);
+  HIP_CHECK_KERNEL();
+  return ws;   // fp32 [Cout, 9*Cin]; caller casts/reshapes
+}
