@@ -1,29 +1,23 @@
-"""NHWC 1x1 convolution on the hand-written MFMA GEMM.
+"""NHWC convolutions on the hand-written MFMA GEMM family
+(SURVEY §2.4 conv worklist).
 
-The north-star conv worklist entry: ResNet's bottleneck
-blocks are dominated by 1x1 convolutions, which in NHWC are exactly
-``out[N*H*W, Cout] = act[N*H*W, Cin] @ W[Cout, Cin]^T`` — the pxgemm_bt
-kernel (ops/hip/csrc/gemm.hip), which measures FASTER than the library
-path at the ResNet shapes (e.g. 56x56 stage: 26.9 us vs 32.0 us).
+1x1: fwd + dgrad on the tile-dispatched GEMMs (_best_gemm), wgrad on
+MIOpen wrw (hipBLASLt picks a ~5x-slower TN split-K schedule there).
+3x3: complete implicit-GEMM path — fwd and stride-1 dgrad on the
+tall-tile kernel over a host-padded halo (dgrad = same kernel on
+pad(dy) with the rotated/transposed weight), wgrad on px_wgrad3x3
+(LDS-transposed staging + split-M fp32 atomics).
 
-Forward runs the hand-written kernel; backward uses library GEMMs (dgrad/
-wgrad are large-K shapes where hipBLASLt is already optimal).  3x3 and
-strided convolutions stay on MIOpen (implicit-GEMM staging for those is
-round-2 work); enable in ResNet with PARALLAX_NATIVE_CONV1X1=1.
+Both stay OPT-IN (PARALLAX_NATIVE_CONV1X1/3X3=1): numerics-tested and
+isolated-faster on the big GEMM shapes, but the exhaustive-find-tuned
+MIOpen conv suite still wins end-to-end ResNet (honest A/Bs in
+doc/design.md).
 """
 
 import torch
 import torch.nn as nn
 
 from parallax_amd import ops as _ops
-
-
-def _pad_rows(x, mult):
-    m = x.shape[0]
-    pad = (-m) % mult
-    if pad == 0:
-        return x, m
-    return torch.nn.functional.pad(x, (0, 0, 0, pad)), m
 
 
 def _best_gemm(e, a, b):
