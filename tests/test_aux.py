@@ -167,6 +167,9 @@ def test_bench_contract_json_line(tmp_path):
     assert d['scaling'] == 'weak' and d['higher_is_better'] is True
     assert d['data'] == 'synthetic'
     assert d['config']['vocab'] == 793470
+    assert d['config']['precision'] in ('fp32', 'pure_bf16',
+                                        'autocast_bf16')
+    assert d['dtype'] in ('fp32', 'bf16')
     assert d['value'] > 0
 
 
