@@ -76,7 +76,20 @@ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ s1,
   float a1[8] = {}, a2[8] = {};
   if (rsub < rpb) {
     const long rstep = (long)gridDim.x * rpb;
-    for (long r = (long)blockIdx.x * rpb + rsub; r < M; r += rstep) {
+    long r = (long)blockIdx.x * rpb + rsub;
+    // 2 rows in flight per thread (independent loads hide latency)
+    for (; r + rstep < M; r += 2 * rstep) {
+      const short8v v0 = *(const short8v*)(x + r * C + cg);
+      const short8v v1 = *(const short8v*)(x + (r + rstep) * C + cg);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float f0 = bn_elt<T>(v0, k);
+        const float f1 = bn_elt<T>(v1, k);
+        a1[k] += f0 + f1;
+        a2[k] += f0 * f0 + f1 * f1;
+      }
+    }
+    for (; r < M; r += rstep) {
       const short8v v = *(const short8v*)(x + r * C + cg);
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
@@ -214,7 +227,31 @@ void bn_bstats_kernel(const T* __restrict__ x, const T* __restrict__ dy,
       is[k] = invstd[cg + k];
     }
     const long rstep = (long)gridDim.x * rpb;
-    for (long r = (long)blockIdx.x * rpb + rsub; r < M; r += rstep) {
+    long r = (long)blockIdx.x * rpb + rsub;
+    for (; r + rstep < M; r += 2 * rstep) {
+      const long b0 = r * C + cg;
+      const long b1 = (r + rstep) * C + cg;
+      const short8v xv0 = *(const short8v*)(x + b0);
+      const short8v dv0 = *(const short8v*)(dy + b0);
+      const short8v xv1 = *(const short8v*)(x + b1);
+      const short8v dv1 = *(const short8v*)(dy + b1);
+      short8v yv0, yv1;
+      if (RELU) {
+        yv0 = *(const short8v*)(y + b0);
+        yv1 = *(const short8v*)(y + b1);
+      }
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float d0 = bn_elt<T>(dv0, k);
+        float d1 = bn_elt<T>(dv1, k);
+        if (RELU && !(bn_elt<T>(yv0, k) > 0.f)) d0 = 0.f;
+        if (RELU && !(bn_elt<T>(yv1, k) > 0.f)) d1 = 0.f;
+        a1[k] += d0 + d1;
+        a2[k] += d0 * (bn_elt<T>(xv0, k) - mn[k]) * is[k] +
+                 d1 * (bn_elt<T>(xv1, k) - mn[k]) * is[k];
+      }
+    }
+    for (; r < M; r += rstep) {
       const long base = r * C + cg;
       const short8v xv = *(const short8v*)(x + base);
       const short8v dv = *(const short8v*)(dy + base);
