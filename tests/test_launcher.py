@@ -27,3 +27,53 @@ def test_simple_driver_end_to_end(tmp_path):
     assert 'final loss:' in proc.stdout
     final = float(proc.stdout.strip().rsplit('final loss:', 1)[1])
     assert final < 1.0  # loss decreased from the random-init value
+
+
+def _run_driver(path, extra, tmp_path, timeout=420):
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    for k in ('WORLD_SIZE', 'RANK', 'LOCAL_RANK', 'MASTER_ADDR',
+              'MASTER_PORT', 'PARALLAX_RESOURCE_INFO'):
+        env.pop(k, None)
+    ri = tmp_path / 'resource_info'
+    ri.write_text('127.0.0.1:\n')
+    env['PYTHONPATH'] = REPO
+    out = subprocess.run(
+        [sys.executable, path, '--resource_info', str(ri),
+         '--max_steps', '2'] + extra,
+        capture_output=True, text=True, timeout=timeout, env=env,
+        cwd=REPO)
+    assert out.returncode == 0, (path, out.stderr[-3000:])
+    return out
+
+
+def test_lm1b_driver_smoke(tmp_path):
+    """Reference-parity driver smoke (lm1b_distributed_driver.py
+    synthetic mode) through the real master->worker launch path."""
+    _run_driver(os.path.join(REPO, 'examples', 'lm1b',
+                             'lm1b_distributed_driver.py'),
+                ['--batch_size', '8', '--num_steps', '4',
+                 '--vocab_size', '1000', '--log_frequency', '1'],
+                tmp_path)
+
+
+def test_skip_thoughts_driver_smoke(tmp_path):
+    _run_driver(os.path.join(REPO, 'examples', 'skip_thoughts',
+                             'skip_thoughts_driver.py'),
+                ['--batch_size', '4', '--seq_len', '6',
+                 '--log_frequency', '1'], tmp_path)
+
+
+def test_cnn_driver_smoke(tmp_path):
+    _run_driver(os.path.join(REPO, 'examples', 'cnn',
+                             'cnn_distributed_driver.py'),
+                ['--model', 'trivial', '--batch_size', '4',
+                 '--log_frequency', '1'], tmp_path)
+
+
+def test_nmt_driver_smoke(tmp_path):
+    _run_driver(os.path.join(REPO, 'examples', 'nmt',
+                             'nmt_distributed_driver.py'),
+                ['--batch_size', '4', '--src_len', '6', '--tgt_len', '6',
+                 '--log_frequency', '1'], tmp_path)
