@@ -172,11 +172,6 @@ MODEL_REGISTRY = {
 }
 
 
-# native input resolutions (reference model classes: overfeat_model.py:32
-# image size 231, inception_model.py 299; everything else 224)
-MODEL_IMAGE_SIZE = {'inception3': 299, 'overfeat': 231}
-
-
 def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
                     optimizer_name='momentum', lr=0.1, momentum=0.9,
                     weight_decay=1e-4, use_bf16=True, channels_last=True,

@@ -24,7 +24,7 @@ for name in names:
             lambda: M.build_cnn_graph(model_name=name, batch_size=64),
             None, parallax_config=ParallaxConfig(run_option='MPI'))
         dev = sess.comm.device
-        isz = M.MODEL_IMAGE_SIZE.get(name, 224)
+        isz = M.image_size_for(name)
         b = R.synthetic_batch(64, image_size=isz, device=dev)
         for _ in range(8):
             sess.run(b)
