@@ -1,25 +1,23 @@
 """Fused short-sequence attention (ops/hip/csrc/attention.hip).
 
-One 256-thread workgroup per (batch, head): q/k/v tiles, scores, softmax
-and the P @ V product all live in LDS; backward recomputes P.  Eligible
-when T, S <= 64 and d_head == 64 (the NMT transformer's 48x48x64 shape,
-where library flash-attention kernels spend 17/28 us on ~2 us of HBM
-traffic); everything else falls back to torch SDPA.
+One 256-thread (4-wave) workgroup per (batch, head); eligible when
+T, S <= 64 and d_head == 64 (the NMT transformer's 48x48x64 shape);
+everything else falls back to torch SDPA.
 
-Measured on MI355X: isolated wall-clock is at parity (fwd+bwd 83.7 vs
-83.8 us, scripts/bench_attn.py) but that comparison flatters us — the
-torch SDPA *call* carries ~15 us of dispatch overhead that disappears
-under hipGraph capture.  In-chain KERNEL time (rocprofv3, captured NMT
-step): ours 28.1 us fwd / 63.9 us bwd vs aotriton's 16.9 / 28.2 — ours
-loses ~1.0 ms/step e2e (308k -> 271k words/s).  Root cause: the
-per-(t,s) dot products re-read both operand rows from LDS for every
-MAC (~0.6 MB of LDS traffic per workgroup fwd, 3x that bwd, against a
-~65 TB/s aggregate LDS ceiling); aotriton's MFMA tiles amortize
-operands in registers.  The fix is a register-/MFMA-tiled rewrite
-(TODO.md), not a split.  DEFAULT-ON since the round-2 MFMA rewrite
-(fwd 6.8 us / bwd 19.3 us vs SDPA-aotriton 33/63 at the NMT shape —
-gpurun_out/attn_ab_mfma.log); PARALLAX_PX_ATTN=0 restores SDPA; was
-torch SDPA.
+DEFAULT-ON since the round-2 MFMA rewrite: every GEMM runs on
+mfma_f32_16x16x32_bf16 over zero-padded 64x64 LDS tiles (reductions
+over the non-contiguous axis read transposed twin images written at
+load; softmax row-reduces are 4-step __shfl_xor inside the C-fragment
+lane groups; the backward's dS images alias dead tiles to keep 2
+workgroups/CU).  Outputs are allocated [B,T,H,D]-permuted so the
+model's transpose+reshape afterwards is a free view.  Measured
+(MI355X, scripts/bench_attn.py): fwd 6.8 us / bwd 12.2 us vs torch
+SDPA (aotriton) 33 / 63; fwd+bwd 19.8 vs 104.6 us; NMT e2e ahead of
+the SDPA path.  PARALLAX_PX_ATTN=0 restores SDPA.
+
+History: the round-1 scalar-VALU version of these kernels lost
+in-chain (28/64 us — every MAC re-read both operand rows from LDS);
+its analysis lives in git and TODO.md item 6b.
 """
 
 import os
