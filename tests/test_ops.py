@@ -421,3 +421,38 @@ def test_px_attention_cpu_fallback_is_sdpa():
     out = px_attention(q, q, q)
     ref = torch.nn.functional.scaled_dot_product_attention(q, q, q)
     assert torch.allclose(out, ref)
+
+
+def test_padded_dedup_property_equivalence():
+    """Property test (hypothesis): unique_sum_rows_padded's valid prefix
+    always equals unique_sum_rows, for any id multiset / row dim /
+    averaging mode — the sync-free fixed-shape path must be a pure
+    re-encoding of the dynamic one."""
+    from hypothesis import given, settings, strategies as st
+    from parallax_amd.ops.sparse import (unique_sum_rows,
+                                         unique_sum_rows_padded)
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(st.integers(0, 12), min_size=1, max_size=40),
+           st.integers(1, 5), st.booleans(), st.integers(0, 2 ** 31 - 1))
+    def check(ids, d, average, seed):
+        g = torch.Generator().manual_seed(seed)
+        idx = torch.tensor(ids, dtype=torch.long)
+        vals = torch.randn(len(ids), d, generator=g)
+        u_ids, u_rows, _ = unique_sum_rows(idx, vals, average=average)
+        p_ids, p_rows, p_counts, nvalid = unique_sum_rows_padded(
+            idx, vals, average=average)
+        n = int(nvalid)
+        assert n == u_ids.numel()
+        assert torch.equal(p_ids[:n], u_ids)
+        assert torch.allclose(p_rows[:n], u_rows, atol=1e-5)
+        # tail is inert: id 0 with zero rows
+        assert (p_ids[n:] == 0).all()
+        assert float(p_rows[n:].abs().sum()) == 0.0
+        # counts = per-id multiplicities
+        import collections
+        c = collections.Counter(ids)
+        for j in range(n):
+            assert int(p_counts[j]) == c[int(p_ids[j])]
+
+    check()
