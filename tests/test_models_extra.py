@@ -143,3 +143,49 @@ def test_cnn_graph_optimizers():
     assert torch.isfinite(loss)
     with pytest.raises(ValueError):
         cnn.build_cnn_graph('trivial', optimizer_name='adagrad')
+
+
+def test_native_conv_state_dict_interchange():
+    """Checkpoints move freely between the nn.Conv2d model and the
+    native-conv swapped model (same parameter names/shapes)."""
+    import os
+    import torch
+    from parallax_amd.models.resnet import ResNet
+    torch.manual_seed(4)
+    ref = ResNet(depth=50, num_classes=10)
+    os.environ['PARALLAX_NATIVE_CONV1X1'] = '1'
+    os.environ['PARALLAX_NATIVE_CONV3X3'] = '1'
+    try:
+        from parallax_amd.ops.conv import swap_conv1x1, swap_conv3x3
+        import copy
+        swapped = swap_conv3x3(swap_conv1x1(copy.deepcopy(ref)))
+        n_swapped = sum(
+            1 for m in swapped.modules()
+            if type(m).__name__ in ('ParallaxConv1x1', 'ParallaxConv3x3'))
+        assert n_swapped > 30, n_swapped
+        sd = ref.state_dict()
+        swapped.load_state_dict(sd)          # nn -> native
+        ref2 = ResNet(depth=50, num_classes=10)
+        ref2.load_state_dict(swapped.state_dict())  # native -> nn
+        for k, v in ref2.state_dict().items():
+            assert torch.equal(v, sd[k]), k
+    finally:
+        del os.environ['PARALLAX_NATIVE_CONV1X1']
+        del os.environ['PARALLAX_NATIVE_CONV3X3']
+
+
+def test_resnet_v2_step():
+    """Pre-activation ResNet v2 (reference resnet_model.py:129-180)
+    builds and trains a step on CPU."""
+    import torch
+    from parallax_amd.models.resnet import ResNet
+    torch.manual_seed(5)
+    m = ResNet(depth=50, num_classes=10, version=2)
+    x = torch.randn(2, 3, 64, 64)
+    y = torch.randint(0, 10, (2,))
+    logits = m(x)
+    loss = torch.nn.functional.cross_entropy(logits, y)
+    loss.backward()
+    assert torch.isfinite(loss)
+    grads = [p.grad for p in m.parameters() if p.grad is not None]
+    assert len(grads) > 100
