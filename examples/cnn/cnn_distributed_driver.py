@@ -21,8 +21,9 @@ from parallax_amd.models import cnn, resnet
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument('--model', default='resnet50',
-                    choices=sorted(cnn.MODEL_REGISTRY))
+    ap.add_argument('--model', default='resnet50')
+    ap.add_argument('--dataset', default='imagenet',
+                    choices=['imagenet', 'cifar10'])
     ap.add_argument('--batch_size', type=int, default=64)
     ap.add_argument('--optimizer', default='momentum',
                     choices=['momentum', 'sgd', 'rmsprop'])
@@ -37,9 +38,12 @@ def main():
 
     sess, num_workers, worker_id, _ = parallax_amd.parallel_run(
         lambda: cnn.build_cnn_graph(args.model,
+                                    num_classes=(11 if args.dataset ==
+                                                 'cifar10' else 1001),
                                     batch_size=args.batch_size,
                                     optimizer_name=args.optimizer,
-                                    lr=args.learning_rate),
+                                    lr=args.learning_rate,
+                                    dataset=args.dataset),
         args.resource_info,
         parallax_config=parallax_amd.Config(run_option=args.run_option))
     device = sess.comm.device
@@ -52,7 +56,9 @@ def main():
     prev = time.time()
     for step in range(args.max_steps):
         x, y = resnet.synthetic_batch(
-            args.batch_size, image_size=cnn.image_size_for(args.model),
+            args.batch_size,
+            image_size=cnn.image_size_for(args.model, args.dataset),
+            num_classes=11 if args.dataset == 'cifar10' else 1001,
             device=device)
         out = sess.run((x, y))
         if step % args.log_frequency == 0 and worker_id == 0:

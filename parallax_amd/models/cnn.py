@@ -175,13 +175,13 @@ MODEL_REGISTRY = {
 def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
                     optimizer_name='momentum', lr=0.1, momentum=0.9,
                     weight_decay=1e-4, use_bf16=True, channels_last=True,
-                    pure_bf16=False):
+                    pure_bf16=False, dataset='imagenet'):
     """pure_bf16: conv/linear weights live in bf16 (no per-step autocast
     weight casts; fp32 update precision via the optimizers' fused master
     weights); BatchNorm modules stay fp32 (weights + running stats — the
     mixed bf16-input/fp32-param BN is the library's standard
     mixed-precision path)."""
-    model = MODEL_REGISTRY[model_name](num_classes)
+    model = get_model(model_name, num_classes, dataset)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
     if pure_bf16:
@@ -353,9 +353,31 @@ class InceptionV3(nn.Module):
 
 MODEL_REGISTRY['inception3'] = lambda nc: InceptionV3(nc)
 
+# vgg11 / lenet / inception4 + the cifar10 registry
+# (models/model_config.py:29-63) live in cnn_extra.py
+from parallax_amd.models.cnn_extra import (  # noqa: E402
+    VGG11, LeNet5, InceptionV4, MODEL_REGISTRY_CIFAR)
+
+MODEL_REGISTRY['vgg11'] = lambda nc: VGG11(nc)
+MODEL_REGISTRY['lenet'] = lambda nc: LeNet5(nc)
+MODEL_REGISTRY['inception4'] = lambda nc: InceptionV4(nc)
+
 # per-model input resolution (reference models/*.py image_size)
-MODEL_IMAGE_SIZE = {'overfeat': 231, 'inception3': 299}
+MODEL_IMAGE_SIZE = {'overfeat': 231, 'inception3': 299, 'inception4': 299,
+                    'lenet': 28}
 
 
-def image_size_for(model_name):
+def image_size_for(model_name, dataset='imagenet'):
+    if dataset == 'cifar10':
+        return 32
     return MODEL_IMAGE_SIZE.get(model_name, 224)
+
+
+def get_model(model_name, num_classes, dataset='imagenet'):
+    """model_config.get_model_config equivalent: dataset-keyed lookup
+    ('alexnet'/'trivial' exist in both registries with different nets)."""
+    reg = MODEL_REGISTRY_CIFAR if dataset == 'cifar10' else MODEL_REGISTRY
+    if model_name not in reg:
+        raise ValueError('unknown model %r for dataset %r'
+                         % (model_name, dataset))
+    return reg[model_name](num_classes)

@@ -189,3 +189,55 @@ def test_resnet_v2_step():
     assert torch.isfinite(loss)
     grads = [p.grad for p in m.parameters() if p.grad is not None]
     assert len(grads) > 100
+
+
+def test_cnn_extra_families():
+    """model_config.py:29-63 parity: vgg11/lenet/inception4 + the
+    cifar10 registry build and produce correct logit shapes."""
+    import torch
+    from parallax_amd.models.cnn import get_model, image_size_for
+    from parallax_amd.models.cnn_extra import MODEL_REGISTRY_CIFAR
+    torch.manual_seed(6)
+    for name in ('vgg11', 'lenet'):
+        s = image_size_for(name)
+        y = get_model(name, 1001)(torch.randn(1, 3, s, s))
+        assert y.shape == (1, 1001), name
+    assert image_size_for('inception4') == 299
+    for name in ('resnet20', 'resnet110_v2', 'densenet40_k12', 'alexnet',
+                 'trivial'):
+        assert name in MODEL_REGISTRY_CIFAR
+        m = get_model(name, 11, dataset='cifar10')
+        y = m(torch.randn(2, 3, 32, 32))
+        assert y.shape == (2, 11), name
+    # dataset-keyed lookup: cifar 'alexnet' is the tutorial net, not the
+    # imagenet one
+    assert type(get_model('alexnet', 11, 'cifar10')).__name__ \
+        == 'AlexNetCifar'
+
+
+def test_inception4_share_columns():
+    """The 'share' split-branch columns in inception_v4_c reuse the
+    previous column's stem (inception_model.py:25-35): the shared stem
+    conv runs once and feeds both heads."""
+    import torch
+    from parallax_amd.models.cnn_extra import _inc4_c
+    m = _inc4_c(1536)
+    x = torch.randn(1, 1536, 4, 4)
+    y = m(x)
+    assert y.shape == (1, 1536, 4, 4)
+    # column 3 shares col 2's first op; columns 5 shares col 4's 3-op stem
+    assert m.n_share == [0, 0, 0, 1, 0, 3]
+    y.sum().backward()
+
+
+def test_cnn_graph_cifar_step():
+    from parallax_amd.models.cnn import build_cnn_graph
+    import torch
+    torch.manual_seed(7)
+    g = build_cnn_graph('resnet20', num_classes=11, batch_size=4,
+                        use_bf16=False, dataset='cifar10')
+    x = torch.randn(4, 3, 32, 32)
+    y = torch.randint(0, 11, (4,))
+    loss, _ = g.step_fn(g.model, (x, y))
+    loss.backward()
+    assert torch.isfinite(loss)
