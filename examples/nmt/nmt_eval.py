@@ -33,6 +33,11 @@ def main():
     ap.add_argument('--batch-size', type=int, default=32)
     ap.add_argument('--max-decode-len', type=int, default=64)
     ap.add_argument('--layers', type=int, default=2)
+    ap.add_argument('--beam-width', type=int, default=0,
+                    help='0 = greedy (reference nmt.py beam_width flag)')
+    ap.add_argument('--length-penalty', type=float, default=0.0)
+    ap.add_argument('--metric', default='bleu',
+                    choices=['bleu', 'rouge', 'accuracy', 'word_accuracy'])
     args = ap.parse_args()
 
     if args.src:
@@ -64,12 +69,21 @@ def main():
     for batch in iterator_utils.bucketed_batches(
             examples, args.batch_size, shuffle=False):
         src, _tgt_in, tgt_out, _sl, _tl = batch
-        out_ids = model.greedy_decode(src, max_len=args.max_decode_len)
+        if args.beam_width > 0:
+            out_ids = model.beam_decode(
+                src, beam_width=args.beam_width,
+                length_penalty_weight=args.length_penalty,
+                max_len=args.max_decode_len)
+        else:
+            out_ids = model.greedy_decode(src,
+                                          max_len=args.max_decode_len)
         for row, ref_row in zip(out_ids, tgt_out.tolist()):
             hyps.append(tv.decode(row))
             refs.append(tv.decode(ref_row))
-    bleu = evaluation_utils.corpus_bleu(refs, hyps, smooth=True)
-    print('sentences=%d BLEU=%.2f' % (len(hyps), 100.0 * bleu))
+    score = evaluation_utils.evaluate(refs, hyps, args.metric)
+    print('sentences=%d %s=%.2f BLEU=%.2f' % (
+        len(hyps), args.metric, 100.0 * score,
+        100.0 * evaluation_utils.corpus_bleu(refs, hyps, smooth=True)))
 
 
 if __name__ == '__main__':

@@ -191,6 +191,80 @@ class NMTTransformer(nn.Module):
             outs.append(hyp)
         return outs
 
+    @torch.no_grad()
+    def beam_decode(self, src, beam_width=4, length_penalty_weight=0.0,
+                    sos_id=1, eos_id=2, max_len=64):
+        """Beam-search decoding with the GNMT length penalty
+        lp(l) = ((5 + l) / 6) ** alpha; candidates are ranked by
+        cumulative log-prob / lp at every step (reference infer path:
+        nmt/model.py:425-439 BeamSearchDecoder with beam_width and
+        length_penalty_weight).  Returns the best hypothesis per
+        sentence (id-lists without SOS/EOS); beam_width=1 with alpha=0
+        reduces to greedy_decode."""
+        mem = self.encode(src)
+        B, S, D = mem.shape
+        K = beam_width
+        dev = src.device
+        W = self.tgt_emb.full_weight().to(mem.dtype)    # [V, D]
+        V = W.shape[0]
+        scale = math.sqrt(self.d_model)
+        mem = mem.unsqueeze(1).expand(B, K, S, D).reshape(B * K, S, D)
+        ys = torch.full((B * K, 1), sos_id, dtype=torch.long, device=dev)
+        # only beam 0 is live at the start (all beams hold identical
+        # prefixes; duplicates would waste the beam)
+        scores = torch.full((B, K), float('-inf'), device=dev)
+        scores[:, 0] = 0.0
+        done = torch.zeros(B * K, dtype=torch.bool, device=dev)
+        # true hypothesis length per beam (finished beams stop growing —
+        # their EOS padding is free AND must not change their penalty)
+        lens = torch.zeros(B * K, dtype=torch.float32, device=dev)
+
+        def lp(length):
+            if length_penalty_weight == 0.0:
+                return torch.ones_like(length)
+            return ((5.0 + length) / 6.0) ** length_penalty_weight
+
+        for _ in range(max_len):
+            d = self.tgt_emb(ys).to(mem.dtype) * scale \
+                + self.pos[:ys.shape[1]]
+            for blk in self.dec:
+                d = blk(d, mem=mem, causal=True)
+            d = self.ln_f(d)[:, -1]                     # [B*K, D]
+            logp = torch.log_softmax((d @ W.t()).float(), dim=-1)
+            # finished beams: the only continuation is EOS at no cost
+            logp = torch.where(
+                done[:, None],
+                torch.full_like(logp, float('-inf')).scatter(
+                    1, torch.full((B * K, 1), eos_id, dtype=torch.long,
+                                  device=dev), 0.0),
+                logp)
+            cum = scores.reshape(B * K, 1) + logp       # [B*K, V]
+            new_len = torch.where(done, lens, lens + 1)  # [B*K]
+            ranked = cum / lp(new_len)[:, None]
+            top = ranked.reshape(B, K * V).topk(K, dim=-1).indices
+            beam_idx = top // V                         # [B, K]
+            tok = top % V                               # [B, K]
+            flat_src = (torch.arange(B, device=dev)[:, None] * K +
+                        beam_idx).reshape(-1)
+            scores = cum.reshape(B, K * V).gather(1, top)
+            ys = torch.cat([ys[flat_src], tok.reshape(-1, 1)], dim=1)
+            lens = new_len[flat_src]
+            done = done[flat_src] | (tok.reshape(-1) == eos_id)
+            if bool(done.all()):
+                break
+
+        final = (scores / lp(lens).reshape(B, K)).argmax(dim=1)  # [B]
+        outs = []
+        rows = ys[:, 1:].reshape(B, K, -1)
+        for b in range(B):
+            hyp = []
+            for t in rows[b, final[b]].tolist():
+                if t == eos_id:
+                    break
+                hyp.append(t)
+            outs.append(hyp)
+        return outs
+
 
 class GNMTAttention(nn.Module):
     """Reference-architecture seq2seq: LSTM encoder, LSTM decoder with

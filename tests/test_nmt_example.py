@@ -204,3 +204,70 @@ def test_nmt_end_to_end_smoke_on_testdata():
     m.eval()
     outs = m.greedy_decode(torch.tensor([exs[0][0]]), max_len=12)
     assert len(outs) == 1 and all(0 <= t < len(tv) for t in outs[0])
+
+
+def test_metric_dispatch_rouge_accuracy():
+    ref = ['the cat sat on the mat'.split()]
+    assert evaluation_utils.evaluate(ref, ref, 'rouge') == \
+        pytest.approx(1.0)
+    hyp = ['the cat sat on mat'.split()]
+    assert 0.5 < evaluation_utils.evaluate(ref, hyp, 'rouge') < 1.0
+    assert evaluation_utils.evaluate(ref, hyp, 'accuracy') == 0.0
+    assert evaluation_utils.evaluate(ref, ref, 'accuracy') == 1.0
+    wa = evaluation_utils.evaluate(ref, hyp, 'word_accuracy')
+    assert wa == pytest.approx(4 / 6)   # positional match, /max(len)
+    with pytest.raises(ValueError):
+        evaluation_utils.evaluate(ref, hyp, 'nope')
+
+
+def _tiny_model(seed=11):
+    from parallax_amd.models.nmt import NMTTransformer
+    torch.manual_seed(seed)
+    m = NMTTransformer(src_vocab=48, tgt_vocab=48, d_model=32, heads=2,
+                       ffn=64, enc_layers=1, dec_layers=1,
+                       min_partitions=2, max_len=40)
+    m.eval()
+    return m
+
+
+def test_beam_width1_equals_greedy():
+    m = _tiny_model()
+    src = torch.randint(3, 48, (4, 6))
+    greedy = m.greedy_decode(src, max_len=10)
+    beam1 = m.beam_decode(src, beam_width=1, max_len=10)
+    assert beam1 == greedy
+
+
+def test_beam_decode_valid_and_scores():
+    import torch.nn.functional as F
+    m = _tiny_model(12)
+    src = torch.randint(3, 48, (3, 5))
+    outs = m.beam_decode(src, beam_width=4, length_penalty_weight=0.6,
+                         max_len=12)
+    assert len(outs) == 3
+    for row in outs:
+        assert len(row) <= 12
+        assert all(0 <= t < 48 and t != 2 for t in row)
+
+    def score(hyp, b):
+        """model log-prob of hyp + EOS given src b (teacher-forced)."""
+        ids = [1] + hyp + [2]
+        mem = m.encode(src[b:b + 1])
+        tot = 0.0
+        import math as _m
+        for t in range(1, len(ids)):
+            ys = torch.tensor([ids[:t]])
+            d = m.tgt_emb(ys) * _m.sqrt(m.d_model) + m.pos[:t]
+            for blk in m.dec:
+                d = blk(d, mem=mem, causal=True)
+            d = m.ln_f(d)[:, -1]
+            logits = d @ m.tgt_emb.full_weight().t()
+            tot += float(F.log_softmax(logits.float(), -1)[0, ids[t]])
+        return tot
+
+    # beam-4 (alpha=0) never scores below greedy on the model's own
+    # likelihood
+    b4 = m.beam_decode(src, beam_width=4, max_len=12)
+    g = m.greedy_decode(src, max_len=12)
+    for b in range(3):
+        assert score(b4[b], b) >= score(g[b], b) - 1e-4
