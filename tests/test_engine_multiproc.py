@@ -21,7 +21,7 @@ def _init_env(rank, world, port):
     os.environ['MASTER_PORT'] = str(port)
 
 
-def _spawn(fn, port, world=WORLD, timeout=180):
+def _spawn_once(fn, port, world, timeout):
     ctx = mp.get_context('spawn')
     procs = []
     err_q = ctx.SimpleQueue()
@@ -34,12 +34,24 @@ def _spawn(fn, port, world=WORLD, timeout=180):
     errs = []
     while not err_q.empty():
         errs.append(err_q.get())
+    infra_only = not errs   # no Python traceback from any worker
     for p in procs:
         if p.is_alive():
             p.terminate()
             errs.append('worker timed out')
+            infra_only = False
         elif p.exitcode != 0:
             errs.append('worker exit %s' % p.exitcode)
+    return errs, infra_only
+
+
+def _spawn(fn, port, world=WORLD, timeout=180):
+    errs, infra_only = _spawn_once(fn, port, world, timeout)
+    if errs and infra_only:
+        # a worker died without raising (rendezvous port clash / gloo
+        # abort): one retry on a fresh port; genuine test failures
+        # always carry a traceback and are never retried
+        errs, _ = _spawn_once(fn, get_free_port(), world, timeout)
     assert not errs, '\n'.join(str(e) for e in errs)
 
 
