@@ -288,6 +288,17 @@ class GNMTAttention(nn.Module):
 
     dropout_p = 0.2   # reference nmt hparams dropout default
 
+    def _dec_step(self, emb_t, ctx, dec_h, dec_c, enc_out):
+        """One decoder step: LSTM cell + Luong multiplicative attention.
+        Returns (attended output, new ctx, new h, new c)."""
+        x = torch.cat([emb_t, ctx], dim=-1)
+        dec_h, dec_c = self.decoder.step(x, dec_h, dec_c)
+        scores = torch.einsum('bu,uv,bsv->bs', dec_h, self.att_w, enc_out)
+        alpha = torch.softmax(scores, dim=-1)
+        ctx = torch.einsum('bs,bsu->bu', alpha, enc_out)
+        out = torch.tanh(self.proj(torch.cat([dec_h, ctx], dim=-1)))
+        return out, ctx, dec_h, dec_c
+
     def forward(self, src, tgt_in, tgt_out):
         semb = self.src_emb(src)
         if self.training and self.dropout_p > 0:
@@ -300,14 +311,9 @@ class GNMTAttention(nn.Module):
         dec_h, dec_c = h, c
         outs = []
         for t in range(T):
-            x = torch.cat([emb[:, t], ctx], dim=-1)
-            dec_h, dec_c = self.decoder.step(x, dec_h, dec_c)
-            scores = torch.einsum('bu,uv,bsv->bs', dec_h, self.att_w,
-                                  enc_out)
-            alpha = torch.softmax(scores, dim=-1)
-            ctx = torch.einsum('bs,bsu->bu', alpha, enc_out)
-            outs.append(torch.tanh(self.proj(
-                torch.cat([dec_h, ctx], dim=-1))))
+            o, ctx, dec_h, dec_c = self._dec_step(emb[:, t], ctx, dec_h,
+                                                  dec_c, enc_out)
+            outs.append(o)
         dec = torch.stack(outs, dim=1).reshape(B * T, U)
         losses = sampled_softmax_loss(self.tgt_emb, None,
                                       tgt_out.reshape(-1), dec,
@@ -315,6 +321,38 @@ class GNMTAttention(nn.Module):
                                                       self.tgt_vocab // 2),
                                       num_classes=self.tgt_vocab)
         return losses.mean()
+
+    @torch.no_grad()
+    def greedy_decode(self, src, sos_id=1, eos_id=2, max_len=64):
+        """Step-wise greedy inference over the full tied softmax
+        (reference gnmt infer path; same output convention as
+        NMTTransformer.greedy_decode)."""
+        enc_out, (h, c) = self.encoder(self.src_emb(src))
+        B, S, U = enc_out.shape
+        W = self.tgt_emb.full_weight()                  # [V, D]
+        ctx = enc_out.new_zeros(B, U)
+        ys = torch.full((B,), sos_id, dtype=torch.long, device=src.device)
+        done = torch.zeros(B, dtype=torch.bool, device=src.device)
+        rows = []
+        for _ in range(max_len):
+            o, ctx, h, c = self._dec_step(self.tgt_emb(ys), ctx, h, c,
+                                          enc_out)
+            nxt = (o @ W.t()).argmax(dim=-1)
+            nxt = torch.where(done, torch.full_like(nxt, eos_id), nxt)
+            rows.append(nxt)
+            done |= nxt == eos_id
+            ys = nxt
+            if bool(done.all()):
+                break
+        outs = []
+        for row in torch.stack(rows, dim=1).tolist():
+            hyp = []
+            for t in row:
+                if t == eos_id:
+                    break
+                hyp.append(t)
+            outs.append(hyp)
+        return outs
 
 
 def build_nmt_graph(arch='transformer', src_vocab=SRC_VOCAB,

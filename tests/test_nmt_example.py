@@ -271,3 +271,26 @@ def test_beam_decode_valid_and_scores():
     g = m.greedy_decode(src, max_len=12)
     for b in range(3):
         assert score(b4[b], b) >= score(g[b], b) - 1e-4
+
+
+def test_gnmt_greedy_decode():
+    from parallax_amd.models.nmt import GNMTAttention
+    torch.manual_seed(13)
+    # tied softmax: units must equal emb_dim (build_nmt_graph defaults)
+    m = GNMTAttention(src_vocab=48, tgt_vocab=48, emb_dim=32, units=32,
+                      min_partitions=2)
+    m.eval()
+    src = torch.randint(3, 48, (3, 6))
+    outs = m.greedy_decode(src, max_len=10)
+    assert len(outs) == 3
+    for row in outs:
+        assert len(row) <= 10
+        assert all(0 <= t < 48 and t != 2 for t in row)
+    # decode is deterministic
+    assert m.greedy_decode(src, max_len=10) == outs
+    # and the training path still steps
+    m.train()
+    loss = m(src, torch.randint(3, 48, (3, 5)),
+             torch.randint(3, 48, (3, 5)))
+    loss.backward()
+    assert torch.isfinite(loss)
