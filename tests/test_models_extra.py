@@ -118,8 +118,9 @@ def test_skip_thoughts_step(monkeypatch, free_port):
     sess.close()
 
 
-@pytest.mark.parametrize('name', ['vgg16', 'vgg19', 'alexnet', 'overfeat',
-                                  'googlenet', 'inception3', 'trivial'])
+@pytest.mark.parametrize('name', ['vgg11', 'vgg16', 'vgg19', 'alexnet',
+                                  'overfeat', 'googlenet', 'inception3',
+                                  'inception4', 'lenet', 'trivial'])
 def test_cnn_family_forward(name):
     torch.manual_seed(0)
     m = cnn.MODEL_REGISTRY[name](11)
