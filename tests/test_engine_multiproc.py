@@ -818,3 +818,40 @@ def test_world8_full_stack(free_port):
     padded-count verification and collective ordering at the exact rank
     count of the round-end 8-GPU run."""
     _spawn(_body_parallel_run_hybrid, free_port, world=8)
+
+
+def _body_densenet_cifar_mpi(rank, world):
+    """All-dense cnn graph (densenet40_k12: ~120 small conv/BN params)
+    through the full parallel_run MPI path — exercises bucket packing
+    over many small tensors and the degenerate no-sparse dispatch."""
+    import parallax_amd
+    from parallax_amd.config import ParallaxConfig
+    from parallax_amd.models.cnn import build_cnn_graph
+
+    def build():
+        torch.manual_seed(31 + rank)   # broadcast must fix divergent init
+        return build_cnn_graph('densenet40_k12', num_classes=11,
+                               batch_size=4, optimizer_name='sgd', lr=0.05,
+                               use_bf16=False, channels_last=False,
+                               dataset='cifar10')
+
+    sess, num_workers, worker_id, _ = parallax_amd.parallel_run(
+        build, None, parallax_config=ParallaxConfig(run_option='MPI'))
+    assert num_workers == world and worker_id == rank
+    torch.manual_seed(200 + rank)      # different data per rank
+    x = torch.randn(4, 3, 32, 32)
+    y = torch.randint(0, 11, (4,))
+    for _ in range(2):
+        out = sess.run((x, y))
+        assert torch.isfinite(torch.as_tensor(out['loss']))
+    model = sess.graph.model
+    # every parameter bitwise-synced across ranks after averaged steps
+    for name, p in list(model.named_parameters())[::17]:
+        ps = [torch.zeros_like(p) for _ in range(world)]
+        torch.distributed.all_gather(ps, p.detach())
+        assert torch.allclose(ps[0], ps[1], atol=1e-5), name
+    sess.close()
+
+
+def test_densenet_cifar_mpi_world2():
+    _spawn(_body_densenet_cifar_mpi, get_free_port(), world=2, timeout=300)
