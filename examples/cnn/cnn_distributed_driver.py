@@ -32,6 +32,10 @@ def main():
     ap.add_argument('--log_frequency', type=int, default=10)
     ap.add_argument('--run_option', default='MPI')
     ap.add_argument('--params_stat', action='store_true')
+    ap.add_argument('--data_dir', default=None,
+                    help='image-folder dataset root (class subdirs); '
+                         'default: synthetic images')
+    ap.add_argument('--distortions', action='store_true')
     ap.add_argument('--resource_info', default=os.path.join(
         os.path.dirname(os.path.abspath(__file__)), 'resource_info'))
     args = ap.parse_args()
@@ -53,13 +57,27 @@ def main():
                               '%.1f MB grads/step', stats['num_tensors'],
                               stats['num_elements'] / 1e6,
                               stats['grad_bytes'] / 1e6)
+    data_it = None
+    if args.data_dir:
+        import preprocessing
+        from parallax_amd.data import PrefetchLoader
+        data_it = iter(PrefetchLoader(
+            preprocessing.minibatch_iterator(
+                args.data_dir, args.batch_size,
+                cnn.image_size_for(args.model, args.dataset),
+                train=True, distortions=args.distortions,
+                seed=worker_id),
+            device))
     prev = time.time()
     for step in range(args.max_steps):
-        x, y = resnet.synthetic_batch(
-            args.batch_size,
-            image_size=cnn.image_size_for(args.model, args.dataset),
-            num_classes=11 if args.dataset == 'cifar10' else 1001,
-            device=device)
+        if data_it is not None:
+            x, y = next(data_it)
+        else:
+            x, y = resnet.synthetic_batch(
+                args.batch_size,
+                image_size=cnn.image_size_for(args.model, args.dataset),
+                num_classes=11 if args.dataset == 'cifar10' else 1001,
+                device=device)
         out = sess.run((x, y))
         if step % args.log_frequency == 0 and worker_id == 0:
             now = time.time()
