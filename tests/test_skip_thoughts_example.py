@@ -124,3 +124,60 @@ def test_perplexity_of_untrained_model_near_vocab_size():
     assert n == 3 * 8 * 9 * 2
     # untrained near-uniform predictions: ppl ~= vocab size
     assert 0.5 * KW['vocab'] < ppl < 2.0 * KW['vocab'], ppl
+
+
+def test_data_utils_vocab_and_triples(tmp_path):
+    import preprocess_dataset as data_utils
+    p = tmp_path / 'corpus.txt'
+    p.write_text('a b c\n' 'b c d\n' 'c d\n' 'a very long sentence '
+                 'with many many words beyond limit here ok\n' 'd a\n')
+    vocab = data_utils.build_vocabulary([str(p)], num_words=8)
+    assert vocab['<eos>'] == 0 and vocab['<unk>'] == 1
+    assert len(vocab) == 8
+    # triples over consecutive sentences, long ones filtered
+    tr = data_utils.make_triples([str(p)], vocab, max_sentence_length=3)
+    # windows: (0,1,2) ok; (1,2,3) has the long sentence; (2,3,4) too
+    assert len(tr) == 1
+    prev, cur, nxt = tr[0]
+    assert prev[-1] == 0 and cur[-1] == 0 and nxt[-1] == 0   # EOS
+    assert cur[:-1] == [vocab['b'], vocab['c'], vocab['d']]
+    # OOV -> UNK
+    assert data_utils.sentence_to_ids(['zzz'], vocab) == [1, 0]
+
+
+def test_data_utils_batches_shard_and_pad(tmp_path):
+    import preprocess_dataset as data_utils
+    lines = '\n'.join('w%d w%d' % (i, i + 1) for i in range(30))
+    p = tmp_path / 'c.txt'
+    p.write_text(lines + '\n')
+    vocab = data_utils.build_vocabulary([str(p)], num_words=40)
+    tr = data_utils.make_triples([str(p)], vocab)
+    assert len(tr) == 28
+    got = 0
+    for s in range(2):
+        for cur, prev, nxt in data_utils.triple_batches(
+                tr, 8, shuffle=False, num_shards=2, shard_id=s):
+            assert cur.dtype == torch.long
+            assert cur.shape == prev.shape == nxt.shape
+            got += cur.shape[0]
+    assert got == 28        # shards disjoint + complete, tails kept
+
+
+def test_driver_real_data_smoke(tmp_path):
+    import subprocess
+    corpus = tmp_path / 'c.txt'
+    corpus.write_text('\n'.join('tok%d tok%d tok%d' % (i % 7, (i + 1) % 7,
+                                                       (i + 2) % 7)
+                                for i in range(20)) + '\n')
+    res_info = tmp_path / 'resource_info'
+    res_info.write_text('127.0.0.1:\n')
+    env = dict(os.environ, PYTHONPATH=REPO)
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, 'examples', 'skip_thoughts',
+                      'skip_thoughts_driver.py'),
+         '--batch_size', '4', '--max_steps', '2',
+         '--input_files', str(corpus), '--num_words', '16',
+         '--resource_info', str(res_info)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
