@@ -294,3 +294,23 @@ def test_gnmt_greedy_decode():
              torch.randint(3, 48, (3, 5)))
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def test_train_driver_real_data_smoke(tmp_path):
+    import subprocess
+    td = os.path.join(REPO, 'examples', 'nmt', 'testdata')
+    res_info = tmp_path / 'resource_info'
+    res_info.write_text('127.0.0.1:\n')
+    env = dict(os.environ, PYTHONPATH=REPO)
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, 'examples', 'nmt',
+                      'nmt_distributed_driver.py'),
+         '--batch_size', '8', '--max_steps', '2',
+         '--src', os.path.join(td, 'test_src.txt'),
+         '--tgt', os.path.join(td, 'test_tgt.txt'),
+         '--src_vocab', os.path.join(td, 'test_vocab_src.txt'),
+         '--tgt_vocab', os.path.join(td, 'test_vocab_tgt.txt'),
+         '--resource_info', str(res_info)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
