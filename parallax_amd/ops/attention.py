@@ -57,6 +57,9 @@ def _eligible(q, k, v):
 
 def px_attention(q, k, v, is_causal=False):
     """[B, H, T/S, D] q/k/v -> [B, H, T, D]; torch-SDPA-compatible."""
-    if _eligible(q, k, v):
+    # causal with T != S is convention-ambiguous (top-left vs
+    # bottom-right alignment); the kernel implements the T == S case the
+    # models use — route anything else to SDPA
+    if _eligible(q, k, v) and not (is_causal and q.shape[2] != k.shape[2]):
         return _PxAttnFn.apply(q, k, v, is_causal)
     return F.scaled_dot_product_attention(q, k, v, is_causal=is_causal)

@@ -74,6 +74,8 @@ class ParallaxBatchNorm2d(nn.Module):
             m.bias.copy_(bn.bias)
             m.running_mean.copy_(bn.running_mean)
             m.running_var.copy_(bn.running_var)
+            if getattr(bn, 'num_batches_tracked', None) is not None:
+                m.num_batches_tracked.copy_(bn.num_batches_tracked)
         return m
 
     def _native_ok(self, x):

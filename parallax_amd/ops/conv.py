@@ -113,7 +113,7 @@ def swap_conv1x1(module):
     for name, child in module.named_children():
         if (isinstance(child, nn.Conv2d) and child.kernel_size == (1, 1)
                 and child.stride == (1, 1) and child.bias is None
-                and child.groups == 1):
+                and child.groups == 1 and child.dilation == (1, 1)):
             setattr(module, name, ParallaxConv1x1.from_conv(child))
         else:
             swap_conv1x1(child)
@@ -185,7 +185,7 @@ class _Conv3x3Fn(torch.autograd.Function):
             if hasattr(e, 'px_wgrad3x3') and cout % 8 == 0:
                 # native wgrad: LDS-transposed implicit GEMM over the
                 # re-padded input; fp32 accumulate, one cast at the end
-                oh = ow = H  # stride-1 branch: output size == input
+                oh, ow = H, W  # stride-1 branch: output size == input
                 xp = torch.nn.functional.pad(
                     x.permute(0, 2, 3, 1), (0, 0, 1, 1, 1, 1))
                 dy2 = (dy_nhwc.reshape(N * H * W, cout).contiguous())
@@ -236,7 +236,7 @@ class ParallaxConv3x3(nn.Module):
     @classmethod
     def from_conv(cls, conv):
         assert conv.kernel_size == (3, 3) and conv.padding == (1, 1) \
-            and conv.bias is None
+            and conv.bias is None and conv.stride[0] == conv.stride[1]
         m = cls(conv.in_channels, conv.out_channels, conv.stride[0])
         with torch.no_grad():
             m.weight.copy_(conv.weight)
@@ -253,7 +253,9 @@ def swap_conv3x3(module):
     for name, child in module.named_children():
         if (isinstance(child, nn.Conv2d) and child.kernel_size == (3, 3)
                 and child.padding == (1, 1) and child.bias is None
-                and child.groups == 1 and child.in_channels >= 64
+                and child.groups == 1 and child.dilation == (1, 1)
+                and child.stride[0] == child.stride[1] <= 2
+                and child.in_channels >= 64
                 and (child.in_channels & (child.in_channels - 1)) == 0):
             setattr(module, name, ParallaxConv3x3.from_conv(child))
         else:
