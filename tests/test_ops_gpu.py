@@ -878,6 +878,30 @@ def test_conv3x3_nhwc_vs_torch():
         assert torch.isfinite(w.grad.float()).all()
 
 
+def test_conv3x3_native_bwd_nonsquare_vs_torch():
+    """Native stride-1 dgrad + wgrad on a NON-SQUARE shape big enough to
+    pass the tall-grid gate (H != W exercises the oh/ow split in the
+    wgrad host path), compared numerically against fp32 conv2d grads."""
+    from parallax_amd.ops.conv import conv3x3_nhwc, _tall_grid
+    torch.manual_seed(63)
+    N, C, H, W, COUT = 4, 64, 56, 112, 64
+    assert _tall_grid(N * H * W, COUT) >= 96   # native path engaged
+    x = (torch.randn(N, C, H, W, device=DEV, dtype=torch.bfloat16)
+         .to(memory_format=torch.channels_last).requires_grad_(True))
+    w = (torch.randn(COUT, C, 3, 3, device=DEV, dtype=torch.bfloat16)
+         * 0.1).requires_grad_(True)
+    y = conv3x3_nhwc(x, w, stride=1)
+    dy = torch.randn_like(y.float())
+    y.backward(dy.to(torch.bfloat16))
+    x2 = x.detach().float().clone().requires_grad_(True)
+    w2 = w.detach().float().clone().requires_grad_(True)
+    torch.nn.functional.conv2d(x2, w2, stride=1, padding=1).backward(dy)
+    for got, ref, nm in ((x.grad, x2.grad, 'dx'), (w.grad, w2.grad, 'dw')):
+        rel = ((got.float() - ref).abs().max()
+               / ref.abs().max().clamp_min(1e-6))
+        assert float(rel) < 0.05, (nm, float(rel))
+
+
 def test_batchnorm_nhwc_vs_torch():
     """Hand-written NHWC bf16 BN (train) fwd/bwd + running stats + fused
     ReLU vs an fp32 torch reference on the same bf16 input."""
