@@ -91,10 +91,17 @@ def word_accuracy(references, hypotheses):
     return total / max(len(references), 1)
 
 
-def evaluate(references, hypotheses, metric='bleu'):
+def evaluate(references, hypotheses, metric='bleu', subword=None):
     """Metric dispatch over token-list corpora (reference
     evaluation_utils.evaluate works on files; callers here already hold
-    token lists)."""
+    token lists).  subword='bpe' de-BPEs both sides before scoring
+    (reference subword_option)."""
+    if subword == 'bpe':
+        from .vocab_utils import restore_bpe
+        references = [restore_bpe(r) for r in references]
+        hypotheses = [restore_bpe(h) for h in hypotheses]
+    elif subword:
+        raise ValueError('unknown subword option %r' % (subword,))
     if metric == 'bleu':
         return corpus_bleu(references, hypotheses, smooth=True)
     if metric == 'rouge':

@@ -314,3 +314,14 @@ def test_train_driver_real_data_smoke(tmp_path):
          '--resource_info', str(res_info)],
         capture_output=True, text=True, timeout=300, env=env)
     assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+
+
+def test_evaluate_subword_debpe():
+    ref = [['lo@@', 'wer', 'newest']]
+    hyp = [['lower', 'newest']]
+    # raw token BLEU sees a mismatch; after de-BPE both sides agree
+    assert evaluation_utils.evaluate(ref, hyp, 'accuracy') == 0.0
+    assert evaluation_utils.evaluate(ref, hyp, 'accuracy',
+                                     subword='bpe') == 1.0
+    with pytest.raises(ValueError):
+        evaluation_utils.evaluate(ref, hyp, 'bleu', subword='wordpiece')
