@@ -38,6 +38,8 @@ def main():
     ap.add_argument('--length-penalty', type=float, default=0.0)
     ap.add_argument('--metric', default='bleu',
                     choices=['bleu', 'rouge', 'accuracy', 'word_accuracy'])
+    ap.add_argument('--subword', default=None, choices=['bpe'],
+                    help='de-BPE both sides before scoring')
     args = ap.parse_args()
 
     if args.src:
@@ -80,7 +82,8 @@ def main():
         for row, ref_row in zip(out_ids, tgt_out.tolist()):
             hyps.append(tv.decode(row))
             refs.append(tv.decode(ref_row))
-    score = evaluation_utils.evaluate(refs, hyps, args.metric)
+    score = evaluation_utils.evaluate(refs, hyps, args.metric,
+                                      subword=args.subword)
     print('sentences=%d %s=%.2f BLEU=%.2f' % (
         len(hyps), args.metric, 100.0 * score,
         100.0 * evaluation_utils.corpus_bleu(refs, hyps, smooth=True)))
