@@ -456,3 +456,48 @@ def test_padded_dedup_property_equivalence():
             assert int(p_counts[j]) == c[int(p_ids[j])]
 
     check()
+
+
+def test_resource_codec_property_roundtrip():
+    """serialize/deserialize_resource_info invert each other for
+    arbitrary machine layouts (reference ^ + : , codec,
+    lib.py:153-176)."""
+    from hypothesis import given, settings, strategies as st
+    from parallax_amd.lib import (serialize_resource_info,
+                                  deserialize_resource_info)
+
+    host = st.from_regex(r'[a-z][a-z0-9.\-]{0,12}', fullmatch=True)
+    machine = st.fixed_dictionaries({
+        'hostname': host,
+        'port': st.lists(st.integers(1024, 65535), max_size=4),
+        'gpus': st.lists(st.integers(0, 15), max_size=8, unique=True),
+    })
+    info = st.dictionaries(st.sampled_from(['master', 'ps', 'worker']),
+                           st.lists(machine, min_size=1, max_size=3),
+                           min_size=1)
+
+    @settings(max_examples=60, deadline=None)
+    @given(info)
+    def roundtrip(ri):
+        assert deserialize_resource_info(serialize_resource_info(ri)) == ri
+
+    roundtrip()
+
+
+def test_fixed_size_partitioner_property():
+    """Shard sizes cover every row exactly once, differ by at most 1,
+    and the first total%p shards carry the extra row
+    (tf.fixed_size_partitioner semantics)."""
+    from hypothesis import given, settings, strategies as st
+    from parallax_amd.partitions import FixedSizePartitioner
+
+    @settings(max_examples=100, deadline=None)
+    @given(st.integers(1, 5000), st.integers(1, 64))
+    def check(total, p):
+        sizes = FixedSizePartitioner(p).shard_sizes(total)
+        assert sum(sizes) == total
+        assert len(sizes) == min(p, total)
+        assert max(sizes) - min(sizes) <= 1
+        assert sorted(sizes, reverse=True) == list(sizes)
+
+    check()
