@@ -501,3 +501,61 @@ def test_fixed_size_partitioner_property():
         assert sorted(sizes, reverse=True) == list(sizes)
 
     check()
+
+
+def test_bucketed_batches_property_nothing_lost():
+    """For arbitrary example sets, bucketed batching over all shards
+    yields every example exactly once (iterator_utils semantics)."""
+    import os
+    import sys
+    from hypothesis import given, settings, strategies as st
+    sys.path.insert(0, os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        'examples', 'nmt'))
+    from utils import iterator_utils
+
+    ex = st.tuples(
+        st.lists(st.integers(3, 30), min_size=1, max_size=12),
+        st.lists(st.integers(3, 30), min_size=1, max_size=12))
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.lists(ex, min_size=1, max_size=40),
+           st.integers(1, 7), st.integers(1, 3), st.booleans())
+    def check(exs, bs, shards, shuffle):
+        got = 0
+        for s in range(shards):
+            for b in iterator_utils.bucketed_batches(
+                    exs, bs, shuffle=shuffle, seed=s,
+                    shard_id=s, num_shards=shards):
+                got += b[0].shape[0]
+        assert got == len(exs)
+
+    check()
+
+
+def test_cost_model_fit_property():
+    """fit_cost_model recovers the argmin of noiseless
+    t(n) = b/n + a(n-1) + c samples for arbitrary positive coefficients
+    (partitions.py:74-170)."""
+    from hypothesis import given, settings, strategies as st
+    from parallax_amd.partitions import fit_cost_model
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.floats(1e-4, 0.1), st.floats(0.5, 50.0),
+           st.floats(0.0, 5.0))
+    def check(a, b, c):
+        ns = [1, 2, 4, 8, 16, 32, 64]
+        ts = [b / n + a * (n - 1) + c for n in ns]
+        n_best = fit_cost_model(ns, ts)
+        # the search is restricted to the measured range (reference
+        # _find_optimal_p semantics)
+        assert 1 <= n_best <= 64
+        true = min(range(1, 65),
+                   key=lambda n: b / n + a * (n - 1) + c)
+        pred_t = b / n_best + a * (n_best - 1) + c
+        true_t = b / true + a * (true - 1) + c
+        # the fitted argmin's PREDICTED time is within 5% of optimal
+        # (the discrete argmin itself may differ by a step)
+        assert pred_t <= true_t * 1.05, (a, b, c, n_best, true)
+
+    check()
