@@ -559,3 +559,21 @@ def test_cost_model_fit_property():
         assert pred_t <= true_t * 1.05, (a, b, c, n_best, true)
 
     check()
+
+
+def test_partitioned_embedding_reshard_property():
+    """A p-shard checkpoint restores bit-identically under ANY p'
+    (arbitrary vocab/partition combinations; §5.4 shard-layout
+    parity)."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.integers(2, 200), st.integers(1, 12), st.integers(1, 12))
+    def check(vocab, p, p2):
+        torch.manual_seed(vocab)
+        a = PartitionedEmbedding(vocab, 3, FixedSizePartitioner(p))
+        b = PartitionedEmbedding(vocab, 3, FixedSizePartitioner(p2))
+        b.load_state_dict(a.state_dict())
+        assert torch.equal(a.full_weight(), b.full_weight())
+
+    check()
