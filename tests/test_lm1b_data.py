@@ -48,3 +48,55 @@ def test_dataset_worker_round_robin():
     assert b[0] == v.encode('a short sentence')
     assert a[1] == v.encode('another one')
     assert b[1] == v.encode('the lazy dog jumps')
+
+
+def test_lm1b_eval_driver_smoke():
+    import subprocess
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, 'examples', 'lm1b',
+                                      'lm1b_eval.py'), '--batches', '4'],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert 'perplexity=' in out.stdout
+
+
+def test_lm1b_eval_perplexity_drops_with_training():
+    """Full-softmax eval perplexity falls after a few training steps on
+    the same corpus (untrained ~= vocab size)."""
+    import math
+    import importlib.util
+    import torch
+    spec = importlib.util.spec_from_file_location(
+        'lm1b_eval', os.path.join(REPO, 'examples', 'lm1b',
+                                  'lm1b_eval.py'))
+    ev = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(ev)
+    import data_utils
+    from parallax_amd.models.lm1b import LM1BModel
+
+    td = os.path.join(REPO, 'examples', 'lm1b', 'testdata')
+    vocab = data_utils.Vocabulary(os.path.join(td, 'test_vocab.txt'))
+    ds = data_utils.LM1BDataset(os.path.join(td, 'test_sentences.txt'),
+                                vocab)
+    torch.manual_seed(1)
+    model = LM1BModel(vocab_size=vocab.size, emb_dim=16, state_size=32,
+                      proj_size=16, num_sampled=8, min_partitions=2)
+    before, _ = ev.evaluate_perplexity(model, ds, batch_size=4,
+                                       num_steps=10, max_batches=5)
+    assert 0.4 * vocab.size < before < 3.0 * vocab.size
+    opt = torch.optim.Adagrad(model.parameters(), lr=0.1)
+    it = ds.iterate_forever(4, 10)
+    model.train()
+    for _ in range(40):
+        x, y, _w = next(it)
+        loss, _st = model(torch.from_numpy(x), torch.from_numpy(y))
+        opt.zero_grad()
+        loss.backward()
+        for p in model.parameters():
+            if p.grad is not None and p.grad.is_sparse:
+                p.grad = p.grad.to_dense()
+        opt.step()
+    after, _ = ev.evaluate_perplexity(model, ds, batch_size=4,
+                                      num_steps=10, max_batches=5)
+    assert after < before * 0.8, (before, after)
+    assert not math.isnan(after)
