@@ -193,7 +193,7 @@ def test_nmt_end_to_end_smoke_on_testdata():
                     if prm.grad is not None and prm.grad.is_sparse:
                         prm.grad = prm.grad.to_dense()
                 opt.step()
-            tot += float(loss)
+            tot += float(loss.detach())
             nb += 1
         return tot / nb
 
