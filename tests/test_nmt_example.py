@@ -262,7 +262,8 @@ def test_beam_decode_valid_and_scores():
                 d = blk(d, mem=mem, causal=True)
             d = m.ln_f(d)[:, -1]
             logits = d @ m.tgt_emb.full_weight().t()
-            tot += float(F.log_softmax(logits.float(), -1)[0, ids[t]])
+            tot += float(F.log_softmax(logits.detach().float(),
+                                       -1)[0, ids[t]])
         return tot
 
     # beam-4 (alpha=0) never scores below greedy on the model's own
