@@ -855,3 +855,42 @@ def _body_densenet_cifar_mpi(rank, world):
 
 def test_densenet_cifar_mpi_world2():
     _spawn(_body_densenet_cifar_mpi, get_free_port(), world=2, timeout=300)
+
+
+def _body_local_aggregation_off(rank, world):
+    """local_aggregation=False (reference graph_transform_lib:1544-1549
+    bypass): raw slices go to the exchange un-deduplicated; the final
+    table must match the aggregated path exactly."""
+    import torch.nn as nn
+    from parallax_amd import optim
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.parallel.sparse_engine import SparsePSEngine
+    from parallax_amd.ops.embedding import ParallaxEmbedding
+    from parallax_amd.tagging import classify_params
+
+    comm = CommContext()
+    tables = {}
+    for local_agg in (True, False):
+        torch.manual_seed(17)
+        model = nn.Module()
+        model.emb = ParallaxEmbedding(30, 4)
+        opt = optim.Adagrad(model.parameters(), lr=0.5)
+        eng = SparsePSEngine(comm, classify_params(model), opt,
+                             local_aggregation=local_agg)
+        eng.prepare_step()
+        ids = torch.tensor([2 + rank, 9, 9, 2 + rank])  # dups + overlap
+        model.emb(ids).sum().backward()
+        eng.reduce_and_apply()
+        tables[local_agg] = model.emb.weight.detach().clone()
+        # ranks agree within each mode
+        ws = [torch.zeros_like(tables[local_agg]) for _ in range(world)]
+        torch.distributed.all_gather(ws, tables[local_agg])
+        assert torch.allclose(ws[0], ws[1], atol=1e-6)
+        eng.close()
+    assert torch.allclose(tables[True], tables[False], atol=1e-6), \
+        (tables[True] - tables[False]).abs().max()
+
+
+def test_local_aggregation_off_world2():
+    _spawn(_body_local_aggregation_off, get_free_port(), world=2,
+           timeout=300)
