@@ -29,6 +29,10 @@ class PSConfig(object):
                  local_aggregation=True,
                  boundary_among_servers=True,
                  boundary_between_workers_and_servers=True):
+        if protocol not in ('rccl', 'grpc', 'grpc+verbs', 'grpc+gdr'):
+            # reference PSConfig.protocol vocabulary (config.py:21-49);
+            # every value maps to the in-process RCCL/xGMI exchange here
+            raise ValueError('unknown PS protocol %r' % (protocol,))
         self.protocol = protocol
         self.replicate_variables = replicate_variables
         self.local_aggregation = local_aggregation

@@ -218,3 +218,11 @@ def test_public_api_surface():
     for name in ('Config', 'PSConfig', 'MPIConfig', 'CommunicationConfig',
                  'CheckPointConfig', 'ProfileConfig'):
         assert hasattr(parallax_amd, name)
+
+
+def test_ps_protocol_validation():
+    from parallax_amd.config import PSConfig
+    for ok in ('rccl', 'grpc', 'grpc+verbs', 'grpc+gdr'):
+        assert PSConfig(protocol=ok).protocol == ok
+    with pytest.raises(ValueError):
+        PSConfig(protocol='carrier-pigeon')
