@@ -27,7 +27,10 @@ def main():
     ap.add_argument('--batch_size', type=int, default=64)
     ap.add_argument('--optimizer', default='momentum',
                     choices=['momentum', 'sgd', 'rmsprop'])
-    ap.add_argument('--learning_rate', type=float, default=0.1)
+    ap.add_argument('--learning_rate', type=float, default=None,
+                    help='fixed LR; default: the model\'s published '
+                         'schedule (warmup+piecewise for resnet, '
+                         'piecewise for cifar resnet/densenet) or 0.1')
     ap.add_argument('--max_steps', type=int, default=1000000)
     ap.add_argument('--log_frequency', type=int, default=10)
     ap.add_argument('--run_option', default='MPI')
@@ -40,13 +43,19 @@ def main():
         os.path.dirname(os.path.abspath(__file__)), 'resource_info'))
     args = ap.parse_args()
 
+    from parallax_amd import lr_schedule
+    schedule = (None if args.learning_rate is not None
+                else lr_schedule.model_lr_schedule(args.model,
+                                                   args.dataset))
+    lr0 = args.learning_rate if args.learning_rate is not None else 0.1
+
     sess, num_workers, worker_id, _ = parallax_amd.parallel_run(
         lambda: cnn.build_cnn_graph(args.model,
                                     num_classes=(11 if args.dataset ==
                                                  'cifar10' else 1001),
                                     batch_size=args.batch_size,
                                     optimizer_name=args.optimizer,
-                                    lr=args.learning_rate,
+                                    lr=lr0,
                                     dataset=args.dataset),
         args.resource_info,
         parallax_config=parallax_amd.Config(run_option=args.run_option))
@@ -70,6 +79,10 @@ def main():
             device))
     prev = time.time()
     for step in range(args.max_steps):
+        if schedule is not None:
+            lr_schedule.set_lr(
+                sess.graph.optimizer,
+                schedule(step, args.batch_size, num_workers))
         if data_it is not None:
             x, y = next(data_it)
         else:

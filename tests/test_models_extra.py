@@ -242,3 +242,47 @@ def test_cnn_graph_cifar_step():
     loss, _ = g.step_fn(g.model, (x, y))
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def test_lr_schedules_reference_values():
+    """resnet_model.py:232-255 warmup+piecewise; cifar schedules
+    :326-331 / densenet_model.py:96-103; tf exponential_decay."""
+    from parallax_amd import lr_schedule as L
+    bpe = L.IMAGENET_NUM_TRAIN_IMAGES / 64.0
+    # warmup: 0.1 at step 0 (batch 64 -> scaled lr stays 0.1, so flat)
+    assert L.resnet_imagenet_lr(0, 64) == pytest.approx(0.1)
+    assert L.resnet_imagenet_lr(int(bpe * 4), 64) == pytest.approx(0.1)
+    # decades at epochs 30/60/80
+    assert L.resnet_imagenet_lr(int(bpe * 31), 64) == pytest.approx(0.01)
+    assert L.resnet_imagenet_lr(int(bpe * 61), 64) == pytest.approx(1e-3)
+    assert L.resnet_imagenet_lr(int(bpe * 81), 64) == pytest.approx(1e-4)
+    # global batch > 256 scales the post-warmup LR linearly
+    big = L.resnet_imagenet_lr(10 ** 9, 64, num_workers=8)  # global 512
+    assert big == pytest.approx(512 / 256.0 * 0.1 / 10 ** 3)
+    # warmup is linear from 0.1 to the scaled LR
+    mid = L.resnet_imagenet_lr(
+        int(L.IMAGENET_NUM_TRAIN_IMAGES / 512.0 * 2.5), 64, 8)
+    assert 0.1 < mid < 0.2
+    # cifar resnet
+    cb = int(L.CIFAR10_NUM_TRAIN_IMAGES / 128)
+    assert L.resnet_cifar_lr(cb * 82, 128) == pytest.approx(0.1)
+    assert L.resnet_cifar_lr(cb * 83, 128) == pytest.approx(0.01)
+    assert L.resnet_cifar_lr(cb * 500, 128) == pytest.approx(0.0002)
+    # densenet cifar
+    assert L.densenet_cifar_lr(cb * 151, 128) == pytest.approx(0.01)
+    # exponential decay (staircase)
+    assert L.exponential_decay(0.1, 99, 100, 0.5) == pytest.approx(0.1)
+    assert L.exponential_decay(0.1, 250, 100, 0.5) == \
+        pytest.approx(0.025)
+    # dispatch
+    assert L.model_lr_schedule('resnet50') is L.resnet_imagenet_lr
+    assert L.model_lr_schedule('resnet20', 'cifar10') is L.resnet_cifar_lr
+    assert L.model_lr_schedule('densenet40_k12', 'cifar10') \
+        is L.densenet_cifar_lr
+    assert L.model_lr_schedule('vgg16') is None
+    # set_lr
+    import torch as T
+    from parallax_amd import optim
+    opt = optim.SGD([T.nn.Parameter(T.zeros(2))], lr=0.5)
+    L.set_lr(opt, 0.125)
+    assert opt.param_groups[0]['lr'] == 0.125
