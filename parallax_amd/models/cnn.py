@@ -172,10 +172,22 @@ MODEL_REGISTRY = {
 }
 
 
+def topk_accuracy(logits, labels, ks=(1, 5)):
+    """Fraction of rows whose label is in the top-k logits
+    (benchmark_cnn.py top_1/top_5 accuracy fetches)."""
+    out = {}
+    top = logits.detach().topk(max(ks), dim=-1).indices
+    for k in ks:
+        hit = (top[:, :k] == labels[:, None]).any(dim=1)
+        out['top_%d' % k] = hit.float().mean()
+    return out
+
+
 def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
                     optimizer_name='momentum', lr=0.1, momentum=0.9,
                     weight_decay=1e-4, use_bf16=True, channels_last=True,
-                    pure_bf16=False, dataset='imagenet'):
+                    pure_bf16=False, dataset='imagenet',
+                    with_accuracy=False):
     """pure_bf16: conv/linear weights live in bf16 (no per-step autocast
     weight casts; fp32 update precision via the optimizers' fused master
     weights); BatchNorm modules stay fp32 (weights + running stats — the
@@ -216,7 +228,8 @@ def build_cnn_graph(model_name='resnet50', num_classes=1001, batch_size=64,
         else:
             logits = m(x)
             loss = softmax_cross_entropy(logits, y)
-        return loss, {}
+        fetches = topk_accuracy(logits.float(), y) if with_accuracy else {}
+        return loss, fetches
 
     return TrainGraph(model, opt, step_fn)
 

@@ -101,3 +101,32 @@ def test_driver_real_data_smoke(image_root):
          '--data_dir', image_root, '--resource_info', res_info],
         capture_output=True, text=True, timeout=300, env=env)
     assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+
+
+def test_driver_train_ckpt_then_eval(image_root, tmp_path):
+    """Train 3 steps with checkpointing, then run the eval mode against
+    the saved checkpoint (benchmark_cnn _eval_once equivalent)."""
+    import subprocess
+    res_info = os.path.join(image_root, 'resource_info')
+    with open(res_info, 'w') as f:
+        f.write('127.0.0.1:\n')
+    ckpt = str(tmp_path / 'ck')
+    env = dict(os.environ, PYTHONPATH=REPO)
+    drv = os.path.join(REPO, 'examples', 'cnn',
+                       'cnn_distributed_driver.py')
+    out = subprocess.run(
+        [sys.executable, drv, '--model', 'trivial', '--dataset',
+         'cifar10', '--batch_size', '4', '--max_steps', '3',
+         '--optimizer', 'sgd', '--print_training_accuracy',
+         '--ckpt_dir', ckpt, '--save_ckpt_steps', '2',
+         '--resource_info', res_info],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    assert os.path.isfile(os.path.join(ckpt, 'checkpoint'))
+    out = subprocess.run(
+        [sys.executable, drv, '--model', 'trivial', '--dataset',
+         'cifar10', '--batch_size', '4', '--eval', '--ckpt_dir', ckpt,
+         '--data_dir', image_root, '--num_eval_batches', '2'],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    assert 'top_1=' in out.stdout and 'top_5=' in out.stdout
