@@ -34,6 +34,9 @@ def main():
     ap.add_argument('--tgt_vocab', default=None)
     ap.add_argument('--vocab_size', type=int, default=32000)
     ap.add_argument('--src_max_len', type=int, default=50)
+    ap.add_argument('--run_option', default='HYBRID')
+    ap.add_argument('--ckpt_dir', default=None)
+    ap.add_argument('--save_ckpt_steps', type=int, default=None)
     ap.add_argument('--resource_info', default=os.path.join(
         os.path.dirname(os.path.abspath(__file__)), 'resource_info'))
     args = ap.parse_args()
@@ -59,7 +62,11 @@ def main():
         lambda: nmt.build_nmt_graph(arch=args.arch,
                                     lr=args.learning_rate, **graph_kw),
         args.resource_info,
-        parallax_config=parallax_amd.Config(run_option='HYBRID'))
+        parallax_config=parallax_amd.Config(
+            run_option=args.run_option,
+            ckpt_config=parallax_amd.CheckPointConfig(
+                ckpt_dir=args.ckpt_dir,
+                save_ckpt_steps=args.save_ckpt_steps)))
     device = sess.comm.device
 
     def real_batches():

@@ -26,6 +26,9 @@ def main():
                          'default: synthetic ids')
     ap.add_argument('--num_words', type=int, default=20000)
     ap.add_argument('--max_sentence_length', type=int, default=30)
+    ap.add_argument('--run_option', default='HYBRID')
+    ap.add_argument('--ckpt_dir', default=None)
+    ap.add_argument('--save_ckpt_steps', type=int, default=None)
     ap.add_argument('--resource_info', default=os.path.join(
         os.path.dirname(os.path.abspath(__file__)), 'resource_info'))
     args = ap.parse_args()
@@ -45,7 +48,11 @@ def main():
             vocab=vocab_size, lr=args.learning_rate,
             num_sampled=min(512, max(2, vocab_size // 2))),
         args.resource_info,
-        parallax_config=parallax_amd.Config(run_option='HYBRID'))
+        parallax_config=parallax_amd.Config(
+            run_option=args.run_option,
+            ckpt_config=parallax_amd.CheckPointConfig(
+                ckpt_dir=args.ckpt_dir,
+                save_ckpt_steps=args.save_ckpt_steps)))
     device = sess.comm.device
 
     def real_batches():

@@ -181,3 +181,34 @@ def test_driver_real_data_smoke(tmp_path):
          '--resource_info', str(res_info)],
         capture_output=True, text=True, timeout=300, env=env)
     assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+
+
+def test_driver_ckpt_feeds_encoder(tmp_path):
+    """Driver trains with checkpointing; the saved checkpoint loads into
+    the inference encoder (train -> encode round trip through the real
+    stack)."""
+    import subprocess
+    corpus = tmp_path / 'c.txt'
+    corpus.write_text('\n'.join('tok%d tok%d' % (i % 5, (i + 1) % 5)
+                                for i in range(12)) + '\n')
+    res_info = tmp_path / 'resource_info'
+    res_info.write_text('127.0.0.1:\n')
+    ckpt = str(tmp_path / 'ck')
+    env = dict(os.environ, PYTHONPATH=REPO)
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, 'examples', 'skip_thoughts',
+                      'skip_thoughts_driver.py'),
+         '--batch_size', '4', '--max_steps', '3',
+         '--input_files', str(corpus), '--num_words', '8',
+         '--ckpt_dir', ckpt, '--save_ckpt_steps', '2',
+         '--resource_info', str(res_info)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    import preprocess_dataset as pd
+    vocab = pd.build_vocabulary([str(corpus)], 8)
+    enc = st_encoder.SkipThoughtsEncoder.from_checkpoint(
+        ckpt, vocab, vocab=len(vocab), emb_dim=256, units=512,
+        num_sampled=min(512, max(2, len(vocab) // 2)), min_partitions=4)
+    vecs = enc.encode(['tok1 tok2', 'tok3'])
+    assert len(vecs) == 2 and vecs[0].shape == (512,)
