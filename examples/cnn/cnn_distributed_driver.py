@@ -68,11 +68,8 @@ def evaluate(args):
     nc = num_classes_for(args.dataset)
     model = cnn.get_model(args.model, nc, args.dataset).eval()
     if args.ckpt_dir:
-        pointer = os.path.join(args.ckpt_dir, 'checkpoint')
-        with open(pointer) as f:
-            name = next(ln.split('"')[1] for ln in f
-                        if ln.startswith('model_checkpoint_path:'))
-        payload = torch.load(os.path.join(args.ckpt_dir, name),
+        from parallax_amd.checkpoint import latest_checkpoint
+        payload = torch.load(latest_checkpoint(args.ckpt_dir),
                              map_location='cpu', weights_only=True)
         model.load_state_dict(payload['model'])
     size = cnn.image_size_for(args.model, args.dataset)

@@ -86,11 +86,8 @@ def main():
                       num_sampled=max(2, vocab.size // 4),
                       min_partitions=2)
     if args.ckpt:
-        pointer = os.path.join(args.ckpt, 'checkpoint')
-        with open(pointer) as f:
-            name = next(ln.split('"')[1] for ln in f
-                        if ln.startswith('model_checkpoint_path:'))
-        payload = torch.load(os.path.join(args.ckpt, name),
+        from parallax_amd.checkpoint import latest_checkpoint
+        payload = torch.load(latest_checkpoint(args.ckpt),
                              map_location='cpu', weights_only=True)
         model.load_state_dict(payload['model'])
     ppl, n = evaluate_perplexity(model, ds, args.batch_size,

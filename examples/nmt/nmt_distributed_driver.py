@@ -34,6 +34,10 @@ def main():
     ap.add_argument('--tgt_vocab', default=None)
     ap.add_argument('--vocab_size', type=int, default=32000)
     ap.add_argument('--src_max_len', type=int, default=50)
+    ap.add_argument('--d_model', type=int, default=512)
+    ap.add_argument('--heads', type=int, default=8)
+    ap.add_argument('--ffn', type=int, default=2048)
+    ap.add_argument('--layers', type=int, default=6)
     ap.add_argument('--run_option', default='HYBRID')
     ap.add_argument('--ckpt_dir', default=None)
     ap.add_argument('--save_ckpt_steps', type=int, default=None)
@@ -42,7 +46,8 @@ def main():
     args = ap.parse_args()
 
     examples = sv = tv = None
-    graph_kw = {}
+    graph_kw = dict(d_model=args.d_model, heads=args.heads, ffn=args.ffn,
+                    enc_layers=args.layers, dec_layers=args.layers)
     if args.src:
         from utils import iterator_utils, vocab_utils
         with open(args.src) as f:
@@ -55,7 +60,7 @@ def main():
               else vocab_utils.build_vocab([args.tgt], args.vocab_size))
         examples = iterator_utils.make_examples(
             src_lines, tgt_lines, sv, tv, src_max_len=args.src_max_len)
-        graph_kw = dict(src_vocab=len(sv), tgt_vocab=len(tv),
+        graph_kw.update(src_vocab=len(sv), tgt_vocab=len(tv),
                         min_partitions=2)
 
     sess, num_workers, worker_id, _ = parallax_amd.parallel_run(

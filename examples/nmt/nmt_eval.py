@@ -33,6 +33,9 @@ def main():
     ap.add_argument('--batch-size', type=int, default=32)
     ap.add_argument('--max-decode-len', type=int, default=64)
     ap.add_argument('--layers', type=int, default=2)
+    ap.add_argument('--d-model', type=int, default=64)
+    ap.add_argument('--heads', type=int, default=4)
+    ap.add_argument('--ffn', type=int, default=128)
     ap.add_argument('--beam-width', type=int, default=0,
                     help='0 = greedy (reference nmt.py beam_width flag)')
     ap.add_argument('--length-penalty', type=float, default=0.0)
@@ -56,7 +59,8 @@ def main():
         tv = vocab_utils.Vocab('x y z'.split())
 
     model = NMTTransformer(src_vocab=len(sv), tgt_vocab=len(tv),
-                           d_model=64, heads=4, ffn=128,
+                           d_model=args.d_model, heads=args.heads,
+                           ffn=args.ffn,
                            enc_layers=args.layers, dec_layers=args.layers,
                            min_partitions=2)
     if args.ckpt:

@@ -45,17 +45,11 @@ class SkipThoughtsEncoder(object):
     @classmethod
     def from_checkpoint(cls, ckpt_dir, vocab_map, embedding_dict=None,
                         **model_kwargs):
-        pointer = os.path.join(ckpt_dir, 'checkpoint')
-        with open(pointer) as f:
-            for line in f:
-                if line.startswith('model_checkpoint_path:'):
-                    name = line.split('"')[1]
-                    break
-            else:
-                raise FileNotFoundError('no model_checkpoint_path in %s'
-                                        % pointer)
-        payload = torch.load(os.path.join(ckpt_dir, name),
-                             map_location='cpu', weights_only=True)
+        from parallax_amd.checkpoint import latest_checkpoint
+        path = latest_checkpoint(ckpt_dir)
+        if path is None:
+            raise FileNotFoundError('no checkpoint under %s' % ckpt_dir)
+        payload = torch.load(path, map_location='cpu', weights_only=True)
         model = SkipThoughts(**model_kwargs)
         model.load_state_dict(payload['model'])
         return cls(model, vocab_map, embedding_dict)

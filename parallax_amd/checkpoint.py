@@ -22,6 +22,20 @@ import torch
 from parallax_amd.lib import parallax_log
 
 
+def latest_checkpoint(ckpt_dir):
+    """Path of the newest checkpoint under ckpt_dir, or None
+    (tf.train.latest_checkpoint equivalent: reads the 'checkpoint'
+    pointer file the chief writes)."""
+    pointer = os.path.join(ckpt_dir, 'checkpoint')
+    if not os.path.isfile(pointer):
+        return None
+    with open(pointer) as f:
+        for line in f:
+            if line.startswith('model_checkpoint_path:'):
+                return os.path.join(ckpt_dir, line.split('"')[1])
+    return None
+
+
 class CheckpointManager(object):
     def __init__(self, ckpt_config, model, optimizer, is_chief):
         self.cfg = ckpt_config
@@ -156,15 +170,7 @@ class CheckpointManager(object):
     def latest_path(self):
         if not self.enabled:
             return None
-        pointer = os.path.join(self.cfg.ckpt_dir, 'checkpoint')
-        if not os.path.isfile(pointer):
-            return None
-        with open(pointer) as f:
-            for line in f:
-                if line.startswith('model_checkpoint_path:'):
-                    name = line.split('"')[1]
-                    return os.path.join(self.cfg.ckpt_dir, name)
-        return None
+        return latest_checkpoint(self.cfg.ckpt_dir)
 
     def maybe_restore(self, map_location='cpu'):
         """Restore latest checkpoint if one exists (all ranks restore the

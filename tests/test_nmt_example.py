@@ -326,3 +326,42 @@ def test_evaluate_subword_debpe():
                                      subword='bpe') == 1.0
     with pytest.raises(ValueError):
         evaluation_utils.evaluate(ref, hyp, 'bleu', subword='wordpiece')
+
+
+def test_train_ckpt_then_eval_chain(tmp_path):
+    """Driver trains with checkpointing on testdata; nmt_eval restores
+    that checkpoint (--ckpt) and scores BLEU — the full train->decode
+    round trip through the real stack."""
+    td = os.path.join(REPO, 'examples', 'nmt', 'testdata')
+    res_info = tmp_path / 'resource_info'
+    res_info.write_text('127.0.0.1:\n')
+    ckpt = str(tmp_path / 'ck')
+    env = dict(os.environ, PYTHONPATH=REPO)
+    dims = ['--d_model', '32', '--heads', '2', '--ffn', '64',
+            '--layers', '1']
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, 'examples', 'nmt',
+                      'nmt_distributed_driver.py'),
+         '--batch_size', '8', '--max_steps', '3',
+         '--src', os.path.join(td, 'test_src.txt'),
+         '--tgt', os.path.join(td, 'test_tgt.txt'),
+         '--src_vocab', os.path.join(td, 'test_vocab_src.txt'),
+         '--tgt_vocab', os.path.join(td, 'test_vocab_tgt.txt'),
+         '--ckpt_dir', ckpt, '--save_ckpt_steps', '2'] + dims +
+        ['--resource_info', str(res_info)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    assert os.path.isfile(os.path.join(ckpt, 'checkpoint'))
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, 'examples', 'nmt', 'nmt_eval.py'),
+         '--src', os.path.join(td, 'test_src.txt'),
+         '--tgt', os.path.join(td, 'test_tgt.txt'),
+         '--src-vocab', os.path.join(td, 'test_vocab_src.txt'),
+         '--tgt-vocab', os.path.join(td, 'test_vocab_tgt.txt'),
+         '--ckpt', ckpt, '--d-model', '32', '--heads', '2',
+         '--ffn', '64', '--layers', '1', '--beam-width', '2'],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    assert 'BLEU=' in out.stdout
