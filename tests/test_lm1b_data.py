@@ -100,3 +100,29 @@ def test_lm1b_eval_perplexity_drops_with_training():
                                       num_steps=10, max_batches=5)
     assert after < before * 0.8, (before, after)
     assert not math.isnan(after)
+
+
+def test_lm1b_eval_ckpt_restore(tmp_path):
+    """lm1b_eval --ckpt restores a CheckpointManager-written checkpoint
+    (dims via flags)."""
+    import subprocess
+    import torch
+    from parallax_amd.config import CheckPointConfig
+    from parallax_amd.checkpoint import CheckpointManager
+    from parallax_amd.models.lm1b import LM1BModel
+    td = os.path.join(REPO, 'examples', 'lm1b', 'testdata')
+    vocab_lines = open(os.path.join(td, 'test_vocab.txt')).read()
+    vsize = len([l for l in vocab_lines.splitlines() if l.strip()])
+    torch.manual_seed(0)
+    model = LM1BModel(vocab_size=vsize, emb_dim=16, state_size=32,
+                      proj_size=16, num_sampled=4, min_partitions=2)
+    cfg = CheckPointConfig(ckpt_dir=str(tmp_path), save_ckpt_steps=1)
+    CheckpointManager(cfg, model, None, is_chief=True).save(5)
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, 'examples', 'lm1b',
+                                      'lm1b_eval.py'),
+         '--ckpt', str(tmp_path), '--emb-dim', '16',
+         '--state-size', '32', '--proj-size', '16', '--batches', '2'],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert 'perplexity=' in out.stdout
