@@ -85,3 +85,7 @@ from parallax_amd.ops.misc import (clip_by_global_norm, global_norm,
                                    ema_update)
 from parallax_amd.ops.linear import PxLinear
 from parallax_amd.ops.attention import px_attention
+from parallax_amd.ops.batchnorm import ParallaxBatchNorm2d
+from parallax_amd.ops.pool import ParallaxMaxPool3x3s2, ParallaxGlobalAvgPool
+from parallax_amd.ops.conv import (ParallaxConv1x1, ParallaxConv3x3,
+                                   swap_conv1x1, swap_conv3x3)
