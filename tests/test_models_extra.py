@@ -286,3 +286,16 @@ def test_lr_schedules_reference_values():
     opt = optim.SGD([T.nn.Parameter(T.zeros(2))], lr=0.5)
     L.set_lr(opt, 0.125)
     assert opt.param_groups[0]['lr'] == 0.125
+
+
+def test_topk_accuracy_values():
+    import torch as T
+    from parallax_amd.models.cnn import topk_accuracy
+    logits = T.tensor([[9., 1., 2., 3., 4., 5., 0.],
+                       [0., 9., 8., 7., 6., 5., 1.]])
+    acc = topk_accuracy(logits, T.tensor([0, 0]))
+    assert float(acc['top_1']) == 0.5        # row0 right, row1 wrong
+    assert float(acc['top_5']) == 0.5        # label 0 not in row1 top-5
+    acc = topk_accuracy(logits, T.tensor([5, 5]))
+    assert float(acc['top_1']) == 0.0
+    assert float(acc['top_5']) == 1.0
