@@ -212,3 +212,38 @@ def test_driver_ckpt_feeds_encoder(tmp_path):
         num_sampled=min(512, max(2, len(vocab) // 2)), min_partitions=4)
     vecs = enc.encode(['tok1 tok2', 'tok3'])
     assert len(vecs) == 2 and vecs[0].shape == (512,)
+
+
+def test_eval_nested_kfold_separable_vs_shuffled():
+    import numpy as np
+    import evaluate as st_eval
+    torch.manual_seed(5)
+    enc = st_encoder.SkipThoughtsEncoder(SkipThoughts(**KW), VOCAB)
+    rng = np.random.RandomState(0)
+    sents, labels = [], []
+    for i in range(40):
+        topic = i % 2
+        base = ['cat', 'dog', 'mat'] if topic else ['ran', 'sat', 'the']
+        sents.append(' '.join(rng.choice(base, 4)))
+        labels.append(topic)
+    acc, folds = st_eval.eval_nested_kfold(enc, sents, labels, k=4)
+    assert acc > 0.9, acc
+    assert len(folds) == 4
+    # shuffled labels: near chance
+    sh = list(labels)
+    rng.shuffle(sh)
+    acc_sh, _ = st_eval.eval_nested_kfold(enc, sents, sh, k=4)
+    assert acc_sh < 0.8, acc_sh
+    with pytest.raises(ValueError):
+        st_eval.eval_nested_kfold(enc, sents[:2], [0, 1], k=4)
+
+
+def test_evaluate_driver_smoke():
+    import subprocess
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, 'examples', 'skip_thoughts',
+                                      'evaluate.py'), '--folds', '3'],
+        capture_output=True, text=True, timeout=300,
+        env=dict(os.environ, PYTHONPATH=REPO))
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert 'accuracy' in out.stdout
