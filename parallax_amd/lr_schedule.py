@@ -8,7 +8,10 @@ decay when --learning_rate/--num_epochs_per_decay are given).
 
 Schedules are plain step -> float functions; drivers apply them with
 ``set_lr(optimizer, lr)`` before each step (the optimizers read
-group['lr'] per launch in eager mode).
+group['lr'] per launch in eager mode).  Not compatible with a captured
+optimizer hipGraph (the capture freezes the scalar LR; bench.py's
+captured paths run at fixed LR, matching the reference's constant-LR
+benchmark setting).
 """
 
 IMAGENET_NUM_TRAIN_IMAGES = 1281167
