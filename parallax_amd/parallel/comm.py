@@ -164,9 +164,14 @@ class CommContext(object):
             out_counts = torch.empty_like(in_counts)
             dist.all_to_all_single(out_counts, in_counts)
             tail_shape = tuple(tensors_per_rank[0].shape[1:])
-            send = torch.cat([t.reshape(t.shape[0], -1)
+            # explicit column count: reshape(0, -1) on a zero-row chunk
+            # (a batch touching none of some remote shard's rows) is
+            # illegal, so never infer with -1
+            cols = 1
+            for d in tail_shape:
+                cols *= int(d)
+            send = torch.cat([t.reshape(t.shape[0], cols)
                               for t in tensors_per_rank], dim=0)
-            cols = send.shape[1] if send.dim() > 1 else 1
             out_counts_l = [int(c) for c in out_counts.tolist()]
             in_counts_l = [int(c) for c in in_counts.tolist()]
             recv = send.new_empty((sum(out_counts_l), cols))
