@@ -488,6 +488,12 @@ static void check_same_layout(const std::vector<at::Tensor>& a,
   for (size_t i = 0; i < a.size(); ++i) {
     TORCH_CHECK(same_dense_layout(a[i], b[i]),
                 name, ": layout mismatch at tensor ", i);
+    // the launch dispatches ONE element type from a[0] and reinterprets
+    // every pointer with it — a mismatched-dtype grad (e.g. an fp32
+    // reduce feeding a bf16 param) must fail loudly, not corrupt
+    TORCH_CHECK(a[i].scalar_type() == b[i].scalar_type(),
+                name, ": dtype mismatch at tensor ", i, " (param ",
+                a[i].scalar_type(), " vs grad ", b[i].scalar_type(), ")");
   }
 }
 

@@ -48,7 +48,11 @@ class _PxLinearFn(torch.autograd.Function):
             e = _ops.ext()
             if (hasattr(e, 'col_sum') and dy.dtype == torch.bfloat16
                     and dy.shape[1] % 8 == 0 and dy.shape[1] <= 2048):
-                db = e.col_sum(dy)
+                # col_sum accumulates fp32; the grad must come back in
+                # the bias dtype (== weight dtype here) — the fused
+                # multi-tensor optimizers dispatch one element type per
+                # launch and would reinterpret an fp32 grad row as bf16
+                db = e.col_sum(dy).to(dy.dtype)
             else:
                 db = dy.sum(0)
         else:
