@@ -997,3 +997,32 @@ def test_resnet_native_bn_step():
         assert torch.isfinite(loss)
     finally:
         del os.environ['PARALLAX_NATIVE_BN']
+
+
+def test_px_linear_bias_through_fused_adam():
+    """Regression: the col_sum bias grad must come back in the bias
+    dtype — a pure-bf16 PxLinear body stepping through the fused Adam
+    (one element type per launch) would otherwise reinterpret an fp32
+    grad row as bf16 and corrupt the update."""
+    from parallax_amd.ops.linear import PxLinear
+    from parallax_amd import optim
+    torch.manual_seed(77)
+    lin = PxLinear(512, 1024, bias=True).to(DEV, torch.bfloat16)
+    opt = optim.Adam(lin.parameters(), lr=1e-2)
+    x = torch.randn(256, 512, device=DEV, dtype=torch.bfloat16)
+    before = lin.bias.detach().float().clone()
+    for _ in range(2):
+        opt.zero_grad(set_to_none=True)
+        y = lin(x)
+        y.float().pow(2).mean().backward()
+        assert lin.bias.grad.dtype == torch.bfloat16
+        opt.step()
+    after = lin.bias.detach().float()
+    assert torch.isfinite(after).all()
+    assert not torch.equal(after, before)
+    # update magnitude sane: Adam with lr 1e-2 moves each coord O(lr)
+    assert float((after - before).abs().max()) < 0.1
+    # master mirrors the rounded param
+    mw = opt.state[lin.bias].get('master')
+    assert mw is not None
+    assert torch.equal(lin.bias.detach(), mw.to(torch.bfloat16))
