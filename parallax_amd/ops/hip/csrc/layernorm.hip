@@ -87,6 +87,10 @@ std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
                                       const at::Tensor& w,
                                       const at::Tensor& b, double eps) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              b.scalar_type() == x.scalar_type(),
+              "layernorm_fwd: weight/bias dtype must match x (the kernel "
+              "reads them with x's element type)");
   const long n = x.size(0), d = x.size(1);
   auto y = at::empty_like(x);
   auto mean = at::empty({n}, x.options().dtype(at::kFloat));
@@ -284,6 +288,8 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
                                       const at::Tensor& mean,
                                       const at::Tensor& rstd) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(w.scalar_type() == x.scalar_type(),
+              "layernorm_bwd: weight dtype must match x");
   const long n = x.size(0), d = x.size(1);
   auto dx = at::empty_like(x);
   auto dw = at::empty_like(w);

@@ -31,7 +31,12 @@ def layer_norm(x, normalized_shape, weight, bias, eps=1e-5):
     if _ops.use_native(x):
         shp = x.shape
         x2 = x.reshape(-1, shp[-1]).contiguous()
-        y = _LayerNormFn.apply(x2, weight, bias, eps)
+        # the kernel reads weight/bias with x's element type: cast when
+        # the params are fp32 over a bf16 body (autocast mode) — the
+        # .to() backward returns the [d] grads to fp32 (differentiable)
+        w = weight.to(x.dtype) if weight.dtype != x.dtype else weight
+        b = bias.to(x.dtype) if bias.dtype != x.dtype else bias
+        y = _LayerNormFn.apply(x2, w, b, eps)
         return y.view(shp)
     return torch.nn.functional.layer_norm(x, normalized_shape, weight, bias,
                                           eps)

@@ -1026,3 +1026,36 @@ def test_px_linear_bias_through_fused_adam():
     mw = opt.state[lin.bias].get('master')
     assert mw is not None
     assert torch.equal(lin.bias.detach(), mw.to(torch.bfloat16))
+
+
+def test_layernorm_fp32_params_bf16_body():
+    """Autocast-arm shape: fp32 LN weights over a bf16 body — the
+    wrapper must cast the [d] params to the kernel's element type and
+    return fp32 weight grads (regression: the kernel would otherwise
+    reinterpret fp32 weight memory as bf16)."""
+    from parallax_amd.ops.layernorm import ParallaxLayerNorm
+    torch.manual_seed(41)
+    ln = ParallaxLayerNorm(512).to(DEV)         # fp32 params
+    with torch.no_grad():
+        ln.weight.mul_(0).add_(torch.randn(512, device=DEV))
+        ln.bias.add_(torch.randn(512, device=DEV) * 0.1)
+    x = torch.randn(64, 512, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ln(x)
+    assert y.dtype == torch.bfloat16
+    x2 = x.detach().cpu().float().requires_grad_(True)
+    w2 = ln.weight.detach().cpu().requires_grad_(True)
+    b2 = ln.bias.detach().cpu().requires_grad_(True)
+    y2 = torch.nn.functional.layer_norm(x2, (512,), w2, b2)
+    assert torch.allclose(y.detach().cpu().float(), y2.detach(),
+                          atol=5e-2, rtol=5e-2)
+    dy = torch.randn(64, 512)
+    y.backward(dy.to(DEV).to(torch.bfloat16))
+    y2.backward(dy)
+    assert ln.weight.grad.dtype == torch.float32
+    assert torch.allclose(ln.weight.grad.cpu(), w2.grad, atol=0.5,
+                          rtol=5e-2)
+    assert torch.allclose(ln.bias.grad.cpu(), b2.grad, atol=0.5,
+                          rtol=5e-2)
+    assert torch.allclose(x.grad.cpu().float(), x2.grad, atol=0.5,
+                          rtol=5e-2)
