@@ -176,9 +176,10 @@ def topk_accuracy(logits, labels, ks=(1, 5)):
     """Fraction of rows whose label is in the top-k logits
     (benchmark_cnn.py top_1/top_5 accuracy fetches)."""
     out = {}
-    top = logits.detach().topk(max(ks), dim=-1).indices
+    kmax = min(max(ks), logits.shape[-1])
+    top = logits.detach().topk(kmax, dim=-1).indices
     for k in ks:
-        hit = (top[:, :k] == labels[:, None]).any(dim=1)
+        hit = (top[:, :min(k, kmax)] == labels[:, None]).any(dim=1)
         out['top_%d' % k] = hit.float().mean()
     return out
 
