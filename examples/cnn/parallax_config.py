@@ -1,4 +1,4 @@
-"""Flags -> ParallaxConfig translation for the LM1B example (the
+"""Flags -> ParallaxConfig translation for the CNN benchmark example (the
 reference's per-example parallax_config.py convention,
 examples/lm1b/parallax_config.py:19-80)."""
 
