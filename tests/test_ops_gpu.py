@@ -1059,3 +1059,35 @@ def test_layernorm_fp32_params_bf16_body():
                           rtol=5e-2)
     assert torch.allclose(x.grad.cpu().float(), x2.grad, atol=0.5,
                           rtol=5e-2)
+
+
+def test_nmt_autocast_arm_learns():
+    """The autocast arm (fp32 params, bf16 body via torch.autocast) —
+    regression for the LayerNorm fp32-params/bf16-x path: loss must be
+    finite and drop on a fixed batch."""
+    import os
+    import parallax_amd
+    from parallax_amd.config import ParallaxConfig
+    from parallax_amd.models import nmt as M
+    os.environ.setdefault('WORLD_SIZE', '1')
+    os.environ.setdefault('RANK', '0')
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    os.environ.setdefault('MASTER_PORT', '29556')
+    torch.manual_seed(0)
+    sess, *_ = parallax_amd.parallel_run(
+        lambda: M.build_nmt_graph(
+            arch='transformer', src_vocab=1000, tgt_vocab=1000,
+            d_model=64, heads=4, ffn=128, enc_layers=2, dec_layers=2,
+            min_partitions=2, lr=2e-3, use_bf16=True, pure_bf16=False),
+        None, parallax_config=ParallaxConfig(run_option='HYBRID',
+                                             search_partitions=False))
+    m = sess.graph.model
+    assert m.pos.dtype == torch.float32          # autocast arm
+    batch = M.synthetic_batch(8, 12, 12, 1000, 1000, device=DEV)
+    losses = [float(sess.run(batch)['loss']) for _ in range(12)]
+    assert all(l == l for l in losses), losses
+    assert sum(losses[-3:]) < sum(losses[:3]), losses
+    sess.close()
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
