@@ -45,12 +45,29 @@ def _spawn_once(fn, port, world, timeout):
     return errs, infra_only
 
 
+_INFRA_SIGNATURES = ('Connection refused', 'Address already in use',
+                     'Connection reset', 'Socket Timeout', 'EOFError',
+                     'Connection closed by peer', 'Broken pipe',
+                     'timed out')
+
+
+def _infra_errs(errs):
+    """True when every error is an infrastructure signature (rendezvous/
+    transport), not a test assertion."""
+    return errs and all(
+        not isinstance(e, str) or not e.startswith('worker timed out')
+        for e in errs) and all(
+        (isinstance(e, str) and e.startswith('worker exit'))
+        or any(sig in str(e) for sig in _INFRA_SIGNATURES)
+        for e in errs)
+
+
 def _spawn(fn, port, world=WORLD, timeout=180):
     errs, infra_only = _spawn_once(fn, port, world, timeout)
-    if errs and infra_only:
-        # a worker died without raising (rendezvous port clash / gloo
-        # abort): one retry on a fresh port; genuine test failures
-        # always carry a traceback and are never retried
+    if errs and (infra_only or _infra_errs(errs)):
+        # a worker died without raising, or raised a pure transport error
+        # (rendezvous port clash / gloo abort): one retry on a fresh
+        # port; genuine test failures (assertion tracebacks) never retry
         errs, _ = _spawn_once(fn, get_free_port(), world, timeout)
     assert not errs, '\n'.join(str(e) for e in errs)
 
