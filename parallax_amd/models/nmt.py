@@ -354,6 +354,75 @@ class GNMTAttention(nn.Module):
             outs.append(hyp)
         return outs
 
+    @torch.no_grad()
+    def beam_decode(self, src, beam_width=10, length_penalty_weight=0.0,
+                    sos_id=1, eos_id=2, max_len=64):
+        """Beam search with the GNMT length penalty (the reference's GNMT
+        infer defaults: beam 10; same candidate-ranking convention as
+        NMTTransformer.beam_decode — cumulative log-prob / lp at every
+        step, finished beams pad with free EOS)."""
+        enc_out, (h, c) = self.encoder(self.src_emb(src))
+        B, S, U = enc_out.shape
+        K = beam_width
+        dev = src.device
+        W = self.tgt_emb.full_weight()
+        V = W.shape[0]
+        rep = lambda t: (t.unsqueeze(1).expand(B, K, *t.shape[1:])
+                         .reshape(B * K, *t.shape[1:]))
+        enc_out = rep(enc_out)
+        h, c = rep(h).clone(), rep(c).clone()
+        ctx = enc_out.new_zeros(B * K, U)
+        cur = torch.full((B * K,), sos_id, dtype=torch.long, device=dev)
+        seqs = torch.zeros(B * K, 0, dtype=torch.long, device=dev)
+        scores = torch.full((B, K), float('-inf'), device=dev)
+        scores[:, 0] = 0.0
+        done = torch.zeros(B * K, dtype=torch.bool, device=dev)
+        lens = torch.zeros(B * K, dtype=torch.float32, device=dev)
+
+        def lp(length):
+            if length_penalty_weight == 0.0:
+                return torch.ones_like(length)
+            return ((5.0 + length) / 6.0) ** length_penalty_weight
+
+        for _ in range(max_len):
+            o, ctx, h, c = self._dec_step(self.tgt_emb(cur), ctx, h, c,
+                                          enc_out)
+            logp = torch.log_softmax((o @ W.t()).float(), dim=-1)
+            logp = torch.where(
+                done[:, None],
+                torch.full_like(logp, float('-inf')).scatter(
+                    1, torch.full((B * K, 1), eos_id, dtype=torch.long,
+                                  device=dev), 0.0),
+                logp)
+            cum = scores.reshape(B * K, 1) + logp
+            new_len = torch.where(done, lens, lens + 1)
+            ranked = cum / lp(new_len)[:, None]
+            top = ranked.reshape(B, K * V).topk(K, dim=-1).indices
+            beam_idx = top // V
+            tok = top % V
+            flat_src = (torch.arange(B, device=dev)[:, None] * K +
+                        beam_idx).reshape(-1)
+            scores = cum.reshape(B, K * V).gather(1, top)
+            seqs = torch.cat([seqs[flat_src], tok.reshape(-1, 1)], dim=1)
+            ctx, h, c = ctx[flat_src], h[flat_src], c[flat_src]
+            lens = new_len[flat_src]
+            done = done[flat_src] | (tok.reshape(-1) == eos_id)
+            cur = tok.reshape(-1)
+            if bool(done.all()):
+                break
+
+        final = (scores / lp(lens).reshape(B, K)).argmax(dim=1)
+        rows = seqs.reshape(B, K, -1)
+        outs = []
+        for b in range(B):
+            hyp = []
+            for t in rows[b, final[b]].tolist():
+                if t == eos_id:
+                    break
+                hyp.append(t)
+            outs.append(hyp)
+        return outs
+
 
 def build_nmt_graph(arch='transformer', src_vocab=SRC_VOCAB,
                     tgt_vocab=TGT_VOCAB, d_model=512, heads=8, ffn=2048,

@@ -365,3 +365,23 @@ def test_train_ckpt_then_eval_chain(tmp_path):
         capture_output=True, text=True, timeout=300, env=env)
     assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
     assert 'BLEU=' in out.stdout
+
+
+def test_gnmt_beam_decode():
+    from parallax_amd.models.nmt import GNMTAttention
+    torch.manual_seed(14)
+    m = GNMTAttention(src_vocab=40, tgt_vocab=40, emb_dim=24, units=24,
+                      min_partitions=2)
+    m.eval()
+    src = torch.randint(3, 40, (3, 5))
+    greedy = m.greedy_decode(src, max_len=8)
+    beam1 = m.beam_decode(src, beam_width=1, max_len=8)
+    assert beam1 == greedy           # K=1, alpha=0 reduces to greedy
+    outs = m.beam_decode(src, beam_width=5, length_penalty_weight=0.6,
+                         max_len=8)
+    assert len(outs) == 3
+    for row in outs:
+        assert len(row) <= 8
+        assert all(0 <= t < 40 and t != 2 for t in row)
+    assert m.beam_decode(src, beam_width=5, length_penalty_weight=0.6,
+                         max_len=8) == outs    # deterministic
