@@ -32,6 +32,8 @@ def main():
     ap.add_argument('--ckpt')
     ap.add_argument('--batch-size', type=int, default=32)
     ap.add_argument('--max-decode-len', type=int, default=64)
+    ap.add_argument('--arch', default='transformer',
+                    choices=['transformer', 'gnmt'])
     ap.add_argument('--layers', type=int, default=2)
     ap.add_argument('--d-model', type=int, default=64)
     ap.add_argument('--heads', type=int, default=4)
@@ -58,11 +60,16 @@ def main():
         sv = vocab_utils.Vocab('a b c d'.split())
         tv = vocab_utils.Vocab('x y z'.split())
 
-    model = NMTTransformer(src_vocab=len(sv), tgt_vocab=len(tv),
-                           d_model=args.d_model, heads=args.heads,
-                           ffn=args.ffn,
-                           enc_layers=args.layers, dec_layers=args.layers,
-                           min_partitions=2)
+    if args.arch == 'gnmt':
+        from parallax_amd.models.nmt import GNMTAttention
+        model = GNMTAttention(src_vocab=len(sv), tgt_vocab=len(tv),
+                              emb_dim=args.d_model, units=args.d_model,
+                              min_partitions=2)
+    else:
+        model = NMTTransformer(
+            src_vocab=len(sv), tgt_vocab=len(tv), d_model=args.d_model,
+            heads=args.heads, ffn=args.ffn, enc_layers=args.layers,
+            dec_layers=args.layers, min_partitions=2)
     if args.ckpt:
         from parallax_amd.checkpoint import latest_checkpoint
         state = torch.load(latest_checkpoint(args.ckpt),
