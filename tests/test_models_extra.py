@@ -299,3 +299,12 @@ def test_topk_accuracy_values():
     acc = topk_accuracy(logits, T.tensor([5, 5]))
     assert float(acc['top_1']) == 0.0
     assert float(acc['top_5']) == 1.0
+
+
+def test_topk_accuracy_clamps_to_classes():
+    import torch as T
+    from parallax_amd.models.cnn import topk_accuracy
+    logits = T.tensor([[1.0, 2.0, 3.0]])          # 3 classes < k=5
+    acc = topk_accuracy(logits, T.tensor([0]))
+    assert float(acc['top_1']) == 0.0
+    assert float(acc['top_5']) == 1.0             # k clamped to 3
