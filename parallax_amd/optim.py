@@ -73,6 +73,11 @@ class ParallaxOptimizer(Optimizer):
             for p in group['params']:
                 if p.grad is None:
                     continue
+                if getattr(p.grad, '_parallax_swallowed', False):
+                    # engine-captured dense grad on a sparse-tagged param
+                    # (sparse_engine._capture_hook dense branch): the
+                    # engine applies the rows itself
+                    continue
                 if p.grad.layout != torch.strided:
                     # sparse grads are handled by the engine via
                     # sparse_row_step; a bare (non-engine) run coalesces here

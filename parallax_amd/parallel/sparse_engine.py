@@ -60,14 +60,15 @@ class _SparseGradSlot(object):
 def _capture_hook(slot):
     def hook(grad):
         g = grad
-        if g.layout == torch.sparse_coo:
-            ids = g._indices().flatten()
-            rows = g._values()
-        else:
-            # a dense grad reached a sparse-tagged param (e.g. weight reuse
-            # in a dense matmul): treat all rows as touched
+        dense_use = g.layout != torch.sparse_coo
+        if dense_use:
+            # a dense grad reached a sparse-tagged param (e.g. weight
+            # reuse in a dense matmul): treat all rows as touched
             ids = torch.arange(g.shape[0], device=g.device)
             rows = g
+        else:
+            ids = g._indices().flatten()
+            rows = g._values()
         if rows.dim() == 1:  # 1-D table (bias): keep rows 2-D internally
             rows = rows.unsqueeze(1)
         if slot.ids is None:
@@ -75,6 +76,14 @@ def _capture_hook(slot):
         else:  # several uses of the table in one step (e.g. sampled softmax)
             slot.ids = torch.cat([slot.ids, ids])
             slot.rows = torch.cat([slot.rows, rows])
+        if dense_use:
+            # autograd forbids changing the grad LAYOUT in a hook, so the
+            # swallowed replacement must stay dense: a zero tensor tagged
+            # for the optimizer to skip (the engine applies the captured
+            # rows itself; ParallaxOptimizer.step honors the tag)
+            z = torch.zeros_like(g)
+            z._parallax_swallowed = True
+            return z
         return torch.sparse_coo_tensor(
             torch.zeros((1, 0), dtype=torch.long, device=grad.device),
             torch.zeros((0,) + tuple(grad.shape[1:]), dtype=grad.dtype,

@@ -331,3 +331,54 @@ def test_dense_engine_overlap_toggle(monkeypatch, free_port):
     assert n_params == len(eng.infos)
     assert set(eng._param_bucket) == {id(gi.param) for gi in eng.infos}
     sess.close()
+
+
+def test_dense_grad_on_sparse_param(monkeypatch, free_port):
+    """A sparse-tagged table used in a DENSE matmul (tied full-softmax
+    eval path) produces a dense grad; the capture hook treats every row
+    as touched and the engine still applies it (sparse_engine.py
+    _capture_hook dense branch)."""
+    import torch
+    import torch.nn as nn
+    import parallax_amd
+    from parallax_amd import optim
+    from parallax_amd.config import ParallaxConfig
+    from parallax_amd.session import TrainGraph
+    from parallax_amd.ops.embedding import ParallaxEmbedding
+
+    monkeypatch.setenv('WORLD_SIZE', '1')
+    monkeypatch.setenv('RANK', '0')
+    monkeypatch.setenv('MASTER_ADDR', '127.0.0.1')
+    monkeypatch.setenv('MASTER_PORT', str(free_port))
+    torch.manual_seed(9)
+
+    def build():
+        model = nn.Module()
+        model.emb = ParallaxEmbedding(20, 6)
+        model.fc = nn.Linear(6, 6)
+        opt = optim.Adagrad(model.parameters(), lr=0.3)
+
+        def step_fn(m, batch):
+            ids, y = batch
+            h = m.fc(m.emb(ids))                   # sparse use
+            logits = h @ m.emb.weight.t()          # DENSE use of the table
+            loss = torch.nn.functional.cross_entropy(logits, y)
+            return loss, {}
+
+        return TrainGraph(model, opt, step_fn)
+
+    sess, *_ = parallax_amd.parallel_run(
+        build, None, parallax_config=ParallaxConfig(run_option='HYBRID'))
+    w0 = sess.graph.model.emb.weight.detach().clone()
+    ids = torch.tensor([1, 3, 3, 7])
+    y = torch.tensor([0, 1, 2, 3])
+    first = None
+    for _ in range(6):
+        out = sess.run((ids, y))
+        if first is None:
+            first = float(out['loss'])
+    assert float(out['loss']) < first
+    w1 = sess.graph.model.emb.weight.detach()
+    # the dense use touches EVERY row: untouched-by-lookup rows moved too
+    assert not torch.allclose(w0[15], w1[15])
+    sess.close()
