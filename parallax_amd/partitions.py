@@ -142,8 +142,16 @@ class PartitionStatCollector(object):
             self.prev_exec_time = exec_time
         else:
             # job died: p too small (reference partitions.py:122-128)
+            self._deaths = getattr(self, '_deaths', 0) + 1
             if self.prev_p:
                 stop = True
+            elif self._deaths >= self.max_trials:
+                # every trial died before a single measurement: a
+                # persistent failure, not an OOM walk — stop doubling
+                raise RuntimeError(
+                    'partition search: %d consecutive job deaths with no '
+                    'successful trial (last p=%d) — the failure is not '
+                    'partition-related' % (self._deaths, self.p_to_test))
             else:
                 self.p_to_test *= 2
                 self.min_partitions = self.p_to_test

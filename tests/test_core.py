@@ -226,3 +226,15 @@ def test_ps_protocol_validation():
         assert PSConfig(protocol=ok).protocol == ok
     with pytest.raises(ValueError):
         PSConfig(protocol='carrier-pigeon')
+
+
+def test_stat_collector_persistent_deaths_raise():
+    """Repeated deaths with no successful trial must not double the
+    partition count forever (the failure is not partition-related)."""
+    from parallax_amd.partitions import PartitionStatCollector
+    c = PartitionStatCollector(4, min_partitions=1, max_trials=5)
+    for _ in range(4):
+        cont, p = c.update(None, alive=False)
+        assert cont
+    with pytest.raises(RuntimeError):
+        c.update(None, alive=False)
