@@ -30,6 +30,11 @@ def main():
     ap.add_argument('--learning_rate', type=float, default=0.2)
     ap.add_argument('--log_frequency', type=int, default=20)
     ap.add_argument('--use_synthetic', action='store_true', default=True)
+    ap.add_argument('--data_path', default=None,
+                    help='sentence corpus (with --vocab_path: real-data '
+                         'training via the round-robin sentence streams; '
+                         'disables --use_synthetic)')
+    ap.add_argument('--vocab_path', default=None)
     ap.add_argument('--run_option', default='HYBRID')
     ap.add_argument('--ckpt_dir', default=None)
     ap.add_argument('--save_ckpt_steps', type=int, default=None)
@@ -41,6 +46,14 @@ def main():
     ap.add_argument('--resource_info', default=os.path.join(
         os.path.dirname(os.path.abspath(__file__)), 'resource_info'))
     args = ap.parse_args()
+
+    dataset = None
+    if args.data_path:
+        import data_utils
+        vocab = data_utils.Vocabulary(args.vocab_path)
+        dataset = data_utils.LM1BDataset(args.data_path, vocab)
+        args.vocab_size = vocab.size
+        args.use_synthetic = False
 
     config = parallax_amd.Config(
         run_option=args.run_option,
@@ -66,12 +79,23 @@ def main():
             args.batch_size, args.num_steps, args.vocab_size, device,
             state0))
 
+    data_it = None
+    if dataset is not None:
+        data_it = dataset.iterate_forever(args.batch_size, args.num_steps,
+                                          num_workers, worker_id)
+
+    def real_batch(state):
+        x, y, _w = next(data_it)
+        return (torch.from_numpy(x).to(device),
+                torch.from_numpy(y).to(device), state)
+
     prev_time, prev_step = time.time(), sess.global_step
     state = None
     for local_step in range(args.max_steps):
-        out = sess.run(lm1b.synthetic_batch(
-            args.batch_size, args.num_steps, args.vocab_size, device,
-            state))
+        batch = (real_batch(state) if data_it is not None else
+                 lm1b.synthetic_batch(args.batch_size, args.num_steps,
+                                      args.vocab_size, device, state))
+        out = sess.run(batch)
         state = out.get('state')
         if local_step % args.log_frequency == 0 and worker_id == 0:
             now = time.time()

@@ -126,3 +126,21 @@ def test_lm1b_eval_ckpt_restore(tmp_path):
         capture_output=True, text=True, timeout=300)
     assert out.returncode == 0, out.stderr[-1500:]
     assert 'perplexity=' in out.stdout
+
+
+def test_lm1b_driver_real_data_smoke(tmp_path):
+    import subprocess
+    td = os.path.join(REPO, 'examples', 'lm1b', 'testdata')
+    res_info = tmp_path / 'resource_info'
+    res_info.write_text('127.0.0.1:\n')
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, 'examples', 'lm1b',
+                      'lm1b_distributed_driver.py'),
+         '--batch_size', '4', '--num_steps', '6', '--max_steps', '2',
+         '--data_path', os.path.join(td, 'test_sentences.txt'),
+         '--vocab_path', os.path.join(td, 'test_vocab.txt'),
+         '--resource_info', str(res_info)],
+        capture_output=True, text=True, timeout=300,
+        env=dict(os.environ, PYTHONPATH=REPO))
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
