@@ -25,14 +25,19 @@
 template <typename T, typename VecT, int VEC>
 __global__ void rows_gather_kernel(const T* __restrict__ w,
                                    const int64_t* __restrict__ ids,
-                                   T* __restrict__ out, long n, long d) {
+                                   T* __restrict__ out, long n, long d,
+                                   long nrows) {
   const long dv = d / VEC;
   const long total = n * dv;
   for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
     const long row = idx / dv;
     const long col = idx - row * dv;
-    const long src = ids[row];
+    // clamp instead of faulting the GPU on an out-of-range id (torch's
+    // index_select raises host-side; a device page fault here would take
+    // the whole process down) — valid ids are unaffected
+    long src = ids[row];
+    src = src < 0 ? 0 : (src >= nrows ? nrows - 1 : src);
     reinterpret_cast<VecT*>(out)[row * dv + col] =
         reinterpret_cast<const VecT*>(w)[src * dv + col];
   }
@@ -60,15 +65,18 @@ at::Tensor rows_gather(const at::Tensor& weight, const at::Tensor& ids) {
         if (sizeof(T) == 4 && d % 4 == 0) {
           long total = n * (d / 4);
           rows_gather_kernel<T, float4, 4>
-              <<<px_grid(total), PX_BLOCK, 0, stream>>>(wp, ip, op, n, d);
+              <<<px_grid(total), PX_BLOCK, 0, stream>>>(wp, ip, op, n, d,
+                                                        w.size(0));
         } else if (sizeof(T) == 2 && d % 8 == 0) {
           long total = n * (d / 8);
           rows_gather_kernel<T, float4, 8>
-              <<<px_grid(total), PX_BLOCK, 0, stream>>>(wp, ip, op, n, d);
+              <<<px_grid(total), PX_BLOCK, 0, stream>>>(wp, ip, op, n, d,
+                                                        w.size(0));
         } else {
           long total = n * d;
           rows_gather_kernel<T, T, 1>
-              <<<px_grid(total), PX_BLOCK, 0, stream>>>(wp, ip, op, n, d);
+              <<<px_grid(total), PX_BLOCK, 0, stream>>>(wp, ip, op, n, d,
+                                                        w.size(0));
         }
         HIP_CHECK_KERNEL();
       });
