@@ -33,7 +33,12 @@ __global__ void softmax_ce_fwd_kernel(const T* __restrict__ logits,
     if (threadIdx.x == 0) {
       const float lse = mx + __logf(se);
       lse_out[row] = lse;
-      loss[row] = lse - to_f32<T>(lr[labels[row]]);
+      // clamp an out-of-range label instead of reading past the row
+      // (torch's CE raises host-side; a device fault would take the
+      // process down)
+      long lb = labels[row];
+      lb = lb < 0 ? 0 : (lb >= c ? c - 1 : lb);
+      loss[row] = lse - to_f32<T>(lr[lb]);
     }
   }
 }
