@@ -385,3 +385,20 @@ def test_gnmt_beam_decode():
         assert all(0 <= t < 40 and t != 2 for t in row)
     assert m.beam_decode(src, beam_width=5, length_penalty_weight=0.6,
                          max_len=8) == outs    # deterministic
+
+
+@pytest.mark.parametrize('seed', [21, 22, 23])
+def test_beam1_equals_greedy_both_archs(seed):
+    """K=1, alpha=0 beam reduces exactly to greedy for both decoders
+    across random models."""
+    from parallax_amd.models.nmt import NMTTransformer, GNMTAttention
+    torch.manual_seed(seed)
+    tr = NMTTransformer(src_vocab=36, tgt_vocab=36, d_model=16, heads=2,
+                        ffn=32, enc_layers=1, dec_layers=1,
+                        min_partitions=2, max_len=24).eval()
+    gn = GNMTAttention(src_vocab=36, tgt_vocab=36, emb_dim=16, units=16,
+                       min_partitions=2).eval()
+    src = torch.randint(3, 36, (2 + seed % 3, 4 + seed % 4))
+    for m in (tr, gn):
+        assert m.beam_decode(src, beam_width=1, max_len=7) == \
+            m.greedy_decode(src, max_len=7), type(m).__name__
