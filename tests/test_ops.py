@@ -577,3 +577,47 @@ def test_partitioned_embedding_reshard_property():
         assert torch.equal(a.full_weight(), b.full_weight())
 
     check()
+
+
+def test_padded_engine_mixed_tables_equivalence():
+    """Padded vs dynamic sparse path over a model with BOTH a 2-D table
+    and a 1-D sparse bias (the lm1b softmax_b shape), multi-step."""
+    import os as _os
+    import torch.nn as nn
+    from parallax_amd.parallel.sparse_engine import SparsePSEngine
+    from parallax_amd.tagging import classify_params, mark_sparse
+
+    class _FakeComm:
+        world_size = 1
+        rank = 0
+        device = torch.device('cpu')
+        backend = 'gloo'
+
+    res = {}
+    for mode in ('force', '0'):
+        _os.environ['PARALLAX_PADDED_SPARSE'] = mode
+        try:
+            torch.manual_seed(21)
+            model = nn.Module()
+            model.emb = ParallaxEmbedding(30, 5)
+            model.bias = nn.Parameter(torch.zeros(30))
+            mark_sparse(model.bias)
+            opt = optim.Adagrad(model.parameters(), lr=0.4)
+            eng = SparsePSEngine(_FakeComm(), classify_params(model), opt,
+                                 average_sparse=True,
+                                 local_aggregation=True)
+            for step in range(3):
+                eng.prepare_step()
+                torch.manual_seed(100 + step)
+                ids = torch.randint(0, 30, (12,))
+                out = model.emb(ids) + embedding_lookup(
+                    model.bias, ids).unsqueeze(1)
+                out.sum().backward()
+                eng.reduce_and_apply()
+            res[mode] = (model.emb.weight.detach().clone(),
+                         model.bias.detach().clone())
+            eng.close()
+        finally:
+            _os.environ.pop('PARALLAX_PADDED_SPARSE', None)
+    for a, b in zip(res['force'], res['0']):
+        assert torch.allclose(a, b, atol=1e-5)
