@@ -77,3 +77,26 @@ def test_nmt_driver_smoke(tmp_path):
                              'nmt_distributed_driver.py'),
                 ['--batch_size', '4', '--src_len', '6', '--tgt_len', '6',
                  '--log_frequency', '1'], tmp_path)
+
+
+def test_simple_driver_world2_spawn(tmp_path):
+    """The REAL master spawner launching TWO workers (gloo on CPU):
+    resource_info with two device slots -> two spawned processes through
+    the env-var role protocol, loss converges, master exits 0."""
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    for k in ('WORLD_SIZE', 'RANK', 'LOCAL_RANK', 'MASTER_ADDR',
+              'MASTER_PORT', 'PARALLAX_RESOURCE_INFO'):
+        env.pop(k, None)
+    env['PYTHONPATH'] = REPO
+    res = tmp_path / 'resource_info'
+    res.write_text('127.0.0.1:0,1\n')
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, 'examples', 'simple', 'simple_driver.py'),
+         '--steps', '8', '--resource_info', str(res)],
+        capture_output=True, text=True, timeout=420, env=env, cwd=REPO)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    assert 'world=2' in out.stdout + out.stderr
+    assert 'final loss' in out.stdout
