@@ -17,7 +17,9 @@ def main():
     y = torch.randn(4, 1)
     for i in range(100):
         sess.run((x, y))
-        if i == 5:
+        if i == 5 and worker_id == num_workers - 1:
+            # only the LAST worker dies: at world>1 the others keep
+            # looping until the master's fail-stop kills the group
             raise RuntimeError('injected worker failure')
 
 
