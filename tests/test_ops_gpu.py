@@ -1091,3 +1091,15 @@ def test_nmt_autocast_arm_learns():
     import torch.distributed as dist
     if dist.is_initialized():
         dist.destroy_process_group()
+
+
+def test_rows_gather_oob_ids_clamped():
+    """Out-of-range ids clamp to the edge rows instead of faulting the
+    device (regression for the crash-robustness clamp)."""
+    e = _ext()
+    w = torch.arange(20, dtype=torch.float32, device=DEV).reshape(5, 4)
+    ids = torch.tensor([-3, 0, 4, 99], device=DEV)
+    out = e.rows_gather(w, ids)
+    assert torch.equal(out[0], w[0])     # negative -> row 0
+    assert torch.equal(out[3], w[4])     # past-end -> last row
+    assert torch.equal(out[1], w[0]) and torch.equal(out[2], w[4])
